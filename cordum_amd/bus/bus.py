@@ -61,6 +61,10 @@ def compute_msg_id(subject: str, pkt: BusPacket) -> Optional[str]:
     """bus/nats.go:404-434: explicit label wins; job requests 'jobreq:<id>';
     other job-bearing packets 'subject:<jobid>'; heartbeats none."""
     override = pkt.labels.get(BUS_MSG_ID_LABEL, "")
+    if not override and pkt.job_request is not None:
+        # the gateway sets the override on the job request's labels
+        # (approval republish gateway.go:3785-3797, remediation :1594-1755)
+        override = pkt.job_request.labels.get(BUS_MSG_ID_LABEL, "")
     if override:
         return override
     if pkt.heartbeat is not None:
@@ -80,6 +84,7 @@ class Subscription:
     queue_group: Optional[str] = None
     id: int = 0
     active: bool = True
+    deferred: bool = False  # deliver via pump() instead of inline (worker-style consumer)
 
     def unsubscribe(self):
         self.active = False
@@ -93,6 +98,7 @@ class _Pending:
     pkt: BusPacket = field(compare=False)
     deliveries: int = field(compare=False, default=0)
     msg_id: Optional[str] = field(compare=False, default=None)
+    target: Optional[Subscription] = field(compare=False, default=None)
 
 
 class Bus:
@@ -130,8 +136,10 @@ class LoopbackBus(Bus):
         self.record_published = False
 
     # -- subscribe -----------------------------------------------------------
-    def subscribe(self, pattern: str, handler, queue_group: Optional[str] = None) -> Subscription:
-        sub = Subscription(pattern, handler, queue_group, id=next(self._sub_seq))
+    def subscribe(
+        self, pattern: str, handler, queue_group: Optional[str] = None, deferred: bool = False
+    ) -> Subscription:
+        sub = Subscription(pattern, handler, queue_group, id=next(self._sub_seq), deferred=deferred)
         with self._mu:
             self._subs.append(sub)
         return sub
@@ -140,8 +148,14 @@ class LoopbackBus(Bus):
     def publish(self, subject: str, pkt: BusPacket) -> None:
         if subj.is_durable_subject(subject):
             msg_id = compute_msg_id(subject, pkt)
-            if msg_id is not None and self._seen(msg_id):
-                return
+            if msg_id is not None:
+                # dedup is per-stream, like JetStream's CORDUM_SYS (sys.>) vs
+                # CORDUM_JOBS (job.>, worker.*.jobs) split (bus/nats.go:339-361):
+                # the same jobreq:<id> msg-id may legally appear on both.
+                stream = "SYS" if subject.startswith("sys.") else "JOBS"
+                msg_id = f"{stream}:{msg_id}"
+                if self._seen(msg_id):
+                    return
         else:
             msg_id = None
         if self.record_published:
@@ -184,8 +198,25 @@ class LoopbackBus(Bus):
                 self._rr[key] = idx + 1
             return out
 
-    def _deliver(self, subject: str, pkt: BusPacket, deliveries: int, msg_id: Optional[str]) -> None:
-        for s in self._targets(subject):
+    def _deliver(
+        self,
+        subject: str,
+        pkt: BusPacket,
+        deliveries: int,
+        msg_id: Optional[str],
+        only: Optional[Subscription] = None,
+    ) -> None:
+        targets = [only] if only is not None else self._targets(subject)
+        for s in targets:
+            if not s.active:
+                continue
+            if s.deferred and only is None:
+                with self._mu:
+                    heapq.heappush(
+                        self._delayed,
+                        _Pending(self._clock.now(), next(self._seq), subject, pkt, deliveries, msg_id, s),
+                    )
+                continue
             try:
                 s.handler(subject, pkt)
             except RetryAfter as ra:
@@ -200,6 +231,7 @@ class LoopbackBus(Bus):
                                 pkt,
                                 deliveries + 1,
                                 msg_id,
+                                s,  # redeliver to the NAKing consumer only
                             ),
                         )
             # other exceptions propagate: handlers are expected to be total
@@ -214,7 +246,7 @@ class LoopbackBus(Bus):
                 if not self._delayed or self._delayed[0].due > t:
                     return n
                 item = heapq.heappop(self._delayed)
-            self._deliver(item.subject, item.pkt, item.deliveries, item.msg_id)
+            self._deliver(item.subject, item.pkt, item.deliveries, item.msg_id, only=item.target)
             n += 1
 
     def pending_count(self) -> int:
