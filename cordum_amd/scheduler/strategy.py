@@ -30,7 +30,6 @@ OVERLOAD_UTILIZATION_THRESHOLD = 0.9
 _NON_PLACEMENT_LABELS = {
     "preferred_worker_id",
     "preferred_pool",
-    "approval_granted",
     "secrets_present",
     "workflow_id",
     "run_id",
@@ -38,6 +37,11 @@ _NON_PLACEMENT_LABELS = {
     "node_id",
     "worker_id",
 }
+# NOTE: the Go filter (strategy_least_loaded.go:196-224) drops only
+# "approval_granted"; the gateway also stamps approval_reason/approval_note on
+# approval (gateway.go:3788-3796), which there become placement constraints
+# that no worker carries. We drop the whole approval_* family — these are
+# audit metadata, not placement labels.
 
 
 @dataclass
@@ -60,7 +64,7 @@ class PoolRouting:
 def filter_placement_labels(labels: Dict[str, str]) -> Dict[str, str]:
     out = {}
     for k, v in (labels or {}).items():
-        if k in _NON_PLACEMENT_LABELS or k.startswith("cordum."):
+        if k in _NON_PLACEMENT_LABELS or k.startswith("cordum.") or k.startswith("approval_"):
             continue
         out[k] = v
     return out
