@@ -1,0 +1,124 @@
+"""Policy compiler tests: the compiled bitset evaluator (torch reference for
+the K1 HIP kernel) must agree with the host first-match evaluator
+(safety/policy.py, oracle safety_policy.go) on randomized policies+inputs."""
+import random
+
+import pytest
+
+from cordum_amd.ops.policy_compile import (
+    CompiledPolicy,
+    JobEncoder,
+    compile_policy,
+    first_match_reference,
+)
+from cordum_amd.safety import policy as pol
+
+
+def random_policy(rng, n_rules=40, vocab=20):
+    tenants = [f"t{i}" for i in range(vocab)]
+    topics = [f"job.a{i}" for i in range(vocab)] + ["job.x.*", "job.*", "job.y?z"]
+    caps = [f"cap{i}" for i in range(vocab)]
+    tags = [f"tag{i}" for i in range(vocab)]
+    reqs = [f"req{i}" for i in range(vocab)]
+    rules = []
+    for i in range(n_rules):
+        m = pol.PolicyMatch()
+        if rng.random() < 0.5:
+            m.tenants = rng.sample(tenants, rng.randint(1, 3))
+        if rng.random() < 0.5:
+            m.topics = rng.sample(topics, rng.randint(1, 3))
+        if rng.random() < 0.3:
+            m.capabilities = rng.sample(caps, rng.randint(1, 2))
+        if rng.random() < 0.3:
+            m.risk_tags = rng.sample(tags, rng.randint(1, 3))
+        if rng.random() < 0.3:
+            m.requires = rng.sample(reqs, rng.randint(1, 2))
+        if rng.random() < 0.2:
+            m.labels = {f"k{rng.randint(0, 5)}": f"v{rng.randint(0, 3)}"}
+        if rng.random() < 0.2:
+            m.secrets_present = rng.random() < 0.5
+        if rng.random() < 0.15:
+            m.mcp = pol.MCPPolicy(
+                deny_tools=rng.sample(["shell", "exec", "rm"], rng.randint(1, 2)),
+                allow_servers=rng.sample(["srv1", "srv2"], rng.randint(0, 2)),
+            )
+        decision = rng.choice(["allow", "deny", "require_approval", "allow_with_constraints", "throttle"])
+        rules.append(pol.PolicyRule(id=f"r{i}", match=m, decision=decision))
+    return pol.SafetyPolicy(version="v1", rules=rules)
+
+
+def random_input(rng, vocab=20):
+    inp = pol.PolicyInput(
+        tenant=f"t{rng.randint(0, vocab - 1)}",
+        topic=rng.choice([f"job.a{rng.randint(0, vocab - 1)}", "job.x.deep", "job.yqz", "job.other"]),
+        capability=rng.choice(["", f"cap{rng.randint(0, vocab - 1)}"]),
+        risk_tags=rng.sample([f"tag{i}" for i in range(vocab)], rng.randint(0, 3)),
+        requires=rng.sample([f"req{i}" for i in range(vocab)], rng.randint(0, 3)),
+        labels={f"k{rng.randint(0, 5)}": f"v{rng.randint(0, 3)}"} if rng.random() < 0.5 else {},
+        secrets_present=rng.random() < 0.3,
+    )
+    if rng.random() < 0.3:
+        inp.mcp = pol.MCPRequest(
+            server=rng.choice(["", "srv1", "srv2", "srv3"]),
+            tool=rng.choice(["", "shell", "search"]),
+        )
+    return inp
+
+
+@pytest.mark.parametrize("seed", [1, 2, 3, 4])
+def test_compiled_matches_host_evaluator(seed):
+    rng = random.Random(seed)
+    policy = random_policy(rng)
+    compiled = compile_policy(policy, words=1)
+    assert compiled.exact, "test vocab should fit one word"
+    encoder = JobEncoder(compiled)
+    inputs = [random_input(rng) for _ in range(300)]
+    batch = encoder.encode(inputs)
+    got = first_match_reference(compiled, batch)
+    for j, inp in enumerate(inputs):
+        want = policy.evaluate(inp)
+        rule_idx = int(got[j])
+        if rule_idx < 0:
+            assert want.rule_id == "", f"job {j}: host matched {want.rule_id}, compiled matched none ({inp})"
+        else:
+            assert compiled.rules[rule_idx].id == want.rule_id, (
+                f"job {j}: host={want.rule_id} compiled={compiled.rules[rule_idx].id} ({inp})"
+            )
+
+
+def test_vocab_overflow_marks_inexact():
+    rules = [
+        pol.PolicyRule(id=f"r{i}", match=pol.PolicyMatch(tenants=[f"tenant-{i}"]), decision="deny")
+        for i in range(80)
+    ]
+    compiled = compile_policy(pol.SafetyPolicy(rules=rules), words=1)
+    assert not compiled.exact
+    compiled2 = compile_policy(pol.SafetyPolicy(rules=rules), words=2)
+    assert compiled2.exact
+
+
+def test_legacy_tenant_policy_compiles():
+    policy = pol.SafetyPolicy(
+        tenants={
+            "t1": pol.TenantPolicy(allow_topics=["job.echo"], deny_topics=["job.admin.*"]),
+        }
+    )
+    compiled = compile_policy(policy)
+    assert compiled.n_rules == 2  # deny + allow legacy rules
+    enc = JobEncoder(compiled)
+    batch = enc.encode([
+        pol.PolicyInput(tenant="t1", topic="job.admin.users"),
+        pol.PolicyInput(tenant="t1", topic="job.echo"),
+        pol.PolicyInput(tenant="other", topic="job.admin.users"),
+    ])
+    got = first_match_reference(compiled, batch)
+    assert compiled.rules[int(got[0])].decision == "deny"
+    assert compiled.rules[int(got[1])].decision == "allow"
+    assert int(got[2]) == -1
+
+
+def test_empty_policy():
+    compiled = compile_policy(None)
+    enc = JobEncoder(compiled)
+    batch = enc.encode([pol.PolicyInput(tenant="x", topic="job.y")])
+    assert int(first_match_reference(compiled, batch)[0]) == -1
