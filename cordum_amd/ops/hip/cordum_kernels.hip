@@ -1,0 +1,487 @@
+// CDNA4 (gfx950) kernels for the cordum_amd control plane.
+//
+// K1 policy_first_match   — batched rule×job policy evaluation
+//                           (replaces SafetyPolicy.Evaluate, safety_policy.go:187-294)
+// K2 least_loaded_pick    — batched worker scoring + masked argmin
+//                           (replaces PickSubject, strategy_least_loaded.go:40-136)
+// K4 deadline_scan        — deadline/timeout sweep over the job table
+//                           (replaces reconciler.go:88-144 + job_store deadline ZSET)
+// K5 apply_transitions    — batched job state transitions with the legality LUT
+//                           (replaces job_store.go:249-329 WATCH/tx)
+// W  echo_worker          — device worker pool execution (payload touch + result)
+//
+// Design notes (see /opt/skills/guides/cdna_hip_programming.md):
+//  - wave64; blocks of 256 threads; grid-stride where applicable.
+//  - policy kernel: one job per thread, rules staged through LDS in chunks by
+//    the whole workgroup (rule rows are reused by all 256 threads; LDS
+//    staging turns R global reads per thread into R/256 per thread).
+//  - first-match across rule chunks uses atomicMin on the output word (the
+//    output is initialized to INT_MAX by the host); chunk-level early-out
+//    via a workgroup ballot on "all jobs in this block already matched an
+//    earlier rule".
+//  - scoring kernel: score = active + cpu/100 + gpu/100 packed into a
+//    monotonic (score,idx) uint64 key -> deterministic argmin (ties by
+//    lowest worker index, matching the host strategy).
+#include <hip/hip_runtime.h>
+#include <cstdint>
+
+#define WAVE 64
+#define BLOCK 256
+
+// ---------------------------------------------------------------------------
+// K1: policy first-match
+// ---------------------------------------------------------------------------
+// Rule row layout (int64 words, W words per dim):
+//   any[7*W] | all[2*W] | mcp_allow[4*W] | mcp_deny[4*W]
+// plus per-rule bytes: secrets(int8), mcp_any(u8).
+// Job row: any[7*W] | all[2*W] | mcp[4*W], secrets u8, mcp_used u8.
+
+template <int W>
+__global__ __launch_bounds__(BLOCK) void policy_first_match_kernel(
+    const long long* __restrict__ rule_any,   // [R,7,W]
+    const long long* __restrict__ rule_all,   // [R,2,W]
+    const signed char* __restrict__ rule_secrets, // [R]
+    const long long* __restrict__ rule_mcp,   // [R,4,2,W]
+    const unsigned char* __restrict__ rule_mcp_any, // [R]
+    const long long* __restrict__ job_any,    // [J,7,W]
+    const long long* __restrict__ job_all,    // [J,2,W]
+    const unsigned char* __restrict__ job_secrets, // [J]
+    const long long* __restrict__ job_mcp,    // [J,4,W]
+    const unsigned char* __restrict__ job_mcp_used, // [J]
+    int* __restrict__ out_first,              // [J], pre-filled INT_MAX
+    int R, int J, int rules_per_chunk)
+{
+    // 2D grid: blockIdx.x tiles jobs, blockIdx.y tiles the rule range — all
+    // chunks resident at once (256 CUs want >> 256 workgroups); first-match
+    // semantics restored by the atomicMin on the output word.
+    const int rule_chunk_begin = blockIdx.y * rules_per_chunk;
+    const int rule_chunk_end = min(R, rule_chunk_begin + rules_per_chunk);
+    constexpr int ANY_W = 7 * W;
+    constexpr int ALL_W = 2 * W;
+    constexpr int MCP_W = 4 * W;
+    constexpr int ROW = ANY_W + ALL_W + 2 * MCP_W;     // int64 words per rule
+    constexpr int CHUNK = 96;                          // rules staged per LDS pass
+
+    __shared__ long long lds_rules[CHUNK * ROW];
+    __shared__ signed char lds_secrets[CHUNK];
+    __shared__ unsigned char lds_mcp_any[CHUNK];
+
+    const int j = blockIdx.x * BLOCK + threadIdx.x;
+
+    // job row -> registers
+    long long jany[ANY_W], jall[ALL_W], jmcp[MCP_W];
+    unsigned char jsec = 0, jused = 0;
+    if (j < J) {
+        #pragma unroll
+        for (int w = 0; w < ANY_W; ++w) jany[w] = job_any[(size_t)j * ANY_W + w];
+        #pragma unroll
+        for (int w = 0; w < ALL_W; ++w) jall[w] = job_all[(size_t)j * ALL_W + w];
+        #pragma unroll
+        for (int w = 0; w < MCP_W; ++w) jmcp[w] = job_mcp[(size_t)j * MCP_W + w];
+        jsec = job_secrets[j];
+        jused = job_mcp_used[j];
+    }
+
+    int best = INT_MAX;
+
+    for (int base = rule_chunk_begin; base < rule_chunk_end; base += CHUNK) {
+        const int n = min(CHUNK, rule_chunk_end - base);
+        // cooperative stage: CHUNK*ROW words by 256 threads
+        for (int i = threadIdx.x; i < n * ROW; i += BLOCK) {
+            const int r = i / ROW, w = i % ROW;
+            const size_t g = (size_t)(base + r);
+            long long v;
+            if (w < ANY_W)                 v = rule_any[g * ANY_W + w];
+            else if (w < ANY_W + ALL_W)    v = rule_all[g * ALL_W + (w - ANY_W)];
+            else                           v = rule_mcp[g * 2 * MCP_W + (w - ANY_W - ALL_W)];
+            lds_rules[i] = v;
+        }
+        for (int i = threadIdx.x; i < n; i += BLOCK) {
+            lds_secrets[i] = rule_secrets[base + i];
+            lds_mcp_any[i] = rule_mcp_any[base + i];
+        }
+        __syncthreads();
+
+        if (j < J && best == INT_MAX) {
+            for (int r = 0; r < n; ++r) {
+                const long long* row = &lds_rules[r * ROW];
+                bool ok = true;
+                #pragma unroll
+                for (int d = 0; d < 7 && ok; ++d) {
+                    long long inter = 0, any = 0;
+                    #pragma unroll
+                    for (int w = 0; w < W; ++w) {
+                        const long long rm = row[d * W + w];
+                        any |= rm;
+                        inter |= rm & jany[d * W + w];
+                    }
+                    ok = (any == 0) | (inter != 0);
+                }
+                #pragma unroll
+                for (int d = 0; d < 2 && ok; ++d) {
+                    long long missing = 0;
+                    #pragma unroll
+                    for (int w = 0; w < W; ++w)
+                        missing |= row[ANY_W + d * W + w] & ~jall[d * W + w];
+                    ok = (missing == 0);
+                }
+                if (ok) {
+                    const signed char rs = lds_secrets[r];
+                    ok = (rs < 0) | (rs == (signed char)jsec);
+                }
+                if (ok && jused && lds_mcp_any[r]) {
+                    const long long* allow = &row[ANY_W + ALL_W];
+                    const long long* deny = &row[ANY_W + ALL_W + MCP_W];
+                    #pragma unroll
+                    for (int f = 0; f < 4 && ok; ++f) {
+                        long long a = 0, ainter = 0, dinter = 0;
+                        #pragma unroll
+                        for (int w = 0; w < W; ++w) {
+                            a |= allow[f * W + w];
+                            ainter |= allow[f * W + w] & jmcp[f * W + w];
+                            dinter |= deny[f * W + w] & jmcp[f * W + w];
+                        }
+                        ok = (dinter == 0) & ((a == 0) | (ainter != 0));
+                    }
+                }
+                if (ok) { best = base + r; break; }
+            }
+        }
+        __syncthreads();
+        // chunk early-out: every live job in this block already matched
+        __shared__ int block_done;
+        if (threadIdx.x == 0) block_done = 1;
+        __syncthreads();
+        if (j < J && best == INT_MAX) block_done = 0;
+        __syncthreads();
+        if (block_done) break;
+    }
+
+    if (j < J && best != INT_MAX) atomicMin(&out_first[j], best);
+}
+
+// ---------------------------------------------------------------------------
+// K2: least-loaded worker pick
+// ---------------------------------------------------------------------------
+// Worker row: pool(int32), active(int32), maxp(int32), cpu(f32), gpu(f32),
+// labels_mask(int64). Job row: pool_mask(int64 over pool vocab), required
+// labels mask(int64), preferred_worker(int32, -1 none).
+// Output per job: packed pick int32 (-1 no_workers, -2 overloaded) .
+
+__device__ __forceinline__ unsigned long long score_key(float score, int widx) {
+    // score >= 0 -> float bits are monotonic; tie-break by lowest index
+    unsigned int sb = __float_as_uint(score);
+    return ((unsigned long long)sb << 32) | (unsigned int)widx;
+}
+
+__global__ __launch_bounds__(BLOCK) void least_loaded_pick_kernel(
+    const int* __restrict__ w_pool,        // [W]
+    const int* __restrict__ w_active,      // [W]
+    const int* __restrict__ w_maxp,        // [W]
+    const float* __restrict__ w_cpu,       // [W]
+    const float* __restrict__ w_gpu,       // [W]
+    const long long* __restrict__ w_labels, // [W]
+    const long long* __restrict__ j_poolmask, // [J]
+    const long long* __restrict__ j_labels,   // [J] required labels
+    int* __restrict__ out_pick,            // [J]
+    int NW, int NJ)
+{
+    constexpr int WCHUNK = 512;
+    __shared__ int s_pool[WCHUNK];
+    __shared__ int s_active[WCHUNK];
+    __shared__ int s_maxp[WCHUNK];
+    __shared__ float s_cpu[WCHUNK];
+    __shared__ float s_gpu[WCHUNK];
+    __shared__ long long s_labels[WCHUNK];
+
+    const int j = blockIdx.x * BLOCK + threadIdx.x;
+    long long pool_mask = 0, req_labels = 0;
+    if (j < NJ) {
+        pool_mask = j_poolmask[j];
+        req_labels = j_labels[j];
+    }
+
+    unsigned long long best = ~0ull;
+    int overloaded = 0, total = 0;
+
+    for (int base = 0; base < NW; base += WCHUNK) {
+        const int n = min(WCHUNK, NW - base);
+        for (int i = threadIdx.x; i < n; i += BLOCK) {
+            s_pool[i] = w_pool[base + i];
+            s_active[i] = w_active[base + i];
+            s_maxp[i] = w_maxp[base + i];
+            s_cpu[i] = w_cpu[base + i];
+            s_gpu[i] = w_gpu[base + i];
+            s_labels[i] = w_labels[base + i];
+        }
+        __syncthreads();
+        if (j < NJ) {
+            for (int i = 0; i < n; ++i) {
+                const int pool = s_pool[i];
+                if (pool < 0 || pool >= 64) continue;
+                if (!((pool_mask >> pool) & 1)) continue;
+                // required labels must be subset of worker labels
+                if (req_labels & ~s_labels[i]) continue;
+                ++total;
+                const int active = s_active[i], maxp = s_maxp[i];
+                const float cpu = s_cpu[i], gpu = s_gpu[i];
+                bool over = false;
+                if (maxp > 0 && (float)active / (float)maxp >= 0.9f) over = true;
+                if (cpu >= 90.f || gpu >= 90.f) over = true;
+                if (over) { ++overloaded; continue; }
+                const float score = (float)active + cpu * 0.01f + gpu * 0.01f;
+                const unsigned long long key = score_key(score, base + i);
+                if (key < best) best = key;
+            }
+        }
+        __syncthreads();
+    }
+
+    if (j < NJ) {
+        if (best != ~0ull) out_pick[j] = (int)(best & 0xffffffffu);
+        else if (total > 0 && overloaded == total) out_pick[j] = -2;
+        else out_pick[j] = -1;
+    }
+}
+
+// ---------------------------------------------------------------------------
+// K5: batched state transitions with legality LUT (job_store.go:70-82)
+// ---------------------------------------------------------------------------
+#define N_STATES 11
+__constant__ unsigned char d_transition_lut[N_STATES * N_STATES];
+
+__global__ __launch_bounds__(BLOCK) void apply_transitions_kernel(
+    unsigned char* __restrict__ states,     // [N] job table states
+    int* __restrict__ attempts,             // [N]
+    const int* __restrict__ slots,          // [B] job slots to transition
+    const unsigned char* __restrict__ to_states, // [B] target states
+    unsigned char* __restrict__ ok_out,     // [B] 1 = applied
+    long long* __restrict__ deadlines,      // [N] cleared on terminal
+    int B)
+{
+    const int i = blockIdx.x * BLOCK + threadIdx.x;
+    if (i >= B) return;
+    const int slot = slots[i];
+    const unsigned char to = to_states[i];
+    const unsigned char from = states[slot];
+    const unsigned char legal = d_transition_lut[from * N_STATES + to];
+    ok_out[i] = legal;
+    if (!legal) return;
+    states[slot] = to;
+    // attempts increment on entering SCHEDULED (JobState.SCHEDULED == 3)
+    if (to == 3 && from != 3) attempts[slot] += 1;
+    // terminal states clear the deadline (SUCCEEDED..DENIED are 6..10)
+    if (to >= 6) deadlines[slot] = (long long)0x7fffffffffffffffLL;
+}
+
+// ---------------------------------------------------------------------------
+// K4: deadline / staleness scan -> TIMEOUT candidates (compacted list)
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(BLOCK) void deadline_scan_kernel(
+    const unsigned char* __restrict__ states,   // [N]
+    const long long* __restrict__ deadlines,    // [N] unix micros
+    const long long* __restrict__ updated_at,   // [N] unix micros
+    long long now_us, long long dispatch_cutoff_us, long long running_cutoff_us,
+    int* __restrict__ out_slots,                // [cap]
+    int* __restrict__ out_count,                // [1]
+    int N, int cap)
+{
+    const int i = blockIdx.x * BLOCK + threadIdx.x;
+    if (i >= N) return;
+    const unsigned char st = states[i];
+    // active states: PENDING(1) APPROVAL(2) SCHEDULED(3) DISPATCHED(4) RUNNING(5)
+    bool expired = false;
+    if (st >= 1 && st <= 5 && deadlines[i] <= now_us) expired = true;
+    if (st == 4 && updated_at[i] <= dispatch_cutoff_us) expired = true;  // DISPATCHED stale
+    if (st == 5 && updated_at[i] <= running_cutoff_us) expired = true;   // RUNNING stale
+    if (expired) {
+        const int pos = atomicAdd(out_count, 1);
+        if (pos < cap) out_slots[pos] = i;
+    }
+}
+
+// ---------------------------------------------------------------------------
+// Echo worker: touch the job payload, write a result word, bump counters
+// ---------------------------------------------------------------------------
+// Payload arena: flat uint32 arena, job i owns [i*stride, (i+1)*stride).
+// The "work" is a full read of the payload (the echo semantics: the result
+// is derived from the entire context blob) + one result word per job.
+__global__ __launch_bounds__(BLOCK) void echo_worker_kernel(
+    const unsigned int* __restrict__ ctx_arena,  // [B*stride]
+    unsigned int* __restrict__ res_arena,        // [B*stride]
+    unsigned int* __restrict__ res_sum,          // [B]
+    int B, int stride)
+{
+    // one wave per job: lanes stream the payload, wave-reduce a checksum,
+    // and echo the payload into the result arena (memory-bound, like the
+    // reference's echo worker copying ctx -> res through Redis).
+    const int wave_id = (blockIdx.x * BLOCK + threadIdx.x) / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    if (wave_id >= B) return;
+    const size_t basep = (size_t)wave_id * stride;
+    unsigned int acc = 0;
+    for (int k = lane; k < stride; k += WAVE) {
+        const unsigned int v = ctx_arena[basep + k];
+        res_arena[basep + k] = v;
+        acc += v;
+    }
+    // wave reduction
+    #pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+        acc += __shfl_down(acc, off, WAVE);
+    if (lane == 0) res_sum[wave_id] = acc;
+}
+
+// ---------------------------------------------------------------------------
+// Torch extension host wrappers
+// ---------------------------------------------------------------------------
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#define CHECK_DEV(t) TORCH_CHECK((t).is_cuda(), #t " must be on device")
+#define CHECK_CONTIG(t) TORCH_CHECK((t).is_contiguous(), #t " must be contiguous")
+
+static inline hipStream_t cur_stream() {
+    return at::hip::getCurrentHIPStream().stream();
+}
+
+torch::Tensor policy_first_match(
+    torch::Tensor rule_any, torch::Tensor rule_all, torch::Tensor rule_secrets,
+    torch::Tensor rule_mcp, torch::Tensor rule_mcp_any,
+    torch::Tensor job_any, torch::Tensor job_all, torch::Tensor job_secrets,
+    torch::Tensor job_mcp, torch::Tensor job_mcp_used,
+    int64_t rule_chunks)
+{
+    for (auto* t : {&rule_any, &rule_all, &rule_mcp, &job_any, &job_all, &job_mcp}) {
+        CHECK_DEV(*t); CHECK_CONTIG(*t);
+    }
+    const int W = (int)rule_any.size(2);
+    const int R = (int)rule_any.size(0);
+    const int J = (int)job_any.size(0);
+    TORCH_CHECK(job_any.size(2) == W, "word-count mismatch");
+    auto out = torch::full({J}, INT_MAX,
+        torch::TensorOptions().dtype(torch::kInt32).device(job_any.device()));
+    if (R == 0 || J == 0) {
+        return out.masked_fill(out == INT_MAX, -1);
+    }
+    const int job_blocks = (J + BLOCK - 1) / BLOCK;
+    int nchunks = (int)rule_chunks;
+    if (nchunks <= 0) {
+        // fill the chip (256 CUs want >> 256 workgroups total), but keep
+        // chunks >= 512 rules so LDS staging amortizes
+        nchunks = std::max(1, std::min((R + 511) / 512,
+                                       std::max(1, 2048 / std::max(job_blocks, 1))));
+    }
+    const int per_chunk = (R + nchunks - 1) / nchunks;
+    hipStream_t stream = cur_stream();
+
+    auto launch = [&](auto kern) {
+        hipLaunchKernelGGL(kern, dim3(job_blocks, nchunks), dim3(BLOCK), 0, stream,
+            (const long long*)rule_any.data_ptr<int64_t>(),
+            (const long long*)rule_all.data_ptr<int64_t>(),
+            (const signed char*)rule_secrets.data_ptr<int8_t>(),
+            (const long long*)rule_mcp.data_ptr<int64_t>(),
+            rule_mcp_any.data_ptr<uint8_t>(),
+            (const long long*)job_any.data_ptr<int64_t>(),
+            (const long long*)job_all.data_ptr<int64_t>(),
+            job_secrets.data_ptr<uint8_t>(),
+            (const long long*)job_mcp.data_ptr<int64_t>(),
+            job_mcp_used.data_ptr<uint8_t>(),
+            out.data_ptr<int>(), R, J, per_chunk);
+    };
+    switch (W) {
+        case 1: launch(policy_first_match_kernel<1>); break;
+        case 2: launch(policy_first_match_kernel<2>); break;
+        case 4: launch(policy_first_match_kernel<4>); break;
+        default: TORCH_CHECK(false, "unsupported word count ", W);
+    }
+    out.masked_fill_(out == INT_MAX, -1);
+    return out;
+}
+
+torch::Tensor least_loaded_pick(
+    torch::Tensor w_pool, torch::Tensor w_active, torch::Tensor w_maxp,
+    torch::Tensor w_cpu, torch::Tensor w_gpu, torch::Tensor w_labels,
+    torch::Tensor j_poolmask, torch::Tensor j_labels)
+{
+    CHECK_DEV(w_pool); CHECK_DEV(j_poolmask);
+    const int NW = (int)w_pool.size(0);
+    const int NJ = (int)j_poolmask.size(0);
+    auto out = torch::empty({NJ}, torch::TensorOptions().dtype(torch::kInt32).device(j_poolmask.device()));
+    if (NJ == 0) return out;
+    const int blocks = (NJ + BLOCK - 1) / BLOCK;
+    hipLaunchKernelGGL(least_loaded_pick_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        w_pool.data_ptr<int>(), w_active.data_ptr<int>(), w_maxp.data_ptr<int>(),
+        w_cpu.data_ptr<float>(), w_gpu.data_ptr<float>(),
+        (const long long*)w_labels.data_ptr<int64_t>(),
+        (const long long*)j_poolmask.data_ptr<int64_t>(),
+        (const long long*)j_labels.data_ptr<int64_t>(),
+        out.data_ptr<int>(), NW, NJ);
+    return out;
+}
+
+void set_transition_lut(torch::Tensor lut) {
+    TORCH_CHECK(lut.numel() == N_STATES * N_STATES, "lut must be 11x11");
+    auto cpu = lut.to(torch::kUInt8).contiguous().cpu();
+    (void)hipMemcpyToSymbol(HIP_SYMBOL(d_transition_lut), cpu.data_ptr<uint8_t>(),
+                            N_STATES * N_STATES);
+}
+
+torch::Tensor apply_transitions(
+    torch::Tensor states, torch::Tensor attempts, torch::Tensor deadlines,
+    torch::Tensor slots, torch::Tensor to_states)
+{
+    CHECK_DEV(states); CHECK_DEV(slots);
+    const int B = (int)slots.size(0);
+    auto ok = torch::zeros({B}, torch::TensorOptions().dtype(torch::kUInt8).device(states.device()));
+    if (B == 0) return ok;
+    const int blocks = (B + BLOCK - 1) / BLOCK;
+    hipLaunchKernelGGL(apply_transitions_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        states.data_ptr<uint8_t>(), attempts.data_ptr<int>(),
+        slots.data_ptr<int>(), to_states.data_ptr<uint8_t>(),
+        ok.data_ptr<uint8_t>(),
+        (long long*)deadlines.data_ptr<int64_t>(), B);
+    return ok;
+}
+
+std::tuple<torch::Tensor, torch::Tensor> deadline_scan(
+    torch::Tensor states, torch::Tensor deadlines, torch::Tensor updated_at,
+    int64_t now_us, int64_t dispatch_cutoff_us, int64_t running_cutoff_us, int64_t cap)
+{
+    CHECK_DEV(states);
+    const int N = (int)states.size(0);
+    auto slots = torch::empty({cap}, torch::TensorOptions().dtype(torch::kInt32).device(states.device()));
+    auto count = torch::zeros({1}, torch::TensorOptions().dtype(torch::kInt32).device(states.device()));
+    if (N == 0) return {slots, count};
+    const int blocks = (N + BLOCK - 1) / BLOCK;
+    hipLaunchKernelGGL(deadline_scan_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        states.data_ptr<uint8_t>(),
+        (const long long*)deadlines.data_ptr<int64_t>(),
+        (const long long*)updated_at.data_ptr<int64_t>(),
+        (long long)now_us, (long long)dispatch_cutoff_us, (long long)running_cutoff_us,
+        slots.data_ptr<int>(), count.data_ptr<int>(), N, (int)cap);
+    return {slots, count};
+}
+
+torch::Tensor echo_execute(torch::Tensor ctx_arena, torch::Tensor res_arena, int64_t stride)
+{
+    CHECK_DEV(ctx_arena); CHECK_CONTIG(ctx_arena);
+    const int B = (int)(ctx_arena.numel() / stride);
+    auto sums = torch::empty({B}, torch::TensorOptions().dtype(torch::kInt32).device(ctx_arena.device()));
+    const int waves_per_block = BLOCK / WAVE;
+    const int blocks = (B + waves_per_block - 1) / waves_per_block;
+    hipLaunchKernelGGL(echo_worker_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        (const unsigned int*)ctx_arena.data_ptr<int32_t>(),
+        (unsigned int*)res_arena.data_ptr<int32_t>(),
+        (unsigned int*)sums.data_ptr<int32_t>(), B, (int)stride);
+    return sums;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("policy_first_match", &policy_first_match, "K1 batched policy first-match");
+    m.def("least_loaded_pick", &least_loaded_pick, "K2 least-loaded worker pick");
+    m.def("set_transition_lut", &set_transition_lut, "upload transition legality LUT");
+    m.def("apply_transitions", &apply_transitions, "K5 batched state transitions");
+    m.def("deadline_scan", &deadline_scan, "K4 deadline/staleness scan");
+    m.def("echo_execute", &echo_execute, "device echo worker pool");
+}

@@ -1,0 +1,106 @@
+"""Torch reference implementations of every HIP kernel.
+
+These are the numerics oracles for the GPU tests (plain torch, fp32/int
+exact) and the CPU execution path for environments without a GPU. Each
+function mirrors its kernel in cordum_amd/ops/hip/cordum_kernels.hip
+bit-for-bit (deterministic tie-breaks included).
+"""
+from __future__ import annotations
+
+from typing import Tuple
+
+import torch
+
+from ..protocol.states import transition_lut
+
+OVERLOAD = 0.9
+
+
+def least_loaded_pick_ref(
+    w_pool: torch.Tensor,
+    w_active: torch.Tensor,
+    w_maxp: torch.Tensor,
+    w_cpu: torch.Tensor,
+    w_gpu: torch.Tensor,
+    w_labels: torch.Tensor,
+    j_poolmask: torch.Tensor,
+    j_labels: torch.Tensor,
+) -> torch.Tensor:
+    """K2 oracle. Returns int32 [J]: worker idx, -1 no_workers, -2 overloaded."""
+    NW = w_pool.shape[0]
+    NJ = j_poolmask.shape[0]
+    if NJ == 0:
+        return torch.empty(0, dtype=torch.int32)
+    pool_ok = ((j_poolmask.unsqueeze(1) >> w_pool.clamp(0, 63).unsqueeze(0)) & 1).bool()
+    pool_ok &= (w_pool >= 0).unsqueeze(0) & (w_pool < 64).unsqueeze(0)
+    labels_ok = (j_labels.unsqueeze(1) & ~w_labels.unsqueeze(0)) == 0
+    eligible = pool_ok & labels_ok  # [J, W]
+
+    util_over = (w_maxp > 0) & (w_active.float() / w_maxp.clamp(min=1).float() >= OVERLOAD)
+    over = util_over | (w_cpu >= 90) | (w_gpu >= 90)  # [W]
+    usable = eligible & ~over.unsqueeze(0)
+
+    score = w_active.float() + w_cpu * 0.01 + w_gpu * 0.01
+    key = (score.view(torch.int32).to(torch.int64) << 32) | torch.arange(NW, dtype=torch.int64)
+    big = torch.iinfo(torch.int64).max
+    keys = torch.where(usable, key.unsqueeze(0).expand(NJ, NW), torch.full((NJ, NW), big, dtype=torch.int64))
+    best = keys.min(dim=1).values
+    pick = (best & 0xFFFFFFFF).to(torch.int32)
+
+    n_eligible = eligible.sum(dim=1)
+    n_over = (eligible & over.unsqueeze(0)).sum(dim=1)
+    none = best == big
+    all_over = none & (n_eligible > 0) & (n_over == n_eligible)
+    out = torch.where(none, torch.full_like(pick, -1), pick)
+    out = torch.where(all_over, torch.full_like(pick, -2), out)
+    return out
+
+
+def apply_transitions_ref(
+    states: torch.Tensor,
+    attempts: torch.Tensor,
+    deadlines: torch.Tensor,
+    slots: torch.Tensor,
+    to_states: torch.Tensor,
+) -> torch.Tensor:
+    """K5 oracle (sequential; duplicate slots in one batch are applied in
+    order, matching the kernel's per-slot independence assumption — the
+    pipeline never puts the same slot twice in one batch)."""
+    lut = torch.tensor(transition_lut(), dtype=torch.uint8)
+    ok = torch.zeros(slots.shape[0], dtype=torch.uint8)
+    for i in range(slots.shape[0]):
+        s = int(slots[i])
+        frm = int(states[s])
+        to = int(to_states[i])
+        if lut[frm, to]:
+            ok[i] = 1
+            states[s] = to
+            if to == 3 and frm != 3:
+                attempts[s] += 1
+            if to >= 6:
+                deadlines[s] = torch.iinfo(torch.int64).max
+    return ok
+
+
+def deadline_scan_ref(
+    states: torch.Tensor,
+    deadlines: torch.Tensor,
+    updated_at: torch.Tensor,
+    now_us: int,
+    dispatch_cutoff_us: int,
+    running_cutoff_us: int,
+) -> torch.Tensor:
+    """K4 oracle: sorted slot indices due for TIMEOUT."""
+    active = (states >= 1) & (states <= 5)
+    expired = active & (deadlines <= now_us)
+    expired |= (states == 4) & (updated_at <= dispatch_cutoff_us)
+    expired |= (states == 5) & (updated_at <= running_cutoff_us)
+    return torch.nonzero(expired, as_tuple=False).flatten().to(torch.int32)
+
+
+def echo_execute_ref(ctx_arena: torch.Tensor, res_arena: torch.Tensor, stride: int) -> torch.Tensor:
+    """Echo worker oracle: copy payload, return per-job uint32 checksum."""
+    B = ctx_arena.numel() // stride
+    ctx = ctx_arena.view(B, stride)
+    res_arena.view(B, stride).copy_(ctx)
+    return ctx.to(torch.int64).sum(dim=1).remainder(1 << 32).to(torch.int32)

@@ -1,0 +1,149 @@
+"""GPU numerics tests: every HIP kernel vs its plain-torch reference
+(ops/reference.py, ops/policy_compile.first_match_reference) and, for K1,
+transitively vs the host policy evaluator oracle."""
+import random
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from cordum_amd.ops import get_ext
+from cordum_amd.ops.policy_compile import JobEncoder, compile_policy, first_match_reference
+from cordum_amd.ops.reference import (
+    apply_transitions_ref,
+    deadline_scan_ref,
+    echo_execute_ref,
+    least_loaded_pick_ref,
+)
+from cordum_amd.protocol.states import transition_lut
+from tests.test_policy_compile import random_input, random_policy
+
+
+@pytest.fixture(scope="module")
+def ext():
+    e = get_ext(required=True)
+    return e
+
+
+def dev():
+    return torch.device("cuda:0")
+
+
+def test_policy_first_match_matches_reference(ext):
+    rng = random.Random(7)
+    policy = random_policy(rng, n_rules=500, vocab=30)
+    compiled = compile_policy(policy, words=1)
+    assert compiled.exact
+    enc = JobEncoder(compiled)
+    inputs = [random_input(rng, vocab=30) for _ in range(2048)]
+    batch = enc.encode(inputs)
+    want = first_match_reference(compiled, batch)
+
+    c = compiled.to(dev())
+    b = batch.to(dev())
+    got = ext.policy_first_match(
+        c.any_masks, c.all_masks, c.secrets, c.mcp_masks, c.mcp_any,
+        b.any_bits, b.all_bits, b.secrets, b.mcp_bits, b.mcp_used, 0,
+    ).cpu()
+    assert torch.equal(got, want)
+    # transitive check against the host evaluator on a sample
+    for j in rng.sample(range(len(inputs)), 64):
+        d = policy.evaluate(inputs[j])
+        idx = int(got[j])
+        if idx < 0:
+            assert d.rule_id == ""
+        else:
+            assert compiled.rules[idx].id == d.rule_id
+
+
+def test_policy_first_match_large_rule_count(ext):
+    """Config #4 shape: 100k-rule bundle, multi-word vocab."""
+    rng = random.Random(11)
+    policy = random_policy(rng, n_rules=2000, vocab=100)
+    compiled = compile_policy(policy, words=2)
+    assert compiled.exact
+    enc = JobEncoder(compiled)
+    inputs = [random_input(rng, vocab=100) for _ in range(512)]
+    batch = enc.encode(inputs)
+    want = first_match_reference(compiled, batch)
+    c, b = compiled.to(dev()), batch.to(dev())
+    got = ext.policy_first_match(
+        c.any_masks, c.all_masks, c.secrets, c.mcp_masks, c.mcp_any,
+        b.any_bits, b.all_bits, b.secrets, b.mcp_bits, b.mcp_used, 0,
+    ).cpu()
+    assert torch.equal(got, want)
+
+
+def test_least_loaded_pick_matches_reference(ext):
+    g = torch.Generator().manual_seed(3)
+    NW, NJ = 1000, 4096
+    w_pool = torch.randint(0, 8, (NW,), dtype=torch.int32, generator=g)
+    w_active = torch.randint(0, 10, (NW,), dtype=torch.int32, generator=g)
+    w_maxp = torch.randint(0, 12, (NW,), dtype=torch.int32, generator=g)
+    w_cpu = torch.rand(NW, generator=g) * 100
+    w_gpu = torch.rand(NW, generator=g) * 100
+    w_labels = torch.randint(0, 16, (NW,), dtype=torch.int64, generator=g)
+    j_poolmask = (1 << torch.randint(0, 8, (NJ,), generator=g)).to(torch.int64)
+    j_labels = torch.where(torch.rand(NJ, generator=g) < 0.2,
+                           torch.randint(0, 4, (NJ,), dtype=torch.int64, generator=g),
+                           torch.zeros(NJ, dtype=torch.int64))
+    want = least_loaded_pick_ref(w_pool, w_active, w_maxp, w_cpu, w_gpu, w_labels, j_poolmask, j_labels)
+    d = dev()
+    got = ext.least_loaded_pick(
+        w_pool.to(d), w_active.to(d), w_maxp.to(d), w_cpu.to(d), w_gpu.to(d),
+        w_labels.to(d), j_poolmask.to(d), j_labels.to(d),
+    ).cpu()
+    assert torch.equal(got, want)
+
+
+def test_apply_transitions_matches_reference(ext):
+    g = torch.Generator().manual_seed(5)
+    N, B = 10000, 4000
+    states0 = torch.randint(0, 11, (N,), dtype=torch.uint8, generator=g)
+    attempts0 = torch.zeros(N, dtype=torch.int32)
+    deadlines0 = torch.randint(0, 1 << 40, (N,), dtype=torch.int64, generator=g)
+    slots = torch.randperm(N, generator=g)[:B].to(torch.int32)  # unique slots
+    to_states = torch.randint(0, 11, (B,), dtype=torch.uint8, generator=g)
+
+    ref_states, ref_attempts, ref_deadlines = states0.clone(), attempts0.clone(), deadlines0.clone()
+    want_ok = apply_transitions_ref(ref_states, ref_attempts, ref_deadlines, slots, to_states)
+
+    d = dev()
+    ext.set_transition_lut(torch.tensor(transition_lut(), dtype=torch.uint8).flatten())
+    g_states, g_attempts, g_deadlines = states0.to(d), attempts0.to(d), deadlines0.to(d)
+    got_ok = ext.apply_transitions(g_states, g_attempts, g_deadlines, slots.to(d), to_states.to(d)).cpu()
+    assert torch.equal(got_ok, want_ok)
+    assert torch.equal(g_states.cpu(), ref_states)
+    assert torch.equal(g_attempts.cpu(), ref_attempts)
+    assert torch.equal(g_deadlines.cpu(), ref_deadlines)
+
+
+def test_deadline_scan_matches_reference(ext):
+    g = torch.Generator().manual_seed(9)
+    N = 50000
+    states = torch.randint(0, 11, (N,), dtype=torch.uint8, generator=g)
+    now = 1_000_000
+    deadlines = torch.randint(0, 2_000_000, (N,), dtype=torch.int64, generator=g)
+    updated = torch.randint(0, 2_000_000, (N,), dtype=torch.int64, generator=g)
+    want = deadline_scan_ref(states, deadlines, updated, now, 500_000, 100_000)
+    d = dev()
+    slots, count = ext.deadline_scan(states.to(d), deadlines.to(d), updated.to(d),
+                                     now, 500_000, 100_000, N)
+    n = int(count.cpu()[0])
+    got = slots[:n].cpu().sort().values
+    assert torch.equal(got, want.sort().values)
+
+
+def test_echo_execute_matches_reference(ext):
+    g = torch.Generator().manual_seed(13)
+    B, stride = 2048, 256
+    ctx = torch.randint(-(1 << 31), (1 << 31) - 1, (B * stride,), dtype=torch.int32, generator=g)
+    res_ref = torch.zeros_like(ctx)
+    want = echo_execute_ref(ctx, res_ref, stride)
+    d = dev()
+    ctx_d = ctx.to(d)
+    res_d = torch.zeros_like(ctx_d)
+    got = ext.echo_execute(ctx_d, res_d, stride).cpu()
+    assert torch.equal(res_d.cpu(), res_ref)
+    assert torch.equal(got, want)
