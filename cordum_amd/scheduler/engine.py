@@ -124,7 +124,9 @@ class Engine:
             self.handle_job_result(pkt.job_result)
             return
         if pkt.job_cancel is not None:
-            self.cancel_job(pkt.job_cancel.job_id, pkt.job_cancel.reason or "cancelled")
+            # bus-delivered cancel: store transition only — republishing here
+            # would echo the broadcast back onto the same subject
+            self.job_store.cancel_job(pkt.job_cancel.job_id)
 
     # -- submit path -----------------------------------------------------------
     def handle_job_request(self, req: JobRequest, trace_id: str = "") -> None:

@@ -1,0 +1,348 @@
+"""Gateway HTTP tests (reference seam: httptest suites — gateway_test.go,
+jobs_http_test.go, approvals_test.go, packs_*_test.go, policy_bundles_test.go)."""
+import io
+import json
+import tarfile
+
+import pytest
+from fastapi.testclient import TestClient
+
+from cordum_amd.gateway import BasicAuthProvider, create_app
+from cordum_amd.protocol import JobState
+from cordum_amd.runtime.node import Node
+from cordum_amd.scheduler import PoolProfile, PoolRouting
+from cordum_amd.utils.clock import ManualClock
+
+POLICY = """
+version: v1
+default_tenant: default
+rules:
+  - id: approve-risky
+    decision: require_approval
+    reason: risky
+    match: {risk_tags: [risky]}
+  - id: deny-bad
+    decision: deny
+    reason: blocked
+    match: {topics: ["job.bad"]}
+"""
+
+
+@pytest.fixture
+def node():
+    clock = ManualClock()
+    routing = PoolRouting(
+        topics={"job.default": ["default"], "job.echo": ["default"]},
+        pools={"default": PoolProfile()},
+    )
+    n = Node(clock=clock, routing=routing, policy_yaml=POLICY).start()
+    n.add_worker("w1", topics=["job.default", "job.echo"])
+    return n
+
+
+@pytest.fixture
+def client(node):
+    auth = BasicAuthProvider(api_keys=["test-key"], tenant="default")
+    app = create_app(node, auth=auth)
+    c = TestClient(app)
+    c.headers.update({"X-API-Key": "test-key", "X-Principal-Id": "tester",
+                      "X-Principal-Role": "admin"})
+    return c
+
+
+def test_auth_required(node):
+    app = create_app(node, auth=BasicAuthProvider(api_keys=["k1"]))
+    c = TestClient(app)
+    assert c.get("/api/v1/status").status_code == 401
+    assert c.get("/api/v1/status", headers={"X-API-Key": "k1"}).status_code == 200
+
+
+def test_submit_and_get_job(client, node):
+    r = client.post("/api/v1/jobs", json={"prompt": "hello world", "topic": "job.echo"})
+    assert r.status_code == 200, r.text
+    job_id = r.json()["job_id"]
+    trace_id = r.json()["trace_id"]
+    r = client.get(f"/api/v1/jobs/{job_id}")
+    assert r.status_code == 200
+    d = r.json()
+    assert d["state"] == "SUCCEEDED"
+    assert d["topic"] == "job.echo"
+    assert d["result"] == {"prompt": "hello world"}
+    assert d["context"]["prompt"] == "hello world"
+    assert d["safety_decision"] == "allow"
+    assert d["trace_id"] == trace_id
+    # trace endpoint
+    r = client.get(f"/api/v1/traces/{trace_id}")
+    assert r.json()["jobs"][0]["id"] == job_id
+
+
+def test_submit_validation(client):
+    assert client.post("/api/v1/jobs", json={}).status_code == 400  # no prompt
+    assert client.post("/api/v1/jobs", json={"prompt": "x", "topic": "sys.bad"}).status_code == 400
+    assert client.post("/api/v1/jobs", json={"prompt": "x", "deadline_ms": -1}).status_code == 400
+    assert client.post("/api/v1/jobs", json={"prompt": "x", "actor_type": "robot"}).status_code == 400
+    labels = {f"k{i}": "v" for i in range(51)}
+    assert client.post("/api/v1/jobs", json={"prompt": "x", "labels": labels}).status_code == 400
+
+
+def test_submit_idempotency(client):
+    r1 = client.post("/api/v1/jobs", json={"prompt": "a", "topic": "job.echo", "idempotency_key": "same"})
+    r2 = client.post("/api/v1/jobs", json={"prompt": "a", "topic": "job.echo", "idempotency_key": "same"})
+    assert r1.json()["job_id"] == r2.json()["job_id"]
+    assert r2.json().get("deduplicated") is True
+
+
+def test_secrets_scan_triggers_risk_tag(client, node):
+    r = client.post("/api/v1/jobs", json={"prompt": "use secret://vault/key", "topic": "job.echo"})
+    job_id = r.json()["job_id"]
+    d = client.get(f"/api/v1/jobs/{job_id}").json()
+    assert "secrets" in d["risk_tags"]
+
+
+def test_job_list_filters(client, node):
+    client.post("/api/v1/jobs", json={"prompt": "a", "topic": "job.echo"})
+    client.post("/api/v1/jobs", json={"prompt": "b", "topic": "job.bad"})
+    r = client.get("/api/v1/jobs", params={"state": "SUCCEEDED"})
+    assert all(i["state"] == "SUCCEEDED" for i in r.json()["items"])
+    r = client.get("/api/v1/jobs", params={"topic": "job.bad"})
+    assert all(i["topic"] == "job.bad" for i in r.json()["items"])
+
+
+def test_deny_path_populates_dlq(client, node):
+    r = client.post("/api/v1/jobs", json={"prompt": "x", "topic": "job.bad"})
+    job_id = r.json()["job_id"]
+    d = client.get(f"/api/v1/jobs/{job_id}").json()
+    assert d["state"] == "DENIED"
+    assert d["error_code"] == "safety_denied"
+    dlq = client.get("/api/v1/dlq").json()["items"]
+    assert any(e["job_id"] == job_id for e in dlq)
+
+
+def test_dlq_retry_endpoint(client, node):
+    r = client.post("/api/v1/jobs", json={"prompt": "x", "topic": "job.bad"})
+    job_id = r.json()["job_id"]
+    r = client.post(f"/api/v1/dlq/{job_id}/retry")
+    assert r.status_code == 200
+    new_id = r.json()["job_id"]
+    assert new_id.startswith(job_id + "-retry-")
+    d = client.get(f"/api/v1/jobs/{new_id}").json()
+    assert d["state"] == "DENIED"  # still denied by policy, but retried through the machine
+    assert d["labels"]["retry_of_job"] == job_id
+
+
+def test_approval_http_flow(client, node):
+    r = client.post("/api/v1/jobs", json={"prompt": "x", "topic": "job.echo", "risk_tags": ["risky"]})
+    job_id = r.json()["job_id"]
+    assert client.get(f"/api/v1/jobs/{job_id}").json()["state"] == "APPROVAL_REQUIRED"
+    items = client.get("/api/v1/approvals").json()["items"]
+    assert any(i["id"] == job_id for i in items)
+    r = client.post(f"/api/v1/approvals/{job_id}/approve", json={"reason": "looks fine"})
+    assert r.status_code == 200, r.text
+    d = client.get(f"/api/v1/jobs/{job_id}").json()
+    assert d["state"] == "SUCCEEDED"
+    assert d["approval_by"] == "tester"
+    assert d["approval_reason"] == "looks fine"
+
+
+def test_approval_requires_admin(node):
+    auth = BasicAuthProvider(api_keys=["k"], tenant="default")
+    app = create_app(node, auth=auth)
+    c = TestClient(app)
+    c.headers.update({"X-API-Key": "k", "X-Principal-Role": "user"})
+    r = c.post("/api/v1/jobs", json={"prompt": "x", "topic": "job.echo", "risk_tags": ["risky"]})
+    job_id = r.json()["job_id"]
+    assert c.post(f"/api/v1/approvals/{job_id}/approve", json={}).status_code == 403
+
+
+def test_approval_reject(client, node):
+    r = client.post("/api/v1/jobs", json={"prompt": "x", "topic": "job.echo", "risk_tags": ["risky"]})
+    job_id = r.json()["job_id"]
+    r = client.post(f"/api/v1/approvals/{job_id}/reject", json={"reason": "nope"})
+    assert r.status_code == 200
+    assert client.get(f"/api/v1/jobs/{job_id}").json()["state"] == "DENIED"
+
+
+def test_cancel_endpoint(client, node):
+    # job stuck pending (no workers for topic after removing worker)
+    node.configsvc.set("system", "default", {"pools": {"topics": {"job.stuck": ["nowhere"]},
+                                                       "pools": {"nowhere": {}}}})
+    r = client.post("/api/v1/jobs", json={"prompt": "x", "topic": "job.stuck"})
+    job_id = r.json()["job_id"]
+    assert client.get(f"/api/v1/jobs/{job_id}").json()["state"] == "PENDING"
+    r = client.post(f"/api/v1/jobs/{job_id}/cancel")
+    assert r.status_code == 200
+    assert client.get(f"/api/v1/jobs/{job_id}").json()["state"] == "CANCELLED"
+
+
+def test_workflow_http_crud_and_run(client, node):
+    wf = {
+        "id": "hello-workflow",
+        "name": "Hello",
+        "input_schema": {"type": "object", "required": ["message"]},
+        "steps": {"echo": {"type": "worker", "topic": "job.echo",
+                           "input": {"message": "${input.message}"}}},
+    }
+    assert client.post("/api/v1/workflows", json=wf).status_code == 200
+    assert any(w["id"] == "hello-workflow" for w in client.get("/api/v1/workflows").json()["items"])
+    # schema validation rejects bad input
+    r = client.post("/api/v1/workflows/hello-workflow/runs", json={"input": {}})
+    assert r.status_code == 400
+    r = client.post("/api/v1/workflows/hello-workflow/runs", json={"input": {"message": "hi"}})
+    assert r.status_code == 200, r.text
+    run_id = r.json()["run_id"]
+    run = client.get(f"/api/v1/workflow-runs/{run_id}").json()
+    assert run["status"] == "succeeded"
+    tl = client.get(f"/api/v1/workflow-runs/{run_id}/timeline").json()["items"]
+    assert any(e["type"] == "step_completed" for e in tl)
+    # rerun
+    r = client.post(f"/api/v1/workflow-runs/{run_id}/rerun", json={})
+    assert r.status_code == 200
+    assert client.get(f"/api/v1/workflow-runs/{r.json()['run_id']}").json()["status"] == "succeeded"
+    # delete
+    assert client.delete("/api/v1/workflows/hello-workflow").status_code == 200
+
+
+def test_policy_endpoints(client, node):
+    r = client.post("/api/v1/policy/evaluate", json={"topic": "job.bad", "tenant": "default"})
+    assert r.json()["decision"] == "DECISION_TYPE_DENY"
+    r = client.post("/api/v1/policy/explain", json={"topic": "job.echo", "tenant": "default"})
+    assert "rules" in r.json()
+    r = client.get("/api/v1/policy/rules")
+    assert any(rule["id"] == "deny-bad" for rule in r.json()["items"])
+    assert client.get("/api/v1/policy/snapshots").json()["current"]
+
+
+def test_policy_bundle_studio(client, node):
+    content = "version: v2\nrules:\n  - id: extra-deny\n    decision: deny\n    match: {topics: [\"job.extra\"]}\n"
+    r = client.put("/api/v1/policy/bundles/team~extra", json={"content": content, "message": "add"})
+    assert r.status_code == 200, r.text
+    items = client.get("/api/v1/policy/bundles").json()["items"]
+    assert any(b["id"] == "team/extra" for b in items)
+    # the kernel picked the bundle up: evaluation now denies job.extra
+    r = client.post("/api/v1/policy/evaluate", json={"topic": "job.extra", "tenant": "default"})
+    assert r.json()["decision"] == "DECISION_TYPE_DENY"
+    # simulate a draft without side effects
+    r = client.post("/api/v1/policy/bundles/team~extra/simulate", json={
+        "content": "version: v3\nrules:\n  - id: sim\n    decision: allow\n    match: {topics: [\"job.extra\"]}\n",
+        "input": {"topic": "job.extra", "tenant": "default"},
+    })
+    assert r.json()["results"][0]["rule_id"] == "sim"
+    # snapshots + publish + rollback
+    snap = client.post("/api/v1/policy/bundles/snapshots", json={"message": "before"}).json()["id"]
+    client.put("/api/v1/policy/bundles/team~extra", json={"content": "", "enabled": False})
+    r = client.post("/api/v1/policy/rollback", json={"snapshot_id": snap})
+    assert r.status_code == 200
+    r = client.post("/api/v1/policy/evaluate", json={"topic": "job.extra", "tenant": "default"})
+    assert r.json()["decision"] == "DECISION_TYPE_DENY"  # rollback restored the deny
+    audit = client.get("/api/v1/policy/audit").json()["items"]
+    assert any(e["action"] == "rollback" for e in audit)
+
+
+def make_pack_tgz(manifest: dict, files: dict) -> bytes:
+    buf = io.BytesIO()
+    with tarfile.open(fileobj=buf, mode="w:gz") as tf:
+        import yaml as _yaml
+
+        def add(name, data: bytes):
+            info = tarfile.TarInfo(name)
+            info.size = len(data)
+            tf.addfile(info, io.BytesIO(data))
+
+        add("pack.yaml", _yaml.safe_dump(manifest).encode())
+        for name, data in files.items():
+            add(name, data)
+    return buf.getvalue()
+
+
+HELLO_PACK = {
+    "apiVersion": "cordum.io/v1alpha1",
+    "metadata": {"name": "hello-pack", "version": "1.0.0", "description": "echo"},
+    "compatibility": {"protocolVersion": 1},
+    "topics": [{"topic": "job.echo", "capability": "echo"}],
+    "resources": {"schemas": ["schemas/echo.json"], "workflows": ["workflows/hello.json"]},
+    "overlays": {
+        "config": [{"scope": "system", "key": "default",
+                    "json_merge_patch": {"pools": {"topics": {"job.echo": ["default"], "job.default": ["default"]},
+                                                   "pools": {"default": {}}}}}],
+        "policy": [{"id": "hello-pack/policy",
+                    "bundle_fragment": "version: hp\nrules:\n  - id: hp-allow\n    decision: allow\n    match: {topics: [\"job.echo\"]}\n"}],
+    },
+}
+
+HELLO_FILES = {
+    "schemas/echo.json": json.dumps({"$id": "echo-input", "type": "object", "required": ["message"]}).encode(),
+    "workflows/hello.json": json.dumps({
+        "id": "hello-workflow", "name": "Hello",
+        "steps": {"echo": {"type": "worker", "topic": "job.echo", "input": {"message": "${input.message}"}}},
+    }).encode(),
+}
+
+
+def test_pack_install_uninstall_verify(client, node):
+    blob = make_pack_tgz(HELLO_PACK, HELLO_FILES)
+    r = client.post("/api/v1/packs/install", content=blob)
+    assert r.status_code == 200, r.text
+    assert r.json()["status"] == "ACTIVE"
+    # resources landed
+    assert client.get("/api/v1/schemas/echo-input").status_code == 200
+    assert client.get("/api/v1/workflows/hello-workflow").status_code == 200
+    packs = client.get("/api/v1/packs").json()["items"]
+    assert any(p["id"] == "hello-pack" and p["status"] == "ACTIVE" for p in packs)
+    # the installed workflow runs end-to-end
+    r = client.post("/api/v1/workflows/hello-workflow/runs", json={"input": {"message": "from pack"}})
+    assert r.status_code == 200
+    assert client.get(f"/api/v1/workflow-runs/{r.json()['run_id']}").json()["status"] == "succeeded"
+    # verify + uninstall (soft)
+    assert client.post("/api/v1/packs/hello-pack/verify").json()["ok"] is True
+    r = client.post("/api/v1/packs/hello-pack/uninstall")
+    assert r.json()["status"] == "INACTIVE"
+    assert client.get("/api/v1/workflows/hello-workflow").status_code == 200  # kept
+
+
+def test_pack_zip_slip_rejected(client):
+    buf = io.BytesIO()
+    with tarfile.open(fileobj=buf, mode="w:gz") as tf:
+        data = b"evil"
+        info = tarfile.TarInfo("../../etc/passwd")
+        info.size = len(data)
+        tf.addfile(info, io.BytesIO(data))
+    r = client.post("/api/v1/packs/install", content=buf.getvalue())
+    assert r.status_code == 400
+
+
+def test_locks_endpoints(client):
+    assert client.post("/api/v1/locks/acquire", json={"resource": "r1", "owner": "a"}).status_code == 200
+    assert client.post("/api/v1/locks/acquire", json={"resource": "r1", "owner": "b"}).status_code == 409
+    assert client.get("/api/v1/locks").json()["items"][0]["resource"] == "r1"
+    assert client.post("/api/v1/locks/renew", json={"resource": "r1", "owner": "a"}).status_code == 200
+    assert client.post("/api/v1/locks/release", json={"resource": "r1", "owner": "a"}).json()["released"]
+
+
+def test_schemas_and_config_endpoints(client):
+    assert client.post("/api/v1/schemas", json={"id": "s1", "schema": {"type": "object"}}).status_code == 200
+    assert client.get("/api/v1/schemas/s1").json()["schema"] == {"type": "object"}
+    assert "s1" in client.get("/api/v1/schemas").json()["items"]
+    assert client.delete("/api/v1/schemas/s1").status_code == 200
+    r = client.post("/api/v1/config", json={"scope": "org", "id": "acme", "config": {"retry": {"max": 2}}})
+    assert r.json()["revision"] == 1
+    eff = client.get("/api/v1/config/effective", params={"org": "acme"}).json()
+    assert eff["config"]["retry"]["max"] == 2 and eff["hash"]
+
+
+def test_status_and_workers(client, node):
+    s = client.get("/api/v1/status").json()
+    assert s["status"] == "ok" and s["workers"] == 1
+    w = client.get("/api/v1/workers").json()
+    assert w["pools"]["default"]["workers"] == 1
+
+
+def test_artifacts_and_memory(client, node):
+    r = client.post("/api/v1/artifacts", content=b'{"x": 1}',
+                    headers={"content-type": "application/json", "x-retention": "audit"})
+    ptr = r.json()["ptr"]
+    art_id = ptr.split("art:")[1]
+    assert client.get(f"/api/v1/artifacts/{art_id}").json() == {"x": 1}
+    # memory pointer reader
+    node.memory.put("ctx:manual", b'{"y": 2}')
+    assert client.get("/api/v1/memory", params={"ptr": "redis://ctx:manual"}).json() == {"y": 2}
+    assert client.get("/api/v1/memory", params={"ptr": "redis://missing"}).status_code == 404
