@@ -1,0 +1,77 @@
+"""Node server entrypoint: assemble the single-process control plane from
+config files and serve the gateway (replaces the reference's five service
+mains + docker-compose)."""
+from __future__ import annotations
+
+import threading
+import time
+from pathlib import Path
+from typing import Optional
+
+
+def build_node(config_dir: str = "", checkpoint_dir: str = ""):
+    from ..config import (
+        NodeConfig,
+        load_pools,
+        load_safety_yaml,
+        load_timeouts,
+        seed_system_config,
+    )
+    from ..runtime.node import Node
+
+    cfg = NodeConfig.from_env(Path(config_dir) if config_dir else None)
+    routing = load_pools(cfg.pool_config_path)
+    timeouts = load_timeouts(cfg.timeout_config_path)
+    policy_yaml = load_safety_yaml(cfg.safety_policy_path)
+    node = Node(routing=routing, policy_yaml=policy_yaml,
+                safety_cache_ttl_s=cfg.safety_decision_cache_ttl_s).start()
+    node.scheduler_reconciler.update_timeouts(timeouts.dispatch_timeout_s, timeouts.running_timeout_s)
+    seed_system_config(node.configsvc, cfg.system_config_path)
+
+    checkpointer = None
+    if checkpoint_dir:
+        from ..store.wal import Checkpointer
+
+        checkpointer = Checkpointer(node, checkpoint_dir)
+        if checkpointer.restore():
+            checkpointer.replay_wal()
+    return node, cfg, checkpointer
+
+
+def serve(host: str = "127.0.0.1", port: int = 8080, config_dir: str = "",
+          workers: int = 2, checkpoint_dir: str = ""):
+    import uvicorn
+
+    from ..gateway import create_app
+    from ..runtime.worker import echo_handler
+
+    node, cfg, checkpointer = build_node(config_dir, checkpoint_dir)
+    for i in range(workers):
+        node.add_worker(f"worker-{i}", handler=echo_handler,
+                        topics=sorted(node.strategy.current_routing().topics))
+
+    app = create_app(node, rate_limit_rps=cfg.api_rate_limit_rps,
+                     rate_limit_burst=cfg.api_rate_limit_burst)
+
+    stop = threading.Event()
+
+    def control_loop():
+        last_reconcile = 0.0
+        last_checkpoint = 0.0
+        while not stop.is_set():
+            node.tick()
+            now = time.time()
+            if now - last_reconcile > 5.0:
+                node.reconcile()
+                last_reconcile = now
+            if checkpointer is not None and now - last_checkpoint > 30.0:
+                checkpointer.checkpoint()
+                last_checkpoint = now
+            time.sleep(0.05)
+
+    t = threading.Thread(target=control_loop, daemon=True)
+    t.start()
+    try:
+        uvicorn.run(app, host=host, port=port, log_level="warning")
+    finally:
+        stop.set()
