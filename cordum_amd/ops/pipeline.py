@@ -37,7 +37,7 @@ import torch.distributed as dist
 from ..protocol.states import transition_lut
 from ..safety import policy as pol
 from . import get_ext
-from .policy_compile import CompiledPolicy, JobBatch, JobEncoder, compile_policy
+from .policy_compile import CompiledPolicy, JobBatch, JobEncoder, compile_policy, first_match_reference
 
 UNSPECIFIED, PENDING, APPROVAL, SCHEDULED, DISPATCHED, RUNNING = 0, 1, 2, 3, 4, 5
 SUCCEEDED, FAILED, CANCELLED, TIMEOUT, DENIED = 6, 7, 8, 9, 10
@@ -95,6 +95,48 @@ def encode_synthetic_jobs(compiled: CompiledPolicy, n_jobs: int, vocab: int = 48
     return enc.encode(inputs)
 
 
+class _RefOps:
+    """CPU backend with the HIP extension's call signatures, built on the
+    torch reference implementations (ops/reference.py). Used for the gloo
+    multi-process tests and CPU-only environments; on a GPU host the HIP
+    extension is mandatory (ops/__init__.get_ext)."""
+
+    def __init__(self):
+        self._lut = None
+
+    def set_transition_lut(self, lut):
+        self._lut = lut
+
+    def policy_first_match(self, rule_any, rule_all, rule_secrets, rule_mcp, rule_mcp_any,
+                           job_any, job_all, job_secrets, job_mcp, job_mcp_used, chunks):
+        from types import SimpleNamespace
+
+        c = SimpleNamespace(any_masks=rule_any, all_masks=rule_all, secrets=rule_secrets,
+                            mcp_masks=rule_mcp, mcp_any=rule_mcp_any,
+                            n_rules=int(rule_any.shape[0]))
+        b = SimpleNamespace(any_bits=job_any, all_bits=job_all, secrets=job_secrets,
+                            mcp_bits=job_mcp, mcp_used=job_mcp_used,
+                            n_jobs=int(job_any.shape[0]))
+        return first_match_reference(c, b)
+
+    def least_loaded_pick(self, w_pool, w_active, w_maxp, w_cpu, w_gpu, w_labels,
+                          j_poolmask, j_labels):
+        from .reference import least_loaded_pick_ref
+
+        return least_loaded_pick_ref(w_pool, w_active, w_maxp, w_cpu, w_gpu, w_labels,
+                                     j_poolmask, j_labels)
+
+    def apply_transitions(self, states, attempts, deadlines, slots, to_states):
+        from .reference import apply_transitions_ref
+
+        return apply_transitions_ref(states, attempts, deadlines, slots, to_states)
+
+    def echo_execute(self, ctx_arena, res_arena, stride):
+        from .reference import echo_execute_ref
+
+        return echo_execute_ref(ctx_arena, res_arena, int(stride))
+
+
 class DevicePipeline:
     def __init__(
         self,
@@ -108,8 +150,10 @@ class DevicePipeline:
         n_batches: int = 8,
         policy: Optional[pol.SafetyPolicy] = None,
         seed: int = 7,
+        backend: str = "ext",
     ):
-        self.ext = get_ext(required=True)
+        device = torch.device(device)
+        self.ext = _RefOps() if backend == "ref" else get_ext(required=True)
         self.device = device
         self.B = batch_size
         self.NWL = n_local_workers
@@ -166,8 +210,8 @@ class DevicePipeline:
                                      dtype=torch.int64, device=device)
         self.j_labels = torch.zeros(self.B, dtype=torch.int64, device=device)
 
-        # result arena
-        self.res_arena = torch.zeros(self.B * payload_words, dtype=torch.int32, device=device)
+        # result arena: a rank can receive up to world*B jobs in one tick
+        self.res_arena = torch.zeros(self.B * self.world * payload_words, dtype=torch.int32, device=device)
         self._tick = 0
         self.total_completed = 0
         self.total_denied = 0
@@ -294,7 +338,8 @@ class DevicePipeline:
             self.w_active_local.scatter_add_(
                 0, recv_widx.long(), torch.ones_like(recv_widx, dtype=torch.int32)
             )
-        torch.cuda.synchronize(dev)
+        if dev.type == "cuda":
+            torch.cuda.synchronize(dev)
         dt = time.perf_counter() - t0
 
         stats = TickStats(
