@@ -174,27 +174,55 @@ __device__ __forceinline__ unsigned long long score_key(float score, int widx) {
     return ((unsigned long long)sb << 32) | (unsigned int)widx;
 }
 
-__global__ __launch_bounds__(BLOCK) void least_loaded_pick_kernel(
+__global__ __launch_bounds__(BLOCK) void worker_precompute_kernel(
     const int* __restrict__ w_pool,        // [W]
     const int* __restrict__ w_active,      // [W]
     const int* __restrict__ w_maxp,        // [W]
     const float* __restrict__ w_cpu,       // [W]
     const float* __restrict__ w_gpu,       // [W]
+    unsigned long long* __restrict__ w_key, // [W] out: (score,idx) or flags
+    int NW)
+{
+    // per-worker tick-invariant precompute: pack (score, idx) into a
+    // monotonic u64 key; overloaded workers get OVERLOADED_KEY so the picker
+    // can still distinguish "all overloaded" from "no workers".
+    const int i = blockIdx.x * BLOCK + threadIdx.x;
+    if (i >= NW) return;
+    const int active = w_active[i], maxp = w_maxp[i];
+    const float cpu = w_cpu[i], gpu = w_gpu[i];
+    bool over = false;
+    if (maxp > 0 && (float)active / (float)maxp >= 0.9f) over = true;
+    if (cpu >= 90.f || gpu >= 90.f) over = true;
+    if (over) {
+        w_key[i] = 0xfffffffe00000000ull | (unsigned int)i;  // OVERLOADED
+    } else {
+        const float score = (float)active + cpu * 0.01f + gpu * 0.01f;
+        w_key[i] = score_key(score, i);
+    }
+}
+
+#define OVERLOADED_TAG 0xfffffffeu
+
+__global__ __launch_bounds__(BLOCK) void least_loaded_pick_kernel(
+    const int* __restrict__ w_pool,         // [W]
+    const unsigned long long* __restrict__ w_key, // [W] from precompute
     const long long* __restrict__ w_labels, // [W]
     const long long* __restrict__ j_poolmask, // [J]
     const long long* __restrict__ j_labels,   // [J] required labels
-    int* __restrict__ out_pick,            // [J]
+    int* __restrict__ out_pick,             // [J]
     int NW, int NJ)
 {
-    constexpr int WCHUNK = 512;
-    __shared__ int s_pool[WCHUNK];
-    __shared__ int s_active[WCHUNK];
-    __shared__ int s_maxp[WCHUNK];
-    __shared__ float s_cpu[WCHUNK];
-    __shared__ float s_gpu[WCHUNK];
+    // one WAVE per job: 64 lanes stride the worker table (staged through LDS
+    // once per workgroup, shared by the 4 waves), then a wave min-reduce.
+    constexpr int WCHUNK = 1024;
+    __shared__ unsigned long long s_key[WCHUNK];
     __shared__ long long s_labels[WCHUNK];
+    __shared__ int s_pool[WCHUNK];
 
-    const int j = blockIdx.x * BLOCK + threadIdx.x;
+    const int wave = threadIdx.x / WAVE;           // 0..3
+    const int lane = threadIdx.x % WAVE;
+    const int j = blockIdx.x * (BLOCK / WAVE) + wave;
+
     long long pool_mask = 0, req_labels = 0;
     if (j < NJ) {
         pool_mask = j_poolmask[j];
@@ -207,37 +235,35 @@ __global__ __launch_bounds__(BLOCK) void least_loaded_pick_kernel(
     for (int base = 0; base < NW; base += WCHUNK) {
         const int n = min(WCHUNK, NW - base);
         for (int i = threadIdx.x; i < n; i += BLOCK) {
-            s_pool[i] = w_pool[base + i];
-            s_active[i] = w_active[base + i];
-            s_maxp[i] = w_maxp[base + i];
-            s_cpu[i] = w_cpu[base + i];
-            s_gpu[i] = w_gpu[base + i];
+            s_key[i] = w_key[base + i];
             s_labels[i] = w_labels[base + i];
+            s_pool[i] = w_pool[base + i];
         }
         __syncthreads();
         if (j < NJ) {
-            for (int i = 0; i < n; ++i) {
+            for (int i = lane; i < n; i += WAVE) {
                 const int pool = s_pool[i];
                 if (pool < 0 || pool >= 64) continue;
                 if (!((pool_mask >> pool) & 1)) continue;
-                // required labels must be subset of worker labels
                 if (req_labels & ~s_labels[i]) continue;
                 ++total;
-                const int active = s_active[i], maxp = s_maxp[i];
-                const float cpu = s_cpu[i], gpu = s_gpu[i];
-                bool over = false;
-                if (maxp > 0 && (float)active / (float)maxp >= 0.9f) over = true;
-                if (cpu >= 90.f || gpu >= 90.f) over = true;
-                if (over) { ++overloaded; continue; }
-                const float score = (float)active + cpu * 0.01f + gpu * 0.01f;
-                const unsigned long long key = score_key(score, base + i);
+                const unsigned long long key = s_key[i];
+                if ((unsigned int)(key >> 32) == OVERLOADED_TAG) { ++overloaded; continue; }
                 if (key < best) best = key;
             }
         }
         __syncthreads();
     }
 
-    if (j < NJ) {
+    // wave reductions (64-lane shuffles)
+    #pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+        const unsigned long long other = __shfl_down(best, off, WAVE);
+        if (other < best) best = other;
+        overloaded += __shfl_down(overloaded, off, WAVE);
+        total += __shfl_down(total, off, WAVE);
+    }
+    if (lane == 0 && j < NJ) {
         if (best != ~0ull) out_pick[j] = (int)(best & 0xffffffffu);
         else if (total > 0 && overloaded == total) out_pick[j] = -2;
         else out_pick[j] = -1;
@@ -332,6 +358,33 @@ __global__ __launch_bounds__(BLOCK) void echo_worker_kernel(
     if (lane == 0) res_sum[wave_id] = acc;
 }
 
+__global__ __launch_bounds__(BLOCK) void echo_worker_indexed_kernel(
+    const unsigned int* __restrict__ ctx_arena,  // [N*stride] home arena
+    const int* __restrict__ slots,               // [B] job slots to execute
+    unsigned int* __restrict__ res_arena,        // [N*stride] result arena
+    unsigned int* __restrict__ res_sum,          // [N] by slot
+    int B, int stride)
+{
+    // local-dispatch echo: read the job's payload in place (no staging copy
+    // — within one GPU "dispatch" is just ownership, exactly as the worker
+    // pool reads the HBM arena directly), write result + checksum.
+    const int w = (blockIdx.x * BLOCK + threadIdx.x) / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    if (w >= B) return;
+    const int slot = slots[w];
+    const size_t basep = (size_t)slot * stride;
+    unsigned int acc = 0;
+    for (int k = lane; k < stride; k += WAVE) {
+        const unsigned int v = ctx_arena[basep + k];
+        res_arena[basep + k] = v;
+        acc += v;
+    }
+    #pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+        acc += __shfl_down(acc, off, WAVE);
+    if (lane == 0) res_sum[slot] = acc;
+}
+
 // ---------------------------------------------------------------------------
 // Torch extension host wrappers
 // ---------------------------------------------------------------------------
@@ -399,9 +452,24 @@ torch::Tensor policy_first_match(
     return out;
 }
 
-torch::Tensor least_loaded_pick(
+torch::Tensor worker_precompute(
     torch::Tensor w_pool, torch::Tensor w_active, torch::Tensor w_maxp,
-    torch::Tensor w_cpu, torch::Tensor w_gpu, torch::Tensor w_labels,
+    torch::Tensor w_cpu, torch::Tensor w_gpu)
+{
+    CHECK_DEV(w_pool);
+    const int NW = (int)w_pool.size(0);
+    auto keys = torch::empty({NW}, torch::TensorOptions().dtype(torch::kInt64).device(w_pool.device()));
+    if (NW == 0) return keys;
+    const int blocks = (NW + BLOCK - 1) / BLOCK;
+    hipLaunchKernelGGL(worker_precompute_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        w_pool.data_ptr<int>(), w_active.data_ptr<int>(), w_maxp.data_ptr<int>(),
+        w_cpu.data_ptr<float>(), w_gpu.data_ptr<float>(),
+        (unsigned long long*)keys.data_ptr<int64_t>(), NW);
+    return keys;
+}
+
+torch::Tensor least_loaded_pick(
+    torch::Tensor w_pool, torch::Tensor w_keys, torch::Tensor w_labels,
     torch::Tensor j_poolmask, torch::Tensor j_labels)
 {
     CHECK_DEV(w_pool); CHECK_DEV(j_poolmask);
@@ -409,10 +477,11 @@ torch::Tensor least_loaded_pick(
     const int NJ = (int)j_poolmask.size(0);
     auto out = torch::empty({NJ}, torch::TensorOptions().dtype(torch::kInt32).device(j_poolmask.device()));
     if (NJ == 0) return out;
-    const int blocks = (NJ + BLOCK - 1) / BLOCK;
+    const int jobs_per_block = BLOCK / WAVE;
+    const int blocks = (NJ + jobs_per_block - 1) / jobs_per_block;
     hipLaunchKernelGGL(least_loaded_pick_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
-        w_pool.data_ptr<int>(), w_active.data_ptr<int>(), w_maxp.data_ptr<int>(),
-        w_cpu.data_ptr<float>(), w_gpu.data_ptr<float>(),
+        w_pool.data_ptr<int>(),
+        (const unsigned long long*)w_keys.data_ptr<int64_t>(),
         (const long long*)w_labels.data_ptr<int64_t>(),
         (const long long*)j_poolmask.data_ptr<int64_t>(),
         (const long long*)j_labels.data_ptr<int64_t>(),
@@ -477,9 +546,28 @@ torch::Tensor echo_execute(torch::Tensor ctx_arena, torch::Tensor res_arena, int
     return sums;
 }
 
+torch::Tensor echo_execute_indexed(torch::Tensor ctx_arena, torch::Tensor slots,
+                                   torch::Tensor res_arena, torch::Tensor res_sum,
+                                   int64_t stride)
+{
+    CHECK_DEV(ctx_arena); CHECK_DEV(slots);
+    const int B = (int)slots.size(0);
+    if (B == 0) return res_sum;
+    const int waves_per_block = BLOCK / WAVE;
+    const int blocks = (B + waves_per_block - 1) / waves_per_block;
+    hipLaunchKernelGGL(echo_worker_indexed_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        (const unsigned int*)ctx_arena.data_ptr<int32_t>(),
+        slots.data_ptr<int>(),
+        (unsigned int*)res_arena.data_ptr<int32_t>(),
+        (unsigned int*)res_sum.data_ptr<int32_t>(), B, (int)stride);
+    return res_sum;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("policy_first_match", &policy_first_match, "K1 batched policy first-match");
+    m.def("worker_precompute", &worker_precompute, "K2a per-worker score/overload precompute");
     m.def("least_loaded_pick", &least_loaded_pick, "K2 least-loaded worker pick");
+    m.def("echo_execute_indexed", &echo_execute_indexed, "device echo worker pool (slot-indexed, in-place)");
     m.def("set_transition_lut", &set_transition_lut, "upload transition legality LUT");
     m.def("apply_transitions", &apply_transitions, "K5 batched state transitions");
     m.def("deadline_scan", &deadline_scan, "K4 deadline/staleness scan");

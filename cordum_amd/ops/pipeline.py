@@ -119,12 +119,20 @@ class _RefOps:
                             n_jobs=int(job_any.shape[0]))
         return first_match_reference(c, b)
 
-    def least_loaded_pick(self, w_pool, w_active, w_maxp, w_cpu, w_gpu, w_labels,
-                          j_poolmask, j_labels):
-        from .reference import least_loaded_pick_ref
+    def worker_precompute(self, w_pool, w_active, w_maxp, w_cpu, w_gpu):
+        from .reference import worker_precompute_ref
 
-        return least_loaded_pick_ref(w_pool, w_active, w_maxp, w_cpu, w_gpu, w_labels,
-                                     j_poolmask, j_labels)
+        return worker_precompute_ref(w_pool, w_active, w_maxp, w_cpu, w_gpu)
+
+    def least_loaded_pick(self, w_pool, w_keys, w_labels, j_poolmask, j_labels):
+        from .reference import least_loaded_pick_keys_ref
+
+        return least_loaded_pick_keys_ref(w_pool, w_keys, w_labels, j_poolmask, j_labels)
+
+    def echo_execute_indexed(self, ctx_arena, slots, res_arena, res_sum, stride):
+        from .reference import echo_execute_indexed_ref
+
+        return echo_execute_indexed_ref(ctx_arena, slots, res_arena, res_sum, int(stride))
 
     def apply_transitions(self, states, attempts, deadlines, slots, to_states):
         from .reference import apply_transitions_ref
@@ -212,6 +220,7 @@ class DevicePipeline:
 
         # result arena: a rank can receive up to world*B jobs in one tick
         self.res_arena = torch.zeros(self.B * self.world * payload_words, dtype=torch.int32, device=device)
+        self.res_sums = torch.zeros(self.B, dtype=torch.int32, device=device)
         self._tick = 0
         self.total_completed = 0
         self.total_denied = 0
@@ -260,10 +269,12 @@ class DevicePipeline:
             self.w_cpu.copy_(self.w_cpu_local)
             self.w_gpu.copy_(self.w_gpu_local)
 
-        # routing (K2) over allowed jobs
+        # routing (K2): per-worker key precompute, then wave-per-job argmin
+        w_keys = self.ext.worker_precompute(
+            self.w_pool, self.w_active, self.w_maxp, self.w_cpu, self.w_gpu
+        )
         pick = self.ext.least_loaded_pick(
-            self.w_pool, self.w_active, self.w_maxp, self.w_cpu, self.w_gpu,
-            self.w_labels, self.j_poolmask, self.j_labels,
+            self.w_pool, w_keys, self.w_labels, self.j_poolmask, self.j_labels,
         )
         routable = allowed & (pick >= 0)
 
@@ -276,13 +287,15 @@ class DevicePipeline:
                 self.ext.apply_transitions(self.states, self.attempts, self.deadlines, r_slots, ts)
 
         # cross-GPU dispatch: group routable jobs by destination rank
-        dest = (pick[routable] // self.NWL).to(torch.int64)
-        order = torch.argsort(dest, stable=True)
-        send_slots = r_slots[order]
-        send_pick = pick[routable][order]
-        send_payload = payload.view(B, self.payload_words)[send_slots.long()].reshape(-1)
-        send_widx = (send_pick % self.NWL).to(torch.int32)
+        # (single-rank: no grouping/copy — the worker pool reads the home
+        # arena in place via the slot-indexed echo kernel)
         if self.world > 1:
+            dest = (pick[routable] // self.NWL).to(torch.int64)
+            order = torch.argsort(dest, stable=True)
+            send_slots = r_slots[order]
+            send_pick = pick[routable][order]
+            send_payload = payload.view(B, self.payload_words)[send_slots.long()].reshape(-1)
+            send_widx = (send_pick % self.NWL).to(torch.int32)
             send_counts = torch.bincount(dest, minlength=self.world)
             recv_counts = torch.empty_like(send_counts)
             dist.all_to_all_single(recv_counts, send_counts)
@@ -300,10 +313,10 @@ class DevicePipeline:
                 [c * self.payload_words for c in sc],
             )
         else:
-            recv_slots = send_slots
-            recv_widx = send_widx
-            recv_payload = send_payload
-            n_recv = int(send_slots.numel())
+            recv_slots = r_slots
+            recv_widx = (pick[routable] % self.NWL).to(torch.int32)
+            recv_payload = None  # in-place indexed execution
+            n_recv = int(r_slots.numel())
 
         # RUNNING on dispatched jobs
         if n_routable:
@@ -311,9 +324,15 @@ class DevicePipeline:
             self.ext.apply_transitions(self.states, self.attempts, self.deadlines, r_slots, ts)
 
         # worker execution: echo (payload copy + checksum) on the receiving rank
-        if n_recv:
+        if n_recv and self.world > 1:
             res = self.res_arena[: n_recv * self.payload_words]
             sums = self.ext.echo_execute(recv_payload, res, self.payload_words)
+        elif n_recv:
+            self.ext.echo_execute_indexed(
+                payload, recv_slots, self.res_arena[: B * self.payload_words],
+                self.res_sums, self.payload_words,
+            )
+            sums = self.res_sums
         else:
             sums = torch.empty(0, dtype=torch.int32, device=dev)
 

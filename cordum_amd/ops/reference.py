@@ -56,6 +56,61 @@ def least_loaded_pick_ref(
     return out
 
 
+def worker_precompute_ref(w_pool, w_active, w_maxp, w_cpu, w_gpu) -> torch.Tensor:
+    """K2a oracle: per-worker (score, idx) key; overloaded -> OVERLOADED tag."""
+    NW = w_pool.shape[0]
+    idx = torch.arange(NW, dtype=torch.int64)
+    util_over = (w_maxp > 0) & (w_active.float() / w_maxp.clamp(min=1).float() >= OVERLOAD)
+    over = util_over | (w_cpu >= 90) | (w_gpu >= 90)
+    score = w_active.float() + w_cpu * 0.01 + w_gpu * 0.01
+    key = (score.view(torch.int32).to(torch.int64) << 32) | idx
+    # 0xFFFFFFFE00000000 as signed int64 (torch has no uint64)
+    over_key = torch.tensor(0xFFFFFFFE00000000 - (1 << 64), dtype=torch.int64) | idx
+    return torch.where(over, over_key, key)
+
+
+def least_loaded_pick_keys_ref(
+    w_pool: torch.Tensor,
+    w_keys: torch.Tensor,
+    w_labels: torch.Tensor,
+    j_poolmask: torch.Tensor,
+    j_labels: torch.Tensor,
+) -> torch.Tensor:
+    """K2 oracle over precomputed keys (must agree with least_loaded_pick_ref)."""
+    NW = w_pool.shape[0]
+    NJ = j_poolmask.shape[0]
+    if NJ == 0:
+        return torch.empty(0, dtype=torch.int32)
+    pool_ok = ((j_poolmask.unsqueeze(1) >> w_pool.clamp(0, 63).unsqueeze(0)) & 1).bool()
+    pool_ok &= (w_pool >= 0).unsqueeze(0) & (w_pool < 64).unsqueeze(0)
+    labels_ok = (j_labels.unsqueeze(1) & ~w_labels.unsqueeze(0)) == 0
+    eligible = pool_ok & labels_ok
+    over = ((w_keys >> 32) & 0xFFFFFFFF) == 0xFFFFFFFE
+    usable = eligible & ~over.unsqueeze(0)
+    big = torch.iinfo(torch.int64).max
+    keys = torch.where(usable, w_keys.unsqueeze(0).expand(NJ, NW), torch.full((NJ, NW), big, dtype=torch.int64))
+    best = keys.min(dim=1).values
+    pick = (best & 0xFFFFFFFF).to(torch.int32)
+    n_eligible = eligible.sum(dim=1)
+    n_over = (eligible & over.unsqueeze(0)).sum(dim=1)
+    none = best == big
+    all_over = none & (n_eligible > 0) & (n_over == n_eligible)
+    out = torch.where(none, torch.full_like(pick, -1), pick)
+    out = torch.where(all_over, torch.full_like(pick, -2), out)
+    return out
+
+
+def echo_execute_indexed_ref(ctx_arena, slots, res_arena, res_sum, stride: int):
+    N = ctx_arena.numel() // stride
+    ctx = ctx_arena.view(N, stride)
+    res = res_arena.view(N, stride)
+    sl = slots.long()
+    res[sl] = ctx[sl]
+    sums = ctx[sl].to(torch.int64).sum(dim=1).remainder(1 << 32).to(torch.int32)
+    res_sum[sl] = sums
+    return res_sum
+
+
 def apply_transitions_ref(
     states: torch.Tensor,
     attempts: torch.Tensor,

@@ -90,9 +90,12 @@ def test_least_loaded_pick_matches_reference(ext):
                            torch.zeros(NJ, dtype=torch.int64))
     want = least_loaded_pick_ref(w_pool, w_active, w_maxp, w_cpu, w_gpu, w_labels, j_poolmask, j_labels)
     d = dev()
+    keys = ext.worker_precompute(w_pool.to(d), w_active.to(d), w_maxp.to(d), w_cpu.to(d), w_gpu.to(d))
+    from cordum_amd.ops.reference import worker_precompute_ref
+
+    assert torch.equal(keys.cpu(), worker_precompute_ref(w_pool, w_active, w_maxp, w_cpu, w_gpu))
     got = ext.least_loaded_pick(
-        w_pool.to(d), w_active.to(d), w_maxp.to(d), w_cpu.to(d), w_gpu.to(d),
-        w_labels.to(d), j_poolmask.to(d), j_labels.to(d),
+        w_pool.to(d), keys, w_labels.to(d), j_poolmask.to(d), j_labels.to(d),
     ).cpu()
     assert torch.equal(got, want)
 
@@ -147,3 +150,21 @@ def test_echo_execute_matches_reference(ext):
     got = ext.echo_execute(ctx_d, res_d, stride).cpu()
     assert torch.equal(res_d.cpu(), res_ref)
     assert torch.equal(got, want)
+
+
+def test_echo_execute_indexed_matches_reference(ext):
+    from cordum_amd.ops.reference import echo_execute_indexed_ref
+
+    g = torch.Generator().manual_seed(21)
+    N, stride = 4096, 64
+    ctx = torch.randint(-(1 << 31), (1 << 31) - 1, (N * stride,), dtype=torch.int32, generator=g)
+    slots = torch.randperm(N, generator=g)[:1500].to(torch.int32)
+    res_ref = torch.zeros_like(ctx)
+    sums_ref = torch.zeros(N, dtype=torch.int32)
+    echo_execute_indexed_ref(ctx, slots, res_ref, sums_ref, stride)
+    d = dev()
+    ctx_d, res_d = ctx.to(d), torch.zeros_like(ctx).to(d)
+    sums_d = torch.zeros(N, dtype=torch.int32, device=d)
+    ext.echo_execute_indexed(ctx_d, slots.to(d), res_d, sums_d, stride)
+    assert torch.equal(res_d.cpu(), res_ref)
+    assert torch.equal(sums_d.cpu(), sums_ref)
