@@ -102,6 +102,14 @@ __global__ __launch_bounds__(BLOCK) void policy_first_match_kernel(
         }
         __syncthreads();
 
+        // cross-chunk early-skip: if another chunk already published a match
+        // below this chunk's range, no rule here can win (atomicMin is
+        // monotonic, so a stale read only costs a wasted scan, never
+        // correctness)
+        if (j < J && best == INT_MAX && base > 0 &&
+            __hip_atomic_load(&out_first[j], __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT) < base) {
+            best = INT_MAX - 1;  // sentinel: done, nothing to publish
+        }
         if (j < J && best == INT_MAX) {
             for (int r = 0; r < n; ++r) {
                 const long long* row = &lds_rules[r * ROW];
@@ -157,7 +165,7 @@ __global__ __launch_bounds__(BLOCK) void policy_first_match_kernel(
         if (block_done) break;
     }
 
-    if (j < J && best != INT_MAX) atomicMin(&out_first[j], best);
+    if (j < J && best < INT_MAX - 1) atomicMin(&out_first[j], best);
 }
 
 // ---------------------------------------------------------------------------
@@ -386,6 +394,112 @@ __global__ __launch_bounds__(BLOCK) void echo_worker_indexed_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// Tick-fusion kernels: device-side compaction + count-pointer variants so a
+// whole single-GPU control-plane tick is a fixed kernel sequence (no host
+// syncs) and can be captured into a hipGraph (guide §Guideline 9).
+// ---------------------------------------------------------------------------
+
+// decisions gather + allow/deny split (replaces 6 torch glue ops)
+__global__ __launch_bounds__(BLOCK) void policy_gate_kernel(
+    const int* __restrict__ first,            // [J] first-match rule (-1 none)
+    const signed char* __restrict__ decisions, // [R]
+    signed char* __restrict__ out_decision,   // [J]
+    int* __restrict__ denied_slots,           // [J] compacted
+    int* __restrict__ denied_count,           // [1]
+    int* __restrict__ allowed_slots,          // [J] compacted
+    int* __restrict__ allowed_count,          // [1]
+    int J)
+{
+    const int j = blockIdx.x * BLOCK + threadIdx.x;
+    if (j >= J) return;
+    const int r = first[j];
+    const signed char d = (r >= 0) ? decisions[r] : (signed char)1;  // default allow
+    out_decision[j] = d;
+    if (d == 1 || d == 5) {
+        allowed_slots[atomicAdd(allowed_count, 1)] = j;
+    } else {
+        denied_slots[atomicAdd(denied_count, 1)] = j;
+    }
+}
+
+// routable compaction: allowed jobs with a worker pick
+__global__ __launch_bounds__(BLOCK) void compact_routable_kernel(
+    const int* __restrict__ allowed_slots,    // [<=J]
+    const int* __restrict__ allowed_count,    // [1]
+    const int* __restrict__ pick,             // [J]
+    int* __restrict__ routable_slots,         // out
+    int* __restrict__ routable_widx,          // out (global worker idx)
+    int* __restrict__ routable_count)         // [1]
+{
+    const int i = blockIdx.x * BLOCK + threadIdx.x;
+    if (i >= *allowed_count) return;
+    const int j = allowed_slots[i];
+    const int w = pick[j];
+    if (w >= 0) {
+        const int pos = atomicAdd(routable_count, 1);
+        routable_slots[pos] = j;
+        routable_widx[pos] = w;
+    }
+}
+
+// K5 with device-resident count + uniform target state
+__global__ __launch_bounds__(BLOCK) void apply_transitions_dyn_kernel(
+    unsigned char* __restrict__ states,
+    int* __restrict__ attempts,
+    long long* __restrict__ deadlines,
+    const int* __restrict__ slots,
+    const int* __restrict__ count,            // [1]
+    unsigned char to)
+{
+    const int i = blockIdx.x * BLOCK + threadIdx.x;
+    if (i >= *count) return;
+    const int slot = slots[i];
+    const unsigned char from = states[slot];
+    if (!d_transition_lut[from * N_STATES + to]) return;
+    states[slot] = to;
+    if (to == 3 && from != 3) attempts[slot] += 1;
+    if (to >= 6) deadlines[slot] = (long long)0x7fffffffffffffffLL;
+}
+
+__global__ __launch_bounds__(BLOCK) void echo_worker_indexed_dyn_kernel(
+    const unsigned int* __restrict__ ctx_arena,
+    const int* __restrict__ slots,
+    const int* __restrict__ count,            // [1]
+    unsigned int* __restrict__ res_arena,
+    unsigned int* __restrict__ res_sum,
+    int stride)
+{
+    const int w = (blockIdx.x * BLOCK + threadIdx.x) / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    if (w >= *count) return;
+    const int slot = slots[w];
+    const size_t basep = (size_t)slot * stride;
+    unsigned int acc = 0;
+    for (int k = lane; k < stride; k += WAVE) {
+        const unsigned int v = ctx_arena[basep + k];
+        res_arena[basep + k] = v;
+        acc += v;
+    }
+    #pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+        acc += __shfl_down(acc, off, WAVE);
+    if (lane == 0) res_sum[slot] = acc;
+}
+
+// per-worker active-count histogram from the routable list
+__global__ __launch_bounds__(BLOCK) void load_feedback_kernel(
+    const int* __restrict__ routable_widx,
+    const int* __restrict__ count,
+    int* __restrict__ w_active_local,         // [NWL] pre-zeroed
+    int nwl, int my_rank)
+{
+    const int i = blockIdx.x * BLOCK + threadIdx.x;
+    if (i >= *count) return;
+    const int w = routable_widx[i];
+    if (w / nwl == my_rank) atomicAdd(&w_active_local[w % nwl], 1);
+}
+
+// ---------------------------------------------------------------------------
 // Torch extension host wrappers
 // ---------------------------------------------------------------------------
 #include <torch/extension.h>
@@ -563,11 +677,72 @@ torch::Tensor echo_execute_indexed(torch::Tensor ctx_arena, torch::Tensor slots,
     return res_sum;
 }
 
+void policy_gate(torch::Tensor first, torch::Tensor decisions, torch::Tensor out_decision,
+                 torch::Tensor denied_slots, torch::Tensor denied_count,
+                 torch::Tensor allowed_slots, torch::Tensor allowed_count)
+{
+    const int J = (int)first.size(0);
+    const int blocks = (J + BLOCK - 1) / BLOCK;
+    hipLaunchKernelGGL(policy_gate_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        first.data_ptr<int>(), (const signed char*)decisions.data_ptr<int8_t>(),
+        (signed char*)out_decision.data_ptr<int8_t>(),
+        denied_slots.data_ptr<int>(), denied_count.data_ptr<int>(),
+        allowed_slots.data_ptr<int>(), allowed_count.data_ptr<int>(), J);
+}
+
+void compact_routable(torch::Tensor allowed_slots, torch::Tensor allowed_count,
+                      torch::Tensor pick, torch::Tensor routable_slots,
+                      torch::Tensor routable_widx, torch::Tensor routable_count)
+{
+    const int J = (int)pick.size(0);
+    const int blocks = (J + BLOCK - 1) / BLOCK;
+    hipLaunchKernelGGL(compact_routable_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        allowed_slots.data_ptr<int>(), allowed_count.data_ptr<int>(),
+        pick.data_ptr<int>(), routable_slots.data_ptr<int>(),
+        routable_widx.data_ptr<int>(), routable_count.data_ptr<int>());
+}
+
+void apply_transitions_dyn(torch::Tensor states, torch::Tensor attempts, torch::Tensor deadlines,
+                           torch::Tensor slots, torch::Tensor count, int64_t to_state, int64_t capacity)
+{
+    const int blocks = ((int)capacity + BLOCK - 1) / BLOCK;
+    hipLaunchKernelGGL(apply_transitions_dyn_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        states.data_ptr<uint8_t>(), attempts.data_ptr<int>(),
+        (long long*)deadlines.data_ptr<int64_t>(),
+        slots.data_ptr<int>(), count.data_ptr<int>(), (unsigned char)to_state);
+}
+
+void echo_execute_indexed_dyn(torch::Tensor ctx_arena, torch::Tensor slots, torch::Tensor count,
+                              torch::Tensor res_arena, torch::Tensor res_sum,
+                              int64_t stride, int64_t capacity)
+{
+    const int waves_per_block = BLOCK / WAVE;
+    const int blocks = ((int)capacity + waves_per_block - 1) / waves_per_block;
+    hipLaunchKernelGGL(echo_worker_indexed_dyn_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        (const unsigned int*)ctx_arena.data_ptr<int32_t>(), slots.data_ptr<int>(),
+        count.data_ptr<int>(), (unsigned int*)res_arena.data_ptr<int32_t>(),
+        (unsigned int*)res_sum.data_ptr<int32_t>(), (int)stride);
+}
+
+void load_feedback(torch::Tensor routable_widx, torch::Tensor count,
+                   torch::Tensor w_active_local, int64_t nwl, int64_t my_rank, int64_t capacity)
+{
+    const int blocks = ((int)capacity + BLOCK - 1) / BLOCK;
+    hipLaunchKernelGGL(load_feedback_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        routable_widx.data_ptr<int>(), count.data_ptr<int>(),
+        w_active_local.data_ptr<int>(), (int)nwl, (int)my_rank);
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("policy_first_match", &policy_first_match, "K1 batched policy first-match");
     m.def("worker_precompute", &worker_precompute, "K2a per-worker score/overload precompute");
     m.def("least_loaded_pick", &least_loaded_pick, "K2 least-loaded worker pick");
     m.def("echo_execute_indexed", &echo_execute_indexed, "device echo worker pool (slot-indexed, in-place)");
+    m.def("policy_gate", &policy_gate, "decision gather + allow/deny compaction");
+    m.def("compact_routable", &compact_routable, "routable-slot compaction");
+    m.def("apply_transitions_dyn", &apply_transitions_dyn, "K5 with device-resident count");
+    m.def("echo_execute_indexed_dyn", &echo_execute_indexed_dyn, "echo worker with device count");
+    m.def("load_feedback", &load_feedback, "per-worker active-count histogram");
     m.def("set_transition_lut", &set_transition_lut, "upload transition legality LUT");
     m.def("apply_transitions", &apply_transitions, "K5 batched state transitions");
     m.def("deadline_scan", &deadline_scan, "K4 deadline/staleness scan");

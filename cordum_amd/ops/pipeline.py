@@ -221,12 +221,127 @@ class DevicePipeline:
         # result arena: a rank can receive up to world*B jobs in one tick
         self.res_arena = torch.zeros(self.B * self.world * payload_words, dtype=torch.int32, device=device)
         self.res_sums = torch.zeros(self.B, dtype=torch.int32, device=device)
+
+        # fused-tick buffers (single-GPU hipGraph path): fixed-capacity
+        # compaction lists + device-resident counts; batch content is copied
+        # into staging tensors so one captured graph serves the whole ring
+        self.out_decision = torch.zeros(self.B, dtype=torch.int8, device=device)
+        self.denied_slots = torch.zeros(self.B, dtype=torch.int32, device=device)
+        self.denied_count = torch.zeros(1, dtype=torch.int32, device=device)
+        self.allowed_slots = torch.zeros(self.B, dtype=torch.int32, device=device)
+        self.allowed_count = torch.zeros(1, dtype=torch.int32, device=device)
+        self.routable_slots = torch.zeros(self.B, dtype=torch.int32, device=device)
+        self.routable_widx = torch.zeros(self.B, dtype=torch.int32, device=device)
+        self.routable_count = torch.zeros(1, dtype=torch.int32, device=device)
+        self._graph = None
+        if device.type == "cuda" and self.world == 1 and backend != "ref":
+            b0 = self.batches[0]
+            self.stage_any = torch.empty_like(b0.any_bits)
+            self.stage_all = torch.empty_like(b0.all_bits)
+            self.stage_secrets = torch.empty_like(b0.secrets)
+            self.stage_mcp = torch.empty_like(b0.mcp_bits)
+            self.stage_mcp_used = torch.empty_like(b0.mcp_used)
+            self.stage_payload = torch.empty_like(self.payloads[0])
         self._tick = 0
         self.total_completed = 0
         self.total_denied = 0
 
+    # -- fused single-GPU tick (hipGraph-captured) --------------------------------
+    def _fused_body(self) -> None:
+        """The whole tick as a fixed kernel sequence over staging buffers —
+        no host decisions, so it is capturable into a hipGraph (the
+        boundary/graph-replay costs in MI355X_MICROARCH.md §price-list are
+        what this amortizes for the launch-bound control-plane tick)."""
+        B = self.B
+        ext = self.ext
+        self.states.zero_()
+        pend = self._pend_states
+        ext.apply_transitions(self.states, self.attempts, self.deadlines, self.slots, pend)
+        first = ext.policy_first_match(
+            self.cpol.any_masks, self.cpol.all_masks, self.cpol.secrets,
+            self.cpol.mcp_masks, self.cpol.mcp_any,
+            self.stage_any, self.stage_all, self.stage_secrets,
+            self.stage_mcp, self.stage_mcp_used, 0,
+        )
+        self.denied_count.zero_()
+        self.allowed_count.zero_()
+        self.routable_count.zero_()
+        ext.policy_gate(first, self.cpol.decisions, self.out_decision,
+                        self.denied_slots, self.denied_count,
+                        self.allowed_slots, self.allowed_count)
+        ext.apply_transitions_dyn(self.states, self.attempts, self.deadlines,
+                                  self.denied_slots, self.denied_count, DENIED, B)
+        self.w_active.copy_(self.w_active_local)
+        self.w_cpu.copy_(self.w_cpu_local)
+        self.w_gpu.copy_(self.w_gpu_local)
+        w_keys = ext.worker_precompute(self.w_pool, self.w_active, self.w_maxp,
+                                       self.w_cpu, self.w_gpu)
+        pick = ext.least_loaded_pick(self.w_pool, w_keys, self.w_labels,
+                                     self.j_poolmask, self.j_labels)
+        ext.compact_routable(self.allowed_slots, self.allowed_count, pick,
+                             self.routable_slots, self.routable_widx, self.routable_count)
+        for st in (SCHEDULED, DISPATCHED, RUNNING):
+            ext.apply_transitions_dyn(self.states, self.attempts, self.deadlines,
+                                      self.routable_slots, self.routable_count, st, B)
+        ext.echo_execute_indexed_dyn(self.stage_payload, self.routable_slots,
+                                     self.routable_count, self.res_arena[: B * self.payload_words],
+                                     self.res_sums, self.payload_words, B)
+        ext.apply_transitions_dyn(self.states, self.attempts, self.deadlines,
+                                  self.routable_slots, self.routable_count, SUCCEEDED, B)
+        self.w_active_local.zero_()
+        ext.load_feedback(self.routable_widx, self.routable_count,
+                          self.w_active_local, self.NWL, self.rank, B)
+
+    def _tick_fused(self) -> TickStats:
+        t0 = time.perf_counter()
+        i = self._tick % len(self.batches)
+        self._tick += 1
+        jb = self.batches[i]
+        self.stage_any.copy_(jb.any_bits, non_blocking=True)
+        self.stage_all.copy_(jb.all_bits, non_blocking=True)
+        self.stage_secrets.copy_(jb.secrets, non_blocking=True)
+        self.stage_mcp.copy_(jb.mcp_bits, non_blocking=True)
+        self.stage_mcp_used.copy_(jb.mcp_used, non_blocking=True)
+        self.stage_payload.copy_(self.payloads[i], non_blocking=True)
+        if self._graph is None:
+            if not hasattr(self, "_pend_states"):
+                self._pend_states = torch.full((self.B,), PENDING, dtype=torch.uint8, device=self.device)
+            # eager warmups on a side stream, then capture
+            side = torch.cuda.Stream(device=self.device)
+            side.wait_stream(torch.cuda.current_stream(self.device))
+            with torch.cuda.stream(side):
+                for _ in range(2):
+                    self._fused_body()
+            torch.cuda.current_stream(self.device).wait_stream(side)
+            torch.cuda.synchronize(self.device)
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                self._fused_body()
+            self._graph = g
+        self._graph.replay()
+        torch.cuda.synchronize(self.device)
+        denied = int(self.denied_count.item())
+        routable = int(self.routable_count.item())
+        dt = time.perf_counter() - t0
+        stats = TickStats(
+            completed=routable,
+            denied=denied,
+            unrouted=self.B - routable - denied,
+            wall_s=dt,
+        )
+        self.total_completed += stats.completed
+        self.total_denied += stats.denied
+        return stats
+
     # -- one control-plane tick -------------------------------------------------
     def tick(self) -> TickStats:
+        if self._graph is not None or (
+            self.device.type == "cuda" and self.world == 1 and hasattr(self, "stage_any")
+        ):
+            return self._tick_fused()
+        return self._tick_eager()
+
+    def _tick_eager(self) -> TickStats:
         t0 = time.perf_counter()
         i = self._tick % len(self.batches)
         self._tick += 1
