@@ -308,6 +308,20 @@ __global__ __launch_bounds__(BLOCK) void apply_transitions_kernel(
     if (to >= 6) deadlines[slot] = (long long)0x7fffffffffffffffLL;
 }
 
+// wave-aggregated compaction append: one atomicAdd per wave instead of one
+// per lane (a single counter word saturates at ~88 atomics/us on this chip —
+// MI355X_MICROARCH.md price-list row "dequeue")
+__device__ __forceinline__ int wave_append_slot(bool pred, int* counter, int lane) {
+    const unsigned long long mask = __ballot(pred);
+    const int nactive = __popcll(mask);
+    int base = 0;
+    const int leader = __ffsll((long long)mask) - 1;
+    if (pred && lane == leader) base = atomicAdd(counter, nactive);
+    base = __shfl(base, leader, WAVE);
+    const int rank = __popcll(mask & ((1ull << lane) - 1ull));
+    return pred ? base + rank : -1;
+}
+
 // ---------------------------------------------------------------------------
 // K4: deadline / staleness scan -> TIMEOUT candidates (compacted list)
 // ---------------------------------------------------------------------------
@@ -328,10 +342,9 @@ __global__ __launch_bounds__(BLOCK) void deadline_scan_kernel(
     if (st >= 1 && st <= 5 && deadlines[i] <= now_us) expired = true;
     if (st == 4 && updated_at[i] <= dispatch_cutoff_us) expired = true;  // DISPATCHED stale
     if (st == 5 && updated_at[i] <= running_cutoff_us) expired = true;   // RUNNING stale
-    if (expired) {
-        const int pos = atomicAdd(out_count, 1);
-        if (pos < cap) out_slots[pos] = i;
-    }
+    const int lane = threadIdx.x % WAVE;
+    const int pos = wave_append_slot(expired, out_count, lane);
+    if (pos >= 0 && pos < cap) out_slots[pos] = i;
 }
 
 // ---------------------------------------------------------------------------
@@ -411,15 +424,20 @@ __global__ __launch_bounds__(BLOCK) void policy_gate_kernel(
     int J)
 {
     const int j = blockIdx.x * BLOCK + threadIdx.x;
-    if (j >= J) return;
-    const int r = first[j];
-    const signed char d = (r >= 0) ? decisions[r] : (signed char)1;  // default allow
-    out_decision[j] = d;
-    if (d == 1 || d == 5) {
-        allowed_slots[atomicAdd(allowed_count, 1)] = j;
-    } else {
-        denied_slots[atomicAdd(denied_count, 1)] = j;
+    const int lane = threadIdx.x % WAVE;
+    const bool live = j < J;
+    signed char d = 0;
+    if (live) {
+        const int r = first[j];
+        d = (r >= 0) ? decisions[r] : (signed char)1;  // default allow
+        out_decision[j] = d;
     }
+    const bool allowed = live && (d == 1 || d == 5);
+    const int apos = wave_append_slot(allowed, allowed_count, lane);
+    if (apos >= 0) allowed_slots[apos] = j;
+    const bool denied = live && !allowed;
+    const int dpos = wave_append_slot(denied, denied_count, lane);
+    if (dpos >= 0) denied_slots[dpos] = j;
 }
 
 // routable compaction: allowed jobs with a worker pick
@@ -432,11 +450,15 @@ __global__ __launch_bounds__(BLOCK) void compact_routable_kernel(
     int* __restrict__ routable_count)         // [1]
 {
     const int i = blockIdx.x * BLOCK + threadIdx.x;
-    if (i >= *allowed_count) return;
-    const int j = allowed_slots[i];
-    const int w = pick[j];
-    if (w >= 0) {
-        const int pos = atomicAdd(routable_count, 1);
+    const int lane = threadIdx.x % WAVE;
+    const bool live = i < *allowed_count;
+    int j = 0, w = -1;
+    if (live) {
+        j = allowed_slots[i];
+        w = pick[j];
+    }
+    const int pos = wave_append_slot(live && w >= 0, routable_count, lane);
+    if (pos >= 0) {
         routable_slots[pos] = j;
         routable_widx[pos] = w;
     }
