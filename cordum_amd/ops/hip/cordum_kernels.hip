@@ -406,6 +406,96 @@ __global__ __launch_bounds__(BLOCK) void echo_worker_indexed_kernel(
     if (lane == 0) res_sum[slot] = acc;
 }
 
+
+// ---------------------------------------------------------------------------
+// K1-MFMA: policy first-match on the matrix cores (comparison variant)
+// ---------------------------------------------------------------------------
+// One v_mfma_i32_16x16x64_i8 per dimension per 16-job x 16-rule tile computes
+// all 256 set-intersection cardinalities of that dim at once. Epilogue turns
+// the 9 dot products into the match predicate and reduces first-match per
+// job across the 16 rule columns with 16-lane shuffles.
+// See ops/policy_mfma.py for the fragment prepack and the measured
+// bitset-vs-MFMA verdict.
+typedef int v4i __attribute__((ext_vector_type(4)));
+
+__global__ __launch_bounds__(BLOCK) void policy_first_match_mfma_kernel(
+    const signed char* __restrict__ a_pack,   // [Jt][9][64][16]
+    const signed char* __restrict__ b_pack,   // [Rt][9][64][16]
+    const int* __restrict__ cards,            // [Rt*16][9]
+    const signed char* __restrict__ rule_secrets, // [Rt*16]
+    const unsigned char* __restrict__ job_secrets, // [J]
+    int* __restrict__ out_first,              // [J] pre-filled INT_MAX
+    int J, int R, int tiles_per_chunk)
+{
+    const int wave = threadIdx.x / WAVE;      // 4 waves per block
+    const int lane = threadIdx.x % WAVE;
+    const int jt = blockIdx.x;                // one job tile per block
+    const int Rt = (R + 15) / 16;
+    const int chunk_begin = blockIdx.y * tiles_per_chunk;
+    const int chunk_end = min(Rt, chunk_begin + tiles_per_chunk);
+
+    // A fragments for the 9 dims: lane-linear 16 B loads
+    v4i afrag[9];
+    #pragma unroll
+    for (int d = 0; d < 9; ++d)
+        afrag[d] = *(const v4i*)&a_pack[(((size_t)jt * 9 + d) * 64 + lane) * 16];
+
+    const int row_base = (lane >> 4) * 4;     // 4 job rows per lane
+    const int col = lane & 15;                // rule column
+    int best[4] = {INT_MAX, INT_MAX, INT_MAX, INT_MAX};
+    unsigned char jsec[4];
+    #pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        const int job = jt * 16 + row_base + r;
+        jsec[r] = (job < J) ? job_secrets[job] : 0;
+    }
+
+    for (int rt = chunk_begin + wave; rt < chunk_end; rt += 4) {
+        v4i acc[9];
+        #pragma unroll
+        for (int d = 0; d < 9; ++d) {
+            const v4i bfrag = *(const v4i*)&b_pack[(((size_t)rt * 9 + d) * 64 + lane) * 16];
+            v4i zero = {0, 0, 0, 0};
+            acc[d] = __builtin_amdgcn_mfma_i32_16x16x64_i8(afrag[d], bfrag, zero, 0, 0, 0);
+        }
+        const int rule = rt * 16 + col;
+        int card[9];
+        #pragma unroll
+        for (int d = 0; d < 9; ++d) card[d] = cards[rule * 9 + d];
+        const signed char rsec = rule_secrets[rule];
+
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            bool ok = true;
+            #pragma unroll
+            for (int d = 0; d < 7; ++d)   // any-of dims
+                ok &= (card[d] == 0) | (acc[d][r] > 0);
+            #pragma unroll
+            for (int d = 7; d < 9; ++d)   // all-of dims
+                ok &= (acc[d][r] == card[d]);
+            ok &= (rsec < 0) | (rsec == (signed char)jsec[r]);
+            if (ok && rule < R) best[r] = min(best[r], rule);
+        }
+    }
+
+    // min across the 16 rule columns (lanes sharing the same lane>>4 group)
+    #pragma unroll
+    for (int off = 8; off > 0; off >>= 1) {
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            const int other = __shfl_xor(best[r], off, WAVE);
+            if ((lane & 15) < 16) best[r] = min(best[r], other);
+        }
+    }
+    if ((lane & 15) == 0) {
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            const int job = jt * 16 + row_base + r;
+            if (job < J && best[r] != INT_MAX) atomicMin(&out_first[job], best[r]);
+        }
+    }
+}
+
 // ---------------------------------------------------------------------------
 // Tick-fusion kernels: device-side compaction + count-pointer variants so a
 // whole single-GPU control-plane tick is a fixed kernel sequence (no host
@@ -699,6 +789,32 @@ torch::Tensor echo_execute_indexed(torch::Tensor ctx_arena, torch::Tensor slots,
     return res_sum;
 }
 
+torch::Tensor policy_first_match_mfma(
+    torch::Tensor a_pack, torch::Tensor b_pack, torch::Tensor cards,
+    torch::Tensor rule_secrets, torch::Tensor job_secrets,
+    int64_t n_jobs, int64_t n_rules)
+{
+    CHECK_DEV(a_pack); CHECK_DEV(b_pack);
+    const int J = (int)n_jobs, R = (int)n_rules;
+    const int Jt = (int)a_pack.size(0);
+    const int Rt = (R + 15) / 16;
+    auto out = torch::full({J}, INT_MAX,
+        torch::TensorOptions().dtype(torch::kInt32).device(a_pack.device()));
+    if (J == 0 || R == 0) return out.masked_fill_(out == INT_MAX, -1);
+    // fill the chip: >=2048 workgroups via rule chunking
+    int nchunks = std::max(1, std::min((Rt + 3) / 4, std::max(1, 2048 / std::max(Jt, 1))));
+    const int tiles_per_chunk = (Rt + nchunks - 1) / nchunks;
+    hipLaunchKernelGGL(policy_first_match_mfma_kernel, dim3(Jt, nchunks), dim3(BLOCK), 0, cur_stream(),
+        (const signed char*)a_pack.data_ptr<int8_t>(),
+        (const signed char*)b_pack.data_ptr<int8_t>(),
+        cards.data_ptr<int>(),
+        (const signed char*)rule_secrets.data_ptr<int8_t>(),
+        job_secrets.data_ptr<uint8_t>(),
+        out.data_ptr<int>(), J, R, tiles_per_chunk);
+    out.masked_fill_(out == INT_MAX, -1);
+    return out;
+}
+
 void policy_gate(torch::Tensor first, torch::Tensor decisions, torch::Tensor out_decision,
                  torch::Tensor denied_slots, torch::Tensor denied_count,
                  torch::Tensor allowed_slots, torch::Tensor allowed_count)
@@ -757,6 +873,7 @@ void load_feedback(torch::Tensor routable_widx, torch::Tensor count,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("policy_first_match", &policy_first_match, "K1 batched policy first-match");
+    m.def("policy_first_match_mfma", &policy_first_match_mfma, "K1 MFMA comparison variant");
     m.def("worker_precompute", &worker_precompute, "K2a per-worker score/overload precompute");
     m.def("least_loaded_pick", &least_loaded_pick, "K2 least-loaded worker pick");
     m.def("echo_execute_indexed", &echo_execute_indexed, "device echo worker pool (slot-indexed, in-place)");

@@ -168,3 +168,22 @@ def test_echo_execute_indexed_matches_reference(ext):
     ext.echo_execute_indexed(ctx_d, slots.to(d), res_d, sums_d, stride)
     assert torch.equal(res_d.cpu(), res_ref)
     assert torch.equal(sums_d.cpu(), sums_ref)
+
+
+def test_policy_mfma_variant_matches_reference(ext):
+    """K1-MFMA must agree with the bitset kernel and the torch oracle
+    (synthetic policies without MCP constraints — the variant's scope)."""
+    from cordum_amd.ops.pipeline import make_synthetic_policy, encode_synthetic_jobs
+    from cordum_amd.ops.policy_mfma import first_match_mfma, pack_jobs_mfma, pack_policy_mfma
+
+    policy = make_synthetic_policy(777, vocab=40, deny_frac=0.05, seed=42)
+    compiled = compile_policy(policy, words=1)
+    assert compiled.exact
+    jobs = encode_synthetic_jobs(compiled, 1000, vocab=40, seed=43)
+    want = first_match_reference(compiled, jobs)
+
+    d = dev()
+    mp = pack_policy_mfma(compiled).to(d)
+    a_pack, job_secrets = pack_jobs_mfma(jobs)
+    got = first_match_mfma(ext, mp, a_pack.to(d), job_secrets.to(d), jobs.n_jobs).cpu()
+    assert torch.equal(got, want)
