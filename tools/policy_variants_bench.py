@@ -4,6 +4,9 @@ several (jobs, rules) shapes. Interleaved timing rounds in one process
 (guide §5.4 rule 24); prints one JSON line per shape."""
 import json
 import sys
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
 import time
 
 
