@@ -234,6 +234,27 @@ class DevicePipeline:
         self.routable_widx = torch.zeros(self.B, dtype=torch.int32, device=device)
         self.routable_count = torch.zeros(1, dtype=torch.int32, device=device)
         self._graph = None
+        # hybrid K1 dispatch (measured, tools/policy_variants_bench.py):
+        # MFMA tiles win up to ~32k rules (occupancy at small R), the bitset
+        # rows win beyond (4.2x smaller rule stream + first-match early-exit)
+        self._use_mfma = (
+            backend != "ref"
+            and device.type == "cuda"
+            and self.compiled.words == 1
+            and self.compiled.n_rules <= 32768
+            and int(self.compiled.mcp_any.sum()) == 0
+        )
+        if self._use_mfma:
+            from .policy_mfma import pack_jobs_mfma, pack_policy_mfma
+
+            self.mfma_policy = pack_policy_mfma(self.compiled).to(device)
+            self.mfma_a_packs = []
+            for jb_host, jb_dev in zip(
+                [encode_synthetic_jobs(self.compiled, self.B, seed=seed * 1000 + i + rank * 77)
+                 for i in range(len(self.batches))], self.batches
+            ):
+                a_pack, _ = pack_jobs_mfma(jb_host)
+                self.mfma_a_packs.append(a_pack.to(device))
         if device.type == "cuda" and self.world == 1 and backend != "ref":
             b0 = self.batches[0]
             self.stage_any = torch.empty_like(b0.any_bits)
@@ -242,6 +263,8 @@ class DevicePipeline:
             self.stage_mcp = torch.empty_like(b0.mcp_bits)
             self.stage_mcp_used = torch.empty_like(b0.mcp_used)
             self.stage_payload = torch.empty_like(self.payloads[0])
+            if self._use_mfma:
+                self.stage_a_pack = torch.empty_like(self.mfma_a_packs[0])
         self._tick = 0
         self.total_completed = 0
         self.total_denied = 0
@@ -257,12 +280,19 @@ class DevicePipeline:
         self.states.zero_()
         pend = self._pend_states
         ext.apply_transitions(self.states, self.attempts, self.deadlines, self.slots, pend)
-        first = ext.policy_first_match(
-            self.cpol.any_masks, self.cpol.all_masks, self.cpol.secrets,
-            self.cpol.mcp_masks, self.cpol.mcp_any,
-            self.stage_any, self.stage_all, self.stage_secrets,
-            self.stage_mcp, self.stage_mcp_used, 0,
-        )
+        if self._use_mfma:
+            first = ext.policy_first_match_mfma(
+                self.stage_a_pack, self.mfma_policy.b_pack, self.mfma_policy.cards,
+                self.mfma_policy.secrets, self.stage_secrets,
+                self.B, self.compiled.n_rules,
+            )
+        else:
+            first = ext.policy_first_match(
+                self.cpol.any_masks, self.cpol.all_masks, self.cpol.secrets,
+                self.cpol.mcp_masks, self.cpol.mcp_any,
+                self.stage_any, self.stage_all, self.stage_secrets,
+                self.stage_mcp, self.stage_mcp_used, 0,
+            )
         self.denied_count.zero_()
         self.allowed_count.zero_()
         self.routable_count.zero_()
@@ -303,6 +333,8 @@ class DevicePipeline:
         self.stage_mcp.copy_(jb.mcp_bits, non_blocking=True)
         self.stage_mcp_used.copy_(jb.mcp_used, non_blocking=True)
         self.stage_payload.copy_(self.payloads[i], non_blocking=True)
+        if self._use_mfma:
+            self.stage_a_pack.copy_(self.mfma_a_packs[i], non_blocking=True)
         if self._graph is None:
             if not hasattr(self, "_pend_states"):
                 self._pend_states = torch.full((self.B,), PENDING, dtype=torch.uint8, device=self.device)
