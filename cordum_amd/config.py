@@ -97,11 +97,44 @@ def load_timeouts(path: str) -> Timeouts:
     )
 
 
-def load_safety_yaml(path: str) -> str:
+def load_safety_yaml(path: str, public_key: str = "", signature: str = "") -> str:
+    """Load the safety policy file; verify its ed25519 signature when a key
+    is configured (oracle: safetykernel bundle load + signature verify,
+    kernel.go:787-868; env SAFETY_POLICY_PUBLIC_KEY / SAFETY_POLICY_SIGNATURE,
+    hex or base64)."""
     p = Path(path)
     if not path or not p.exists():
         return ""
-    return p.read_text()
+    data = p.read_bytes()
+    public_key = public_key or os.environ.get("SAFETY_POLICY_PUBLIC_KEY", "")
+    signature = signature or os.environ.get("SAFETY_POLICY_SIGNATURE", "")
+    if public_key:
+        from .utils.ed25519 import verify
+
+        if not signature:
+            sig_file = Path(str(p) + ".sig")
+            if sig_file.exists():
+                signature = sig_file.read_text().strip()
+        if not signature:
+            raise ValueError("safety policy signature required but missing")
+        if not verify(_decode_key(public_key), _decode_key(signature), data):
+            raise ValueError("safety policy signature verification failed")
+    return data.decode("utf-8")
+
+
+def _decode_key(s: str) -> bytes:
+    import base64
+    import binascii
+
+    s = s.strip()
+    try:
+        return bytes.fromhex(s)
+    except ValueError:
+        pass
+    try:
+        return base64.b64decode(s, validate=True)
+    except (binascii.Error, ValueError):
+        return base64.urlsafe_b64decode(s + "=" * (-len(s) % 4))
 
 
 def seed_system_config(configsvc, path: str) -> None:

@@ -199,3 +199,26 @@ def test_cli_init_and_pack_create(tmp_path):
     r = c.post("/api/v1/packs/install", content=out.read_bytes())
     assert r.status_code == 200, r.text
     assert r.json()["pack_id"] == "hello-pack"
+
+
+# --- ed25519 policy signatures -------------------------------------------------
+
+
+def test_signed_policy_load(tmp_path):
+    from cordum_amd.config import load_safety_yaml
+    from cordum_amd.utils import ed25519
+
+    policy = tmp_path / "safety.yaml"
+    policy.write_text("version: signed\ntenants: {}\n")
+    sk = b"\x01" * 32
+    pk = ed25519.public_key(sk)
+    sig = ed25519.sign(sk, policy.read_bytes())
+    # valid signature
+    text = load_safety_yaml(str(policy), public_key=pk.hex(), signature=sig.hex())
+    assert "signed" in text
+    # tampered content fails
+    policy.write_text("version: tampered\ntenants: {}\n")
+    with pytest.raises(ValueError):
+        load_safety_yaml(str(policy), public_key=pk.hex(), signature=sig.hex())
+    # no key configured -> no verification
+    assert "tampered" in load_safety_yaml(str(policy))
