@@ -322,6 +322,67 @@ __device__ __forceinline__ int wave_append_slot(bool pred, int* counter, int lan
     return pred ? base + rank : -1;
 }
 
+
+// ---------------------------------------------------------------------------
+// K3: run/step readiness sweep over HBM-resident run state
+// ---------------------------------------------------------------------------
+// Batched equivalent of scheduleReady's dependency gate (workflow/engine.go
+// :453-827, depsSatisfied :1231-1242) for runs with <= 64 steps: per run,
+// step s is dispatchable when
+//   step_state[s] == PENDING(0)  AND  deps_mask[s] subset-of succeeded_mask
+// where succeeded_mask bit t = (step_state[t] == SUCCEEDED). Emits a
+// compacted (run, step) dispatch list. One thread per run; 64-step bitmask
+// per run row (larger workflows stay on the host path).
+#define STEP_PENDING_C 0
+#define STEP_SUCCEEDED_C 3
+
+__global__ __launch_bounds__(BLOCK) void run_readiness_kernel(
+    const unsigned char* __restrict__ step_state, // [NR][64]
+    const long long* __restrict__ deps_mask,      // [NR][64]
+    const unsigned char* __restrict__ n_steps,    // [NR]
+    const unsigned char* __restrict__ run_active, // [NR] 1 = pending/running
+    long long* __restrict__ ready_mask_out,       // [NR]
+    int* __restrict__ dispatch_runs,              // [cap]
+    int* __restrict__ dispatch_steps,             // [cap]
+    int* __restrict__ dispatch_count,             // [1]
+    int NR, int cap)
+{
+    const int run = blockIdx.x * BLOCK + threadIdx.x;
+    const int lane = threadIdx.x % WAVE;
+    long long ready = 0;
+    int ns = 0;
+    if (run < NR && run_active[run]) {
+        ns = n_steps[run];
+        const unsigned char* st = &step_state[(size_t)run * 64];
+        unsigned long long succeeded = 0;
+        #pragma unroll 8
+        for (int t = 0; t < ns; ++t)
+            succeeded |= (unsigned long long)(st[t] == STEP_SUCCEEDED_C) << t;
+        const long long* dm = &deps_mask[(size_t)run * 64];
+        for (int s = 0; s < ns; ++s) {
+            if (st[s] != STEP_PENDING_C) continue;
+            const unsigned long long need = (unsigned long long)dm[s];
+            if ((need & ~succeeded) == 0) ready |= 1ll << s;
+        }
+    }
+    if (run < NR) ready_mask_out[run] = ready;
+    // compact the dispatch list, one wave-aggregated append per ready step
+    while (true) {
+        const bool has = ready != 0;
+        if (!__any(has)) break;
+        int step = -1;
+        if (has) {
+            step = __ffsll(ready) - 1;
+            ready &= ready - 1;
+        }
+        const int pos = wave_append_slot(step >= 0, dispatch_count, lane);
+        if (pos >= 0 && pos < cap) {
+            dispatch_runs[pos] = run;
+            dispatch_steps[pos] = step;
+        }
+    }
+}
+
 // ---------------------------------------------------------------------------
 // K4: deadline / staleness scan -> TIMEOUT candidates (compacted list)
 // ---------------------------------------------------------------------------
@@ -871,6 +932,29 @@ void load_feedback(torch::Tensor routable_widx, torch::Tensor count,
         w_active_local.data_ptr<int>(), (int)nwl, (int)my_rank);
 }
 
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor> run_readiness(
+    torch::Tensor step_state, torch::Tensor deps_mask, torch::Tensor n_steps,
+    torch::Tensor run_active, int64_t cap)
+{
+    CHECK_DEV(step_state);
+    const int NR = (int)step_state.size(0);
+    auto opts_i = torch::TensorOptions().dtype(torch::kInt32).device(step_state.device());
+    auto ready = torch::zeros({NR}, torch::TensorOptions().dtype(torch::kInt64).device(step_state.device()));
+    auto runs = torch::empty({cap}, opts_i);
+    auto steps = torch::empty({cap}, opts_i);
+    auto count = torch::zeros({1}, opts_i);
+    if (NR == 0) return {ready, runs, steps, count};
+    const int blocks = (NR + BLOCK - 1) / BLOCK;
+    hipLaunchKernelGGL(run_readiness_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        step_state.data_ptr<uint8_t>(),
+        (const long long*)deps_mask.data_ptr<int64_t>(),
+        n_steps.data_ptr<uint8_t>(), run_active.data_ptr<uint8_t>(),
+        (long long*)ready.data_ptr<int64_t>(),
+        runs.data_ptr<int>(), steps.data_ptr<int>(), count.data_ptr<int>(),
+        NR, (int)cap);
+    return {ready, runs, steps, count};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("policy_first_match", &policy_first_match, "K1 batched policy first-match");
     m.def("policy_first_match_mfma", &policy_first_match_mfma, "K1 MFMA comparison variant");
@@ -882,6 +966,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("apply_transitions_dyn", &apply_transitions_dyn, "K5 with device-resident count");
     m.def("echo_execute_indexed_dyn", &echo_execute_indexed_dyn, "echo worker with device count");
     m.def("load_feedback", &load_feedback, "per-worker active-count histogram");
+    m.def("run_readiness", &run_readiness, "K3 run/step readiness sweep");
     m.def("set_transition_lut", &set_transition_lut, "upload transition legality LUT");
     m.def("apply_transitions", &apply_transitions, "K5 batched state transitions");
     m.def("deadline_scan", &deadline_scan, "K4 deadline/staleness scan");

@@ -159,3 +159,34 @@ def echo_execute_ref(ctx_arena: torch.Tensor, res_arena: torch.Tensor, stride: i
     ctx = ctx_arena.view(B, stride)
     res_arena.view(B, stride).copy_(ctx)
     return ctx.to(torch.int64).sum(dim=1).remainder(1 << 32).to(torch.int32)
+
+
+# device run-table step-state encoding (K3): 0 pending, 1 running, 2 waiting,
+# 3 succeeded, 4 failed, 5 cancelled, 6 timed_out
+RUN_STEP_PENDING, RUN_STEP_SUCCEEDED = 0, 3
+
+
+def run_readiness_ref(step_state, deps_mask, n_steps, run_active):
+    """K3 oracle: per-run 64-bit ready mask + (run, step) dispatch pairs."""
+    NR = step_state.shape[0]
+    ready = torch.zeros(NR, dtype=torch.int64)
+    pairs = []
+    for r in range(NR):
+        if not bool(run_active[r]):
+            continue
+        ns = int(n_steps[r])
+        succeeded = 0
+        for t in range(ns):
+            if int(step_state[r, t]) == RUN_STEP_SUCCEEDED:
+                succeeded |= 1 << t
+        m = 0
+        for s_i in range(ns):
+            if int(step_state[r, s_i]) != RUN_STEP_PENDING:
+                continue
+            need = int(deps_mask[r, s_i]) & ((1 << 64) - 1)
+            if need & ~succeeded & ((1 << 64) - 1):
+                continue
+            m |= 1 << s_i
+            pairs.append((r, s_i))
+        ready[r] = m - (1 << 64) if m >= (1 << 63) else m
+    return ready, pairs

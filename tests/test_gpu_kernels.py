@@ -187,3 +187,29 @@ def test_policy_mfma_variant_matches_reference(ext):
     a_pack, job_secrets = pack_jobs_mfma(jobs)
     got = first_match_mfma(ext, mp, a_pack.to(d), job_secrets.to(d), jobs.n_jobs).cpu()
     assert torch.equal(got, want)
+
+
+def test_run_readiness_matches_reference(ext):
+    from cordum_amd.ops.reference import run_readiness_ref
+
+    g = torch.Generator().manual_seed(31)
+    NR = 3000
+    n_steps = torch.randint(1, 17, (NR,), dtype=torch.uint8, generator=g)
+    step_state = torch.randint(0, 7, (NR, 64), dtype=torch.uint8, generator=g)
+    deps_mask = torch.zeros(NR, 64, dtype=torch.int64)
+    for r in range(NR):
+        ns = int(n_steps[r])
+        for s_i in range(ns):
+            if s_i > 0 and torch.rand(1, generator=g).item() < 0.6:
+                k = torch.randint(0, s_i, (min(3, s_i),), generator=g)
+                for d_ in k.tolist():
+                    deps_mask[r, s_i] |= 1 << d_
+    run_active = (torch.rand(NR, generator=g) < 0.8).to(torch.uint8)
+    want_ready, want_pairs = run_readiness_ref(step_state, deps_mask, n_steps, run_active)
+    d = dev()
+    ready, runs, steps, count = ext.run_readiness(
+        step_state.to(d), deps_mask.to(d), n_steps.to(d), run_active.to(d), NR * 64)
+    assert torch.equal(ready.cpu(), want_ready)
+    n = int(count.cpu()[0])
+    got_pairs = sorted(zip(runs[:n].cpu().tolist(), steps[:n].cpu().tolist()))
+    assert got_pairs == sorted(want_pairs)
