@@ -107,6 +107,15 @@ def main() -> int:
             dlq_head.add_(n)
     t_dlq = timed(dlq)
 
+    # K3 readiness sweep over 1M runs (16-step workflows, dep chains)
+    g2 = torch.Generator().manual_seed(7)
+    step_state = torch.randint(0, 5, (N, 64), dtype=torch.uint8, generator=g2).to(dev)
+    deps_mask = torch.zeros(N, 64, dtype=torch.int64, device=dev)
+    deps_mask[:, 1:16] = (1 << torch.arange(15, device=dev)).view(1, 15)  # chain deps
+    n_steps_t = torch.full((N,), 16, dtype=torch.uint8, device=dev)
+    run_active = torch.ones(N, dtype=torch.uint8, device=dev)
+    t_ready = timed(lambda: ext.run_readiness(step_state, deps_mask, n_steps_t, run_active, N))
+
     # payload touch at scale (1/16 of the arena per "tick")
     window = torch.arange(0, N, 16, dtype=torch.int32, device=dev)
     t_echo = timed(lambda: ext.echo_execute_indexed(payload, window, res_arena, res_sums, W))
@@ -129,6 +138,7 @@ def main() -> int:
         "retry_wave_ms_125k": round(t_retry * 1e3, 3),
         "dlq_append_ms": round(t_dlq * 1e3, 3),
         "echo_62k_jobs_ms": round(t_echo * 1e3, 3),
+        "readiness_sweep_1M_runs_ms": round(t_ready * 1e3, 3),
     }
     print(json.dumps(out))
     return 0
