@@ -666,10 +666,27 @@ __global__ __launch_bounds__(BLOCK) void load_feedback_kernel(
     int* __restrict__ w_active_local,         // [NWL] pre-zeroed
     int nwl, int my_rank)
 {
+    // wave match-any aggregation: a frozen-snapshot least-loaded pick is
+    // highly concentrated (often ONE worker for a whole homogeneous batch),
+    // so per-lane atomics serialize on a single word (~88 atomics/us).
+    // Group lanes by equal widx and issue one atomicAdd per distinct value.
     const int i = blockIdx.x * BLOCK + threadIdx.x;
-    if (i >= *count) return;
-    const int w = routable_widx[i];
-    if (w / nwl == my_rank) atomicAdd(&w_active_local[w % nwl], 1);
+    const int lane = threadIdx.x % WAVE;
+    bool live = i < *count;
+    int bin = -1;
+    if (live) {
+        const int w = routable_widx[i];
+        if (w / nwl == my_rank) bin = w % nwl;
+        else live = false;
+    }
+    unsigned long long pending = __ballot(live);
+    while (pending) {
+        const int leader = __ffsll((long long)pending) - 1;
+        const int leader_bin = __shfl(bin, leader, WAVE);
+        const unsigned long long same = __ballot(live && bin == leader_bin);
+        if (lane == leader) atomicAdd(&w_active_local[leader_bin], __popcll(same));
+        pending &= ~same;
+    }
 }
 
 // ---------------------------------------------------------------------------
