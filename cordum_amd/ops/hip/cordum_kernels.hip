@@ -689,6 +689,140 @@ __global__ __launch_bounds__(BLOCK) void load_feedback_kernel(
     }
 }
 
+
+// ---------------------------------------------------------------------------
+// Padded cross-rank dispatch (multi-GPU path): fixed-capacity per-destination
+// segments so the all_to_all shapes are static and the tick needs NO host
+// splits sync. Capacity per destination = B (worst case: the whole batch to
+// one rank); validity travels as a device-resident count vector.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(BLOCK) void pack_by_dest_kernel(
+    const int* __restrict__ routable_slots,   // [B]
+    const int* __restrict__ routable_widx,    // [B] global worker idx
+    const int* __restrict__ routable_count,   // [1]
+    int* __restrict__ send_slots,             // [world*B]
+    int* __restrict__ send_widx,              // [world*B] local worker idx
+    int* __restrict__ send_cnt,               // [world] pre-zeroed
+    int nwl, int cap)
+{
+    const int i = blockIdx.x * BLOCK + threadIdx.x;
+    const int lane = threadIdx.x % WAVE;
+    bool live = i < *routable_count;
+    int slot = 0, widx = 0, dest = -1;
+    if (live) {
+        slot = routable_slots[i];
+        widx = routable_widx[i];
+        dest = widx / nwl;
+    }
+    // match-any aggregation on dest: one atomicAdd per distinct dest per wave
+    unsigned long long pending = __ballot(live);
+    int my_pos = -1;
+    while (pending) {
+        const int leader = __ffsll((long long)pending) - 1;
+        const int leader_dest = __shfl(dest, leader, WAVE);
+        const unsigned long long same = __ballot(live && dest == leader_dest);
+        int base = 0;
+        if (lane == leader) base = atomicAdd(&send_cnt[leader_dest], __popcll(same));
+        base = __shfl(base, leader, WAVE);
+        if (live && dest == leader_dest) {
+            my_pos = base + __popcll(same & ((1ull << lane) - 1ull));
+            live = false;
+        }
+        pending &= ~same;
+    }
+    if (my_pos >= 0 && my_pos < cap) {
+        send_slots[(size_t)dest * cap + my_pos] = slot;
+        send_widx[(size_t)dest * cap + my_pos] = widx % nwl;
+    }
+}
+
+// gather payload rows into the padded send arena, region-valid
+__global__ __launch_bounds__(BLOCK) void gather_payload_padded_kernel(
+    const unsigned int* __restrict__ payload,  // [B*stride] home arena
+    const int* __restrict__ send_slots,        // [world*cap]
+    const int* __restrict__ send_cnt,          // [world]
+    unsigned int* __restrict__ send_payload,   // [world*cap*stride]
+    int stride, int cap, int world)
+{
+    const int w = (blockIdx.x * BLOCK + threadIdx.x) / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    if (w >= world * cap) return;
+    const int region = w / cap, e = w % cap;
+    if (e >= send_cnt[region]) return;
+    const size_t src = (size_t)send_slots[w] * stride;
+    const size_t dst = (size_t)w * stride;
+    for (int k = lane; k < stride; k += WAVE)
+        send_payload[dst + k] = payload[src + k];
+}
+
+// region-valid echo over the padded receive arena
+__global__ __launch_bounds__(BLOCK) void echo_padded_kernel(
+    const unsigned int* __restrict__ recv_payload, // [world*cap*stride]
+    const int* __restrict__ recv_cnt,              // [world]
+    unsigned int* __restrict__ res_arena,          // [world*cap*stride]
+    unsigned int* __restrict__ res_sums,           // [world*cap]
+    int stride, int cap, int world)
+{
+    const int w = (blockIdx.x * BLOCK + threadIdx.x) / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    if (w >= world * cap) return;
+    const int region = w / cap, e = w % cap;
+    if (e >= recv_cnt[region]) return;
+    const size_t basep = (size_t)w * stride;
+    unsigned int acc = 0;
+    for (int k = lane; k < stride; k += WAVE) {
+        const unsigned int v = recv_payload[basep + k];
+        res_arena[basep + k] = v;
+        acc += v;
+    }
+    #pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+        acc += __shfl_down(acc, off, WAVE);
+    if (lane == 0) res_sums[w] = acc;
+}
+
+// region-valid transitions over the padded slot list
+__global__ __launch_bounds__(BLOCK) void apply_transitions_padded_kernel(
+    unsigned char* __restrict__ states,
+    int* __restrict__ attempts,
+    long long* __restrict__ deadlines,
+    const int* __restrict__ slots_pad,        // [world*cap]
+    const int* __restrict__ cnt,              // [world]
+    unsigned char to, int cap, int world)
+{
+    const int i = blockIdx.x * BLOCK + threadIdx.x;
+    if (i >= world * cap) return;
+    if ((i % cap) >= cnt[i / cap]) return;
+    const int slot = slots_pad[i];
+    const unsigned char from = states[slot];
+    if (!d_transition_lut[from * N_STATES + to]) return;
+    states[slot] = to;
+    if (to == 3 && from != 3) attempts[slot] += 1;
+    if (to >= 6) deadlines[slot] = (long long)0x7fffffffffffffffLL;
+}
+
+// region-valid per-worker load histogram from padded recv widx
+__global__ __launch_bounds__(BLOCK) void load_feedback_padded_kernel(
+    const int* __restrict__ recv_widx,        // [world*cap] local worker idx
+    const int* __restrict__ recv_cnt,         // [world]
+    int* __restrict__ w_active_local,         // [NWL] pre-zeroed
+    int cap, int world)
+{
+    const int i = blockIdx.x * BLOCK + threadIdx.x;
+    const int lane = threadIdx.x % WAVE;
+    bool live = (i < world * cap) && ((i % cap) < recv_cnt[i / cap]);
+    int bin = live ? recv_widx[i] : -1;
+    unsigned long long pending = __ballot(live);
+    while (pending) {
+        const int leader = __ffsll((long long)pending) - 1;
+        const int leader_bin = __shfl(bin, leader, WAVE);
+        const unsigned long long same = __ballot(live && bin == leader_bin);
+        if (lane == leader) atomicAdd(&w_active_local[leader_bin], __popcll(same));
+        pending &= ~same;
+        live = live && bin != leader_bin;
+    }
+}
+
 // ---------------------------------------------------------------------------
 // Torch extension host wrappers
 // ---------------------------------------------------------------------------
@@ -972,6 +1106,65 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor> run_readi
     return {ready, runs, steps, count};
 }
 
+
+void pack_by_dest(torch::Tensor routable_slots, torch::Tensor routable_widx,
+                  torch::Tensor routable_count, torch::Tensor send_slots,
+                  torch::Tensor send_widx, torch::Tensor send_cnt,
+                  int64_t nwl, int64_t cap, int64_t capacity)
+{
+    const int blocks = ((int)capacity + BLOCK - 1) / BLOCK;
+    hipLaunchKernelGGL(pack_by_dest_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        routable_slots.data_ptr<int>(), routable_widx.data_ptr<int>(),
+        routable_count.data_ptr<int>(), send_slots.data_ptr<int>(),
+        send_widx.data_ptr<int>(), send_cnt.data_ptr<int>(), (int)nwl, (int)cap);
+}
+
+void gather_payload_padded(torch::Tensor payload, torch::Tensor send_slots,
+                           torch::Tensor send_cnt, torch::Tensor send_payload,
+                           int64_t stride, int64_t cap, int64_t world)
+{
+    const int waves = (int)(world * cap);
+    const int blocks = (waves + (BLOCK / WAVE) - 1) / (BLOCK / WAVE);
+    hipLaunchKernelGGL(gather_payload_padded_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        (const unsigned int*)payload.data_ptr<int32_t>(), send_slots.data_ptr<int>(),
+        send_cnt.data_ptr<int>(), (unsigned int*)send_payload.data_ptr<int32_t>(),
+        (int)stride, (int)cap, (int)world);
+}
+
+void echo_padded(torch::Tensor recv_payload, torch::Tensor recv_cnt,
+                 torch::Tensor res_arena, torch::Tensor res_sums,
+                 int64_t stride, int64_t cap, int64_t world)
+{
+    const int waves = (int)(world * cap);
+    const int blocks = (waves + (BLOCK / WAVE) - 1) / (BLOCK / WAVE);
+    hipLaunchKernelGGL(echo_padded_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        (const unsigned int*)recv_payload.data_ptr<int32_t>(), recv_cnt.data_ptr<int>(),
+        (unsigned int*)res_arena.data_ptr<int32_t>(),
+        (unsigned int*)res_sums.data_ptr<int32_t>(), (int)stride, (int)cap, (int)world);
+}
+
+void apply_transitions_padded(torch::Tensor states, torch::Tensor attempts,
+                              torch::Tensor deadlines, torch::Tensor slots_pad,
+                              torch::Tensor cnt, int64_t to_state, int64_t cap, int64_t world)
+{
+    const int n = (int)(world * cap);
+    const int blocks = (n + BLOCK - 1) / BLOCK;
+    hipLaunchKernelGGL(apply_transitions_padded_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        states.data_ptr<uint8_t>(), attempts.data_ptr<int>(),
+        (long long*)deadlines.data_ptr<int64_t>(), slots_pad.data_ptr<int>(),
+        cnt.data_ptr<int>(), (unsigned char)to_state, (int)cap, (int)world);
+}
+
+void load_feedback_padded(torch::Tensor recv_widx, torch::Tensor recv_cnt,
+                          torch::Tensor w_active_local, int64_t cap, int64_t world)
+{
+    const int n = (int)(world * cap);
+    const int blocks = (n + BLOCK - 1) / BLOCK;
+    hipLaunchKernelGGL(load_feedback_padded_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        recv_widx.data_ptr<int>(), recv_cnt.data_ptr<int>(),
+        w_active_local.data_ptr<int>(), (int)cap, (int)world);
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("policy_first_match", &policy_first_match, "K1 batched policy first-match");
     m.def("policy_first_match_mfma", &policy_first_match_mfma, "K1 MFMA comparison variant");
@@ -984,6 +1177,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("echo_execute_indexed_dyn", &echo_execute_indexed_dyn, "echo worker with device count");
     m.def("load_feedback", &load_feedback, "per-worker active-count histogram");
     m.def("run_readiness", &run_readiness, "K3 run/step readiness sweep");
+    m.def("pack_by_dest", &pack_by_dest, "padded per-destination dispatch pack");
+    m.def("gather_payload_padded", &gather_payload_padded, "payload gather into padded send arena");
+    m.def("echo_padded", &echo_padded, "region-valid echo over padded recv arena");
+    m.def("apply_transitions_padded", &apply_transitions_padded, "K5 over padded slot list");
+    m.def("load_feedback_padded", &load_feedback_padded, "padded per-worker load histogram");
     m.def("set_transition_lut", &set_transition_lut, "upload transition legality LUT");
     m.def("apply_transitions", &apply_transitions, "K5 batched state transitions");
     m.def("deadline_scan", &deadline_scan, "K4 deadline/staleness scan");

@@ -134,6 +134,61 @@ class _RefOps:
 
         return echo_execute_indexed_ref(ctx_arena, slots, res_arena, res_sum, int(stride))
 
+    def apply_transitions_dyn(self, states, attempts, deadlines, slots, count, to_state, capacity):
+        from .reference import apply_transitions_ref
+
+        n = int(count[0])
+        to = torch.full((n,), int(to_state), dtype=torch.uint8)
+        apply_transitions_ref(states, attempts, deadlines, slots[:n], to)
+
+    def pack_by_dest(self, rs, rw, rc, ss, sw, sc, nwl, cap, capacity):
+        from .reference import pack_by_dest_ref
+
+        pack_by_dest_ref(rs, rw, rc, ss, sw, sc, int(nwl), int(cap))
+
+    def gather_payload_padded(self, payload, ss, sc, sp, stride, cap, world):
+        from .reference import gather_payload_padded_ref
+
+        gather_payload_padded_ref(payload, ss, sc, sp, int(stride), int(cap), int(world))
+
+    def echo_padded(self, rp, rc, ra, rsums, stride, cap, world):
+        from .reference import echo_padded_ref
+
+        echo_padded_ref(rp, rc, ra, rsums, int(stride), int(cap), int(world))
+
+    def apply_transitions_padded(self, states, attempts, deadlines, sp, cnt, to, cap, world):
+        from .reference import apply_transitions_padded_ref
+
+        apply_transitions_padded_ref(states, attempts, deadlines, sp, cnt, int(to), int(cap), int(world))
+
+    def load_feedback_padded(self, rw, rc, wal, cap, world):
+        from .reference import load_feedback_padded_ref
+
+        load_feedback_padded_ref(rw, rc, wal, int(cap), int(world))
+
+    def policy_gate(self, first, decisions, out_decision, ds, dc, als, alc):
+        J = first.shape[0]
+        dec = torch.where(first >= 0, decisions.to(torch.int32)[first.clamp(min=0).long()],
+                          torch.ones_like(first))
+        out_decision.copy_(dec.to(torch.int8))
+        allowed = (dec == 1) | (dec == 5)
+        a_idx = torch.nonzero(allowed).flatten().to(torch.int32)
+        d_idx = torch.nonzero(~allowed).flatten().to(torch.int32)
+        als[: a_idx.numel()] = a_idx
+        ds[: d_idx.numel()] = d_idx
+        alc[0] = a_idx.numel()
+        dc[0] = d_idx.numel()
+
+    def compact_routable(self, als, alc, pick, rs, rw, rc):
+        n = int(alc[0])
+        sel = als[:n].long()
+        p = pick[sel]
+        ok = p >= 0
+        slots = als[:n][ok]
+        rs[: slots.numel()] = slots
+        rw[: slots.numel()] = p[ok]
+        rc[0] = slots.numel()
+
     def apply_transitions(self, states, attempts, deadlines, slots, to_states):
         from .reference import apply_transitions_ref
 
@@ -371,7 +426,103 @@ class DevicePipeline:
             self.device.type == "cuda" and self.world == 1 and hasattr(self, "stage_any")
         ):
             return self._tick_fused()
+        if self.world > 1:
+            return self._tick_padded()
         return self._tick_eager()
+
+    # -- multi-rank tick: fixed-capacity padded all-to-all ------------------------
+    def _tick_padded(self) -> TickStats:
+        """Cross-GPU dispatch with static shapes: per-destination segments of
+        capacity B, counts as a device-resident vector exchanged with the
+        payload — no host splits sync anywhere in the tick (the one
+        `.item()` at the end is the stats read the bench needs anyway)."""
+        t0 = time.perf_counter()
+        i = self._tick % len(self.batches)
+        self._tick += 1
+        jb = self.batches[i]
+        payload = self.payloads[i]
+        B, W, world, dev = self.B, self.payload_words, self.world, self.device
+        ext = self.ext
+
+        if not hasattr(self, "pad_send_slots"):
+            def zi(n):
+                return torch.zeros(n, dtype=torch.int32, device=dev)
+
+            self.pad_send_slots = zi(world * B)
+            self.pad_send_widx = zi(world * B)
+            self.pad_send_cnt = zi(world)
+            self.pad_recv_cnt = zi(world)
+            self.pad_recv_widx = zi(world * B)
+            self.pad_send_payload = torch.zeros(world * B * W, dtype=torch.int32, device=dev)
+            self.pad_recv_payload = torch.zeros(world * B * W, dtype=torch.int32, device=dev)
+            self.pad_res = torch.zeros(world * B * W, dtype=torch.int32, device=dev)
+            self.pad_sums = zi(world * B)
+            self.pad_sums_back = zi(world * B)
+            self._pend_states = torch.full((B,), PENDING, dtype=torch.uint8, device=dev)
+
+        # submit
+        self.states.zero_()
+        ext.apply_transitions(self.states, self.attempts, self.deadlines, self.slots, self._pend_states)
+        # safety gate (bitset path; the hybrid MFMA path is single-GPU fused)
+        first = ext.policy_first_match(
+            self.cpol.any_masks, self.cpol.all_masks, self.cpol.secrets,
+            self.cpol.mcp_masks, self.cpol.mcp_any,
+            jb.any_bits, jb.all_bits, jb.secrets, jb.mcp_bits, jb.mcp_used, 0,
+        )
+        self.denied_count.zero_()
+        self.allowed_count.zero_()
+        self.routable_count.zero_()
+        ext.policy_gate(first, self.cpol.decisions, self.out_decision,
+                        self.denied_slots, self.denied_count,
+                        self.allowed_slots, self.allowed_count)
+        ext.apply_transitions_dyn(self.states, self.attempts, self.deadlines,
+                                  self.denied_slots, self.denied_count, DENIED, B)
+        # heartbeat all-gather
+        dist.all_gather_into_tensor(self.w_active, self.w_active_local)
+        dist.all_gather_into_tensor(self.w_cpu, self.w_cpu_local)
+        dist.all_gather_into_tensor(self.w_gpu, self.w_gpu_local)
+        # routing
+        w_keys = ext.worker_precompute(self.w_pool, self.w_active, self.w_maxp,
+                                       self.w_cpu, self.w_gpu)
+        pick = ext.least_loaded_pick(self.w_pool, w_keys, self.w_labels,
+                                     self.j_poolmask, self.j_labels)
+        ext.compact_routable(self.allowed_slots, self.allowed_count, pick,
+                             self.routable_slots, self.routable_widx, self.routable_count)
+        for st in (SCHEDULED, DISPATCHED):
+            ext.apply_transitions_dyn(self.states, self.attempts, self.deadlines,
+                                      self.routable_slots, self.routable_count, st, B)
+        # padded pack + exchange
+        self.pad_send_cnt.zero_()
+        ext.pack_by_dest(self.routable_slots, self.routable_widx, self.routable_count,
+                         self.pad_send_slots, self.pad_send_widx, self.pad_send_cnt,
+                         self.NWL, B, B)
+        ext.gather_payload_padded(payload, self.pad_send_slots, self.pad_send_cnt,
+                                  self.pad_send_payload, W, B, world)
+        dist.all_to_all_single(self.pad_recv_cnt, self.pad_send_cnt)
+        dist.all_to_all_single(self.pad_recv_widx, self.pad_send_widx)
+        dist.all_to_all_single(self.pad_recv_payload, self.pad_send_payload)
+        ext.apply_transitions_padded(self.states, self.attempts, self.deadlines,
+                                     self.pad_send_slots, self.pad_send_cnt, RUNNING, B, world)
+        # worker execution on the receiving rank + result return
+        ext.echo_padded(self.pad_recv_payload, self.pad_recv_cnt, self.pad_res,
+                        self.pad_sums, W, B, world)
+        dist.all_to_all_single(self.pad_sums_back, self.pad_sums)
+        ext.apply_transitions_padded(self.states, self.attempts, self.deadlines,
+                                     self.pad_send_slots, self.pad_send_cnt, SUCCEEDED, B, world)
+        # load feedback from the receive side
+        self.w_active_local.zero_()
+        ext.load_feedback_padded(self.pad_recv_widx, self.pad_recv_cnt,
+                                 self.w_active_local, B, world)
+        if dev.type == "cuda":
+            torch.cuda.synchronize(dev)
+        denied = int(self.denied_count.cpu()[0])
+        routable = int(self.routable_count.cpu()[0])
+        dt = time.perf_counter() - t0
+        stats = TickStats(completed=routable, denied=denied,
+                          unrouted=B - routable - denied, wall_s=dt)
+        self.total_completed += stats.completed
+        self.total_denied += stats.denied
+        return stats
 
     def _tick_eager(self) -> TickStats:
         t0 = time.perf_counter()

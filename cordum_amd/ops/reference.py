@@ -190,3 +190,66 @@ def run_readiness_ref(step_state, deps_mask, n_steps, run_active):
             pairs.append((r, s_i))
         ready[r] = m - (1 << 64) if m >= (1 << 63) else m
     return ready, pairs
+
+
+# -- padded cross-rank dispatch references ------------------------------------
+
+
+def pack_by_dest_ref(routable_slots, routable_widx, routable_count,
+                     send_slots, send_widx, send_cnt, nwl: int, cap: int):
+    n = int(routable_count[0])
+    for i in range(n):
+        slot = int(routable_slots[i])
+        widx = int(routable_widx[i])
+        dest = widx // nwl
+        pos = int(send_cnt[dest])
+        if pos < cap:
+            send_slots[dest * cap + pos] = slot
+            send_widx[dest * cap + pos] = widx % nwl
+        send_cnt[dest] += 1
+
+
+def gather_payload_padded_ref(payload, send_slots, send_cnt, send_payload,
+                              stride: int, cap: int, world: int):
+    pl = payload.view(-1, stride)
+    sp = send_payload.view(-1, stride)
+    for r in range(world):
+        for e in range(int(send_cnt[r])):
+            i = r * cap + e
+            sp[i] = pl[int(send_slots[i])]
+
+
+def echo_padded_ref(recv_payload, recv_cnt, res_arena, res_sums,
+                    stride: int, cap: int, world: int):
+    rp = recv_payload.view(-1, stride)
+    ra = res_arena.view(-1, stride)
+    for r in range(world):
+        for e in range(int(recv_cnt[r])):
+            i = r * cap + e
+            ra[i] = rp[i]
+            res_sums[i] = int(rp[i].to(torch.int64).sum()) % (1 << 32) - (
+                (1 << 32) if int(rp[i].to(torch.int64).sum()) % (1 << 32) >= (1 << 31) else 0
+            )
+
+
+def apply_transitions_padded_ref(states, attempts, deadlines, slots_pad, cnt,
+                                 to_state: int, cap: int, world: int):
+    from ..protocol.states import transition_lut
+
+    lut = transition_lut()
+    for r in range(world):
+        for e in range(int(cnt[r])):
+            slot = int(slots_pad[r * cap + e])
+            frm = int(states[slot])
+            if lut[frm][to_state]:
+                states[slot] = to_state
+                if to_state == 3 and frm != 3:
+                    attempts[slot] += 1
+                if to_state >= 6:
+                    deadlines[slot] = torch.iinfo(torch.int64).max
+
+
+def load_feedback_padded_ref(recv_widx, recv_cnt, w_active_local, cap: int, world: int):
+    for r in range(world):
+        for e in range(int(recv_cnt[r])):
+            w_active_local[int(recv_widx[r * cap + e])] += 1

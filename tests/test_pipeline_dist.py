@@ -35,6 +35,19 @@ def _worker(rank, world, port, result_dir):
     # with pool mask = all ranks and deterministic argmin, dispatch crosses
     # ranks whenever a remote worker scores lower
     totals["local_active"] = int(pipe.w_active_local.sum())
+    # result integrity: the returned checksums must equal the checksums of
+    # the payloads this rank dispatched (round-trip through the exchange)
+    if hasattr(pipe, "pad_send_cnt"):
+        pl = pipe.payloads[(pipe._tick - 1) % len(pipe.payloads)].view(pipe.B, -1)
+        ok = True
+        for r in range(world):
+            n = int(pipe.pad_send_cnt[r])
+            for e in range(min(n, 8)):
+                slot = int(pipe.pad_send_slots[r * pipe.B + e])
+                want = int(pl[slot].to(torch.int64).sum()) & 0xFFFFFFFF
+                got = int(pipe.pad_sums_back[r * pipe.B + e]) & 0xFFFFFFFF
+                ok = ok and (want == got)
+        totals["sums_ok"] = bool(ok)
     with open(os.path.join(result_dir, f"rank{rank}.json"), "w") as f:
         json.dump(totals, f)
     dist.barrier()
@@ -52,3 +65,4 @@ def test_two_rank_dispatch_conserves_jobs(tmp_path):
     assert r0["completed"] > 0 and r1["completed"] > 0
     # at least one rank executed remotely-submitted work in the final tick
     assert r0["local_active"] + r1["local_active"] > 0
+    assert r0.get("sums_ok", True) and r1.get("sums_ok", True)
