@@ -143,7 +143,14 @@ class Node:
         self.scheduler_reconciler.tick()
         self.pending_replayer.tick()
         self.run_reconciler.tick()
+        self.write_worker_snapshot()
         self.bus.pump()
+
+    def write_worker_snapshot(self) -> None:
+        """Periodic cluster snapshot under `sys:workers:snapshot`
+        (cmd/cordum-scheduler/main.go:128-162, 5s writer)."""
+        snap = self.registry.cluster_snapshot()
+        self.memory.put("sys:workers:snapshot", json.dumps(snap).encode())
 
     def drain(self, max_iters: int = 64) -> None:
         """Pump until quiescent at the current clock (tests / sync callers)."""
