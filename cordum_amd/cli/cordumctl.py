@@ -45,7 +45,8 @@ def cmd_serve(args):
     from ..cli.serve import serve
 
     serve(host=args.host, port=args.port, config_dir=args.config_dir,
-          workers=args.workers, checkpoint_dir=args.checkpoint_dir)
+          workers=args.workers, checkpoint_dir=args.checkpoint_dir,
+          checkpoint_interval_s=args.checkpoint_interval)
 
 
 def cmd_init(args):
@@ -179,6 +180,7 @@ def main(argv=None) -> int:
     p.add_argument("--config-dir", default="")
     p.add_argument("--workers", type=int, default=2)
     p.add_argument("--checkpoint-dir", default="")
+    p.add_argument("--checkpoint-interval", type=float, default=30.0)
     p.set_defaults(fn=cmd_serve)
 
     p = sub.add_parser("init")

@@ -35,11 +35,12 @@ def build_node(config_dir: str = "", checkpoint_dir: str = ""):
         checkpointer = Checkpointer(node, checkpoint_dir)
         if checkpointer.restore():
             checkpointer.replay_wal()
+        node.wal = checkpointer
     return node, cfg, checkpointer
 
 
 def serve(host: str = "127.0.0.1", port: int = 8080, config_dir: str = "",
-          workers: int = 2, checkpoint_dir: str = ""):
+          workers: int = 2, checkpoint_dir: str = "", checkpoint_interval_s: float = 30.0):
     import uvicorn
 
     from ..gateway import create_app
@@ -64,7 +65,7 @@ def serve(host: str = "127.0.0.1", port: int = 8080, config_dir: str = "",
             if now - last_reconcile > 5.0:
                 node.reconcile()
                 last_reconcile = now
-            if checkpointer is not None and now - last_checkpoint > 30.0:
+            if checkpointer is not None and now - last_checkpoint > checkpoint_interval_s:
                 checkpointer.checkpoint()
                 last_checkpoint = now
             time.sleep(0.05)

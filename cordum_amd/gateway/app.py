@@ -785,6 +785,18 @@ def create_app(
     def health():
         return {"status": "ok"}
 
+    @app.get("/metrics")
+    def metrics():
+        from ..utils.metrics import Metrics as _M
+
+        m = getattr(node, "metrics_exporter", None)
+        if m is None:
+            return Response(b"", media_type="text/plain")
+        return Response(m.exposition(), media_type="text/plain; version=0.0.4")
+
+    from .dashboard import add_dashboard
+
+    add_dashboard(app)
     return app
 
 

@@ -1,0 +1,62 @@
+"""Parallel/distribution layer — the RCCL-over-xGMI fabric.
+
+The reference distributes with NATS queue groups + scheduler replicas
+(SURVEY.md §2.3); here distribution is one rank per GPU over
+torch.distributed (backend "nccl" == RCCL on ROCm):
+
+ - heartbeat fan-in  -> all_gather_into_tensor of per-rank worker loads
+ - job dispatch      -> fixed-capacity padded all_to_all_single
+                        (descriptors + payload; per-destination counts are a
+                        device-resident vector, so the tick has no host
+                        splits sync)
+ - result return     -> mirrored all_to_all_single
+
+The implementation lives in ops/pipeline.py (DevicePipeline._tick_padded and
+the pack/echo/load-feedback kernels in ops/hip/cordum_kernels.hip); this
+package exposes the fabric helpers.
+"""
+from __future__ import annotations
+
+import os
+
+import torch
+import torch.distributed as dist
+
+
+def init_fabric(backend: str | None = None) -> tuple[int, int, torch.device]:
+    """Initialise the per-GPU process group from torchrun env vars.
+
+    Returns (rank, world_size, device). Safe to call with WORLD_SIZE=1 (no
+    process group is created)."""
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    if torch.cuda.is_available():
+        torch.cuda.set_device(local_rank)
+        device = torch.device(f"cuda:{local_rank}")
+    else:
+        device = torch.device("cpu")
+    if world > 1 and not dist.is_initialized():
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        dist.init_process_group(
+            backend=backend or ("nccl" if device.type == "cuda" else "gloo"),
+            rank=rank,
+            world_size=world,
+        )
+    return rank, world, device
+
+
+def heartbeat_all_gather(global_buf: torch.Tensor, local: torch.Tensor) -> None:
+    """sys.heartbeat -> all-gather of per-rank worker load vectors."""
+    if dist.is_initialized() and dist.get_world_size() > 1:
+        dist.all_gather_into_tensor(global_buf, local)
+    else:
+        global_buf.copy_(local)
+
+
+def padded_all_to_all(recv: torch.Tensor, send: torch.Tensor) -> None:
+    """Equal-split all_to_all (the padded dispatch exchange)."""
+    if dist.is_initialized() and dist.get_world_size() > 1:
+        dist.all_to_all_single(recv, send)
+    else:
+        recv.copy_(send)

@@ -41,6 +41,7 @@ from ..store import (
     SchemaRegistry,
 )
 from ..utils.clock import Clock, SYSTEM_CLOCK
+from ..utils.metrics import Metrics as PromMetrics
 from ..workflow import Engine as WorkflowEngine, RunReconciler, WorkflowService, WorkflowStore
 from .worker import Worker, echo_handler
 
@@ -71,6 +72,7 @@ class Node:
             base_policy, configsvc=self.configsvc, cache_ttl_s=safety_cache_ttl_s, clock=clock
         )
         self.strategy = LeastLoadedStrategy(routing or DEFAULT_ROUTING)
+        self.metrics_exporter = PromMetrics()
         self.scheduler = SchedulerEngine(
             self.bus,
             self.job_store,
@@ -78,6 +80,7 @@ class Node:
             self.registry,
             self.strategy,
             configsvc=self.configsvc,
+            metrics=self.metrics_exporter,
             clock=clock,
         )
         self.workflow_store = WorkflowStore(clock=clock)
@@ -94,6 +97,7 @@ class Node:
         self.pending_replayer = PendingReplayer(self.scheduler, self.job_store, clock=clock)
         self.run_reconciler = RunReconciler(self.workflow, self.workflow_store, self.job_store, clock=clock)
         self.workers: List[Worker] = []
+        self.wal = None  # optional Checkpointer: submissions are WAL-appended
         self._started = False
         self._route_config_watch()
 
@@ -167,6 +171,8 @@ class Node:
         trace_id = trace_id or new_trace_id()
         if context is not None:
             req.context_ptr = self.memory.put_context(req.job_id, context)
+        if self.wal is not None:
+            self.wal.wal_append(req, trace_id)
         self.bus.publish(subj.SUBJECT_SUBMIT, BusPacket(trace_id=trace_id, job_request=req))
         return trace_id
 

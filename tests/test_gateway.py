@@ -346,3 +346,10 @@ def test_artifacts_and_memory(client, node):
     node.memory.put("ctx:manual", b'{"y": 2}')
     assert client.get("/api/v1/memory", params={"ptr": "redis://ctx:manual"}).json() == {"y": 2}
     assert client.get("/api/v1/memory", params={"ptr": "redis://missing"}).status_code == 404
+
+
+def test_dashboard_and_health(client):
+    assert client.get("/health").json() == {"status": "ok"}
+    r = client.get("/dashboard")
+    assert r.status_code == 200 and "cordum-mi355x" in r.text
+    assert client.get("/metrics").status_code == 200
