@@ -228,9 +228,12 @@ class DevicePipeline:
 
         # --- policy (compiled once; swapped on reload by the kernel watch hook)
         policy = policy or make_synthetic_policy(n_rules, seed=seed)
-        self.compiled = compile_policy(policy, words=1 if n_rules < 4096 else 2)
-        if not self.compiled.exact:
-            self.compiled = compile_policy(policy, words=4)
+        # word count follows the policy's VOCABULARY, not its rule count:
+        # every extra word is 9*8 B more rule-stream traffic per rule
+        for w in (1, 2, 4):
+            self.compiled = compile_policy(policy, words=w)
+            if self.compiled.exact:
+                break
         assert self.compiled.exact, "bench policy must compile exactly"
         self.cpol = self.compiled.to(device)
 
