@@ -36,7 +36,7 @@
 // plus per-rule bytes: secrets(int8), mcp_any(u8).
 // Job row: any[7*W] | all[2*W] | mcp[4*W], secrets u8, mcp_used u8.
 
-template <int W>
+template <int W, int JPT>
 __global__ __launch_bounds__(BLOCK) void policy_first_match_kernel(
     const long long* __restrict__ rule_any,   // [R,7,W]
     const long long* __restrict__ rule_all,   // [R,2,W]
@@ -51,9 +51,10 @@ __global__ __launch_bounds__(BLOCK) void policy_first_match_kernel(
     int* __restrict__ out_first,              // [J], pre-filled INT_MAX
     int R, int J, int rules_per_chunk)
 {
-    // 2D grid: blockIdx.x tiles jobs, blockIdx.y tiles the rule range — all
-    // chunks resident at once (256 CUs want >> 256 workgroups); first-match
-    // semantics restored by the atomicMin on the output word.
+    // 2D grid: blockIdx.x tiles jobs (JPT jobs per thread — the rule stream
+    // is the traffic bound at large R, and every extra job per block divides
+    // it), blockIdx.y tiles the rule range; first-match semantics restored
+    // by the atomicMin on the output word.
     const int rule_chunk_begin = blockIdx.y * rules_per_chunk;
     const int rule_chunk_end = min(R, rule_chunk_begin + rules_per_chunk);
     constexpr int ANY_W = 7 * W;
@@ -65,28 +66,33 @@ __global__ __launch_bounds__(BLOCK) void policy_first_match_kernel(
     __shared__ long long lds_rules[CHUNK * ROW];
     __shared__ signed char lds_secrets[CHUNK];
     __shared__ unsigned char lds_mcp_any[CHUNK];
+    __shared__ int block_done;
 
-    const int j = blockIdx.x * BLOCK + threadIdx.x;
-
-    // job row -> registers
-    long long jany[ANY_W], jall[ALL_W], jmcp[MCP_W];
-    unsigned char jsec = 0, jused = 0;
-    if (j < J) {
-        #pragma unroll
-        for (int w = 0; w < ANY_W; ++w) jany[w] = job_any[(size_t)j * ANY_W + w];
-        #pragma unroll
-        for (int w = 0; w < ALL_W; ++w) jall[w] = job_all[(size_t)j * ALL_W + w];
-        #pragma unroll
-        for (int w = 0; w < MCP_W; ++w) jmcp[w] = job_mcp[(size_t)j * MCP_W + w];
-        jsec = job_secrets[j];
-        jused = job_mcp_used[j];
+    int jj[JPT];
+    long long jany[JPT][ANY_W], jall[JPT][ALL_W], jmcp[JPT][MCP_W];
+    unsigned char jsec[JPT], jused[JPT];
+    int best[JPT];
+    #pragma unroll
+    for (int k = 0; k < JPT; ++k) {
+        const int j = (blockIdx.x * JPT + k) * BLOCK + threadIdx.x;
+        jj[k] = j;
+        best[k] = INT_MAX;
+        if (j < J) {
+            #pragma unroll
+            for (int w = 0; w < ANY_W; ++w) jany[k][w] = job_any[(size_t)j * ANY_W + w];
+            #pragma unroll
+            for (int w = 0; w < ALL_W; ++w) jall[k][w] = job_all[(size_t)j * ALL_W + w];
+            #pragma unroll
+            for (int w = 0; w < MCP_W; ++w) jmcp[k][w] = job_mcp[(size_t)j * MCP_W + w];
+            jsec[k] = job_secrets[j];
+            jused[k] = job_mcp_used[j];
+        } else {
+            best[k] = INT_MAX - 1;  // dead
+        }
     }
-
-    int best = INT_MAX;
 
     for (int base = rule_chunk_begin; base < rule_chunk_end; base += CHUNK) {
         const int n = min(CHUNK, rule_chunk_end - base);
-        // cooperative stage: CHUNK*ROW words by 256 threads
         for (int i = threadIdx.x; i < n * ROW; i += BLOCK) {
             const int r = i / ROW, w = i % ROW;
             const size_t g = (size_t)(base + r);
@@ -102,15 +108,15 @@ __global__ __launch_bounds__(BLOCK) void policy_first_match_kernel(
         }
         __syncthreads();
 
-        // cross-chunk early-skip: if another chunk already published a match
-        // below this chunk's range, no rule here can win (atomicMin is
-        // monotonic, so a stale read only costs a wasted scan, never
-        // correctness)
-        if (j < J && best == INT_MAX && base > 0 &&
-            __hip_atomic_load(&out_first[j], __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT) < base) {
-            best = INT_MAX - 1;  // sentinel: done, nothing to publish
-        }
-        if (j < J && best == INT_MAX) {
+        #pragma unroll
+        for (int k = 0; k < JPT; ++k) {
+            // cross-chunk early-skip (atomicMin output is monotonic; a stale
+            // read only wastes a scan, never correctness)
+            if (best[k] == INT_MAX && base > 0 &&
+                __hip_atomic_load(&out_first[jj[k]], __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT) < base) {
+                best[k] = INT_MAX - 1;
+            }
+            if (best[k] != INT_MAX) continue;
             for (int r = 0; r < n; ++r) {
                 const long long* row = &lds_rules[r * ROW];
                 bool ok = true;
@@ -121,7 +127,7 @@ __global__ __launch_bounds__(BLOCK) void policy_first_match_kernel(
                     for (int w = 0; w < W; ++w) {
                         const long long rm = row[d * W + w];
                         any |= rm;
-                        inter |= rm & jany[d * W + w];
+                        inter |= rm & jany[k][d * W + w];
                     }
                     ok = (any == 0) | (inter != 0);
                 }
@@ -130,14 +136,14 @@ __global__ __launch_bounds__(BLOCK) void policy_first_match_kernel(
                     long long missing = 0;
                     #pragma unroll
                     for (int w = 0; w < W; ++w)
-                        missing |= row[ANY_W + d * W + w] & ~jall[d * W + w];
+                        missing |= row[ANY_W + d * W + w] & ~jall[k][d * W + w];
                     ok = (missing == 0);
                 }
                 if (ok) {
                     const signed char rs = lds_secrets[r];
-                    ok = (rs < 0) | (rs == (signed char)jsec);
+                    ok = (rs < 0) | (rs == (signed char)jsec[k]);
                 }
-                if (ok && jused && lds_mcp_any[r]) {
+                if (ok && jused[k] && lds_mcp_any[r]) {
                     const long long* allow = &row[ANY_W + ALL_W];
                     const long long* deny = &row[ANY_W + ALL_W + MCP_W];
                     #pragma unroll
@@ -146,26 +152,29 @@ __global__ __launch_bounds__(BLOCK) void policy_first_match_kernel(
                         #pragma unroll
                         for (int w = 0; w < W; ++w) {
                             a |= allow[f * W + w];
-                            ainter |= allow[f * W + w] & jmcp[f * W + w];
-                            dinter |= deny[f * W + w] & jmcp[f * W + w];
+                            ainter |= allow[f * W + w] & jmcp[k][f * W + w];
+                            dinter |= deny[f * W + w] & jmcp[k][f * W + w];
                         }
                         ok = (dinter == 0) & ((a == 0) | (ainter != 0));
                     }
                 }
-                if (ok) { best = base + r; break; }
+                if (ok) { best[k] = base + r; break; }
             }
         }
         __syncthreads();
         // chunk early-out: every live job in this block already matched
-        __shared__ int block_done;
         if (threadIdx.x == 0) block_done = 1;
         __syncthreads();
-        if (j < J && best == INT_MAX) block_done = 0;
+        #pragma unroll
+        for (int k = 0; k < JPT; ++k)
+            if (best[k] == INT_MAX) block_done = 0;
         __syncthreads();
         if (block_done) break;
     }
 
-    if (j < J && best < INT_MAX - 1) atomicMin(&out_first[j], best);
+    #pragma unroll
+    for (int k = 0; k < JPT; ++k)
+        if (jj[k] < J && best[k] < INT_MAX - 1) atomicMin(&out_first[jj[k]], best[k]);
 }
 
 // ---------------------------------------------------------------------------
@@ -855,11 +864,13 @@ torch::Tensor policy_first_match(
     if (R == 0 || J == 0) {
         return out.masked_fill(out == INT_MAX, -1);
     }
-    const int job_blocks = (J + BLOCK - 1) / BLOCK;
+    // JPT: divide the rule-stream traffic at large R (each block reads the
+    // whole rule range once; 2 jobs/thread = half the job blocks)
+    const int JPT = (R >= 16384) ? 2 : 1;
+    const int jobs_per_block = BLOCK * JPT;
+    const int job_blocks = (J + jobs_per_block - 1) / jobs_per_block;
     int nchunks = (int)rule_chunks;
     if (nchunks <= 0) {
-        // fill the chip (256 CUs want >> 256 workgroups total), but keep
-        // chunks >= 512 rules so LDS staging amortizes
         nchunks = std::max(1, std::min((R + 511) / 512,
                                        std::max(1, 2048 / std::max(job_blocks, 1))));
     }
@@ -880,11 +891,20 @@ torch::Tensor policy_first_match(
             job_mcp_used.data_ptr<uint8_t>(),
             out.data_ptr<int>(), R, J, per_chunk);
     };
-    switch (W) {
-        case 1: launch(policy_first_match_kernel<1>); break;
-        case 2: launch(policy_first_match_kernel<2>); break;
-        case 4: launch(policy_first_match_kernel<4>); break;
-        default: TORCH_CHECK(false, "unsupported word count ", W);
+    if (JPT == 2) {
+        switch (W) {
+            case 1: launch(policy_first_match_kernel<1, 2>); break;
+            case 2: launch(policy_first_match_kernel<2, 2>); break;
+            case 4: launch(policy_first_match_kernel<4, 2>); break;
+            default: TORCH_CHECK(false, "unsupported word count ", W);
+        }
+    } else {
+        switch (W) {
+            case 1: launch(policy_first_match_kernel<1, 1>); break;
+            case 2: launch(policy_first_match_kernel<2, 1>); break;
+            case 4: launch(policy_first_match_kernel<4, 1>); break;
+            default: TORCH_CHECK(false, "unsupported word count ", W);
+        }
     }
     out.masked_fill_(out == INT_MAX, -1);
     return out;
