@@ -864,9 +864,12 @@ torch::Tensor policy_first_match(
     if (R == 0 || J == 0) {
         return out.masked_fill(out == INT_MAX, -1);
     }
-    // JPT: divide the rule-stream traffic at large R (each block reads the
-    // whole rule range once; 2 jobs/thread = half the job blocks)
-    const int JPT = (R >= 16384) ? 2 : 1;
+    // JPT: 2 jobs/thread halves the rule-stream traffic per block, but
+    // measured NET-NEGATIVE at every shape tried (16k rules: 162->260 us;
+    // 100k: 244->323 us — the doubled job-row registers cost occupancy and
+    // the per-chunk early-out needs BOTH jobs finished). Kept selectable for
+    // future shapes; disabled by measurement.
+    const int JPT = 1;
     const int jobs_per_block = BLOCK * JPT;
     const int job_blocks = (J + jobs_per_block - 1) / jobs_per_block;
     int nchunks = (int)rule_chunks;
