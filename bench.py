@@ -34,6 +34,9 @@ def main() -> int:
     ap.add_argument("--rules", type=int, default=1024, help="policy rules in the compiled bundle")
     ap.add_argument("--workers", type=int, default=1000, help="workers per rank")
     ap.add_argument("--payload-bytes", type=int, default=256)
+    ap.add_argument("--allow-cpu", action="store_true",
+                    help="CI validation: run the identical pipeline on CPU "
+                         "(gloo + torch reference ops) — not a perf mode")
     args = ap.parse_args()
 
     import torch
@@ -49,7 +52,8 @@ def main() -> int:
         )
         return 2
 
-    if not torch.cuda.is_available():
+    use_gpu = torch.cuda.is_available()
+    if not use_gpu and not args.allow_cpu:
         print("error: bench.py requires an MI355X (no GPU visible)", file=sys.stderr)
         return 1
 
@@ -57,9 +61,13 @@ def main() -> int:
 
     if world_size > 1:
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-        dist.init_process_group(backend="nccl", rank=rank, world_size=world_size)
-    torch.cuda.set_device(local_rank)
-    device = torch.device(f"cuda:{local_rank}")
+        dist.init_process_group(backend="nccl" if use_gpu else "gloo",
+                                rank=rank, world_size=world_size)
+    if use_gpu:
+        torch.cuda.set_device(local_rank)
+        device = torch.device(f"cuda:{local_rank}")
+    else:
+        device = torch.device("cpu")
 
     from cordum_amd.ops.pipeline import DevicePipeline
 
@@ -71,12 +79,14 @@ def main() -> int:
         payload_words=max(1, args.payload_bytes // 4),
         world_size=world_size,
         rank=rank,
+        backend="ext" if use_gpu else "ref",
     )
 
     def barrier():
         if world_size > 1:
             dist.barrier()
-        torch.cuda.synchronize(device)
+        if use_gpu:
+            torch.cuda.synchronize(device)
 
     # warmup
     for _ in range(args.warmup):

@@ -66,3 +66,29 @@ def test_two_rank_dispatch_conserves_jobs(tmp_path):
     # at least one rank executed remotely-submitted work in the final tick
     assert r0["local_active"] + r1["local_active"] > 0
     assert r0.get("sums_ok", True) and r1.get("sums_ok", True)
+
+
+def test_bench_contract_torchrun_cpu(tmp_path):
+    """Run bench.py exactly as the driver does (torch.distributed.run, one
+    rank per 'GPU') on CPU/gloo and validate the JSON contract line."""
+    import subprocess
+    from pathlib import Path
+
+    repo = Path(__file__).resolve().parent.parent
+    env = dict(os.environ, PYTHONPATH=str(repo))
+    res = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29731", str(repo / "bench.py"),
+         "--gpus", "2", "--steps", "3", "--warmup", "1",
+         "--batch", "256", "--workers", "16", "--rules", "64", "--allow-cpu"],
+        cwd=str(repo), env=env, capture_output=True, text=True, timeout=150,
+    )
+    assert res.returncode == 0, res.stdout + res.stderr
+    line = [l for l in res.stdout.splitlines() if l.startswith("{")][-1]
+    out = json.loads(line)
+    assert out["metric"] == "jobs dispatched/sec (whole node)"
+    assert out["n_gpus"] == 2
+    assert out["value"] > 0
+    assert out["scaling"] == "weak"
+    assert out["config"]["global_batch"] == 512
