@@ -285,13 +285,16 @@ class DevicePipeline:
         # into staging tensors so one captured graph serves the whole ring
         self.out_decision = torch.zeros(self.B, dtype=torch.int8, device=device)
         self.denied_slots = torch.zeros(self.B, dtype=torch.int32, device=device)
-        self.denied_count = torch.zeros(1, dtype=torch.int32, device=device)
+        # one backing tensor for the three counters -> one D2H read per tick
+        self._counts = torch.zeros(3, dtype=torch.int32, device=device)
+        self.denied_count = self._counts[0:1]
+        self.allowed_count = self._counts[1:2]
+        self.routable_count = self._counts[2:3]
         self.allowed_slots = torch.zeros(self.B, dtype=torch.int32, device=device)
-        self.allowed_count = torch.zeros(1, dtype=torch.int32, device=device)
         self.routable_slots = torch.zeros(self.B, dtype=torch.int32, device=device)
         self.routable_widx = torch.zeros(self.B, dtype=torch.int32, device=device)
-        self.routable_count = torch.zeros(1, dtype=torch.int32, device=device)
         self._graph = None
+        self._graphs = {}
         # hybrid K1 dispatch (measured, tools/policy_variants_bench.py):
         # MFMA tiles win up to ~32k rules (occupancy at small R), the bitset
         # rows win beyond (4.2x smaller rule stream + first-match early-exit)
@@ -313,16 +316,7 @@ class DevicePipeline:
             ):
                 a_pack, _ = pack_jobs_mfma(jb_host)
                 self.mfma_a_packs.append(a_pack.to(device))
-        if device.type == "cuda" and self.world == 1 and backend != "ref":
-            b0 = self.batches[0]
-            self.stage_any = torch.empty_like(b0.any_bits)
-            self.stage_all = torch.empty_like(b0.all_bits)
-            self.stage_secrets = torch.empty_like(b0.secrets)
-            self.stage_mcp = torch.empty_like(b0.mcp_bits)
-            self.stage_mcp_used = torch.empty_like(b0.mcp_used)
-            self.stage_payload = torch.empty_like(self.payloads[0])
-            if self._use_mfma:
-                self.stage_a_pack = torch.empty_like(self.mfma_a_packs[0])
+        self._fused_capable = device.type == "cuda" and self.world == 1 and backend != "ref"
         self._tick = 0
         self.total_completed = 0
         self.total_denied = 0
@@ -371,7 +365,7 @@ class DevicePipeline:
         for st in (SCHEDULED, DISPATCHED, RUNNING):
             ext.apply_transitions_dyn(self.states, self.attempts, self.deadlines,
                                       self.routable_slots, self.routable_count, st, B)
-        ext.echo_execute_indexed_dyn(self.stage_payload, self.routable_slots,
+        ext.echo_execute_indexed_dyn(self.payloads[slot], self.routable_slots,
                                      self.routable_count, self.res_arena[: B * self.payload_words],
                                      self.res_sums, self.payload_words, B)
         ext.apply_transitions_dyn(self.states, self.attempts, self.deadlines,
@@ -384,34 +378,27 @@ class DevicePipeline:
         t0 = time.perf_counter()
         i = self._tick % len(self.batches)
         self._tick += 1
-        jb = self.batches[i]
-        self.stage_any.copy_(jb.any_bits, non_blocking=True)
-        self.stage_all.copy_(jb.all_bits, non_blocking=True)
-        self.stage_secrets.copy_(jb.secrets, non_blocking=True)
-        self.stage_mcp.copy_(jb.mcp_bits, non_blocking=True)
-        self.stage_mcp_used.copy_(jb.mcp_used, non_blocking=True)
-        self.stage_payload.copy_(self.payloads[i], non_blocking=True)
-        if self._use_mfma:
-            self.stage_a_pack.copy_(self.mfma_a_packs[i], non_blocking=True)
-        if self._graph is None:
+        g = self._graphs.get(i)
+        if g is None:
             if not hasattr(self, "_pend_states"):
                 self._pend_states = torch.full((self.B,), PENDING, dtype=torch.uint8, device=self.device)
-            # eager warmups on a side stream, then capture
+            # eager warmups on a side stream, then capture this slot's graph
             side = torch.cuda.Stream(device=self.device)
             side.wait_stream(torch.cuda.current_stream(self.device))
             with torch.cuda.stream(side):
                 for _ in range(2):
-                    self._fused_body()
+                    self._fused_body(i)
             torch.cuda.current_stream(self.device).wait_stream(side)
             torch.cuda.synchronize(self.device)
             g = torch.cuda.CUDAGraph()
             with torch.cuda.graph(g):
-                self._fused_body()
-            self._graph = g
-        self._graph.replay()
-        torch.cuda.synchronize(self.device)
-        denied = int(self.denied_count.item())
-        routable = int(self.routable_count.item())
+                self._fused_body(i)
+            self._graphs[i] = g
+            self._graph = g  # marker: fused path active
+        g.replay()
+        counts = self._counts.cpu()  # one D2H read = the tick's only sync
+        denied = int(counts[0])
+        routable = int(counts[2])
         dt = time.perf_counter() - t0
         stats = TickStats(
             completed=routable,
@@ -425,9 +412,7 @@ class DevicePipeline:
 
     # -- one control-plane tick -------------------------------------------------
     def tick(self) -> TickStats:
-        if self._graph is not None or (
-            self.device.type == "cuda" and self.world == 1 and hasattr(self, "stage_any")
-        ):
+        if getattr(self, "_fused_capable", False):
             return self._tick_fused()
         if self.world > 1:
             return self._tick_padded()

@@ -56,8 +56,7 @@ def test_eager_and_fused_agree():
     fused = DevicePipeline(device=torch.device("cuda:0"), **kw)
     s1 = [fused.tick() for _ in range(2)]
     eager = DevicePipeline(device=torch.device("cuda:0"), **kw)
-    eager._graph = None
-    del eager.stage_any  # force the eager path
+    eager._fused_capable = False  # force the eager path
     s2 = [eager._tick_eager() for _ in range(2)]
     for a, b in zip(s1, s2):
         assert (a.completed, a.denied, a.unrouted) == (b.completed, b.denied, b.unrouted)
