@@ -321,33 +321,32 @@ class DevicePipeline:
         self.total_completed = 0
         self.total_denied = 0
 
-    # -- fused single-GPU tick (hipGraph-captured) --------------------------------
-    def _fused_body(self) -> None:
-        """The whole tick as a fixed kernel sequence over staging buffers —
-        no host decisions, so it is capturable into a hipGraph (the
-        boundary/graph-replay costs in MI355X_MICROARCH.md §price-list are
-        what this amortizes for the launch-bound control-plane tick)."""
+    # -- fused single-GPU tick (hipGraph-captured per ring slot) -------------------
+    def _fused_body(self, slot: int) -> None:
+        """The whole tick as a fixed kernel sequence over ring slot `slot`'s
+        tensors — no host decisions, no staging copies; one captured graph
+        per ring slot (the boundary/graph-replay costs in
+        MI355X_MICROARCH.md §price-list are what this amortizes for the
+        launch-bound control-plane tick)."""
         B = self.B
         ext = self.ext
+        jb = self.batches[slot]
         self.states.zero_()
         pend = self._pend_states
         ext.apply_transitions(self.states, self.attempts, self.deadlines, self.slots, pend)
         if self._use_mfma:
             first = ext.policy_first_match_mfma(
-                self.stage_a_pack, self.mfma_policy.b_pack, self.mfma_policy.cards,
-                self.mfma_policy.secrets, self.stage_secrets,
+                self.mfma_a_packs[slot], self.mfma_policy.b_pack, self.mfma_policy.cards,
+                self.mfma_policy.secrets, jb.secrets,
                 self.B, self.compiled.n_rules,
             )
         else:
             first = ext.policy_first_match(
                 self.cpol.any_masks, self.cpol.all_masks, self.cpol.secrets,
                 self.cpol.mcp_masks, self.cpol.mcp_any,
-                self.stage_any, self.stage_all, self.stage_secrets,
-                self.stage_mcp, self.stage_mcp_used, 0,
+                jb.any_bits, jb.all_bits, jb.secrets, jb.mcp_bits, jb.mcp_used, 0,
             )
-        self.denied_count.zero_()
-        self.allowed_count.zero_()
-        self.routable_count.zero_()
+        self._counts.zero_()
         ext.policy_gate(first, self.cpol.decisions, self.out_decision,
                         self.denied_slots, self.denied_count,
                         self.allowed_slots, self.allowed_count)
