@@ -373,28 +373,34 @@ class DevicePipeline:
         ext.load_feedback(self.routable_widx, self.routable_count,
                           self.w_active_local, self.NWL, self.rank, B)
 
+    def _ensure_graphs(self) -> None:
+        """Capture every ring slot's graph once (first tick), so no capture
+        cost ever lands inside a timed step."""
+        if self._graphs:
+            return
+        if not hasattr(self, "_pend_states"):
+            self._pend_states = torch.full((self.B,), PENDING, dtype=torch.uint8, device=self.device)
+        side = torch.cuda.Stream(device=self.device)
+        side.wait_stream(torch.cuda.current_stream(self.device))
+        with torch.cuda.stream(side):
+            for slot in range(len(self.batches)):
+                self._fused_body(slot)
+        torch.cuda.current_stream(self.device).wait_stream(side)
+        torch.cuda.synchronize(self.device)
+        for slot in range(len(self.batches)):
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                self._fused_body(slot)
+            self._graphs[slot] = g
+        self._graph = self._graphs[0]  # marker: fused path active
+        torch.cuda.synchronize(self.device)
+
     def _tick_fused(self) -> TickStats:
+        self._ensure_graphs()
         t0 = time.perf_counter()
         i = self._tick % len(self.batches)
         self._tick += 1
-        g = self._graphs.get(i)
-        if g is None:
-            if not hasattr(self, "_pend_states"):
-                self._pend_states = torch.full((self.B,), PENDING, dtype=torch.uint8, device=self.device)
-            # eager warmups on a side stream, then capture this slot's graph
-            side = torch.cuda.Stream(device=self.device)
-            side.wait_stream(torch.cuda.current_stream(self.device))
-            with torch.cuda.stream(side):
-                for _ in range(2):
-                    self._fused_body(i)
-            torch.cuda.current_stream(self.device).wait_stream(side)
-            torch.cuda.synchronize(self.device)
-            g = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g):
-                self._fused_body(i)
-            self._graphs[i] = g
-            self._graph = g  # marker: fused path active
-        g.replay()
+        self._graphs[i].replay()
         counts = self._counts.cpu()  # one D2H read = the tick's only sync
         denied = int(counts[0])
         routable = int(counts[2])
