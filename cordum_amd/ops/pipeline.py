@@ -295,6 +295,7 @@ class DevicePipeline:
         self.routable_widx = torch.zeros(self.B, dtype=torch.int32, device=device)
         self._graph = None
         self._graphs = {}
+        self._pad_graphs = None
         # hybrid K1 dispatch (measured, tools/policy_variants_bench.py):
         # MFMA tiles win up to ~32k rules (occupancy at small R), the bitset
         # rows win beyond (4.2x smaller rule stream + first-match early-exit)
@@ -424,57 +425,50 @@ class DevicePipeline:
         return self._tick_eager()
 
     # -- multi-rank tick: fixed-capacity padded all-to-all ------------------------
-    def _tick_padded(self) -> TickStats:
-        """Cross-GPU dispatch with static shapes: per-destination segments of
-        capacity B, counts as a device-resident vector exchanged with the
-        payload — no host splits sync anywhere in the tick (the one
-        `.item()` at the end is the stats read the bench needs anyway)."""
-        t0 = time.perf_counter()
-        i = self._tick % len(self.batches)
-        self._tick += 1
-        jb = self.batches[i]
-        payload = self.payloads[i]
+    def _pad_alloc(self) -> None:
+        if hasattr(self, "pad_send_slots"):
+            return
         B, W, world, dev = self.B, self.payload_words, self.world, self.device
-        ext = self.ext
 
-        if not hasattr(self, "pad_send_slots"):
-            def zi(n):
-                return torch.zeros(n, dtype=torch.int32, device=dev)
+        def zi(n):
+            return torch.zeros(n, dtype=torch.int32, device=dev)
 
-            self.pad_send_slots = zi(world * B)
-            self.pad_send_widx = zi(world * B)
-            self.pad_send_cnt = zi(world)
-            self.pad_recv_cnt = zi(world)
-            self.pad_recv_widx = zi(world * B)
-            self.pad_send_payload = torch.zeros(world * B * W, dtype=torch.int32, device=dev)
-            self.pad_recv_payload = torch.zeros(world * B * W, dtype=torch.int32, device=dev)
-            self.pad_res = torch.zeros(world * B * W, dtype=torch.int32, device=dev)
-            self.pad_sums = zi(world * B)
-            self.pad_sums_back = zi(world * B)
+        self.pad_send_slots = zi(world * B)
+        self.pad_send_widx = zi(world * B)
+        self.pad_send_cnt = zi(world)
+        self.pad_recv_cnt = zi(world)
+        self.pad_recv_widx = zi(world * B)
+        self.pad_send_payload = torch.zeros(world * B * W, dtype=torch.int32, device=dev)
+        self.pad_recv_payload = torch.zeros(world * B * W, dtype=torch.int32, device=dev)
+        self.pad_res = torch.zeros(world * B * W, dtype=torch.int32, device=dev)
+        self.pad_sums = zi(world * B)
+        self.pad_sums_back = zi(world * B)
+        if not hasattr(self, "_pend_states"):
             self._pend_states = torch.full((B,), PENDING, dtype=torch.uint8, device=dev)
 
-        # submit
+    # the tick is cut at the collective seams into four fixed-shape segments
+    # (G1/G2 depend on the ring slot; G3/G4 do not) so each segment can be
+    # hipGraph-captured — the collectives themselves stay eager, which keeps
+    # RCCL out of graph capture entirely.
+    def _pad_g1(self, slot: int) -> None:
+        ext, B = self.ext, self.B
+        jb = self.batches[slot]
         self.states.zero_()
         ext.apply_transitions(self.states, self.attempts, self.deadlines, self.slots, self._pend_states)
-        # safety gate (bitset path; the hybrid MFMA path is single-GPU fused)
         first = ext.policy_first_match(
             self.cpol.any_masks, self.cpol.all_masks, self.cpol.secrets,
             self.cpol.mcp_masks, self.cpol.mcp_any,
             jb.any_bits, jb.all_bits, jb.secrets, jb.mcp_bits, jb.mcp_used, 0,
         )
-        self.denied_count.zero_()
-        self.allowed_count.zero_()
-        self.routable_count.zero_()
+        self._counts.zero_()
         ext.policy_gate(first, self.cpol.decisions, self.out_decision,
                         self.denied_slots, self.denied_count,
                         self.allowed_slots, self.allowed_count)
         ext.apply_transitions_dyn(self.states, self.attempts, self.deadlines,
                                   self.denied_slots, self.denied_count, DENIED, B)
-        # heartbeat all-gather
-        dist.all_gather_into_tensor(self.w_active, self.w_active_local)
-        dist.all_gather_into_tensor(self.w_cpu, self.w_cpu_local)
-        dist.all_gather_into_tensor(self.w_gpu, self.w_gpu_local)
-        # routing
+
+    def _pad_g2(self, slot: int) -> None:
+        ext, B, world = self.ext, self.B, self.world
         w_keys = ext.worker_precompute(self.w_pool, self.w_active, self.w_maxp,
                                        self.w_cpu, self.w_gpu)
         pick = ext.least_loaded_pick(self.w_pool, w_keys, self.w_labels,
@@ -484,32 +478,126 @@ class DevicePipeline:
         for st in (SCHEDULED, DISPATCHED):
             ext.apply_transitions_dyn(self.states, self.attempts, self.deadlines,
                                       self.routable_slots, self.routable_count, st, B)
-        # padded pack + exchange
         self.pad_send_cnt.zero_()
         ext.pack_by_dest(self.routable_slots, self.routable_widx, self.routable_count,
                          self.pad_send_slots, self.pad_send_widx, self.pad_send_cnt,
                          self.NWL, B, B)
-        ext.gather_payload_padded(payload, self.pad_send_slots, self.pad_send_cnt,
-                                  self.pad_send_payload, W, B, world)
-        dist.all_to_all_single(self.pad_recv_cnt, self.pad_send_cnt)
-        dist.all_to_all_single(self.pad_recv_widx, self.pad_send_widx)
-        dist.all_to_all_single(self.pad_recv_payload, self.pad_send_payload)
+        ext.gather_payload_padded(self.payloads[slot], self.pad_send_slots, self.pad_send_cnt,
+                                  self.pad_send_payload, self.payload_words, B, world)
+
+    def _pad_g3(self) -> None:
+        ext, B, world = self.ext, self.B, self.world
         ext.apply_transitions_padded(self.states, self.attempts, self.deadlines,
                                      self.pad_send_slots, self.pad_send_cnt, RUNNING, B, world)
-        # worker execution on the receiving rank + result return
         ext.echo_padded(self.pad_recv_payload, self.pad_recv_cnt, self.pad_res,
-                        self.pad_sums, W, B, world)
-        dist.all_to_all_single(self.pad_sums_back, self.pad_sums)
+                        self.pad_sums, self.payload_words, B, world)
+
+    def _pad_g4(self) -> None:
+        ext, B, world = self.ext, self.B, self.world
         ext.apply_transitions_padded(self.states, self.attempts, self.deadlines,
                                      self.pad_send_slots, self.pad_send_cnt, SUCCEEDED, B, world)
-        # load feedback from the receive side
         self.w_active_local.zero_()
         ext.load_feedback_padded(self.pad_recv_widx, self.pad_recv_cnt,
                                  self.w_active_local, B, world)
-        if dev.type == "cuda":
-            torch.cuda.synchronize(dev)
-        denied = int(self.denied_count.cpu()[0])
-        routable = int(self.routable_count.cpu()[0])
+
+    def _pad_heartbeats(self) -> None:
+        if self.world > 1:
+            dist.all_gather_into_tensor(self.w_active, self.w_active_local)
+            dist.all_gather_into_tensor(self.w_cpu, self.w_cpu_local)
+            dist.all_gather_into_tensor(self.w_gpu, self.w_gpu_local)
+        else:
+            self.w_active.copy_(self.w_active_local)
+            self.w_cpu.copy_(self.w_cpu_local)
+            self.w_gpu.copy_(self.w_gpu_local)
+
+    def _pad_exchange_out(self) -> None:
+        if self.world > 1:
+            dist.all_to_all_single(self.pad_recv_cnt, self.pad_send_cnt)
+            dist.all_to_all_single(self.pad_recv_widx, self.pad_send_widx)
+            dist.all_to_all_single(self.pad_recv_payload, self.pad_send_payload)
+        else:
+            self.pad_recv_cnt.copy_(self.pad_send_cnt)
+            self.pad_recv_widx.copy_(self.pad_send_widx)
+            self.pad_recv_payload.copy_(self.pad_send_payload)
+
+    def _pad_exchange_back(self) -> None:
+        if self.world > 1:
+            dist.all_to_all_single(self.pad_sums_back, self.pad_sums)
+        else:
+            self.pad_sums_back.copy_(self.pad_sums)
+
+    def _ensure_pad_graphs(self) -> None:
+        """Capture G1/G2 per ring slot and G3/G4 once, with eager warmups.
+        On any capture failure (e.g. an allocator/backend corner) fall back
+        to the uncaptured segments — same kernels, same semantics."""
+        if self._pad_graphs is not None:
+            return
+        try:
+            side = torch.cuda.Stream(device=self.device)
+            side.wait_stream(torch.cuda.current_stream(self.device))
+            with torch.cuda.stream(side):
+                for slot in range(len(self.batches)):
+                    self._pad_g1(slot)
+                    self._pad_g2(slot)
+                self._pad_g3()
+                self._pad_g4()
+            torch.cuda.current_stream(self.device).wait_stream(side)
+            torch.cuda.synchronize(self.device)
+            g1s, g2s = [], []
+            for slot in range(len(self.batches)):
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    self._pad_g1(slot)
+                g1s.append(g)
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    self._pad_g2(slot)
+                g2s.append(g)
+            g3 = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g3):
+                self._pad_g3()
+            g4 = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g4):
+                self._pad_g4()
+            torch.cuda.synchronize(self.device)
+            self._pad_graphs = (g1s, g2s, g3, g4)
+        except Exception:
+            self._pad_graphs = ()  # capture unsupported: stay eager
+
+    def _tick_padded(self) -> TickStats:
+        """Cross-GPU dispatch with static shapes: per-destination segments of
+        capacity B, counts as a device-resident vector exchanged with the
+        payload — no host splits sync anywhere in the tick (the one
+        `.cpu()` at the end is the stats read the bench needs anyway)."""
+        self._pad_alloc()
+        use_graphs = False
+        if self.device.type == "cuda":
+            self._ensure_pad_graphs()
+            use_graphs = bool(self._pad_graphs)
+        t0 = time.perf_counter()
+        i = self._tick % len(self.batches)
+        self._tick += 1
+        B = self.B
+        if use_graphs:
+            g1s, g2s, g3, g4 = self._pad_graphs
+            g1s[i].replay()
+            self._pad_heartbeats()
+            g2s[i].replay()
+            self._pad_exchange_out()
+            g3.replay()
+            self._pad_exchange_back()
+            g4.replay()
+        else:
+            self._pad_g1(i)
+            self._pad_heartbeats()
+            self._pad_g2(i)
+            self._pad_exchange_out()
+            self._pad_g3()
+            self._pad_exchange_back()
+            self._pad_g4()
+        counts = self._counts.cpu()  # the tick's only host sync
+        denied = int(counts[0])
+        routable = int(counts[2])
         dt = time.perf_counter() - t0
         stats = TickStats(completed=routable, denied=denied,
                           unrouted=B - routable - denied, wall_s=dt)

@@ -60,3 +60,25 @@ def test_eager_and_fused_agree():
     s2 = [eager._tick_eager() for _ in range(2)]
     for a, b in zip(s1, s2):
         assert (a.completed, a.denied, a.unrouted) == (b.completed, b.denied, b.unrouted)
+
+
+def test_padded_graphed_path_matches_fused_world1():
+    """Run the multi-rank (padded, segment-graphed) tick on one GPU with
+    world=1 (collectives degrade to copies) and require identical per-tick
+    counts to the fused single-GPU path — this validates the exact code the
+    8-GPU bench runs, minus RCCL."""
+    from cordum_amd.ops.pipeline import DevicePipeline
+
+    kw = dict(batch_size=4096, n_local_workers=64, n_rules=512, n_batches=3,
+              payload_words=16, seed=9)
+    fused = DevicePipeline(device=torch.device("cuda:0"), **kw)
+    s1 = [fused.tick() for _ in range(4)]
+
+    padded = DevicePipeline(device=torch.device("cuda:0"), **kw)
+    padded._fused_capable = False
+    s2 = [padded._tick_padded() for _ in range(4)]
+    assert padded._pad_graphs, "segment graphs must capture on GPU"
+    for a, b in zip(s1, s2):
+        assert (a.completed, a.denied, a.unrouted) == (b.completed, b.denied, b.unrouted)
+    # state tables agree too
+    assert torch.equal(fused.states.cpu(), padded.states.cpu())
