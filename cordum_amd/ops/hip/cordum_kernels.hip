@@ -287,6 +287,29 @@ __global__ __launch_bounds__(BLOCK) void least_loaded_pick_kernel(
     }
 }
 
+// K2c: batch spreading — unconstrained jobs round-robin over the K least
+// loaded workers. A frozen-snapshot argmin sends a whole homogeneous batch
+// to ONE worker (the reference has the same pathology between heartbeats:
+// its scheduler replica piles onto the lowest scorer until the next 10 s
+// heartbeat); queue-group fan-in is the intended behavior, and spreading a
+// batch across the least-loaded set IS that behavior for a batched tick.
+// Constrained jobs (labels / narrowed pool mask) keep their exact scan pick.
+__global__ __launch_bounds__(BLOCK) void spread_pick_kernel(
+    const int* __restrict__ order,        // [NW] worker idx sorted by key asc
+    const int* __restrict__ valid_count,  // [1] non-overloaded workers
+    const long long* __restrict__ j_poolmask,
+    const long long* __restrict__ j_labels,
+    long long full_mask,
+    int* __restrict__ pick,               // in-out
+    int NJ, int K)
+{
+    const int j = blockIdx.x * BLOCK + threadIdx.x;
+    if (j >= NJ) return;
+    if (j_labels[j] != 0 || j_poolmask[j] != full_mask) return;  // constrained
+    const int V = min(*valid_count, K);
+    if (V > 0) pick[j] = order[j % V];
+}
+
 // ---------------------------------------------------------------------------
 // K5: batched state transitions with legality LUT (job_store.go:70-82)
 // ---------------------------------------------------------------------------
@@ -950,6 +973,19 @@ torch::Tensor least_loaded_pick(
     return out;
 }
 
+void spread_pick(torch::Tensor order, torch::Tensor valid_count,
+                 torch::Tensor j_poolmask, torch::Tensor j_labels,
+                 int64_t full_mask, torch::Tensor pick, int64_t K)
+{
+    const int NJ = (int)pick.size(0);
+    const int blocks = (NJ + BLOCK - 1) / BLOCK;
+    hipLaunchKernelGGL(spread_pick_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        order.data_ptr<int>(), valid_count.data_ptr<int>(),
+        (const long long*)j_poolmask.data_ptr<int64_t>(),
+        (const long long*)j_labels.data_ptr<int64_t>(),
+        (long long)full_mask, pick.data_ptr<int>(), NJ, (int)K);
+}
+
 void set_transition_lut(torch::Tensor lut) {
     TORCH_CHECK(lut.numel() == N_STATES * N_STATES, "lut must be 11x11");
     auto cpu = lut.to(torch::kUInt8).contiguous().cpu();
@@ -1193,6 +1229,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("policy_first_match_mfma", &policy_first_match_mfma, "K1 MFMA comparison variant");
     m.def("worker_precompute", &worker_precompute, "K2a per-worker score/overload precompute");
     m.def("least_loaded_pick", &least_loaded_pick, "K2 least-loaded worker pick");
+    m.def("spread_pick", &spread_pick, "K2c batch spreading over the least-loaded set");
     m.def("echo_execute_indexed", &echo_execute_indexed, "device echo worker pool (slot-indexed, in-place)");
     m.def("policy_gate", &policy_gate, "decision gather + allow/deny compaction");
     m.def("compact_routable", &compact_routable, "routable-slot compaction");

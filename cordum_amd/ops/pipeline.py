@@ -166,6 +166,16 @@ class _RefOps:
 
         load_feedback_padded_ref(rw, rc, wal, int(cap), int(world))
 
+    def spread_pick(self, order, valid_count, j_poolmask, j_labels, full_mask, pick, K):
+        V = min(int(valid_count[0]), int(K))
+        if V <= 0:
+            return
+        NJ = pick.shape[0]
+        default = (j_labels == 0) & (j_poolmask == full_mask)
+        idx = torch.arange(NJ) % V
+        spread = order[idx.long()].to(torch.int32)
+        pick[default] = spread[default]
+
     def policy_gate(self, first, decisions, out_decision, ds, dc, als, alc):
         J = first.shape[0]
         dec = torch.where(first >= 0, decisions.to(torch.int32)[first.clamp(min=0).long()],
@@ -322,6 +332,21 @@ class DevicePipeline:
         self.total_completed = 0
         self.total_denied = 0
 
+
+    def _spread(self, w_keys, pick):
+        """K2c batch spreading: unconstrained jobs round-robin over the K
+        least-loaded workers (see the kernel comment for why). All torch ops
+        here are capture-safe."""
+        order = torch.argsort(w_keys).to(torch.int32)
+        # a key is overloaded iff its top 32 bits carry the OVERLOADED tag
+        valid = ((w_keys >> 32) & 0xFFFFFFFF).ne(0xFFFFFFFE).to(torch.int32)
+        valid_count = valid.sum().to(torch.int32).reshape(1)
+        full_mask = (1 << self.world) - 1
+        K = min(int(w_keys.shape[0]), 1024)
+        self.ext.spread_pick(order, valid_count, self.j_poolmask, self.j_labels,
+                             full_mask, pick, K)
+        return pick
+
     # -- fused single-GPU tick (hipGraph-captured per ring slot) -------------------
     def _fused_body(self, slot: int) -> None:
         """The whole tick as a fixed kernel sequence over ring slot `slot`'s
@@ -360,6 +385,7 @@ class DevicePipeline:
                                        self.w_cpu, self.w_gpu)
         pick = ext.least_loaded_pick(self.w_pool, w_keys, self.w_labels,
                                      self.j_poolmask, self.j_labels)
+        self._spread(w_keys, pick)
         ext.compact_routable(self.allowed_slots, self.allowed_count, pick,
                              self.routable_slots, self.routable_widx, self.routable_count)
         for st in (SCHEDULED, DISPATCHED, RUNNING):
@@ -473,6 +499,7 @@ class DevicePipeline:
                                        self.w_cpu, self.w_gpu)
         pick = ext.least_loaded_pick(self.w_pool, w_keys, self.w_labels,
                                      self.j_poolmask, self.j_labels)
+        self._spread(w_keys, pick)
         ext.compact_routable(self.allowed_slots, self.allowed_count, pick,
                              self.routable_slots, self.routable_widx, self.routable_count)
         for st in (SCHEDULED, DISPATCHED):
