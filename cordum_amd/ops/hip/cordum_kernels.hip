@@ -973,6 +973,19 @@ torch::Tensor least_loaded_pick(
     return out;
 }
 
+void worker_precompute_into(
+    torch::Tensor w_pool, torch::Tensor w_active, torch::Tensor w_maxp,
+    torch::Tensor w_cpu, torch::Tensor w_gpu, torch::Tensor out_keys)
+{
+    const int NW = (int)w_pool.size(0);
+    if (NW == 0) return;
+    const int blocks = (NW + BLOCK - 1) / BLOCK;
+    hipLaunchKernelGGL(worker_precompute_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        w_pool.data_ptr<int>(), w_active.data_ptr<int>(), w_maxp.data_ptr<int>(),
+        w_cpu.data_ptr<float>(), w_gpu.data_ptr<float>(),
+        (unsigned long long*)out_keys.data_ptr<int64_t>(), NW);
+}
+
 void spread_pick(torch::Tensor order, torch::Tensor valid_count,
                  torch::Tensor j_poolmask, torch::Tensor j_labels,
                  int64_t full_mask, torch::Tensor pick, int64_t K)
@@ -1230,6 +1243,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("worker_precompute", &worker_precompute, "K2a per-worker score/overload precompute");
     m.def("least_loaded_pick", &least_loaded_pick, "K2 least-loaded worker pick");
     m.def("spread_pick", &spread_pick, "K2c batch spreading over the least-loaded set");
+    m.def("worker_precompute_into", &worker_precompute_into, "K2a into a persistent key tensor");
     m.def("echo_execute_indexed", &echo_execute_indexed, "device echo worker pool (slot-indexed, in-place)");
     m.def("policy_gate", &policy_gate, "decision gather + allow/deny compaction");
     m.def("compact_routable", &compact_routable, "routable-slot compaction");

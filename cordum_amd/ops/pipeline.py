@@ -166,6 +166,11 @@ class _RefOps:
 
         load_feedback_padded_ref(rw, rc, wal, int(cap), int(world))
 
+    def worker_precompute_into(self, w_pool, w_active, w_maxp, w_cpu, w_gpu, out_keys):
+        from .reference import worker_precompute_ref
+
+        out_keys.copy_(worker_precompute_ref(w_pool, w_active, w_maxp, w_cpu, w_gpu))
+
     def spread_pick(self, order, valid_count, j_poolmask, j_labels, full_mask, pick, K):
         V = min(int(valid_count[0]), int(K))
         if V <= 0:
@@ -281,6 +286,14 @@ class DevicePipeline:
         self.w_cpu = torch.zeros(NWG, dtype=torch.float32, device=device)
         self.w_gpu = torch.zeros(NWG, dtype=torch.float32, device=device)
 
+        # persistent K2 buffers: keys written in-graph; the spread ORDER is
+        # refreshed eagerly once per tick from them (1-tick-stale ordering —
+        # the reference tolerates 10 s-stale heartbeats)
+        NWG_ = self.NWL * self.world
+        self.w_keys = torch.zeros(NWG_, dtype=torch.int64, device=device)
+        self.order_buf = torch.arange(NWG_, dtype=torch.int32, device=device)
+        self.valid_buf = torch.tensor([NWG_], dtype=torch.int32, device=device)
+
         # job routing masks: any pool (any rank)
         self.j_poolmask = torch.full((self.B,), (1 << self.world) - 1,
                                      dtype=torch.int64, device=device)
@@ -333,19 +346,21 @@ class DevicePipeline:
         self.total_denied = 0
 
 
-    def _spread(self, w_keys, pick):
+    def _spread(self, pick):
         """K2c batch spreading: unconstrained jobs round-robin over the K
-        least-loaded workers (see the kernel comment for why). All torch ops
-        here are capture-safe."""
-        order = torch.argsort(w_keys).to(torch.int32)
-        # a key is overloaded iff its top 32 bits carry the OVERLOADED tag
-        valid = ((w_keys >> 32) & 0xFFFFFFFF).ne(0xFFFFFFFE).to(torch.int32)
-        valid_count = valid.sum().to(torch.int32).reshape(1)
+        least-loaded workers, using the persistent (last-refresh) order."""
         full_mask = (1 << self.world) - 1
-        K = min(int(w_keys.shape[0]), 1024)
-        self.ext.spread_pick(order, valid_count, self.j_poolmask, self.j_labels,
-                             full_mask, pick, K)
+        K = min(int(self.order_buf.shape[0]), 1024)
+        self.ext.spread_pick(self.order_buf, self.valid_buf, self.j_poolmask,
+                             self.j_labels, full_mask, pick, K)
         return pick
+
+    def _refresh_order(self) -> None:
+        """Eager (non-captured) refresh of the spread order from the key
+        tensor the captured tick just wrote."""
+        self.order_buf.copy_(torch.argsort(self.w_keys).to(torch.int32))
+        valid = ((self.w_keys >> 32) & 0xFFFFFFFF).ne(0xFFFFFFFE)
+        self.valid_buf.copy_(valid.sum().to(torch.int32).reshape(1))
 
     # -- fused single-GPU tick (hipGraph-captured per ring slot) -------------------
     def _fused_body(self, slot: int) -> None:
@@ -381,11 +396,11 @@ class DevicePipeline:
         self.w_active.copy_(self.w_active_local)
         self.w_cpu.copy_(self.w_cpu_local)
         self.w_gpu.copy_(self.w_gpu_local)
-        w_keys = ext.worker_precompute(self.w_pool, self.w_active, self.w_maxp,
-                                       self.w_cpu, self.w_gpu)
-        pick = ext.least_loaded_pick(self.w_pool, w_keys, self.w_labels,
+        ext.worker_precompute_into(self.w_pool, self.w_active, self.w_maxp,
+                                   self.w_cpu, self.w_gpu, self.w_keys)
+        pick = ext.least_loaded_pick(self.w_pool, self.w_keys, self.w_labels,
                                      self.j_poolmask, self.j_labels)
-        self._spread(w_keys, pick)
+        self._spread(pick)
         ext.compact_routable(self.allowed_slots, self.allowed_count, pick,
                              self.routable_slots, self.routable_widx, self.routable_count)
         for st in (SCHEDULED, DISPATCHED, RUNNING):
@@ -428,6 +443,7 @@ class DevicePipeline:
         i = self._tick % len(self.batches)
         self._tick += 1
         self._graphs[i].replay()
+        self._refresh_order()
         counts = self._counts.cpu()  # one D2H read = the tick's only sync
         denied = int(counts[0])
         routable = int(counts[2])
@@ -495,11 +511,11 @@ class DevicePipeline:
 
     def _pad_g2(self, slot: int) -> None:
         ext, B, world = self.ext, self.B, self.world
-        w_keys = ext.worker_precompute(self.w_pool, self.w_active, self.w_maxp,
-                                       self.w_cpu, self.w_gpu)
-        pick = ext.least_loaded_pick(self.w_pool, w_keys, self.w_labels,
+        ext.worker_precompute_into(self.w_pool, self.w_active, self.w_maxp,
+                                   self.w_cpu, self.w_gpu, self.w_keys)
+        pick = ext.least_loaded_pick(self.w_pool, self.w_keys, self.w_labels,
                                      self.j_poolmask, self.j_labels)
-        self._spread(w_keys, pick)
+        self._spread(pick)
         ext.compact_routable(self.allowed_slots, self.allowed_count, pick,
                              self.routable_slots, self.routable_widx, self.routable_count)
         for st in (SCHEDULED, DISPATCHED):
@@ -614,6 +630,7 @@ class DevicePipeline:
             g3.replay()
             self._pad_exchange_back()
             g4.replay()
+            self._refresh_order()
         else:
             self._pad_g1(i)
             self._pad_heartbeats()
@@ -622,6 +639,7 @@ class DevicePipeline:
             self._pad_g3()
             self._pad_exchange_back()
             self._pad_g4()
+            self._refresh_order()
         counts = self._counts.cpu()  # the tick's only host sync
         denied = int(counts[0])
         routable = int(counts[2])
