@@ -280,7 +280,11 @@ class DevicePipeline:
         self.w_gpu_local = (torch.rand(NW, generator=g) * 50).to(device)
         NWG = NW * self.world
         self.w_pool = (torch.arange(NWG, device=device, dtype=torch.int32) // NW)
-        self.w_maxp = torch.full((NWG,), 64, dtype=torch.int32, device=device)
+        # worker capacity scales with the fair batch share: a tick assigns
+        # ~B/NWL jobs per worker, and "overloaded" should mean imbalance
+        # (>= 0.9 * 4x fair share), not a full healthy batch
+        maxp_val = max(64, (4 * batch_size) // max(1, n_local_workers))
+        self.w_maxp = torch.full((NWG,), maxp_val, dtype=torch.int32, device=device)
         self.w_labels = torch.zeros(NWG, dtype=torch.int64, device=device)
         self.w_active = torch.zeros(NWG, dtype=torch.int32, device=device)
         self.w_cpu = torch.zeros(NWG, dtype=torch.float32, device=device)
