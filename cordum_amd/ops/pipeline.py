@@ -312,8 +312,9 @@ class DevicePipeline:
         # into staging tensors so one captured graph serves the whole ring
         self.out_decision = torch.zeros(self.B, dtype=torch.int8, device=device)
         self.denied_slots = torch.zeros(self.B, dtype=torch.int32, device=device)
-        # one backing tensor for the three counters -> one D2H read per tick
-        self._counts = torch.zeros(3, dtype=torch.int32, device=device)
+        # one backing tensor for the counters -> one D2H read per tick
+        # [0]=denied [1]=allowed [2]=routable [3]=dispatched (padded path)
+        self._counts = torch.zeros(4, dtype=torch.int32, device=device)
         self.denied_count = self._counts[0:1]
         self.allowed_count = self._counts[1:2]
         self.routable_count = self._counts[2:3]
@@ -480,20 +481,25 @@ class DevicePipeline:
         if hasattr(self, "pad_send_slots"):
             return
         B, W, world, dev = self.B, self.payload_words, self.world, self.device
+        # per-destination capacity: spreading makes the expected share B/world;
+        # 4x headroom covers imbalance, and overflow (counted as unrouted
+        # backpressure, like a worker-at-capacity NAK) is accounted below
+        self.pad_cap = min(B, max(64, (4 * B) // max(1, world)))
+        cap = self.pad_cap
 
         def zi(n):
             return torch.zeros(n, dtype=torch.int32, device=dev)
 
-        self.pad_send_slots = zi(world * B)
-        self.pad_send_widx = zi(world * B)
+        self.pad_send_slots = zi(world * cap)
+        self.pad_send_widx = zi(world * cap)
         self.pad_send_cnt = zi(world)
         self.pad_recv_cnt = zi(world)
-        self.pad_recv_widx = zi(world * B)
-        self.pad_send_payload = torch.zeros(world * B * W, dtype=torch.int32, device=dev)
-        self.pad_recv_payload = torch.zeros(world * B * W, dtype=torch.int32, device=dev)
-        self.pad_res = torch.zeros(world * B * W, dtype=torch.int32, device=dev)
-        self.pad_sums = zi(world * B)
-        self.pad_sums_back = zi(world * B)
+        self.pad_recv_widx = zi(world * cap)
+        self.pad_send_payload = torch.zeros(world * cap * W, dtype=torch.int32, device=dev)
+        self.pad_recv_payload = torch.zeros(world * cap * W, dtype=torch.int32, device=dev)
+        self.pad_res = torch.zeros(world * cap * W, dtype=torch.int32, device=dev)
+        self.pad_sums = zi(world * cap)
+        self.pad_sums_back = zi(world * cap)
         if not hasattr(self, "_pend_states"):
             self._pend_states = torch.full((B,), PENDING, dtype=torch.uint8, device=dev)
 
@@ -530,27 +536,32 @@ class DevicePipeline:
         for st in (SCHEDULED, DISPATCHED):
             ext.apply_transitions_dyn(self.states, self.attempts, self.deadlines,
                                       self.routable_slots, self.routable_count, st, B)
+        cap = self.pad_cap
         self.pad_send_cnt.zero_()
         ext.pack_by_dest(self.routable_slots, self.routable_widx, self.routable_count,
                          self.pad_send_slots, self.pad_send_widx, self.pad_send_cnt,
-                         self.NWL, B, B)
+                         self.NWL, cap, B)
+        # clamp: entries beyond capacity were not written (overflow = dropped
+        # for retry, the at-capacity NAK analog); dispatched = sum of clamped
+        self.pad_send_cnt.clamp_(max=cap)
+        self._counts[3:4].copy_(self.pad_send_cnt.sum().reshape(1))
         ext.gather_payload_padded(self.payloads[slot], self.pad_send_slots, self.pad_send_cnt,
-                                  self.pad_send_payload, self.payload_words, B, world)
+                                  self.pad_send_payload, self.payload_words, cap, world)
 
     def _pad_g3(self) -> None:
-        ext, B, world = self.ext, self.B, self.world
+        ext, world, cap = self.ext, self.world, self.pad_cap
         ext.apply_transitions_padded(self.states, self.attempts, self.deadlines,
-                                     self.pad_send_slots, self.pad_send_cnt, RUNNING, B, world)
+                                     self.pad_send_slots, self.pad_send_cnt, RUNNING, cap, world)
         ext.echo_padded(self.pad_recv_payload, self.pad_recv_cnt, self.pad_res,
-                        self.pad_sums, self.payload_words, B, world)
+                        self.pad_sums, self.payload_words, cap, world)
 
     def _pad_g4(self) -> None:
-        ext, B, world = self.ext, self.B, self.world
+        ext, world, cap = self.ext, self.world, self.pad_cap
         ext.apply_transitions_padded(self.states, self.attempts, self.deadlines,
-                                     self.pad_send_slots, self.pad_send_cnt, SUCCEEDED, B, world)
+                                     self.pad_send_slots, self.pad_send_cnt, SUCCEEDED, cap, world)
         self.w_active_local.zero_()
         ext.load_feedback_padded(self.pad_recv_widx, self.pad_recv_cnt,
-                                 self.w_active_local, B, world)
+                                 self.w_active_local, cap, world)
 
     def _pad_heartbeats(self) -> None:
         if self.world > 1:
@@ -651,10 +662,10 @@ class DevicePipeline:
             self._refresh_order()
         counts = self._counts.cpu()  # the tick's only host sync
         denied = int(counts[0])
-        routable = int(counts[2])
+        dispatched = int(counts[3])  # routable minus capacity overflow
         dt = time.perf_counter() - t0
-        stats = TickStats(completed=routable, denied=denied,
-                          unrouted=B - routable - denied, wall_s=dt)
+        stats = TickStats(completed=dispatched, denied=denied,
+                          unrouted=B - dispatched - denied, wall_s=dt)
         self.total_completed += stats.completed
         self.total_denied += stats.denied
         return stats
