@@ -65,6 +65,13 @@ def create_app(
     rate_limit_burst: int = 100,
 ) -> FastAPI:
     app = FastAPI(title="cordum-mi355x gateway", version="0.1.0")
+    # CORS for the dashboard (gateway.go:2051-2143)
+    from fastapi.middleware.cors import CORSMiddleware
+
+    app.add_middleware(
+        CORSMiddleware, allow_origins=["*"], allow_methods=["*"],
+        allow_headers=["*"], expose_headers=["*"],
+    )
     auth = auth or BasicAuthProvider()
     bucket = TokenBucket(rate_limit_rps, rate_limit_burst, clock=node.clock)
     app.state.node = node

@@ -1,0 +1,56 @@
+#!/usr/bin/env python3
+"""Sustained-load soak (reference analog: BENCHMARKS.md §2 '8h sustained'):
+run the flagship tick continuously, report sustained jobs/s, p50/p99 step
+latency, and host RSS growth."""
+import argparse
+import json
+import resource
+import statistics
+import sys
+import time
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--seconds", type=float, default=60.0)
+    ap.add_argument("--batch", type=int, default=16384)
+    args = ap.parse_args()
+    import torch
+
+    if not torch.cuda.is_available():
+        print("needs GPU", file=sys.stderr)
+        return 1
+    from cordum_amd.ops.pipeline import DevicePipeline
+
+    pipe = DevicePipeline(device=torch.device("cuda:0"), batch_size=args.batch)
+    for _ in range(10):
+        pipe.tick()
+    rss0 = resource.getrusage(resource.RUSAGE_SELF).ru_maxrss
+    times = []
+    completed = 0
+    t0 = time.perf_counter()
+    while time.perf_counter() - t0 < args.seconds:
+        st = pipe.tick()
+        times.append(st.wall_s)
+        completed += st.completed + st.denied
+    elapsed = time.perf_counter() - t0
+    rss1 = resource.getrusage(resource.RUSAGE_SELF).ru_maxrss
+    times.sort()
+    print(json.dumps({
+        "soak_seconds": round(elapsed, 1),
+        "steps": len(times),
+        "jobs_completed": completed,
+        "sustained_jobs_per_sec": round(completed / elapsed),
+        "p50_ms": round(times[len(times) // 2] * 1e3, 3),
+        "p99_ms": round(times[int(len(times) * 0.99)] * 1e3, 3),
+        "max_ms": round(times[-1] * 1e3, 3),
+        "rss_growth_mb": round((rss1 - rss0) / 1024, 1),
+    }))
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())
