@@ -6,7 +6,6 @@ from __future__ import annotations
 import threading
 import time
 from pathlib import Path
-from typing import Optional
 
 
 def build_node(config_dir: str = "", checkpoint_dir: str = ""):

@@ -10,7 +10,7 @@ system.yaml seeded into the config service as cfg:system:default
 from __future__ import annotations
 
 import os
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from pathlib import Path
 from typing import Any, Dict, Optional
 

@@ -24,7 +24,7 @@ from typing import Any, Dict, List, Optional
 from fastapi import APIRouter, Depends, FastAPI, HTTPException, Query, Request, Response
 from fastapi.responses import JSONResponse
 
-from ..protocol import JobState, is_terminal, parse_state
+from ..protocol import JobState
 from ..protocol import subjects as subj
 from ..protocol.capv2 import (
     ActorType,
@@ -38,11 +38,11 @@ from ..protocol.capv2 import (
     PolicyCheckRequest,
 )
 from ..runtime.node import Node
-from ..store.memory_store import key_from_pointer, pointer_for_key
+from ..store.memory_store import pointer_for_key
 from ..utils.hashing import BUS_MSG_ID_LABEL, job_hash
 from ..utils.ids import new_id, new_trace_id, short_id
 from ..utils.secrets import contains_secret_refs
-from ..workflow import Step, Workflow, WorkflowRun
+from ..workflow import Workflow, WorkflowRun
 from ..store.job_store import ApprovalRecord
 from .auth import BasicAuthProvider, Principal, TokenBucket
 

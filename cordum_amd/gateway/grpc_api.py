@@ -26,10 +26,7 @@ from ..protocol.capv2 import (
     ActorType,
     Message,
     PolicyCheckRequest,
-    PolicyCheckResponse,
-)
-from ..protocol.states import parse_state
-from ..protocol import capv2
+    )
 
 
 # -- api.proto messages (field numbers from core/protocol/proto/v1/api.proto) --

@@ -17,7 +17,7 @@ import hashlib
 import io
 import json
 import tarfile
-from typing import Any, Callable, Dict, List, Optional
+from typing import Any, Callable, Dict, List
 
 import yaml
 from fastapi import APIRouter, Depends, HTTPException, Request

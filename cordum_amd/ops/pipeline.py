@@ -28,7 +28,7 @@ tests/test_gpu_kernels.py.
 from __future__ import annotations
 
 import time
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import List, Optional
 
 import torch

@@ -7,7 +7,6 @@ bit-for-bit (deterministic tie-breaks included).
 """
 from __future__ import annotations
 
-from typing import Tuple
 
 import torch
 

@@ -14,7 +14,7 @@ trimmed, empty string never matches — so interning keys are
 """
 from __future__ import annotations
 
-from typing import Dict, Iterable, List, Optional
+from typing import Dict, Iterable, List
 
 
 class Interner:

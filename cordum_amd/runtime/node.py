@@ -13,13 +13,12 @@ The HTTP/gRPC gateway (gateway/app.py) and the GPU data plane
 from __future__ import annotations
 
 import json
-from dataclasses import dataclass, field
-from typing import Dict, List, Optional
+from typing import List, Optional
 
 from ..bus import LoopbackBus
 from ..protocol import subjects as subj
-from ..protocol.capv2 import BusPacket, Heartbeat, JobRequest
-from ..safety import AllowAllSafety, SafetyKernel, parse_safety_policy
+from ..protocol.capv2 import BusPacket, JobRequest
+from ..safety import SafetyKernel, parse_safety_policy
 from ..scheduler import (
     Engine as SchedulerEngine,
     LeastLoadedStrategy,

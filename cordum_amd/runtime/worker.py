@@ -13,7 +13,6 @@ is ops/worker_pool.py; it presents the same Heartbeat rows to the registry.
 from __future__ import annotations
 
 import threading
-import time
 from dataclasses import dataclass, field
 from typing import Callable, Dict, List, Optional
 

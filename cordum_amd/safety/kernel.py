@@ -18,15 +18,14 @@ from __future__ import annotations
 
 import hashlib
 import threading
-from dataclasses import dataclass, field
-from typing import Any, Dict, List, Optional, Tuple
+from dataclasses import dataclass
+from typing import Any, Dict, List, Optional
 
 from ..protocol.capv2 import (
     DecisionType,
     PolicyCheckRequest,
     PolicyCheckResponse,
-    PolicyConstraints,
-    PolicyRemediation,
+        PolicyRemediation,
 )
 from ..store.configsvc import ConfigService
 from ..utils.clock import Clock, SYSTEM_CLOCK

@@ -20,12 +20,11 @@ config, remediation).
 """
 from __future__ import annotations
 
-import time
 from dataclasses import dataclass, field
-from typing import Callable, Dict, Optional
+from typing import Dict, Optional
 
-from ..bus import Bus, LoopbackBus, RetryAfter
-from ..protocol import JobState, TERMINAL_STATES, is_terminal
+from ..bus import Bus, RetryAfter
+from ..protocol import JobState, is_terminal
 from ..protocol import subjects as subj
 from ..protocol.capv2 import (
     BusPacket,
@@ -35,7 +34,7 @@ from ..protocol.capv2 import (
     JobStatus,
     PolicyConstraints,
 )
-from ..store import DLQEntry, DLQStore, InvalidTransition, JobStore, SafetyDecisionRecord
+from ..store import InvalidTransition, JobStore, SafetyDecisionRecord
 from ..store.configsvc import ConfigService
 from ..utils.canonical_json import canonical_json
 from ..utils.clock import Clock, SYSTEM_CLOCK

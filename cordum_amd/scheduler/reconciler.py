@@ -13,7 +13,6 @@ device job table.
 """
 from __future__ import annotations
 
-from typing import Optional
 
 from ..protocol import JobState
 from ..store import InvalidTransition, JobStore

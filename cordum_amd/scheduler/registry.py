@@ -12,7 +12,7 @@ the seam unit tests drive.
 from __future__ import annotations
 
 import threading
-from typing import Dict, List, Optional
+from typing import Dict
 
 from ..protocol.capv2 import Heartbeat
 from ..utils.clock import Clock, SYSTEM_CLOCK

@@ -13,8 +13,6 @@ behavior is part of the tested contract, safety_client_test.go).
 from __future__ import annotations
 
 import threading
-from dataclasses import dataclass
-from typing import Callable, Optional
 
 from ..protocol.capv2 import (
     DecisionType,

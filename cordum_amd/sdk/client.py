@@ -7,8 +7,7 @@ the compat HTTP API, so either client works against either implementation.
 """
 from __future__ import annotations
 
-import json
-from typing import Any, Dict, List, Optional
+from typing import Any, Dict, Optional
 
 import requests
 

@@ -11,7 +11,7 @@ from typing import Dict, Optional, Tuple
 
 from ..utils.clock import Clock, SYSTEM_CLOCK
 from ..utils.ids import new_id
-from .memory_store import MemoryStore, pointer_for_key
+from .memory_store import MemoryStore
 
 RETENTION_TTLS = {
     "short": 24 * 3600,

@@ -7,7 +7,7 @@ Oracle: core/infra/memory/dlq_store.go:14-180 — `dlq:entry:<job_id>` JSON
 from __future__ import annotations
 
 import threading
-from dataclasses import dataclass, field, asdict
+from dataclasses import dataclass, asdict
 from typing import Dict, List, Optional, Tuple
 
 from ..utils.clock import Clock, SYSTEM_CLOCK

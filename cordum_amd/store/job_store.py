@@ -18,7 +18,7 @@ from __future__ import annotations
 import threading
 from contextlib import contextmanager
 from dataclasses import dataclass, field
-from typing import Any, Dict, Iterable, List, Optional, Tuple
+from typing import Any, Dict, List, Optional, Tuple
 
 from ..protocol import JobState, can_transition, is_terminal, parse_state
 from ..protocol.capv2 import JobRequest, PolicyConstraints, PolicyRemediation

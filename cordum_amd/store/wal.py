@@ -18,7 +18,6 @@ import json
 import os
 import tempfile
 from pathlib import Path
-from typing import Optional
 
 from ..protocol.capv2 import JobRequest
 

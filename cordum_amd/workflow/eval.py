@@ -7,7 +7,7 @@ mixed strings stringify each expression; nil renders as empty).
 """
 from __future__ import annotations
 
-from typing import Any, Dict, List, Optional, Tuple
+from typing import Any, Dict, List
 
 
 class EvalError(Exception):

@@ -17,7 +17,7 @@ from ..protocol import subjects as subj
 from ..protocol.capv2 import BusPacket, JobResult, JobStatus
 from ..store import JobStore
 from ..utils.clock import Clock, SYSTEM_CLOCK
-from .engine import Engine, split_job_id
+from .engine import Engine
 from .models import RUN_PENDING, RUN_RUNNING, RUN_WAITING, STEP_RUNNING
 from .store import WorkflowStore
 
