@@ -287,6 +287,22 @@ __global__ __launch_bounds__(BLOCK) void least_loaded_pick_kernel(
     }
 }
 
+// K7: DLQ ring append — denied/failed slots into a capped device ring
+// (dlq_store.go's capped index as an HBM ring with an atomic head)
+__global__ __launch_bounds__(BLOCK) void dlq_ring_append_kernel(
+    const int* __restrict__ slots,     // [<=cap]
+    const int* __restrict__ count,     // [1]
+    int* __restrict__ ring,            // [ring_size]
+    int* __restrict__ head,            // [1] monotonically increasing
+    int ring_size)
+{
+    const int i = blockIdx.x * BLOCK + threadIdx.x;
+    const int lane = threadIdx.x % WAVE;
+    const bool live = i < *count;
+    const int pos = wave_append_slot(live, head, lane);
+    if (pos >= 0) ring[pos % ring_size] = live ? slots[i] : -1;
+}
+
 // K2c: batch spreading — unconstrained jobs round-robin over the K least
 // loaded workers. A frozen-snapshot argmin sends a whole homogeneous batch
 // to ONE worker (the reference has the same pathology between heartbeats:
@@ -973,6 +989,15 @@ torch::Tensor least_loaded_pick(
     return out;
 }
 
+void dlq_ring_append(torch::Tensor slots, torch::Tensor count,
+                     torch::Tensor ring, torch::Tensor head, int64_t capacity)
+{
+    const int blocks = ((int)capacity + BLOCK - 1) / BLOCK;
+    hipLaunchKernelGGL(dlq_ring_append_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        slots.data_ptr<int>(), count.data_ptr<int>(), ring.data_ptr<int>(),
+        head.data_ptr<int>(), (int)ring.size(0));
+}
+
 void worker_precompute_into(
     torch::Tensor w_pool, torch::Tensor w_active, torch::Tensor w_maxp,
     torch::Tensor w_cpu, torch::Tensor w_gpu, torch::Tensor out_keys)
@@ -1244,6 +1269,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("least_loaded_pick", &least_loaded_pick, "K2 least-loaded worker pick");
     m.def("spread_pick", &spread_pick, "K2c batch spreading over the least-loaded set");
     m.def("worker_precompute_into", &worker_precompute_into, "K2a into a persistent key tensor");
+    m.def("dlq_ring_append", &dlq_ring_append, "K7 device DLQ ring append");
     m.def("echo_execute_indexed", &echo_execute_indexed, "device echo worker pool (slot-indexed, in-place)");
     m.def("policy_gate", &policy_gate, "decision gather + allow/deny compaction");
     m.def("compact_routable", &compact_routable, "routable-slot compaction");

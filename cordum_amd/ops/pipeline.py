@@ -181,6 +181,14 @@ class _RefOps:
         spread = order[idx.long()].to(torch.int32)
         pick[default] = spread[default]
 
+    def dlq_ring_append(self, slots, count, ring, head, capacity):
+        n = int(count[0])
+        h = int(head[0])
+        size = ring.shape[0]
+        for i in range(n):
+            ring[(h + i) % size] = slots[i]
+        head[0] = h + n
+
     def policy_gate(self, first, decisions, out_decision, ds, dc, als, alc):
         J = first.shape[0]
         dec = torch.where(first >= 0, decisions.to(torch.int32)[first.clamp(min=0).long()],
@@ -289,6 +297,10 @@ class DevicePipeline:
         self.w_active = torch.zeros(NWG, dtype=torch.int32, device=device)
         self.w_cpu = torch.zeros(NWG, dtype=torch.float32, device=device)
         self.w_gpu = torch.zeros(NWG, dtype=torch.float32, device=device)
+
+        # device DLQ ring (K7): denied slots, capped, host-drainable
+        self.dlq_ring = torch.full((1 << 16,), -1, dtype=torch.int32, device=device)
+        self.dlq_head = torch.zeros(1, dtype=torch.int32, device=device)
 
         # persistent K2 buffers: keys written in-graph; the spread ORDER is
         # refreshed eagerly once per tick from them (1-tick-stale ordering —
@@ -403,6 +415,8 @@ class DevicePipeline:
                         self.allowed_slots, self.allowed_count)
         ext.apply_transitions_dyn(self.states, self.attempts, self.deadlines,
                                   self.denied_slots, self.denied_count, DENIED, B)
+        ext.dlq_ring_append(self.denied_slots, self.denied_count,
+                            self.dlq_ring, self.dlq_head, B)
         self.w_active.copy_(self.w_active_local)
         self.w_cpu.copy_(self.w_cpu_local)
         self.w_gpu.copy_(self.w_gpu_local)
@@ -523,6 +537,8 @@ class DevicePipeline:
                         self.allowed_slots, self.allowed_count)
         ext.apply_transitions_dyn(self.states, self.attempts, self.deadlines,
                                   self.denied_slots, self.denied_count, DENIED, B)
+        ext.dlq_ring_append(self.denied_slots, self.denied_count,
+                            self.dlq_ring, self.dlq_head, B)
 
     def _pad_g2(self, slot: int) -> None:
         ext, B, world = self.ext, self.B, self.world

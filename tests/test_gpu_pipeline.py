@@ -82,3 +82,23 @@ def test_padded_graphed_path_matches_fused_world1():
         assert (a.completed, a.denied, a.unrouted) == (b.completed, b.denied, b.unrouted)
     # state tables agree too
     assert torch.equal(fused.states.cpu(), padded.states.cpu())
+
+
+def test_device_dlq_ring_collects_denied():
+    from cordum_amd.ops.pipeline import DevicePipeline
+
+    pipe = DevicePipeline(device=torch.device("cuda:0"), batch_size=4096,
+                          n_local_workers=64, n_rules=512, n_batches=2, payload_words=8)
+    total_denied = 0
+    for _ in range(3):
+        st = pipe.tick()
+        total_denied += st.denied
+    assert int(pipe.dlq_head.cpu()[0]) == total_denied
+    ring = pipe.dlq_ring.cpu()
+    entries = ring[ring >= 0]
+    assert entries.numel() == min(total_denied, ring.numel())
+    # every DLQ'd slot really is DENIED in the job table (last tick's entries)
+    states = pipe.states.cpu()
+    last_tick_entries = entries[-(total_denied // 3):] if total_denied else entries
+    for slot in last_tick_entries.tolist():
+        assert states[slot] == 10  # DENIED

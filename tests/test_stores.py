@@ -243,3 +243,39 @@ def test_json_merge_patch():
     assert json_merge_patch({"a": {"b": 1}}, {"a": {"c": 2}}) == {"a": {"b": 1, "c": 2}}
     assert json_merge_patch({"a": 1}, {"a": None}) == {}
     assert json_merge_patch({"a": 1}, "str") == "str"
+
+
+# --- pagination / cursors ----------------------------------------------------
+
+
+def test_job_store_recent_pagination(clock):
+    js = JobStore(clock=clock)
+    for i in range(10):
+        clock.advance(1)
+        js.set_state(f"p{i}", JobState.PENDING)
+    page1, cursor = js.list_recent(limit=4)
+    assert page1 == [f"p{i}" for i in range(9, 5, -1)]
+    assert cursor is not None
+    page2, cursor2 = js.list_recent(limit=4, cursor=cursor)
+    assert page2 == [f"p{i}" for i in range(5, 1, -1)]
+    page3, cursor3 = js.list_recent(limit=4, cursor=cursor2)
+    assert page3 == ["p1", "p0"] and cursor3 is None
+
+
+def test_workflow_store_run_filters_and_pagination(clock):
+    from cordum_amd.workflow import Workflow, WorkflowRun, WorkflowStore
+
+    ws = WorkflowStore(clock=clock)
+    for i in range(6):
+        clock.advance(1)
+        ws.create_run(WorkflowRun(id=f"r{i}", workflow_id="wfA" if i % 2 else "wfB",
+                                  org_id="org", status="running" if i < 4 else "succeeded"))
+    runs, cursor = ws.list_runs(limit=3)
+    assert [r.id for r in runs] == ["r5", "r4", "r3"]
+    runs2, _ = ws.list_runs(limit=10, cursor=cursor)
+    assert [r.id for r in runs2] == ["r2", "r1", "r0"]
+    only_a, _ = ws.list_runs(workflow_id="wfA")
+    assert all(r.workflow_id == "wfA" for r in only_a)
+    running, _ = ws.list_runs(status="running")
+    assert len(running) == 4
+    assert ws.count_active_runs("org") == 4
