@@ -287,22 +287,6 @@ __global__ __launch_bounds__(BLOCK) void least_loaded_pick_kernel(
     }
 }
 
-// K7: DLQ ring append — denied/failed slots into a capped device ring
-// (dlq_store.go's capped index as an HBM ring with an atomic head)
-__global__ __launch_bounds__(BLOCK) void dlq_ring_append_kernel(
-    const int* __restrict__ slots,     // [<=cap]
-    const int* __restrict__ count,     // [1]
-    int* __restrict__ ring,            // [ring_size]
-    int* __restrict__ head,            // [1] monotonically increasing
-    int ring_size)
-{
-    const int i = blockIdx.x * BLOCK + threadIdx.x;
-    const int lane = threadIdx.x % WAVE;
-    const bool live = i < *count;
-    const int pos = wave_append_slot(live, head, lane);
-    if (pos >= 0) ring[pos % ring_size] = live ? slots[i] : -1;
-}
-
 // K2c: batch spreading — unconstrained jobs round-robin over the K least
 // loaded workers. A frozen-snapshot argmin sends a whole homogeneous batch
 // to ONE worker (the reference has the same pathology between heartbeats:
@@ -368,6 +352,22 @@ __device__ __forceinline__ int wave_append_slot(bool pred, int* counter, int lan
     base = __shfl(base, leader, WAVE);
     const int rank = __popcll(mask & ((1ull << lane) - 1ull));
     return pred ? base + rank : -1;
+}
+
+// K7: DLQ ring append — denied/failed slots into a capped device ring
+// (dlq_store.go's capped index as an HBM ring with an atomic head)
+__global__ __launch_bounds__(BLOCK) void dlq_ring_append_kernel(
+    const int* __restrict__ slots,     // [<=cap]
+    const int* __restrict__ count,     // [1]
+    int* __restrict__ ring,            // [ring_size]
+    int* __restrict__ head,            // [1] monotonically increasing
+    int ring_size)
+{
+    const int i = blockIdx.x * BLOCK + threadIdx.x;
+    const int lane = threadIdx.x % WAVE;
+    const bool live = i < *count;
+    const int pos = wave_append_slot(live, head, lane);
+    if (pos >= 0) ring[pos % ring_size] = live ? slots[i] : -1;
 }
 
 
