@@ -90,15 +90,22 @@ def test_device_dlq_ring_collects_denied():
     pipe = DevicePipeline(device=torch.device("cuda:0"), batch_size=4096,
                           n_local_workers=64, n_rules=512, n_batches=2, payload_words=8)
     total_denied = 0
+    last_denied = 0
     for _ in range(3):
         st = pipe.tick()
         total_denied += st.denied
-    assert int(pipe.dlq_head.cpu()[0]) == total_denied
+        last_denied = st.denied
+    head = int(pipe.dlq_head.cpu()[0])
+    assert head == total_denied
     ring = pipe.dlq_ring.cpu()
     entries = ring[ring >= 0]
     assert entries.numel() == min(total_denied, ring.numel())
-    # every DLQ'd slot really is DENIED in the job table (last tick's entries)
+    # the last tick's appends (ring[head-last : head]) point at slots that are
+    # DENIED in the job table (earlier ticks' slots were recycled since —
+    # states is per-tick working state, the ring is the durable record)
     states = pipe.states.cpu()
-    last_tick_entries = entries[-(total_denied // 3):] if total_denied else entries
-    for slot in last_tick_entries.tolist():
+    rs = ring.numel()
+    last_slots = [int(ring[p % rs]) for p in range(head - last_denied, head)]
+    assert sorted(set(last_slots)) == sorted(last_slots)  # unique slots
+    for slot in last_slots:
         assert states[slot] == 10  # DENIED
