@@ -459,6 +459,10 @@ class DevicePipeline:
                 self._fused_body(slot)
             self._graphs[slot] = g
         self._graph = self._graphs[0]  # marker: fused path active
+        # the eager warmup pass above ran the body for real, so it appended to
+        # the durable DLQ ring — reset so the ring only records counted ticks
+        self.dlq_head.zero_()
+        self.dlq_ring.fill_(-1)
         torch.cuda.synchronize(self.device)
 
     def _tick_fused(self) -> TickStats:
@@ -638,6 +642,9 @@ class DevicePipeline:
             g4 = torch.cuda.CUDAGraph()
             with torch.cuda.graph(g4):
                 self._pad_g4()
+            # reset the durable DLQ ring: the eager warmup appended for real
+            self.dlq_head.zero_()
+            self.dlq_ring.fill_(-1)
             torch.cuda.synchronize(self.device)
             self._pad_graphs = (g1s, g2s, g3, g4)
         except Exception:
