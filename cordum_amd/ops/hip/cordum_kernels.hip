@@ -639,6 +639,66 @@ __global__ __launch_bounds__(BLOCK) void policy_gate_kernel(
     if (dpos >= 0) denied_slots[dpos] = j;
 }
 
+// gate + DENIED transition + DLQ ring in ONE launch: the profile showed the
+// split (gate, apply_transitions_dyn[DENIED], dlq_ring_append) spending
+// ~14 us/tick on three ~4.6 us launches over the same denied set
+// (profiles/r14_bench_ktrace_stats.txt) — at a 150 us tick that is pure
+// launch-boundary tax (MI355X_MICROARCH.md price-list "kernel launch").
+__global__ __launch_bounds__(BLOCK) void policy_gate_full_kernel(
+    const int* __restrict__ first,            // [J]
+    const signed char* __restrict__ decisions, // [R]
+    signed char* __restrict__ out_decision,   // [J]
+    int* __restrict__ denied_slots,           // [J] compacted
+    int* __restrict__ denied_count,           // [1]
+    int* __restrict__ allowed_slots,          // [J] compacted
+    int* __restrict__ allowed_count,          // [1]
+    unsigned char* __restrict__ states,
+    long long* __restrict__ deadlines,
+    int* __restrict__ dlq_ring,               // [ring_size]
+    int* __restrict__ dlq_head,               // [1] monotonic
+    int ring_size,
+    int J)
+{
+    const int j = blockIdx.x * BLOCK + threadIdx.x;
+    const int lane = threadIdx.x % WAVE;
+    const bool live = j < J;
+    signed char d = 0;
+    if (live) {
+        const int r = first[j];
+        d = (r >= 0) ? decisions[r] : (signed char)1;  // default allow
+        out_decision[j] = d;
+    }
+    const bool allowed = live && (d == 1 || d == 5);
+    const int apos = wave_append_slot(allowed, allowed_count, lane);
+    if (apos >= 0) allowed_slots[apos] = j;
+    const bool denied = live && !allowed;
+    const int dpos = wave_append_slot(denied, denied_count, lane);
+    if (dpos >= 0) denied_slots[dpos] = j;
+    // DENIED transition (LUT-checked) + terminal deadline clear, fused
+    bool dlq = false;
+    if (denied) {
+        const unsigned char from = states[j];
+        if (d_transition_lut[from * N_STATES + 10]) {  // -> DENIED
+            states[j] = 10;
+            deadlines[j] = (long long)0x7fffffffffffffffLL;
+            dlq = true;
+        }
+    }
+    const int rpos = wave_append_slot(dlq, dlq_head, lane);
+    if (rpos >= 0) dlq_ring[rpos % ring_size] = j;
+}
+
+// one-launch per-tick reset: states + the small counter words (replaces two
+// torch zero_() elementwise launches inside the captured tick)
+__global__ __launch_bounds__(BLOCK) void tick_reset_kernel(
+    unsigned char* __restrict__ states, int B,
+    int* __restrict__ counts, int ncounts)
+{
+    const int i = blockIdx.x * BLOCK + threadIdx.x;
+    if (i < B) states[i] = 0;
+    if (i < ncounts) counts[i] = 0;
+}
+
 // routable compaction: allowed jobs with a worker pick
 __global__ __launch_bounds__(BLOCK) void compact_routable_kernel(
     const int* __restrict__ allowed_slots,    // [<=J]
@@ -680,6 +740,40 @@ __global__ __launch_bounds__(BLOCK) void apply_transitions_dyn_kernel(
     states[slot] = to;
     if (to == 3 && from != 3) attempts[slot] += 1;
     if (to >= 6) deadlines[slot] = (long long)0x7fffffffffffffffLL;
+}
+
+// chained K5: march each routed slot through up to 4 states in ONE launch
+// (SCHEDULED -> DISPATCHED -> RUNNING -> SUCCEEDED after the echo worker; the
+// intermediate states are unobservable inside a captured graph, so four
+// launches over the identical slot set were pure launch tax — see
+// profiles/r14_bench_ktrace_stats.txt: apply_transitions_dyn 5x/tick, 13.6%).
+// Also zeroes an extra int region (the per-tick local-load accumulator) so
+// the load_feedback pass starts clean without its own zero_() launch.
+__global__ __launch_bounds__(BLOCK) void apply_transitions_chain_dyn_kernel(
+    unsigned char* __restrict__ states,
+    int* __restrict__ attempts,
+    long long* __restrict__ deadlines,
+    const int* __restrict__ slots,
+    const int* __restrict__ count,            // [1]
+    int to0, int to1, int to2, int to3,       // -1 = unused
+    int* __restrict__ extra_zero, int n_extra)
+{
+    const int i = blockIdx.x * BLOCK + threadIdx.x;
+    if (i < n_extra) extra_zero[i] = 0;
+    if (i >= *count) return;
+    const int slot = slots[i];
+    unsigned char from = states[slot];
+    const int chain[4] = {to0, to1, to2, to3};
+    #pragma unroll
+    for (int c = 0; c < 4; ++c) {
+        if (chain[c] < 0) break;
+        const unsigned char to = (unsigned char)chain[c];
+        if (!d_transition_lut[from * N_STATES + to]) break;
+        if (to == 3 && from != 3) attempts[slot] += 1;
+        from = to;
+    }
+    states[slot] = from;
+    if (from >= 6) deadlines[slot] = (long long)0x7fffffffffffffffLL;
 }
 
 __global__ __launch_bounds__(BLOCK) void echo_worker_indexed_dyn_kernel(
@@ -1149,6 +1243,47 @@ void compact_routable(torch::Tensor allowed_slots, torch::Tensor allowed_count,
         routable_widx.data_ptr<int>(), routable_count.data_ptr<int>());
 }
 
+void policy_gate_full(torch::Tensor first, torch::Tensor decisions, torch::Tensor out_decision,
+                      torch::Tensor denied_slots, torch::Tensor denied_count,
+                      torch::Tensor allowed_slots, torch::Tensor allowed_count,
+                      torch::Tensor states, torch::Tensor deadlines,
+                      torch::Tensor dlq_ring, torch::Tensor dlq_head)
+{
+    const int J = (int)first.size(0);
+    const int blocks = (J + BLOCK - 1) / BLOCK;
+    hipLaunchKernelGGL(policy_gate_full_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        first.data_ptr<int>(), (const signed char*)decisions.data_ptr<int8_t>(),
+        (signed char*)out_decision.data_ptr<int8_t>(),
+        denied_slots.data_ptr<int>(), denied_count.data_ptr<int>(),
+        allowed_slots.data_ptr<int>(), allowed_count.data_ptr<int>(),
+        states.data_ptr<uint8_t>(), (long long*)deadlines.data_ptr<int64_t>(),
+        dlq_ring.data_ptr<int>(), dlq_head.data_ptr<int>(),
+        (int)dlq_ring.size(0), J);
+}
+void tick_reset(torch::Tensor states, torch::Tensor counts)
+{
+    const int B = (int)states.size(0);
+    const int n = (int)counts.size(0);
+    const int blocks = (std::max(B, n) + BLOCK - 1) / BLOCK;
+    hipLaunchKernelGGL(tick_reset_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        states.data_ptr<uint8_t>(), B, counts.data_ptr<int>(), n);
+}
+void apply_transitions_chain_dyn(torch::Tensor states, torch::Tensor attempts,
+                                 torch::Tensor deadlines, torch::Tensor slots,
+                                 torch::Tensor count, std::vector<int64_t> chain,
+                                 torch::Tensor extra_zero, int64_t capacity)
+{
+    int to[4] = {-1, -1, -1, -1};
+    for (size_t c = 0; c < chain.size() && c < 4; ++c) to[c] = (int)chain[c];
+    const int n_extra = (int)extra_zero.size(0);
+    const int blocks = (std::max((int)capacity, n_extra) + BLOCK - 1) / BLOCK;
+    hipLaunchKernelGGL(apply_transitions_chain_dyn_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        states.data_ptr<uint8_t>(), attempts.data_ptr<int>(),
+        (long long*)deadlines.data_ptr<int64_t>(),
+        slots.data_ptr<int>(), count.data_ptr<int>(),
+        to[0], to[1], to[2], to[3],
+        extra_zero.data_ptr<int>(), n_extra);
+}
 void apply_transitions_dyn(torch::Tensor states, torch::Tensor attempts, torch::Tensor deadlines,
                            torch::Tensor slots, torch::Tensor count, int64_t to_state, int64_t capacity)
 {
@@ -1272,6 +1407,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("dlq_ring_append", &dlq_ring_append, "K7 device DLQ ring append");
     m.def("echo_execute_indexed", &echo_execute_indexed, "device echo worker pool (slot-indexed, in-place)");
     m.def("policy_gate", &policy_gate, "decision gather + allow/deny compaction");
+    m.def("policy_gate_full", &policy_gate_full,
+          "gate + DENIED transition + DLQ ring append in one launch");
+    m.def("tick_reset", &tick_reset, "one-launch per-tick state/counter reset");
+    m.def("apply_transitions_chain_dyn", &apply_transitions_chain_dyn,
+          "K5 chained targets in one launch (+ extra zero region)");
     m.def("compact_routable", &compact_routable, "routable-slot compaction");
     m.def("apply_transitions_dyn", &apply_transitions_dyn, "K5 with device-resident count");
     m.def("echo_execute_indexed_dyn", &echo_execute_indexed_dyn, "echo worker with device count");

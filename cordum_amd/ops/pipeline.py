@@ -202,6 +202,47 @@ class _RefOps:
         alc[0] = a_idx.numel()
         dc[0] = d_idx.numel()
 
+    def policy_gate_full(self, first, decisions, out_decision, ds, dc, als, alc,
+                         states, deadlines, dlq_ring, dlq_head):
+        self.policy_gate(first, decisions, out_decision, ds, dc, als, alc)
+        lut = self._lut
+        n = int(dc[0])
+        appended = []
+        for i in range(n):
+            j = int(ds[i])
+            frm = int(states[j])
+            if lut[frm * 11 + 10]:
+                states[j] = 10
+                deadlines[j] = 0x7FFFFFFFFFFFFFFF
+                appended.append(j)
+        h = int(dlq_head[0])
+        size = dlq_ring.shape[0]
+        for k, j in enumerate(appended):
+            dlq_ring[(h + k) % size] = j
+        dlq_head[0] = h + len(appended)
+
+    def tick_reset(self, states, counts):
+        states.zero_()
+        counts.zero_()
+
+    def apply_transitions_chain_dyn(self, states, attempts, deadlines, slots, count,
+                                    chain, extra_zero, capacity):
+        extra_zero.zero_()
+        lut = self._lut
+        n = int(count[0])
+        for i in range(n):
+            slot = int(slots[i])
+            frm = int(states[slot])
+            for to in chain:
+                if to < 0 or not lut[frm * 11 + to]:
+                    break
+                if to == 3 and frm != 3:
+                    attempts[slot] += 1
+                frm = to
+            states[slot] = frm
+            if frm >= 6:
+                deadlines[slot] = 0x7FFFFFFFFFFFFFFF
+
     def compact_routable(self, als, alc, pick, rs, rw, rc):
         n = int(alc[0])
         sel = als[:n].long()
@@ -394,7 +435,7 @@ class DevicePipeline:
         B = self.B
         ext = self.ext
         jb = self.batches[slot]
-        self.states.zero_()
+        ext.tick_reset(self.states, self._counts)
         pend = self._pend_states
         ext.apply_transitions(self.states, self.attempts, self.deadlines, self.slots, pend)
         if self._use_mfma:
@@ -409,33 +450,33 @@ class DevicePipeline:
                 self.cpol.mcp_masks, self.cpol.mcp_any,
                 jb.any_bits, jb.all_bits, jb.secrets, jb.mcp_bits, jb.mcp_used, 0,
             )
-        self._counts.zero_()
-        ext.policy_gate(first, self.cpol.decisions, self.out_decision,
-                        self.denied_slots, self.denied_count,
-                        self.allowed_slots, self.allowed_count)
-        ext.apply_transitions_dyn(self.states, self.attempts, self.deadlines,
-                                  self.denied_slots, self.denied_count, DENIED, B)
-        ext.dlq_ring_append(self.denied_slots, self.denied_count,
-                            self.dlq_ring, self.dlq_head, B)
-        self.w_active.copy_(self.w_active_local)
-        self.w_cpu.copy_(self.w_cpu_local)
-        self.w_gpu.copy_(self.w_gpu_local)
-        ext.worker_precompute_into(self.w_pool, self.w_active, self.w_maxp,
-                                   self.w_cpu, self.w_gpu, self.w_keys)
+        # gate + DENIED + DLQ ring fused into one launch; worker precompute
+        # reads the local load tensors directly (world==1, so the global view
+        # IS the local view — the three D2D copies were 12% of the tick,
+        # profiles/r14_bench_ktrace_stats.txt)
+        ext.policy_gate_full(first, self.cpol.decisions, self.out_decision,
+                             self.denied_slots, self.denied_count,
+                             self.allowed_slots, self.allowed_count,
+                             self.states, self.deadlines,
+                             self.dlq_ring, self.dlq_head)
+        ext.worker_precompute_into(self.w_pool, self.w_active_local, self.w_maxp,
+                                   self.w_cpu_local, self.w_gpu_local, self.w_keys)
         pick = ext.least_loaded_pick(self.w_pool, self.w_keys, self.w_labels,
                                      self.j_poolmask, self.j_labels)
         self._spread(pick)
         ext.compact_routable(self.allowed_slots, self.allowed_count, pick,
                              self.routable_slots, self.routable_widx, self.routable_count)
-        for st in (SCHEDULED, DISPATCHED, RUNNING):
-            ext.apply_transitions_dyn(self.states, self.attempts, self.deadlines,
-                                      self.routable_slots, self.routable_count, st, B)
         ext.echo_execute_indexed_dyn(self.payloads[slot], self.routable_slots,
                                      self.routable_count, self.res_arena[: B * self.payload_words],
                                      self.res_sums, self.payload_words, B)
-        ext.apply_transitions_dyn(self.states, self.attempts, self.deadlines,
-                                  self.routable_slots, self.routable_count, SUCCEEDED, B)
-        self.w_active_local.zero_()
+        # the intermediate SCHEDULED/DISPATCHED/RUNNING states are not
+        # observable inside a captured graph: march the chain in one launch
+        # (LUT-checked hop by hop, attempts++ on SCHEDULED entry), and zero
+        # the local-load accumulator in the same grid for load_feedback
+        ext.apply_transitions_chain_dyn(self.states, self.attempts, self.deadlines,
+                                        self.routable_slots, self.routable_count,
+                                        [SCHEDULED, DISPATCHED, RUNNING, SUCCEEDED],
+                                        self.w_active_local, B)
         ext.load_feedback(self.routable_widx, self.routable_count,
                           self.w_active_local, self.NWL, self.rank, B)
 
