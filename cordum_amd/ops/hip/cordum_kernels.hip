@@ -699,6 +699,55 @@ __global__ __launch_bounds__(BLOCK) void tick_reset_kernel(
     if (i < ncounts) counts[i] = 0;
 }
 
+// tick prologue in one launch: every ring slot is re-admitted as PENDING
+// (CREATED -> PENDING is the only LUT edge exercised by the old zero_() +
+// apply_transitions pair, so write it directly) and the counter words reset
+__global__ __launch_bounds__(BLOCK) void begin_tick_kernel(
+    unsigned char* __restrict__ states, int B,
+    int* __restrict__ counts, int ncounts)
+{
+    const int i = blockIdx.x * BLOCK + threadIdx.x;
+    if (i < B) states[i] = 1;  // PENDING
+    if (i < ncounts) counts[i] = 0;
+}
+
+// routable compaction with the K2c spread computed inline: spread_pick keyed
+// the round-robin on the slot index j, which this kernel already has — so a
+// separate spread launch over all B slots was redundant work + launch tax
+__global__ __launch_bounds__(BLOCK) void compact_routable_spread_kernel(
+    const int* __restrict__ allowed_slots,    // [<=J]
+    const int* __restrict__ allowed_count,    // [1]
+    const int* __restrict__ pick,             // [J] exact least-loaded pick
+    const int* __restrict__ order,            // [NW] workers by key asc
+    const int* __restrict__ valid_count,      // [1] non-overloaded workers
+    const long long* __restrict__ j_poolmask,
+    const long long* __restrict__ j_labels,
+    long long full_mask,
+    int K,
+    int* __restrict__ routable_slots,         // out
+    int* __restrict__ routable_widx,          // out (global worker idx)
+    int* __restrict__ routable_count)         // [1]
+{
+    const int i = blockIdx.x * BLOCK + threadIdx.x;
+    const int lane = threadIdx.x % WAVE;
+    const bool live = i < *allowed_count;
+    int j = 0, w = -1;
+    if (live) {
+        j = allowed_slots[i];
+        if (j_labels[j] == 0 && j_poolmask[j] == full_mask) {
+            const int V = min(*valid_count, K);
+            w = (V > 0) ? order[j % V] : pick[j];
+        } else {
+            w = pick[j];  // constrained: keep the exact scan pick
+        }
+    }
+    const int pos = wave_append_slot(live && w >= 0, routable_count, lane);
+    if (pos >= 0) {
+        routable_slots[pos] = j;
+        routable_widx[pos] = w;
+    }
+}
+
 // routable compaction: allowed jobs with a worker pick
 __global__ __launch_bounds__(BLOCK) void compact_routable_kernel(
     const int* __restrict__ allowed_slots,    // [<=J]
@@ -1268,6 +1317,32 @@ void tick_reset(torch::Tensor states, torch::Tensor counts)
     hipLaunchKernelGGL(tick_reset_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
         states.data_ptr<uint8_t>(), B, counts.data_ptr<int>(), n);
 }
+void begin_tick(torch::Tensor states, torch::Tensor counts)
+{
+    const int B = (int)states.size(0);
+    const int n = (int)counts.size(0);
+    const int blocks = (std::max(B, n) + BLOCK - 1) / BLOCK;
+    hipLaunchKernelGGL(begin_tick_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        states.data_ptr<uint8_t>(), B, counts.data_ptr<int>(), n);
+}
+void compact_routable_spread(torch::Tensor allowed_slots, torch::Tensor allowed_count,
+                             torch::Tensor pick, torch::Tensor order, torch::Tensor valid_count,
+                             torch::Tensor j_poolmask, torch::Tensor j_labels,
+                             int64_t full_mask, int64_t K,
+                             torch::Tensor routable_slots, torch::Tensor routable_widx,
+                             torch::Tensor routable_count)
+{
+    const int J = (int)allowed_slots.size(0);
+    const int blocks = (J + BLOCK - 1) / BLOCK;
+    hipLaunchKernelGGL(compact_routable_spread_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        allowed_slots.data_ptr<int>(), allowed_count.data_ptr<int>(),
+        pick.data_ptr<int>(), order.data_ptr<int>(), valid_count.data_ptr<int>(),
+        (const long long*)j_poolmask.data_ptr<int64_t>(),
+        (const long long*)j_labels.data_ptr<int64_t>(),
+        (long long)full_mask, (int)K,
+        routable_slots.data_ptr<int>(), routable_widx.data_ptr<int>(),
+        routable_count.data_ptr<int>());
+}
 void apply_transitions_chain_dyn(torch::Tensor states, torch::Tensor attempts,
                                  torch::Tensor deadlines, torch::Tensor slots,
                                  torch::Tensor count, std::vector<int64_t> chain,
@@ -1410,6 +1485,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("policy_gate_full", &policy_gate_full,
           "gate + DENIED transition + DLQ ring append in one launch");
     m.def("tick_reset", &tick_reset, "one-launch per-tick state/counter reset");
+    m.def("begin_tick", &begin_tick, "tick prologue: PENDING re-admit + counter reset");
+    m.def("compact_routable_spread", &compact_routable_spread,
+          "routable compaction with inline K2c spread");
     m.def("apply_transitions_chain_dyn", &apply_transitions_chain_dyn,
           "K5 chained targets in one launch (+ extra zero region)");
     m.def("compact_routable", &compact_routable, "routable-slot compaction");

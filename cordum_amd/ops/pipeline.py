@@ -225,6 +225,20 @@ class _RefOps:
         states.zero_()
         counts.zero_()
 
+    def begin_tick(self, states, counts):
+        states.fill_(1)  # PENDING
+        counts.zero_()
+
+    def compact_routable_spread(self, als, alc, pick, order, valid_count,
+                                j_poolmask, j_labels, full_mask, K, rs, rw, rc):
+        eff = pick.clone()
+        V = min(int(valid_count[0]), K)
+        if V > 0:
+            uncon = (j_labels == 0) & (j_poolmask == full_mask)
+            idx = torch.arange(pick.shape[0], dtype=torch.int64)
+            eff[uncon] = order[(idx % V)[uncon]]
+        self.compact_routable(als, alc, eff, rs, rw, rc)
+
     def apply_transitions_chain_dyn(self, states, attempts, deadlines, slots, count,
                                     chain, extra_zero, capacity):
         extra_zero.zero_()
@@ -435,9 +449,10 @@ class DevicePipeline:
         B = self.B
         ext = self.ext
         jb = self.batches[slot]
-        ext.tick_reset(self.states, self._counts)
-        pend = self._pend_states
-        ext.apply_transitions(self.states, self.attempts, self.deadlines, self.slots, pend)
+        # prologue: every ring slot re-admitted PENDING + counters reset, one
+        # launch (replaces zero_() + full-LUT apply_transitions: the only edge
+        # that pair ever exercised was CREATED -> PENDING)
+        ext.begin_tick(self.states, self._counts)
         if self._use_mfma:
             first = ext.policy_first_match_mfma(
                 self.mfma_a_packs[slot], self.mfma_policy.b_pack, self.mfma_policy.cards,
@@ -463,9 +478,15 @@ class DevicePipeline:
                                    self.w_cpu_local, self.w_gpu_local, self.w_keys)
         pick = ext.least_loaded_pick(self.w_pool, self.w_keys, self.w_labels,
                                      self.j_poolmask, self.j_labels)
-        self._spread(pick)
-        ext.compact_routable(self.allowed_slots, self.allowed_count, pick,
-                             self.routable_slots, self.routable_widx, self.routable_count)
+        # K2c spread computed inline in the compaction (keyed on the slot
+        # index, bit-identical to the standalone spread_pick launch)
+        full_mask = (1 << self.world) - 1
+        K = min(int(self.order_buf.shape[0]), 1024)
+        ext.compact_routable_spread(self.allowed_slots, self.allowed_count, pick,
+                                    self.order_buf, self.valid_buf,
+                                    self.j_poolmask, self.j_labels, full_mask, K,
+                                    self.routable_slots, self.routable_widx,
+                                    self.routable_count)
         ext.echo_execute_indexed_dyn(self.payloads[slot], self.routable_slots,
                                      self.routable_count, self.res_arena[: B * self.payload_words],
                                      self.res_sums, self.payload_words, B)
