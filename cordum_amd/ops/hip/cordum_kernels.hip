@@ -227,8 +227,34 @@ __global__ __launch_bounds__(BLOCK) void least_loaded_pick_kernel(
     const long long* __restrict__ j_poolmask, // [J]
     const long long* __restrict__ j_labels,   // [J] required labels
     int* __restrict__ out_pick,             // [J]
-    int NW, int NJ)
+    int NW, int NJ,
+    const int* __restrict__ valid_count,    // [1] or nullptr: skip gate
+    long long full_mask)
 {
+    // K2c short-circuit: when any worker is non-overloaded, unconstrained
+    // jobs take the spread (round-robin over the sorted order) in the
+    // compaction kernel — their exact scan pick is dead work. The full
+    // 1000-worker scan was 15.5% of the tick for picks that were
+    // overwritten (profiles/r15_bench_ktrace_stats.txt). -1 here matches
+    // the all-overloaded scan outcome (-2): both are "no direct route".
+    // (block-uniform check: the scan below has __syncthreads() barriers
+    // shared by the block's 4 waves, so only a whole-block skip is legal)
+    if (valid_count != nullptr && *valid_count > 0) {
+        const int jb = blockIdx.x * (BLOCK / WAVE);
+        bool all_skip = true;
+        for (int k = 0; k < BLOCK / WAVE; ++k) {
+            const int jj = jb + k;
+            if (jj < NJ && (j_labels[jj] != 0 || j_poolmask[jj] != full_mask)) {
+                all_skip = false;
+                break;
+            }
+        }
+        if (all_skip) {
+            const int jj = jb + threadIdx.x / WAVE;
+            if (jj < NJ && threadIdx.x % WAVE == 0) out_pick[jj] = -1;
+            return;
+        }
+    }
     // one WAVE per job: 64 lanes stride the worker table (staged through LDS
     // once per workgroup, shared by the 4 waves), then a wave min-reduce.
     constexpr int WCHUNK = 1024;
@@ -657,7 +683,7 @@ __global__ __launch_bounds__(BLOCK) void policy_gate_full_kernel(
     int* __restrict__ dlq_ring,               // [ring_size]
     int* __restrict__ dlq_head,               // [1] monotonic
     int ring_size,
-    int J)
+    int J, int R)
 {
     const int j = blockIdx.x * BLOCK + threadIdx.x;
     const int lane = threadIdx.x % WAVE;
@@ -665,7 +691,9 @@ __global__ __launch_bounds__(BLOCK) void policy_gate_full_kernel(
     signed char d = 0;
     if (live) {
         const int r = first[j];
-        d = (r >= 0) ? decisions[r] : (signed char)1;  // default allow
+        // no-match is -1 (host-masked) OR the raw INT_MAX atomicMin identity
+        // when the producer skipped the mask pass (_into variants)
+        d = (r >= 0 && r < R) ? decisions[r] : (signed char)1;  // default allow
         out_decision[j] = d;
     }
     const bool allowed = live && (d == 1 || d == 5);
@@ -704,10 +732,14 @@ __global__ __launch_bounds__(BLOCK) void tick_reset_kernel(
 // apply_transitions pair, so write it directly) and the counter words reset
 __global__ __launch_bounds__(BLOCK) void begin_tick_kernel(
     unsigned char* __restrict__ states, int B,
-    int* __restrict__ counts, int ncounts)
+    int* __restrict__ counts, int ncounts,
+    int* __restrict__ first)               // [B] primed for atomicMin, or null
 {
     const int i = blockIdx.x * BLOCK + threadIdx.x;
-    if (i < B) states[i] = 1;  // PENDING
+    if (i < B) {
+        states[i] = 1;  // PENDING
+        if (first != nullptr) first[i] = INT_MAX;
+    }
     if (i < ncounts) counts[i] = 0;
 }
 
@@ -1128,8 +1160,28 @@ torch::Tensor least_loaded_pick(
         (const long long*)w_labels.data_ptr<int64_t>(),
         (const long long*)j_poolmask.data_ptr<int64_t>(),
         (const long long*)j_labels.data_ptr<int64_t>(),
-        out.data_ptr<int>(), NW, NJ);
+        out.data_ptr<int>(), NW, NJ, (const int*)nullptr, 0LL);
     return out;
+}
+void least_loaded_pick_into(
+    torch::Tensor w_pool, torch::Tensor w_keys, torch::Tensor w_labels,
+    torch::Tensor j_poolmask, torch::Tensor j_labels,
+    torch::Tensor valid_count, int64_t full_mask, torch::Tensor out)
+{
+    CHECK_DEV(w_pool); CHECK_DEV(j_poolmask);
+    const int NW = (int)w_pool.size(0);
+    const int NJ = (int)j_poolmask.size(0);
+    if (NJ == 0) return;
+    const int jobs_per_block = BLOCK / WAVE;
+    const int blocks = (NJ + jobs_per_block - 1) / jobs_per_block;
+    hipLaunchKernelGGL(least_loaded_pick_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        w_pool.data_ptr<int>(),
+        (const unsigned long long*)w_keys.data_ptr<int64_t>(),
+        (const long long*)w_labels.data_ptr<int64_t>(),
+        (const long long*)j_poolmask.data_ptr<int64_t>(),
+        (const long long*)j_labels.data_ptr<int64_t>(),
+        out.data_ptr<int>(), NW, NJ,
+        valid_count.data_ptr<int>(), (long long)full_mask);
 }
 
 void dlq_ring_append(torch::Tensor slots, torch::Tensor count,
@@ -1307,7 +1359,7 @@ void policy_gate_full(torch::Tensor first, torch::Tensor decisions, torch::Tenso
         allowed_slots.data_ptr<int>(), allowed_count.data_ptr<int>(),
         states.data_ptr<uint8_t>(), (long long*)deadlines.data_ptr<int64_t>(),
         dlq_ring.data_ptr<int>(), dlq_head.data_ptr<int>(),
-        (int)dlq_ring.size(0), J);
+        (int)dlq_ring.size(0), J, (int)decisions.size(0));
 }
 void tick_reset(torch::Tensor states, torch::Tensor counts)
 {
@@ -1323,7 +1375,35 @@ void begin_tick(torch::Tensor states, torch::Tensor counts)
     const int n = (int)counts.size(0);
     const int blocks = (std::max(B, n) + BLOCK - 1) / BLOCK;
     hipLaunchKernelGGL(begin_tick_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
-        states.data_ptr<uint8_t>(), B, counts.data_ptr<int>(), n);
+        states.data_ptr<uint8_t>(), B, counts.data_ptr<int>(), n, (int*)nullptr);
+}
+void begin_tick_first(torch::Tensor states, torch::Tensor counts, torch::Tensor first)
+{
+    const int B = (int)states.size(0);
+    const int n = (int)counts.size(0);
+    const int blocks = (std::max(B, n) + BLOCK - 1) / BLOCK;
+    hipLaunchKernelGGL(begin_tick_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        states.data_ptr<uint8_t>(), B, counts.data_ptr<int>(), n, first.data_ptr<int>());
+}
+void policy_first_match_mfma_into(
+    torch::Tensor a_pack, torch::Tensor b_pack, torch::Tensor cards,
+    torch::Tensor rule_secrets, torch::Tensor job_secrets,
+    int64_t n_jobs, int64_t n_rules, torch::Tensor out)
+{
+    CHECK_DEV(a_pack); CHECK_DEV(b_pack);
+    const int J = (int)n_jobs, R = (int)n_rules;
+    const int Jt = (int)a_pack.size(0);
+    const int Rt = (R + 15) / 16;
+    if (J == 0 || R == 0) return;
+    int nchunks = std::max(1, std::min((Rt + 3) / 4, std::max(1, 2048 / std::max(Jt, 1))));
+    const int tiles_per_chunk = (Rt + nchunks - 1) / nchunks;
+    hipLaunchKernelGGL(policy_first_match_mfma_kernel, dim3(Jt, nchunks), dim3(BLOCK), 0, cur_stream(),
+        (const signed char*)a_pack.data_ptr<int8_t>(),
+        (const signed char*)b_pack.data_ptr<int8_t>(),
+        cards.data_ptr<int>(),
+        (const signed char*)rule_secrets.data_ptr<int8_t>(),
+        job_secrets.data_ptr<uint8_t>(),
+        out.data_ptr<int>(), J, R, tiles_per_chunk);
 }
 void compact_routable_spread(torch::Tensor allowed_slots, torch::Tensor allowed_count,
                              torch::Tensor pick, torch::Tensor order, torch::Tensor valid_count,
@@ -1486,6 +1566,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "gate + DENIED transition + DLQ ring append in one launch");
     m.def("tick_reset", &tick_reset, "one-launch per-tick state/counter reset");
     m.def("begin_tick", &begin_tick, "tick prologue: PENDING re-admit + counter reset");
+    m.def("begin_tick_first", &begin_tick_first,
+          "tick prologue that also primes the first-match buffer (INT_MAX)");
+    m.def("least_loaded_pick_into", &least_loaded_pick_into,
+          "K2 pick into a persistent buffer, skipping unconstrained jobs");
+    m.def("policy_first_match_mfma_into", &policy_first_match_mfma_into,
+          "K1 MFMA into a pre-primed buffer (no alloc/fill launches)");
     m.def("compact_routable_spread", &compact_routable_spread,
           "routable compaction with inline K2c spread");
     m.def("apply_transitions_chain_dyn", &apply_transitions_chain_dyn,

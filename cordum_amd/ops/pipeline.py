@@ -191,7 +191,10 @@ class _RefOps:
 
     def policy_gate(self, first, decisions, out_decision, ds, dc, als, alc):
         J = first.shape[0]
-        dec = torch.where(first >= 0, decisions.to(torch.int32)[first.clamp(min=0).long()],
+        R = decisions.shape[0]
+        matched = (first >= 0) & (first < R)  # INT_MAX = unprimed no-match
+        dec = torch.where(matched,
+                          decisions.to(torch.int32)[first.clamp(min=0, max=R - 1).long()],
                           torch.ones_like(first))
         out_decision.copy_(dec.to(torch.int8))
         allowed = (dec == 1) | (dec == 5)
@@ -228,6 +231,18 @@ class _RefOps:
     def begin_tick(self, states, counts):
         states.fill_(1)  # PENDING
         counts.zero_()
+
+    def begin_tick_first(self, states, counts, first):
+        self.begin_tick(states, counts)
+        first.fill_(2**31 - 1)
+
+    def least_loaded_pick_into(self, w_pool, w_keys, w_labels, j_poolmask,
+                               j_labels, valid_count, full_mask, out):
+        pick = self.least_loaded_pick(w_pool, w_keys, w_labels, j_poolmask, j_labels)
+        if int(valid_count[0]) > 0:
+            uncon = (j_labels == 0) & (j_poolmask == full_mask)
+            pick[uncon] = -1
+        out.copy_(pick)
 
     def compact_routable_spread(self, als, alc, pick, order, valid_count,
                                 j_poolmask, j_labels, full_mask, K, rs, rw, rc):
@@ -378,6 +393,9 @@ class DevicePipeline:
         # compaction lists + device-resident counts; batch content is copied
         # into staging tensors so one captured graph serves the whole ring
         self.out_decision = torch.zeros(self.B, dtype=torch.int8, device=device)
+        # persistent kernel outputs: primed/overwritten in-graph each tick
+        self.first_buf = torch.full((self.B,), -1, dtype=torch.int32, device=device)
+        self.pick_buf = torch.full((self.B,), -1, dtype=torch.int32, device=device)
         self.denied_slots = torch.zeros(self.B, dtype=torch.int32, device=device)
         # one backing tensor for the counters -> one D2H read per tick
         # [0]=denied [1]=allowed [2]=routable [3]=dispatched (padded path)
@@ -449,17 +467,22 @@ class DevicePipeline:
         B = self.B
         ext = self.ext
         jb = self.batches[slot]
-        # prologue: every ring slot re-admitted PENDING + counters reset, one
-        # launch (replaces zero_() + full-LUT apply_transitions: the only edge
-        # that pair ever exercised was CREATED -> PENDING)
-        ext.begin_tick(self.states, self._counts)
+        # prologue: every ring slot re-admitted PENDING + counters reset +
+        # first-match buffer primed to the atomicMin identity, one launch
+        # (replaces zero_() + full-LUT apply_transitions + the MFMA wrapper's
+        # alloc-fill: the only LUT edge the old pair exercised was
+        # CREATED -> PENDING, and the gate now treats out-of-range rule ids
+        # as no-match so the host-side mask pass is gone too)
         if self._use_mfma:
-            first = ext.policy_first_match_mfma(
+            ext.begin_tick_first(self.states, self._counts, self.first_buf)
+            ext.policy_first_match_mfma_into(
                 self.mfma_a_packs[slot], self.mfma_policy.b_pack, self.mfma_policy.cards,
                 self.mfma_policy.secrets, jb.secrets,
-                self.B, self.compiled.n_rules,
+                self.B, self.compiled.n_rules, self.first_buf,
             )
+            first = self.first_buf
         else:
+            ext.begin_tick(self.states, self._counts)
             first = ext.policy_first_match(
                 self.cpol.any_masks, self.cpol.all_masks, self.cpol.secrets,
                 self.cpol.mcp_masks, self.cpol.mcp_any,
@@ -476,11 +499,16 @@ class DevicePipeline:
                              self.dlq_ring, self.dlq_head)
         ext.worker_precompute_into(self.w_pool, self.w_active_local, self.w_maxp,
                                    self.w_cpu_local, self.w_gpu_local, self.w_keys)
-        pick = ext.least_loaded_pick(self.w_pool, self.w_keys, self.w_labels,
-                                     self.j_poolmask, self.j_labels)
+        # K2 exact pick only for constrained jobs (the spread overwrites the
+        # unconstrained ones; -1 for a skipped job ≡ the all-overloaded scan
+        # outcome, both unroutable when valid_count==0)
+        full_mask = (1 << self.world) - 1
+        ext.least_loaded_pick_into(self.w_pool, self.w_keys, self.w_labels,
+                                   self.j_poolmask, self.j_labels,
+                                   self.valid_buf, full_mask, self.pick_buf)
+        pick = self.pick_buf
         # K2c spread computed inline in the compaction (keyed on the slot
         # index, bit-identical to the standalone spread_pick launch)
-        full_mask = (1 << self.world) - 1
         K = min(int(self.order_buf.shape[0]), 1024)
         ext.compact_routable_spread(self.allowed_slots, self.allowed_count, pick,
                                     self.order_buf, self.valid_buf,
