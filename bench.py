@@ -93,17 +93,36 @@ def main() -> int:
         pipe.tick()
     barrier()
 
-    t0 = time.perf_counter()
-    step_times = []
-    completed = 0
-    denied = 0
-    for _ in range(args.steps):
-        st = pipe.tick()
-        step_times.append(st.wall_s)
-        completed += st.completed + st.denied  # denied jobs also fully decided
-        denied += st.denied
-    barrier()
-    elapsed = time.perf_counter() - t0
+    if use_gpu:
+        # sync-free timed window: stats fold into a device accumulator inside
+        # the captured tick (read ONCE after the closing barrier), per-step
+        # latency from hipEvents — the contract brackets the WHOLE window
+        # with barrier+synchronize, not every step
+        pipe.reset_stats()
+        events = [torch.cuda.Event(enable_timing=True) for _ in range(args.steps + 1)]
+        t0 = time.perf_counter()
+        events[0].record()
+        for s in range(args.steps):
+            pipe.tick_async()
+            events[s + 1].record()
+        barrier()
+        elapsed = time.perf_counter() - t0
+        completed, denied = pipe.collect_stats()
+        completed += denied  # denied jobs are also fully decided
+        step_times = [events[s].elapsed_time(events[s + 1]) / 1000.0
+                      for s in range(args.steps)]
+    else:
+        t0 = time.perf_counter()
+        step_times = []
+        completed = 0
+        denied = 0
+        for _ in range(args.steps):
+            st = pipe.tick()
+            step_times.append(st.wall_s)
+            completed += st.completed + st.denied  # denied jobs also fully decided
+            denied += st.denied
+        barrier()
+        elapsed = time.perf_counter() - t0
 
     # MAX elapsed over ranks; SUM of completed jobs over ranks
     if world_size > 1:

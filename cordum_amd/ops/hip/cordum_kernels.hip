@@ -743,6 +743,16 @@ __global__ __launch_bounds__(BLOCK) void begin_tick_kernel(
     if (i < ncounts) counts[i] = 0;
 }
 
+// end-of-tick stats fold: the per-tick D2H counts read was the tick's only
+// host sync (~15 us graph-replay + copy stall); accumulate on-device and let
+// the host read ONCE per timed window instead
+__global__ void accumulate_counts_kernel(
+    const int* __restrict__ counts, long long* __restrict__ acc, int n)
+{
+    const int t = threadIdx.x;
+    if (t < n) acc[t] += (long long)counts[t];
+}
+
 // routable compaction with the K2c spread computed inline: spread_pick keyed
 // the round-robin on the slot index j, which this kernel already has — so a
 // separate spread launch over all B slots was redundant work + launch tax
@@ -1369,6 +1379,12 @@ void tick_reset(torch::Tensor states, torch::Tensor counts)
     hipLaunchKernelGGL(tick_reset_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
         states.data_ptr<uint8_t>(), B, counts.data_ptr<int>(), n);
 }
+void accumulate_counts(torch::Tensor counts, torch::Tensor acc)
+{
+    const int n = (int)counts.size(0);
+    hipLaunchKernelGGL(accumulate_counts_kernel, dim3(1), dim3(WAVE), 0, cur_stream(),
+        counts.data_ptr<int>(), (long long*)acc.data_ptr<int64_t>(), n);
+}
 void begin_tick(torch::Tensor states, torch::Tensor counts)
 {
     const int B = (int)states.size(0);
@@ -1566,6 +1582,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "gate + DENIED transition + DLQ ring append in one launch");
     m.def("tick_reset", &tick_reset, "one-launch per-tick state/counter reset");
     m.def("begin_tick", &begin_tick, "tick prologue: PENDING re-admit + counter reset");
+    m.def("accumulate_counts", &accumulate_counts, "fold tick counters into device accumulator");
     m.def("begin_tick_first", &begin_tick_first,
           "tick prologue that also primes the first-match buffer (INT_MAX)");
     m.def("least_loaded_pick_into", &least_loaded_pick_into,

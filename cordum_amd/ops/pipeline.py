@@ -228,6 +228,9 @@ class _RefOps:
         states.zero_()
         counts.zero_()
 
+    def accumulate_counts(self, counts, acc):
+        acc += counts.to(torch.int64)
+
     def begin_tick(self, states, counts):
         states.fill_(1)  # PENDING
         counts.zero_()
@@ -393,6 +396,10 @@ class DevicePipeline:
         # compaction lists + device-resident counts; batch content is copied
         # into staging tensors so one captured graph serves the whole ring
         self.out_decision = torch.zeros(self.B, dtype=torch.int8, device=device)
+        # device-side running totals of the tick counters (read once per
+        # timed window by collect_stats — the per-tick D2H read was the
+        # tick's only host sync)
+        self.acc_counts = torch.zeros(4, dtype=torch.int64, device=device)
         # persistent kernel outputs: primed/overwritten in-graph each tick
         self.first_buf = torch.full((self.B,), -1, dtype=torch.int32, device=device)
         self.pick_buf = torch.full((self.B,), -1, dtype=torch.int32, device=device)
@@ -528,6 +535,7 @@ class DevicePipeline:
                                         self.w_active_local, B)
         ext.load_feedback(self.routable_widx, self.routable_count,
                           self.w_active_local, self.NWL, self.rank, B)
+        ext.accumulate_counts(self._counts, self.acc_counts)
 
     def _ensure_graphs(self) -> None:
         """Capture every ring slot's graph once (first tick), so no capture
@@ -550,9 +558,11 @@ class DevicePipeline:
             self._graphs[slot] = g
         self._graph = self._graphs[0]  # marker: fused path active
         # the eager warmup pass above ran the body for real, so it appended to
-        # the durable DLQ ring — reset so the ring only records counted ticks
+        # the durable DLQ ring and stats accumulator — reset so they only
+        # record counted ticks
         self.dlq_head.zero_()
         self.dlq_ring.fill_(-1)
+        self.acc_counts.zero_()
         torch.cuda.synchronize(self.device)
 
     def _tick_fused(self) -> TickStats:
@@ -583,6 +593,66 @@ class DevicePipeline:
         if self.world > 1:
             return self._tick_padded()
         return self._tick_eager()
+
+    # -- sync-free tick: stats stay on-device until collect_stats() ---------------
+    def reset_stats(self) -> None:
+        self.acc_counts.zero_()
+        self.total_completed = 0
+        self.total_denied = 0
+
+    def tick_async(self) -> None:
+        """One tick with NO host synchronization: counters fold into
+        acc_counts on-device (accumulate_counts kernel at the tail of the
+        captured body). The driver contract only requires a barrier +
+        synchronize around the whole timed window — the per-tick stats read
+        tick() does is a self-imposed ~15 us graph-replay stall."""
+        if getattr(self, "_fused_capable", False):
+            self._ensure_graphs()
+            i = self._tick % len(self.batches)
+            self._tick += 1
+            self._graphs[i].replay()
+            self._refresh_order()
+            return
+        if self.world > 1 and self.device.type == "cuda":
+            self._pad_alloc()
+            self._ensure_pad_graphs()
+            i = self._tick % len(self.batches)
+            self._tick += 1
+            if self._pad_graphs:
+                g1s, g2s, g3, g4 = self._pad_graphs
+                g1s[i].replay()
+                self._pad_heartbeats()
+                g2s[i].replay()
+                self._pad_exchange_out()
+                g3.replay()
+                self._pad_exchange_back()
+                g4.replay()
+            else:
+                self._pad_g1(i)
+                self._pad_heartbeats()
+                self._pad_g2(i)
+                self._pad_exchange_out()
+                self._pad_g3()
+                self._pad_exchange_back()
+                self._pad_g4()
+            self._refresh_order()
+            return
+        # CPU backends: run the stats-bearing tick and fold into the same
+        # accumulator so collect_stats() is backend-uniform (the padded path
+        # already accumulates inside _pad_g4 — only the eager path doesn't)
+        st = self.tick()
+        if self.world == 1:
+            self.acc_counts[0] += st.denied
+            self.acc_counts[2] += st.completed
+            self.acc_counts[3] += st.completed
+
+    def collect_stats(self):
+        """One D2H read of the device accumulator: (completed, denied) totals
+        since the last reset_stats()."""
+        acc = self.acc_counts.cpu()
+        denied = int(acc[0])
+        completed = int(acc[3]) if self.world > 1 else int(acc[2])
+        return completed, denied
 
     # -- multi-rank tick: fixed-capacity padded all-to-all ------------------------
     def _pad_alloc(self) -> None:
@@ -672,6 +742,7 @@ class DevicePipeline:
         self.w_active_local.zero_()
         ext.load_feedback_padded(self.pad_recv_widx, self.pad_recv_cnt,
                                  self.w_active_local, cap, world)
+        ext.accumulate_counts(self._counts, self.acc_counts)
 
     def _pad_heartbeats(self) -> None:
         if self.world > 1:
@@ -732,9 +803,11 @@ class DevicePipeline:
             g4 = torch.cuda.CUDAGraph()
             with torch.cuda.graph(g4):
                 self._pad_g4()
-            # reset the durable DLQ ring: the eager warmup appended for real
+            # reset the durable DLQ ring + stats accumulator: the eager
+            # warmup ran for real
             self.dlq_head.zero_()
             self.dlq_ring.fill_(-1)
+            self.acc_counts.zero_()
             torch.cuda.synchronize(self.device)
             self._pad_graphs = (g1s, g2s, g3, g4)
         except Exception:

@@ -109,3 +109,25 @@ def test_device_dlq_ring_collects_denied():
     assert sorted(set(last_slots)) == sorted(last_slots)  # unique slots
     for slot in last_slots:
         assert states[slot] == 10  # DENIED
+
+
+def test_tick_async_matches_tick_stats():
+    from cordum_amd.ops.pipeline import DevicePipeline
+
+    mk = lambda: DevicePipeline(device=torch.device("cuda:0"), batch_size=4096,
+                                n_local_workers=64, n_rules=512, n_batches=2,
+                                payload_words=8, seed=7)
+    a = mk()
+    sync_tot = [0, 0]
+    for _ in range(6):
+        st = a.tick()
+        sync_tot[0] += st.completed
+        sync_tot[1] += st.denied
+    b = mk()
+    b.tick()  # trigger capture (warmup resets the accumulator)
+    b.reset_stats()
+    for _ in range(6):
+        b.tick_async()
+    torch.cuda.synchronize()
+    completed, denied = b.collect_stats()
+    assert (completed, denied) == tuple(sync_tot)
