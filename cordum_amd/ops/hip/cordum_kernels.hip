@@ -611,6 +611,25 @@ __global__ __launch_bounds__(BLOCK) void policy_first_match_mfma_kernel(
             ok &= (rsec < 0) | (rsec == (signed char)jsec[r]);
             if (ok && rule < R) best[r] = min(best[r], rule);
         }
+
+        // first-match early exit: fold in other blocks'/waves' published
+        // finds (out_first only ever decreases, so a stale read just skips
+        // less) and stop this wave once every job row of the tile has SOME
+        // match — later tiles only yield larger rule ids, and a smaller id
+        // found elsewhere wins at the atomicMin anyway. Typical first match
+        // lands within a few tiles, so this cuts most of the R-dimension.
+        bool all_matched = true;
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            const int job = jt * 16 + row_base + r;
+            if (job < J)
+                best[r] = min(best[r], __hip_atomic_load(&out_first[job],
+                    __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT));
+            const unsigned long long m = __ballot(best[r] != INT_MAX);
+            all_matched &= ((m & 0xFFFFull) != 0) & (((m >> 16) & 0xFFFFull) != 0)
+                         & (((m >> 32) & 0xFFFFull) != 0) & (((m >> 48) & 0xFFFFull) != 0);
+        }
+        if (all_matched) break;
     }
 
     // min across the 16 rule columns (lanes sharing the same lane>>4 group)
