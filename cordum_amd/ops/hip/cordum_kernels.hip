@@ -612,19 +612,15 @@ __global__ __launch_bounds__(BLOCK) void policy_first_match_mfma_kernel(
             if (ok && rule < R) best[r] = min(best[r], rule);
         }
 
-        // first-match early exit: fold in other blocks'/waves' published
-        // finds (out_first only ever decreases, so a stale read just skips
-        // less) and stop this wave once every job row of the tile has SOME
-        // match — later tiles only yield larger rule ids, and a smaller id
-        // found elsewhere wins at the atomicMin anyway. Typical first match
-        // lands within a few tiles, so this cuts most of the R-dimension.
+        // first-match early exit (wave-local, register-only): stop once every
+        // job row of the tile has SOME match — later tiles only yield larger
+        // rule ids, and a smaller id found by another wave/block wins at the
+        // atomicMin anyway. A per-tile fold of out_first was measured NET
+        // NEGATIVE here (L2 atomic-load latency > tiles saved: 171.6M ->
+        // 145.4M jobs/s on the default config), so the check is 4 ballots.
         bool all_matched = true;
         #pragma unroll
         for (int r = 0; r < 4; ++r) {
-            const int job = jt * 16 + row_base + r;
-            if (job < J)
-                best[r] = min(best[r], __hip_atomic_load(&out_first[job],
-                    __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT));
             const unsigned long long m = __ballot(best[r] != INT_MAX);
             all_matched &= ((m & 0xFFFFull) != 0) & (((m >> 16) & 0xFFFFull) != 0)
                          & (((m >> 32) & 0xFFFFull) != 0) & (((m >> 48) & 0xFFFFull) != 0);
