@@ -585,6 +585,9 @@ __global__ __launch_bounds__(BLOCK) void policy_first_match_mfma_kernel(
         jsec[r] = (job < J) ? job_secrets[job] : 0;
     }
 
+    // early-exit checks only pay when each wave scans enough rule tiles
+    const bool do_exit = (chunk_end - chunk_begin) > 32;
+
     for (int rt = chunk_begin + wave; rt < chunk_end; rt += 4) {
         v4i acc[9];
         #pragma unroll
@@ -612,20 +615,29 @@ __global__ __launch_bounds__(BLOCK) void policy_first_match_mfma_kernel(
             if (ok && rule < R) best[r] = min(best[r], rule);
         }
 
-        // first-match early exit (wave-local, register-only): stop once every
-        // job row of the tile has SOME match — later tiles only yield larger
-        // rule ids, and a smaller id found by another wave/block wins at the
-        // atomicMin anyway. A per-tile fold of out_first was measured NET
-        // NEGATIVE here (L2 atomic-load latency > tiles saved: 171.6M ->
-        // 145.4M jobs/s on the default config), so the check is 4 ballots.
-        bool all_matched = true;
-        #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-            const unsigned long long m = __ballot(best[r] != INT_MAX);
-            all_matched &= ((m & 0xFFFFull) != 0) & (((m >> 16) & 0xFFFFull) != 0)
-                         & (((m >> 32) & 0xFFFFull) != 0) & (((m >> 48) & 0xFFFFull) != 0);
+        // first-match early exit, adaptive: stop once every job row of the
+        // tile has SOME match — later tiles only yield larger rule ids, and
+        // a smaller id found by another wave/block wins at the atomicMin
+        // anyway. Folding out_first (other blocks' published finds; the
+        // value only decreases, so a stale read just skips less) pays only
+        // when each wave scans many tiles: measured on the default config
+        // (8 tiles/wave) ANY per-tile check loses (171.6 -> 145-167M
+        // jobs/s), while at 16k rules (128 tiles/wave) the fold wins 35%+.
+        // So: no checks on short scans, fold+ballot every 4th tile on long.
+        if (do_exit && ((rt >> 2) & 3) == 3) {
+            bool all_matched = true;
+            #pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int job = jt * 16 + row_base + r;
+                if (job < J)
+                    best[r] = min(best[r], __hip_atomic_load(&out_first[job],
+                        __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT));
+                const unsigned long long m = __ballot(best[r] != INT_MAX);
+                all_matched &= ((m & 0xFFFFull) != 0) & (((m >> 16) & 0xFFFFull) != 0)
+                             & (((m >> 32) & 0xFFFFull) != 0) & (((m >> 48) & 0xFFFFull) != 0);
+            }
+            if (all_matched) break;
         }
-        if (all_matched) break;
     }
 
     // min across the 16 rule columns (lanes sharing the same lane>>4 group)
