@@ -29,14 +29,24 @@ def main():
     for _ in range(10):
         pipe.tick()
     rss0 = resource.getrusage(resource.RUSAGE_SELF).ru_maxrss
+    # production mode: sync-free ticks in windows of 256, stats from the
+    # device accumulator, per-step latency via hipEvents
+    pipe.reset_stats()
     times = []
     completed = 0
+    WINDOW = 256
     t0 = time.perf_counter()
     while time.perf_counter() - t0 < args.seconds:
-        st = pipe.tick()
-        times.append(st.wall_s)
-        completed += st.completed + st.denied
+        events = [torch.cuda.Event(enable_timing=True) for _ in range(WINDOW + 1)]
+        events[0].record()
+        for s_ in range(WINDOW):
+            pipe.tick_async()
+            events[s_ + 1].record()
+        torch.cuda.synchronize()
+        times.extend(events[k].elapsed_time(events[k + 1]) / 1e3 for k in range(WINDOW))
     elapsed = time.perf_counter() - t0
+    c, d = pipe.collect_stats()
+    completed = c + d
     rss1 = resource.getrusage(resource.RUSAGE_SELF).ru_maxrss
     times.sort()
     print(json.dumps({
