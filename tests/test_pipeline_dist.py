@@ -128,3 +128,46 @@ def test_bench_contract_torchrun_cpu(tmp_path):
     assert out["value"] > 0
     assert out["scaling"] == "weak"
     assert out["config"]["global_batch"] == 512
+
+
+def _worker_async(rank, world, port, result_dir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    import torch.distributed as dist
+
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    from cordum_amd.ops.pipeline import DevicePipeline
+
+    def mk():
+        return DevicePipeline(
+            device="cpu", batch_size=256, n_local_workers=16, n_rules=64,
+            payload_words=8, world_size=world, rank=rank, n_batches=2, backend="ref",
+        )
+
+    a = mk()
+    sync_tot = [0, 0]
+    for _ in range(3):
+        st = a.tick()
+        sync_tot[0] += st.completed
+        sync_tot[1] += st.denied
+    b = mk()
+    b.reset_stats()
+    for _ in range(3):
+        b.tick_async()
+    c, d = b.collect_stats()
+    with open(os.path.join(result_dir, f"rank{rank}.json"), "w") as f:
+        json.dump({"sync": sync_tot, "async": [c, d]}, f)
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+def test_tick_async_matches_tick_across_ranks(tmp_path):
+    """The sync-free accumulator path (what bench.py times) must count
+    exactly what the per-tick stats path counts, on the world>1 padded
+    exchange as well."""
+    port = 29531
+    mp.spawn(_worker_async, args=(2, port, str(tmp_path)), nprocs=2, join=True)
+    for rank in range(2):
+        with open(tmp_path / f"rank{rank}.json") as f:
+            r = json.load(f)
+        assert r["async"] == r["sync"], r
