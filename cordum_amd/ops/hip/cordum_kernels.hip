@@ -559,6 +559,7 @@ __global__ __launch_bounds__(BLOCK) void policy_first_match_mfma_kernel(
     const int* __restrict__ cards,            // [Rt*16][9]
     const signed char* __restrict__ rule_secrets, // [Rt*16]
     const unsigned char* __restrict__ job_secrets, // [J]
+    const int* __restrict__ tile_dims,        // [Rt] bitmask of live dims
     int* __restrict__ out_first,              // [J] pre-filled INT_MAX
     int J, int R, int tiles_per_chunk)
 {
@@ -589,30 +590,31 @@ __global__ __launch_bounds__(BLOCK) void policy_first_match_mfma_kernel(
     const bool do_exit = (chunk_end - chunk_begin) > 32;
 
     for (int rt = chunk_begin + wave; rt < chunk_end; rt += 4) {
-        v4i acc[9];
+        // dim skipping: most rules constrain only 2-4 of the 9 set dims; a
+        // dim whose card is 0 for EVERY rule in the tile auto-passes (any-of:
+        // card==0 | _; all-of: 0==0), so its MFMA + B-fragment load is dead.
+        // tile_dims is the per-tile union, uniform across the wave.
+        const unsigned dmask = (unsigned)tile_dims[rt];
+        const int rule = rt * 16 + col;
+        const signed char rsec = rule_secrets[rule];
+        bool ok[4] = {true, true, true, true};
         #pragma unroll
         for (int d = 0; d < 9; ++d) {
+            if (!(dmask & (1u << d))) continue;
             const v4i bfrag = *(const v4i*)&b_pack[(((size_t)rt * 9 + d) * 64 + lane) * 16];
             v4i zero = {0, 0, 0, 0};
-            acc[d] = __builtin_amdgcn_mfma_i32_16x16x64_i8(afrag[d], bfrag, zero, 0, 0, 0);
+            const v4i acc = __builtin_amdgcn_mfma_i32_16x16x64_i8(afrag[d], bfrag, zero, 0, 0, 0);
+            const int card = cards[rule * 9 + d];
+            #pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                if (d < 7) ok[r] &= (card == 0) | (acc[r] > 0);   // any-of
+                else       ok[r] &= (acc[r] == card);             // all-of
+            }
         }
-        const int rule = rt * 16 + col;
-        int card[9];
-        #pragma unroll
-        for (int d = 0; d < 9; ++d) card[d] = cards[rule * 9 + d];
-        const signed char rsec = rule_secrets[rule];
-
         #pragma unroll
         for (int r = 0; r < 4; ++r) {
-            bool ok = true;
-            #pragma unroll
-            for (int d = 0; d < 7; ++d)   // any-of dims
-                ok &= (card[d] == 0) | (acc[d][r] > 0);
-            #pragma unroll
-            for (int d = 7; d < 9; ++d)   // all-of dims
-                ok &= (acc[d][r] == card[d]);
-            ok &= (rsec < 0) | (rsec == (signed char)jsec[r]);
-            if (ok && rule < R) best[r] = min(best[r], rule);
+            const bool pass = ok[r] & ((rsec < 0) | (rsec == (signed char)jsec[r]));
+            if (pass && rule < R) best[r] = min(best[r], rule);
         }
 
         // first-match early exit, adaptive: stop once every job row of the
@@ -1333,7 +1335,7 @@ torch::Tensor echo_execute_indexed(torch::Tensor ctx_arena, torch::Tensor slots,
 torch::Tensor policy_first_match_mfma(
     torch::Tensor a_pack, torch::Tensor b_pack, torch::Tensor cards,
     torch::Tensor rule_secrets, torch::Tensor job_secrets,
-    int64_t n_jobs, int64_t n_rules)
+    torch::Tensor tile_dims, int64_t n_jobs, int64_t n_rules)
 {
     CHECK_DEV(a_pack); CHECK_DEV(b_pack);
     const int J = (int)n_jobs, R = (int)n_rules;
@@ -1351,6 +1353,7 @@ torch::Tensor policy_first_match_mfma(
         cards.data_ptr<int>(),
         (const signed char*)rule_secrets.data_ptr<int8_t>(),
         job_secrets.data_ptr<uint8_t>(),
+        tile_dims.data_ptr<int>(),
         out.data_ptr<int>(), J, R, tiles_per_chunk);
     out.masked_fill_(out == INT_MAX, -1);
     return out;
@@ -1431,7 +1434,7 @@ void begin_tick_first(torch::Tensor states, torch::Tensor counts, torch::Tensor 
 void policy_first_match_mfma_into(
     torch::Tensor a_pack, torch::Tensor b_pack, torch::Tensor cards,
     torch::Tensor rule_secrets, torch::Tensor job_secrets,
-    int64_t n_jobs, int64_t n_rules, torch::Tensor out)
+    torch::Tensor tile_dims, int64_t n_jobs, int64_t n_rules, torch::Tensor out)
 {
     CHECK_DEV(a_pack); CHECK_DEV(b_pack);
     const int J = (int)n_jobs, R = (int)n_rules;
@@ -1446,6 +1449,7 @@ void policy_first_match_mfma_into(
         cards.data_ptr<int>(),
         (const signed char*)rule_secrets.data_ptr<int8_t>(),
         job_secrets.data_ptr<uint8_t>(),
+        tile_dims.data_ptr<int>(),
         out.data_ptr<int>(), J, R, tiles_per_chunk);
 }
 void compact_routable_spread(torch::Tensor allowed_slots, torch::Tensor allowed_count,

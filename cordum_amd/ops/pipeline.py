@@ -484,7 +484,7 @@ class DevicePipeline:
             ext.begin_tick_first(self.states, self._counts, self.first_buf)
             ext.policy_first_match_mfma_into(
                 self.mfma_a_packs[slot], self.mfma_policy.b_pack, self.mfma_policy.cards,
-                self.mfma_policy.secrets, jb.secrets,
+                self.mfma_policy.secrets, jb.secrets, self.mfma_policy.tile_dims,
                 self.B, self.compiled.n_rules, self.first_buf,
             )
             first = self.first_buf

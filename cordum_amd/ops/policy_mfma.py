@@ -79,11 +79,13 @@ class MfmaPolicy:
     b_pack: torch.Tensor  # [Rt, 9, 64, 16] int8
     cards: torch.Tensor  # [Rt*16, 9] int32 (padded rules have impossible cards)
     secrets: torch.Tensor  # [Rt*16] int8
+    tile_dims: torch.Tensor  # [Rt] int32 bitmask: dims any rule in the tile uses
     n_rules: int
 
     def to(self, device):
         return MfmaPolicy(self.b_pack.to(device), self.cards.to(device),
-                          self.secrets.to(device), self.n_rules)
+                          self.secrets.to(device), self.tile_dims.to(device),
+                          self.n_rules)
 
 
 def pack_policy_mfma(c: CompiledPolicy) -> MfmaPolicy:
@@ -100,7 +102,12 @@ def pack_policy_mfma(c: CompiledPolicy) -> MfmaPolicy:
         cards[R:, 7] = 127
     secrets = torch.full((Rt * 16,), -1, dtype=torch.int8)
     secrets[:R] = c.secrets
-    return MfmaPolicy(b_pack, cards, secrets, R)
+    # per-tile union of constrained dims: a dim with card==0 for every rule
+    # in the tile auto-passes, so the kernel skips its MFMA + B load
+    used = (cards.view(Rt, 16, N_DIMS) != 0).any(dim=1)  # [Rt, 9]
+    tile_dims = (used.to(torch.int32) << torch.arange(N_DIMS, dtype=torch.int32)).sum(
+        dim=1).to(torch.int32)
+    return MfmaPolicy(b_pack, cards, secrets, tile_dims, R)
 
 
 def pack_jobs_mfma(jobs: JobBatch) -> Tuple[torch.Tensor, torch.Tensor]:
@@ -112,6 +119,6 @@ def pack_jobs_mfma(jobs: JobBatch) -> Tuple[torch.Tensor, torch.Tensor]:
 def first_match_mfma(ext, mp: MfmaPolicy, a_pack: torch.Tensor,
                      job_secrets: torch.Tensor, n_jobs: int) -> torch.Tensor:
     return ext.policy_first_match_mfma(
-        a_pack, mp.b_pack, mp.cards, mp.secrets, job_secrets,
+        a_pack, mp.b_pack, mp.cards, mp.secrets, job_secrets, mp.tile_dims,
         int(n_jobs), int(mp.n_rules),
     )

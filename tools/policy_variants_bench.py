@@ -42,7 +42,7 @@ def main():
 
         def run_mfma():
             return ext.policy_first_match_mfma(a_pack, mp.b_pack, mp.cards,
-                                               mp.secrets, jsec, J, R)
+                                               mp.secrets, jsec, mp.tile_dims, J, R)
 
         o1, o2 = run_bitset(), run_mfma()
         torch.cuda.synchronize()
