@@ -745,10 +745,17 @@ class DevicePipeline:
         ext.accumulate_counts(self._counts, self.acc_counts)
 
     def _pad_heartbeats(self) -> None:
+        """Worker-load fan-in. Only the active-job vector is live per tick;
+        cpu/gpu utilization changes on the reference's 10 s heartbeat cadence,
+        so those two all-gathers run once and then only on the every-8-tick
+        refresh beat (xGMI ring collectives are per-link bound — fewer, larger
+        gathers beat three small ones every tick)."""
         if self.world > 1:
             dist.all_gather_into_tensor(self.w_active, self.w_active_local)
-            dist.all_gather_into_tensor(self.w_cpu, self.w_cpu_local)
-            dist.all_gather_into_tensor(self.w_gpu, self.w_gpu_local)
+            if not getattr(self, "_hb_static_done", False) or self._tick % 8 == 1:
+                dist.all_gather_into_tensor(self.w_cpu, self.w_cpu_local)
+                dist.all_gather_into_tensor(self.w_gpu, self.w_gpu_local)
+                self._hb_static_done = True
         else:
             self.w_active.copy_(self.w_active_local)
             self.w_cpu.copy_(self.w_cpu_local)
