@@ -660,124 +660,6 @@ __global__ __launch_bounds__(BLOCK) void policy_first_match_mfma_kernel(
     }
 }
 
-// Block-cooperative MFMA K1: 4 job tiles per block share ONE rule-tile
-// stream staged through LDS. The strided variant above re-reads each B
-// fragment once per wave (4x), and the kernel is L2-bandwidth-bound on that
-// stream (dim-skip barely moved it: profiles/r18_bench_ktrace_stats.txt) —
-// staging the live dims in LDS divides B traffic by the 4 waves.
-// Early exit: per-wave ballot as above; a wave that finishes raises s_done
-// and keeps hitting the barriers (never a divergent __syncthreads), the
-// whole block exits once all 4 waves are done.
-__global__ __launch_bounds__(BLOCK) void policy_first_match_mfma_lds_kernel(
-    const signed char* __restrict__ a_pack,   // [Jt][9][64][16]
-    const signed char* __restrict__ b_pack,   // [Rt][9][64][16]
-    const int* __restrict__ cards,            // [Rt*16][9]
-    const signed char* __restrict__ rule_secrets, // [Rt*16]
-    const unsigned char* __restrict__ job_secrets, // [J]
-    const int* __restrict__ tile_dims,        // [Rt]
-    int* __restrict__ out_first,              // [J] pre-filled INT_MAX
-    int J, int R, int tiles_per_chunk, int Jt)
-{
-    const int wave = threadIdx.x / WAVE;
-    const int lane = threadIdx.x % WAVE;
-    const int jt_raw = blockIdx.x * 4 + wave;
-    const bool jt_live = jt_raw < Jt;
-    const int jt = jt_live ? jt_raw : (Jt - 1);   // clamp: loads stay in-range
-    const int Rt = (R + 15) / 16;
-    const int chunk_begin = blockIdx.y * tiles_per_chunk;
-    const int chunk_end = min(Rt, chunk_begin + tiles_per_chunk);
-
-    __shared__ signed char s_b[9][64 * 16];       // 9 KB staged rule tile
-    __shared__ int s_done;
-
-    v4i afrag[9];
-    #pragma unroll
-    for (int d = 0; d < 9; ++d)
-        afrag[d] = *(const v4i*)&a_pack[(((size_t)jt * 9 + d) * 64 + lane) * 16];
-
-    const int row_base = (lane >> 4) * 4;
-    const int col = lane & 15;
-    int best[4] = {INT_MAX, INT_MAX, INT_MAX, INT_MAX};
-    unsigned char jsec[4];
-    #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-        const int job = jt * 16 + row_base + r;
-        jsec[r] = (job < J) ? job_secrets[job] : 0;
-    }
-
-    if (threadIdx.x == 0) s_done = 0;
-    __syncthreads();
-    bool wave_done = !jt_live;
-    if (wave_done && lane == 0) atomicAdd(&s_done, 1);
-
-    int it = 0;
-    for (int rt = chunk_begin; rt < chunk_end; ++rt, ++it) {
-        const unsigned dmask = (unsigned)tile_dims[rt];
-        __syncthreads();                          // previous tile's reads done
-        if (s_done == 4) break;                   // uniform: read post-barrier
-        #pragma unroll
-        for (int d = 0; d < 9; ++d) {             // cooperative 1 KB/dim loads
-            if (!(dmask & (1u << d))) continue;
-            ((int*)s_b[d])[threadIdx.x] =
-                ((const int*)&b_pack[(((size_t)rt * 9 + d) * 64) * 16])[threadIdx.x];
-        }
-        __syncthreads();
-        if (wave_done) continue;
-
-        const int rule = rt * 16 + col;
-        const signed char rsec = rule_secrets[rule];
-        bool ok[4] = {true, true, true, true};
-        #pragma unroll
-        for (int d = 0; d < 9; ++d) {
-            if (!(dmask & (1u << d))) continue;
-            const v4i bfrag = *(const v4i*)&s_b[d][lane * 16];
-            v4i zero = {0, 0, 0, 0};
-            const v4i acc = __builtin_amdgcn_mfma_i32_16x16x64_i8(afrag[d], bfrag, zero, 0, 0, 0);
-            const int card = cards[rule * 9 + d];
-            #pragma unroll
-            for (int r = 0; r < 4; ++r) {
-                if (d < 7) ok[r] &= (card == 0) | (acc[r] > 0);
-                else       ok[r] &= (acc[r] == card);
-            }
-        }
-        bool all_matched = true;
-        #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-            const bool pass = ok[r] & ((rsec < 0) | (rsec == (signed char)jsec[r]));
-            if (pass && rule < R) best[r] = min(best[r], rule);
-            if ((it & 3) == 3) {
-                const int job = jt * 16 + row_base + r;
-                if (jt_live && job < J)
-                    best[r] = min(best[r], __hip_atomic_load(&out_first[job],
-                        __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT));
-            }
-            const unsigned long long m = __ballot(best[r] != INT_MAX);
-            all_matched &= ((m & 0xFFFFull) != 0) & (((m >> 16) & 0xFFFFull) != 0)
-                         & (((m >> 32) & 0xFFFFull) != 0) & (((m >> 48) & 0xFFFFull) != 0);
-        }
-        if (all_matched) {
-            wave_done = true;
-            if (lane == 0) atomicAdd(&s_done, 1);
-        }
-    }
-
-    #pragma unroll
-    for (int off = 8; off > 0; off >>= 1) {
-        #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-            const int other = __shfl_xor(best[r], off, WAVE);
-            best[r] = min(best[r], other);
-        }
-    }
-    if ((lane & 15) == 0 && jt_live) {
-        #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-            const int job = jt * 16 + row_base + r;
-            if (job < J && best[r] != INT_MAX) atomicMin(&out_first[job], best[r]);
-        }
-    }
-}
-
 // ---------------------------------------------------------------------------
 // Tick-fusion kernels: device-side compaction + count-pointer variants so a
 // whole single-GPU control-plane tick is a fixed kernel sequence (no host
@@ -1462,19 +1344,17 @@ torch::Tensor policy_first_match_mfma(
     auto out = torch::full({J}, INT_MAX,
         torch::TensorOptions().dtype(torch::kInt32).device(a_pack.device()));
     if (J == 0 || R == 0) return out.masked_fill_(out == INT_MAX, -1);
-    // 4 job tiles per block (LDS-shared rule stream); fill the chip with
-    // >=2048 workgroups via rule chunking
-    const int gx = (Jt + 3) / 4;
-    int nchunks = std::max(1, std::min(Rt, (2048 + gx - 1) / std::max(gx, 1)));
+    // fill the chip: >=2048 workgroups via rule chunking
+    int nchunks = std::max(1, std::min((Rt + 3) / 4, std::max(1, 2048 / std::max(Jt, 1))));
     const int tiles_per_chunk = (Rt + nchunks - 1) / nchunks;
-    hipLaunchKernelGGL(policy_first_match_mfma_lds_kernel, dim3(gx, nchunks), dim3(BLOCK), 0, cur_stream(),
+    hipLaunchKernelGGL(policy_first_match_mfma_kernel, dim3(Jt, nchunks), dim3(BLOCK), 0, cur_stream(),
         (const signed char*)a_pack.data_ptr<int8_t>(),
         (const signed char*)b_pack.data_ptr<int8_t>(),
         cards.data_ptr<int>(),
         (const signed char*)rule_secrets.data_ptr<int8_t>(),
         job_secrets.data_ptr<uint8_t>(),
         tile_dims.data_ptr<int>(),
-        out.data_ptr<int>(), J, R, tiles_per_chunk, Jt);
+        out.data_ptr<int>(), J, R, tiles_per_chunk);
     out.masked_fill_(out == INT_MAX, -1);
     return out;
 }
@@ -1561,17 +1441,16 @@ void policy_first_match_mfma_into(
     const int Jt = (int)a_pack.size(0);
     const int Rt = (R + 15) / 16;
     if (J == 0 || R == 0) return;
-    const int gx = (Jt + 3) / 4;
-    int nchunks = std::max(1, std::min(Rt, (2048 + gx - 1) / std::max(gx, 1)));
+    int nchunks = std::max(1, std::min((Rt + 3) / 4, std::max(1, 2048 / std::max(Jt, 1))));
     const int tiles_per_chunk = (Rt + nchunks - 1) / nchunks;
-    hipLaunchKernelGGL(policy_first_match_mfma_lds_kernel, dim3(gx, nchunks), dim3(BLOCK), 0, cur_stream(),
+    hipLaunchKernelGGL(policy_first_match_mfma_kernel, dim3(Jt, nchunks), dim3(BLOCK), 0, cur_stream(),
         (const signed char*)a_pack.data_ptr<int8_t>(),
         (const signed char*)b_pack.data_ptr<int8_t>(),
         cards.data_ptr<int>(),
         (const signed char*)rule_secrets.data_ptr<int8_t>(),
         job_secrets.data_ptr<uint8_t>(),
         tile_dims.data_ptr<int>(),
-        out.data_ptr<int>(), J, R, tiles_per_chunk, Jt);
+        out.data_ptr<int>(), J, R, tiles_per_chunk);
 }
 void compact_routable_spread(torch::Tensor allowed_slots, torch::Tensor allowed_count,
                              torch::Tensor pick, torch::Tensor order, torch::Tensor valid_count,
