@@ -278,3 +278,86 @@ def test_padded_dispatch_kernels_match_reference(ext):
     wal = torch.zeros(nwl, dtype=torch.int32, device=d)
     ext.load_feedback_padded(sw_d, sc_d, wal, B, world)
     assert torch.equal(wal.cpu(), wal_r)
+
+
+def test_fused_tick_kernels_match_refops(ext):
+    """Direct unit checks of the launch-fusion kernels (policy_gate_full,
+    apply_transitions_chain_dyn, begin_tick, compact_routable_spread,
+    accumulate_counts) against the CPU _RefOps oracles on random data."""
+    from cordum_amd.ops.pipeline import _RefOps
+
+    torch.manual_seed(5)
+    d = dev()
+    ref = _RefOps()
+    lut = torch.tensor(transition_lut(), dtype=torch.uint8).flatten()
+    ref.set_transition_lut(lut)
+    ext.set_transition_lut(lut)
+    B, R = 2048, 96
+
+    # -- policy_gate_full ----------------------------------------------------
+    first = torch.where(torch.rand(B) < 0.7,
+                        torch.randint(0, R, (B,), dtype=torch.int32),
+                        torch.tensor(2**31 - 1, dtype=torch.int32))
+    decisions = torch.randint(1, 6, (R,), dtype=torch.int8)
+
+    def gate_state():
+        states = torch.ones(B, dtype=torch.uint8)  # PENDING
+        deadlines = torch.randint(1, 2**40, (B,), dtype=torch.int64)
+        out_dec = torch.zeros(B, dtype=torch.int8)
+        ds = torch.zeros(B, dtype=torch.int32)
+        als = torch.zeros(B, dtype=torch.int32)
+        cnt = torch.zeros(4, dtype=torch.int32)
+        ring = torch.full((4096,), -1, dtype=torch.int32)  # > B: no wrap in-test
+        head = torch.zeros(1, dtype=torch.int32)
+        return states, deadlines, out_dec, ds, als, cnt, ring, head
+
+    s1 = gate_state()
+    ref.policy_gate_full(first, decisions, s1[2], s1[3], s1[5][0:1], s1[4], s1[5][1:2],
+                         s1[0], s1[1], s1[6], s1[7])
+    s2 = [t.to(d) for t in gate_state()]
+    ext.policy_gate_full(first.to(d), decisions.to(d), s2[2], s2[3], s2[5][0:1],
+                         s2[4], s2[5][1:2], s2[0], s2[1], s2[6], s2[7])
+    assert torch.equal(s1[0], s2[0].cpu())          # states (DENIED applied)
+    assert torch.equal(s1[1], s2[1].cpu())          # deadlines cleared
+    assert torch.equal(s1[2], s2[2].cpu())          # decisions
+    assert int(s1[5][0]) == int(s2[5][0].cpu())     # denied count
+    assert int(s1[5][1]) == int(s2[5][1].cpu())     # allowed count
+    assert int(s1[7][0]) == int(s2[7][0].cpu())     # dlq head
+    # ring contents as sets (append order differs across waves)
+    n = int(s1[7][0])
+    assert set(s1[6][:n].tolist()) == set(s2[6].cpu()[s2[6].cpu() >= 0].tolist())
+    # compacted slot sets match
+    assert set(s1[3][: int(s1[5][0])].tolist()) == set(s2[3].cpu()[: int(s1[5][0])].tolist())
+    assert set(s1[4][: int(s1[5][1])].tolist()) == set(s2[4].cpu()[: int(s1[5][1])].tolist())
+
+    # -- apply_transitions_chain_dyn ----------------------------------------
+    slots = torch.randperm(B, dtype=torch.int64)[: B // 2].to(torch.int32)
+    count = torch.tensor([slots.numel()], dtype=torch.int32)
+    chain = [3, 4, 5, 6]  # SCHEDULED..SUCCEEDED
+
+    def chain_state():
+        states = torch.ones(B, dtype=torch.uint8)
+        states[::7] = 10  # some terminal slots: chain must not move them
+        attempts = torch.zeros(B, dtype=torch.int32)
+        deadlines = torch.randint(1, 2**40, (B,), dtype=torch.int64)
+        extra = torch.randint(1, 100, (64,), dtype=torch.int32)
+        return states, attempts, deadlines, extra
+
+    c1 = chain_state()
+    ref.apply_transitions_chain_dyn(c1[0], c1[1], c1[2], slots, count, chain, c1[3], B)
+    c2 = [t.to(d) for t in chain_state()]
+    ext.apply_transitions_chain_dyn(c2[0], c2[1], c2[2], slots.to(d), count.to(d),
+                                    chain, c2[3], B)
+    for i in range(4):
+        assert torch.equal(c1[i], c2[i].cpu()), f"chain tensor {i}"
+
+    # -- begin_tick + accumulate_counts -------------------------------------
+    st = torch.zeros(B, dtype=torch.uint8, device=d)
+    cnts = torch.tensor([3, 5, 7, 9], dtype=torch.int32, device=d)
+    ext.begin_tick(st, cnts)
+    assert int(st.sum().cpu()) == B and int(cnts.abs().sum().cpu()) == 0
+    acc = torch.zeros(4, dtype=torch.int64, device=d)
+    src = torch.tensor([1, 2, 3, 4], dtype=torch.int32, device=d)
+    ext.accumulate_counts(src, acc)
+    ext.accumulate_counts(src, acc)
+    assert acc.cpu().tolist() == [2, 4, 6, 8]
