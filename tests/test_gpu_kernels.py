@@ -311,9 +311,11 @@ def test_fused_tick_kernels_match_refops(ext):
         head = torch.zeros(1, dtype=torch.int32)
         return states, deadlines, out_dec, ds, als, cnt, ring, head
 
+    torch.manual_seed(11)
     s1 = gate_state()
     ref.policy_gate_full(first, decisions, s1[2], s1[3], s1[5][0:1], s1[4], s1[5][1:2],
                          s1[0], s1[1], s1[6], s1[7])
+    torch.manual_seed(11)
     s2 = [t.to(d) for t in gate_state()]
     ext.policy_gate_full(first.to(d), decisions.to(d), s2[2], s2[3], s2[5][0:1],
                          s2[4], s2[5][1:2], s2[0], s2[1], s2[6], s2[7])
@@ -343,8 +345,10 @@ def test_fused_tick_kernels_match_refops(ext):
         extra = torch.randint(1, 100, (64,), dtype=torch.int32)
         return states, attempts, deadlines, extra
 
+    torch.manual_seed(13)
     c1 = chain_state()
     ref.apply_transitions_chain_dyn(c1[0], c1[1], c1[2], slots, count, chain, c1[3], B)
+    torch.manual_seed(13)
     c2 = [t.to(d) for t in chain_state()]
     ext.apply_transitions_chain_dyn(c2[0], c2[1], c2[2], slots.to(d), count.to(d),
                                     chain, c2[3], B)
