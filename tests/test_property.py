@@ -56,3 +56,79 @@ def test_glob_match_never_crashes_and_agrees_with_fnmatch_subset(pattern, name):
     # patterns without specials behave as equality
     if not any(c in pattern for c in "*?[]\\"):
         assert got == (pattern == name)
+
+
+# --- workflow expression language -------------------------------------------
+
+from cordum_amd.workflow.eval import (  # noqa: E402
+    EvalError,
+    eval_expr,
+    eval_template_string,
+    eval_templates,
+)
+
+scalar = st.one_of(st.integers(min_value=-10**6, max_value=10**6),
+                   st.booleans(), st.text(max_size=8))
+json_val = st.recursive(
+    scalar,
+    lambda inner: st.one_of(st.lists(inner, max_size=3),
+                            st.dictionaries(st.text(min_size=1, max_size=6), inner, max_size=3)),
+    max_leaves=8,
+)
+
+
+@settings(max_examples=200, deadline=None)
+@given(v=json_val)
+def test_eval_templates_preserves_shape_without_placeholders(v):
+    # values containing no ${...} must round-trip unchanged
+    out = eval_templates(v, {"input": {}, "ctx": {}, "steps": {}})
+    def strip(x):
+        if isinstance(x, str) and "${" in x:
+            return None
+        if isinstance(x, list):
+            return [strip(i) for i in x]
+        if isinstance(x, dict):
+            return {k: strip(u) for k, u in x.items()}
+        return x
+    assert strip(out) == strip(v) or any(
+        isinstance(s, str) and "${" in s for s in _strings(v))
+
+
+def _strings(v):
+    if isinstance(v, str):
+        yield v
+    elif isinstance(v, list):
+        for i in v:
+            yield from _strings(i)
+    elif isinstance(v, dict):
+        for i in v.values():
+            yield from _strings(i)
+
+
+@settings(max_examples=300, deadline=None)
+@given(a=st.integers(min_value=-10**6, max_value=10**6),
+       b=st.integers(min_value=-10**6, max_value=10**6),
+       op=st.sampled_from(["==", "!=", ">", "<", ">=", "<="]))
+def test_eval_numeric_comparisons_match_python(a, b, op):
+    got = eval_expr(f"{a} {op} {b}", {})
+    want = {"==": a == b, "!=": a != b, ">": a > b,
+            "<": a < b, ">=": a >= b, "<=": a <= b}[op]
+    assert got == want
+
+
+@settings(max_examples=200, deadline=None)
+@given(path=st.lists(st.text(alphabet="abc", min_size=1, max_size=4),
+                     min_size=1, max_size=4),
+       val=st.integers(min_value=0, max_value=1000))
+def test_eval_dot_path_resolution(path, val):
+    ctx = cur = {}
+    for part in path[:-1]:
+        cur[part] = {}
+        cur = cur[part]
+    cur[path[-1]] = val
+    assert eval_expr(".".join(path), ctx) == val
+    # template wrapping preserves type
+    assert eval_template_string("${" + ".".join(path) + "}", ctx) == val
+    # mixed template stringifies
+    out = eval_template_string("v=${" + ".".join(path) + "}!", ctx)
+    assert out == f"v={val}!"
