@@ -49,7 +49,7 @@ def subject_matches(pattern: str, subject: str) -> bool:
     st = subject.split(".")
     for i, p in enumerate(pt):
         if p == ">":
-            return True
+            return len(st) > i  # NATS '>': one or MORE remaining tokens
         if i >= len(st):
             return False
         if p != "*" and p != st[i]:

@@ -132,3 +132,52 @@ def test_eval_dot_path_resolution(path, val):
     # mixed template stringifies
     out = eval_template_string("v=${" + ".".join(path) + "}!", ctx)
     assert out == f"v={val}!"
+
+
+# --- bus subject matching + canonical JSON ----------------------------------
+
+from cordum_amd.bus.bus import subject_matches  # noqa: E402
+from cordum_amd.utils.canonical_json import canonical_json  # noqa: E402
+
+token = st.text(alphabet="abcz0", min_size=1, max_size=4)
+subject_s = st.lists(token, min_size=1, max_size=4).map(".".join)
+pattern_tok = st.one_of(token, st.just("*"))
+
+
+@settings(max_examples=300, deadline=None)
+@given(subj=subject_s)
+def test_subject_matching_nats_semantics(subj):
+    parts = subj.split(".")
+    assert subject_matches(subj, subj)                      # exact
+    assert subject_matches(">", subj)                       # full wildcard
+    assert subject_matches(parts[0] + ".>", subj) == (len(parts) > 1)
+    star = ".".join(["*"] * len(parts))
+    assert subject_matches(star, subj)                      # per-token star
+    assert not subject_matches(subj + ".x", subj)           # longer pattern
+    if len(parts) > 1:
+        assert not subject_matches(parts[0], subj)          # shorter pattern
+
+
+@settings(max_examples=300, deadline=None)
+@given(pattern=st.lists(pattern_tok, min_size=1, max_size=4),
+       subj=subject_s)
+def test_subject_star_matches_iff_tokenwise(pattern, subj):
+    p = ".".join(pattern)
+    parts = subj.split(".")
+    want = len(pattern) == len(parts) and all(
+        pt == "*" or pt == sp for pt, sp in zip(pattern, parts))
+    assert subject_matches(p, subj) == want
+
+
+@settings(max_examples=200, deadline=None)
+@given(v=json_val)
+def test_canonical_json_stable_and_order_insensitive(v):
+    import json as _json
+
+    s1 = canonical_json(v)
+    # round-trip through a parse (which scrambles nothing semantic)
+    assert canonical_json(_json.loads(s1)) == s1
+    if isinstance(v, dict):
+        # rebuilding the dict in reversed insertion order must not change it
+        rev = dict(reversed(list(v.items())))
+        assert canonical_json(rev) == s1
