@@ -1,5 +1,6 @@
 """Property-based tests (hypothesis): codec round-trips and compiled-policy
 agreement on adversarial inputs."""
+import pytest
 from hypothesis import given, settings, strategies as st
 
 from cordum_amd.protocol.capv2 import Budget, JobMetadata, JobRequest, JobPriority
@@ -181,3 +182,35 @@ def test_canonical_json_stable_and_order_insensitive(v):
         # rebuilding the dict in reversed insertion order must not change it
         rev = dict(reversed(list(v.items())))
         assert canonical_json(rev) == s1
+
+
+# --- job state machine random walk ------------------------------------------
+
+from cordum_amd.protocol import JobState  # noqa: E402
+from cordum_amd.protocol.states import ALLOWED_TRANSITIONS, is_terminal  # noqa: E402
+from cordum_amd.store import InvalidTransition, JobStore  # noqa: E402
+from cordum_amd.utils.clock import ManualClock  # noqa: E402
+
+
+@settings(max_examples=150, deadline=None)
+@given(walk=st.lists(st.sampled_from([s for s in JobState if s != JobState.UNSPECIFIED]),
+                     min_size=1, max_size=12))
+def test_job_store_enforces_transition_table_on_random_walks(walk):
+    js = JobStore(clock=ManualClock())
+    cur = JobState.UNSPECIFIED
+    attempts = 0
+    touched = False
+    for nxt in walk:
+        legal = nxt in ALLOWED_TRANSITIONS.get(cur, frozenset())
+        if legal:
+            js.set_state("j", nxt)
+            if nxt == JobState.SCHEDULED and cur != JobState.SCHEDULED:
+                attempts += 1
+            cur = nxt
+            touched = True
+        else:
+            with pytest.raises(InvalidTransition):
+                js.set_state("j", nxt)
+    if touched:
+        assert js.get_state("j") == cur
+        assert int(js.get_job_meta("j").get("attempts", 0)) == attempts
