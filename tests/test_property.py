@@ -214,3 +214,42 @@ def test_job_store_enforces_transition_table_on_random_walks(walk):
     if touched:
         assert js.get_state("j") == cur
         assert int(js.get_job_meta("j").get("attempts", 0)) == attempts
+
+
+# --- store snapshot round-trips ----------------------------------------------
+
+import json as _json  # noqa: E402
+
+
+@settings(max_examples=100, deadline=None)
+@given(
+    jobs=st.lists(
+        st.tuples(st.text(alphabet="jk0", min_size=1, max_size=5),   # id
+                  st.sampled_from(["t1", "t2", ""]),                 # tenant
+                  st.sampled_from([JobState.PENDING, JobState.SCHEDULED])),
+        min_size=0, max_size=8, unique_by=lambda t: t[0]),
+)
+def test_job_store_snapshot_roundtrip_and_json_clean(jobs):
+    clock = ManualClock()
+    js = JobStore(clock=clock)
+    for jid, tenant, state in jobs:
+        if tenant:
+            js.set_tenant(jid, tenant)
+        js.set_state(jid, state)
+        if state == JobState.SCHEDULED:
+            js.set_deadline(jid, clock.now_micros() + 1_000_000)
+        clock.advance(1)
+    snap = js.snapshot()
+    # checkpoint constraint: the snapshot must be plain JSON (the WAL file is)
+    snap2 = _json.loads(_json.dumps(snap))
+    js2 = JobStore(clock=clock)
+    js2.restore(snap2)
+    assert js2.snapshot() == snap
+    for jid, tenant, state in jobs:
+        assert js2.get_state(jid) == state
+        if tenant:
+            assert js2.get_job_meta(jid).get("tenant") == tenant
+    # indexes rebuilt: per-state listings agree
+    for s in (JobState.PENDING, JobState.SCHEDULED):
+        assert sorted(js.list_jobs_by_state(s)) == sorted(js2.list_jobs_by_state(s))
+    assert sorted(js.list_expired_deadlines()) == sorted(js2.list_expired_deadlines())
