@@ -38,7 +38,36 @@ def build_extension(verbose: bool = False):
         with_cuda=True,
     )
     _load_error = None
+    build_capv2_native()
     return _ext
+
+
+def build_capv2_native(verbose: bool = False) -> bool:
+    """Compile the native CAP v2 codec (plain C++/pybind11, no torch/HIP dep)
+    in-tree next to its source so the .so travels with the snapshot."""
+    import subprocess
+    import sys
+    import sysconfig
+
+    import pybind11
+
+    native_dir = _HERE.parent / "protocol" / "native"
+    src = native_dir / "capv2_codec.cpp"
+    out = native_dir / "_capv2_native.so"
+    if out.exists() and out.stat().st_mtime >= src.stat().st_mtime:
+        return True
+    cmd = [
+        "g++", "-O3", "-shared", "-fPIC", "-std=c++17",
+        f"-I{sysconfig.get_paths()['include']}",
+        f"-I{pybind11.get_include()}",
+        str(src), "-o", str(out),
+    ]
+    r = subprocess.run(cmd, capture_output=True, text=True)
+    if r.returncode != 0:
+        if verbose:
+            print(r.stderr, file=sys.stderr)
+        return False
+    return True
 
 
 def _try_import_prebuilt():

@@ -779,3 +779,104 @@ def state_name_from_status(status: JobStatus) -> str:
     if status == JobStatus.UNSPECIFIED:
         return ""
     return JobStatus(status).name
+
+
+# ---------------------------------------------------------------------------
+# Native codec engine (cordum_amd/protocol/native/capv2_codec.cpp)
+# ---------------------------------------------------------------------------
+# The C++ engine is schema-driven from the SAME FIELDS tables, so the wire
+# layout has one source of truth; tests/test_property.py asserts byte
+# equality against the pure-Python path on random messages. Loading is
+# best-effort: without the built .so every path below stays pure Python.
+
+_NATIVE = None
+_KIND_CODES = {"int": 0, "sint64": 1, "bool": 2, "enum": 3, "str": 4,
+               "bytes": 5, "double": 6, "msg": 7, "rep_str": 8, "map_ss": 9}
+
+
+def _all_message_classes():
+    seen, out, stack = set(), [], [Message]
+    while stack:
+        cls = stack.pop()
+        for sub in cls.__subclasses__():
+            if sub not in seen:
+                seen.add(sub)
+                out.append(sub)
+                stack.append(sub)
+    return out
+
+
+def _load_native():
+    global _NATIVE
+    if _NATIVE is not None:
+        return _NATIVE
+    try:
+        import importlib.util
+        from pathlib import Path
+
+        so = Path(__file__).resolve().parent / "native" / "_capv2_native.so"
+        if not so.exists():
+            return None
+        spec = importlib.util.spec_from_file_location("_capv2_native", so)
+        mod = importlib.util.module_from_spec(spec)
+        spec.loader.exec_module(mod)
+        for cls in _all_message_classes():
+            if any(fs.kind not in _KIND_CODES for fs in cls.FIELDS.values()):
+                continue  # custom-kind classes (rep_msg) keep their own codec
+            fields = [
+                (name, fs.num, _KIND_CODES[fs.kind],
+                 fs.sub.__name__ if fs.sub is not None else "",
+                 fs.enum)
+                for name, fs in cls.FIELDS.items()
+            ]
+            mod.register_message(cls.__name__, cls, fields)
+        _NATIVE = mod
+        return mod
+    except Exception:
+        return None
+
+
+def _native_encode(msg: "Message"):
+    n = _NATIVE
+    if n is None:
+        return None
+    try:
+        return n.encode(type(msg).__name__, msg)
+    except Exception:
+        return None
+
+
+def _native_decode(cls, data: bytes):
+    n = _NATIVE
+    if n is None:
+        return None
+    try:
+        return n.decode(cls.__name__, bytes(data))
+    except Exception:
+        return None
+
+
+_PY_ENCODE = Message.encode
+_PY_DECODE = Message.decode.__func__
+
+
+def _encode_dispatch(self) -> bytes:
+    out = _native_encode(self)
+    if out is not None:
+        return out
+    return _PY_ENCODE(self)
+
+
+def _decode_dispatch(cls, data: bytes):
+    out = _native_decode(cls, data)
+    if out is not None:
+        return out
+    return _PY_DECODE(cls, data)
+
+
+Message.encode = _encode_dispatch
+Message.decode = classmethod(_decode_dispatch)
+Message.encode_py = _PY_ENCODE            # pure-Python paths kept addressable
+Message.decode_py = classmethod(_PY_DECODE)  # (equality tests + fallback)
+
+_load_native()

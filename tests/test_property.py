@@ -253,3 +253,32 @@ def test_job_store_snapshot_roundtrip_and_json_clean(jobs):
     for s in (JobState.PENDING, JobState.SCHEDULED):
         assert sorted(js.list_jobs_by_state(s)) == sorted(js2.list_jobs_by_state(s))
     assert sorted(js.list_expired_deadlines()) == sorted(js2.list_expired_deadlines())
+
+
+# --- native codec equivalence -------------------------------------------------
+
+
+@settings(max_examples=300, deadline=None)
+@given(
+    job_id=texts, topic=texts,
+    labels=st.dictionaries(label_keys, texts, max_size=5),
+    risk=st.lists(texts, max_size=4),
+    tokens=st.integers(min_value=-2**53, max_value=2**53),
+    prio=st.sampled_from(list(JobPriority)),
+)
+def test_native_codec_bytes_identical_to_python(job_id, topic, labels, risk, tokens, prio):
+    from cordum_amd.protocol import capv2
+
+    if capv2._NATIVE is None:
+        pytest.skip("native codec not built")
+    req = JobRequest(job_id=job_id, topic=topic, labels=labels, priority=prio,
+                     meta=JobMetadata(risk_tags=risk),
+                     budget=Budget(max_tokens=tokens))
+    nb = req.encode()
+    assert nb == req.encode_py()
+    # native decode == python decode, and both re-encode to the same bytes
+    d_native = JobRequest.decode(nb)
+    d_python = JobRequest.decode_py(nb)
+    assert d_native.encode() == d_python.encode_py() == nb
+    assert d_native.labels == d_python.labels == labels
+    assert d_native.meta.risk_tags == risk
