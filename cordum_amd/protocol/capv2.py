@@ -156,6 +156,9 @@ class Message:
             if ent is None:
                 continue  # unknown field: skip (forward compat)
             name, spec = ent
+            if _WIRE_TYPE.get(spec.kind, _WT_LEN) != wt:
+                raise ValueError(
+                    f"wire type {wt} invalid for field {num} ({spec.kind})")
             _decode_into(msg, name, spec, raw)
         return msg
 
@@ -214,6 +217,13 @@ class Message:
 
     def __eq__(self, other):
         return type(self) is type(other) and self.encode() == other.encode()
+
+
+_WIRE_TYPE = {
+    "int": _WT_VARINT, "sint64": _WT_VARINT, "bool": _WT_VARINT,
+    "enum": _WT_VARINT, "double": _WT_I64, "str": _WT_LEN, "bytes": _WT_LEN,
+    "msg": _WT_LEN, "rep_str": _WT_LEN, "map_ss": _WT_LEN, "rep_msg": _WT_LEN,
+}
 
 
 def _is_default(spec: F, val: Any) -> bool:
@@ -684,6 +694,9 @@ class PolicyCheckResponse(Message):
             if ent is None:
                 continue
             name, spec = ent
+            if _WIRE_TYPE.get(spec.kind, _WT_LEN) != wt:
+                raise ValueError(
+                    f"wire type {wt} invalid for field {num} ({spec.kind})")
             if spec.kind == "rep_msg":
                 msg.remediations.append(PolicyRemediation.decode(raw))
             else:

@@ -298,7 +298,14 @@ static py::object decode_msg(const std::string& key, const unsigned char* buf,
     }
     auto fit = it->second.by_num.find(num);
     if (fit == it->second.by_num.end()) continue;  // unknown: skip
-    decode_field(msg, it->second.fields[fit->second], raw, rn, vint, wt);
+    const FieldSpec& fs = it->second.fields[fit->second];
+    const int want_wt =
+        (fs.kind == K_INT || fs.kind == K_SINT64 || fs.kind == K_BOOL ||
+         fs.kind == K_ENUM) ? 0 : (fs.kind == K_DOUBLE ? 1 : 2);
+    if (wt != want_wt)
+      throw std::runtime_error("wire type mismatch for field " +
+                               std::to_string(num));
+    decode_field(msg, fs, raw, rn, vint, wt);
   }
   return msg;
 }

@@ -282,3 +282,23 @@ def test_native_codec_bytes_identical_to_python(job_id, topic, labels, risk, tok
     assert d_native.encode() == d_python.encode_py() == nb
     assert d_native.labels == d_python.labels == labels
     assert d_native.meta.risk_tags == risk
+
+
+@settings(max_examples=500, deadline=None)
+@given(junk=st.binary(max_size=64))
+def test_codec_decode_rejects_or_parses_arbitrary_bytes(junk):
+    """API-boundary robustness: decoding attacker-controlled bytes must either
+    produce a message or raise a clean error — never crash the process or
+    hang (both the native engine and the Python fallback)."""
+    from cordum_amd.protocol import capv2
+
+    from cordum_amd.protocol.capv2 import PolicyCheckResponse
+
+    for decoder in (JobRequest.decode, JobRequest.decode_py,
+                    PolicyCheckResponse.decode):
+        try:
+            msg = decoder(junk)
+        except (ValueError, RuntimeError, IndexError, UnicodeDecodeError):
+            continue
+        # decodable junk must re-encode deterministically
+        assert isinstance(msg.encode(), bytes)
