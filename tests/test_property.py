@@ -302,3 +302,96 @@ def test_codec_decode_rejects_or_parses_arbitrary_bytes(junk):
             continue
         # decodable junk must re-encode deterministically
         assert isinstance(msg.encode(), bytes)
+
+
+# --- workflow engine: random DAGs --------------------------------------------
+
+from cordum_amd.bus import LoopbackBus  # noqa: E402
+from cordum_amd.protocol import subjects as subj  # noqa: E402
+from cordum_amd.protocol.capv2 import JobResult, JobStatus  # noqa: E402
+from cordum_amd.workflow import (  # noqa: E402
+    Engine,
+    RUN_FAILED,
+    RUN_SUCCEEDED,
+    Step,
+    Workflow,
+    WorkflowRun,
+    WorkflowStore,
+)
+from cordum_amd.store import MemoryStore  # noqa: E402
+
+
+@settings(max_examples=60, deadline=None)
+@given(
+    n=st.integers(min_value=1, max_value=7),
+    edge_bits=st.integers(min_value=0, max_value=2**21 - 1),
+    fail_step=st.integers(min_value=-1, max_value=6),
+)
+def test_workflow_random_dag_terminates_with_correct_statuses(n, edge_bits, fail_step):
+    """Any random dependency DAG must drive to a terminal run status with
+    every reachable step completed; a failing step must block its downstream
+    closure (deps gate workflow/engine.go depsSatisfied:1231-1242) and fail
+    the run."""
+    clock = ManualClock()
+    bus = LoopbackBus(clock=clock)
+    store = WorkflowStore(clock=clock)
+    memory = MemoryStore(clock=clock)
+    engine = Engine(store, bus, memory=memory, clock=clock)
+    submitted = []
+    bus.subscribe(subj.SUBJECT_SUBMIT, lambda s, p: submitted.append(p.job_request))
+
+    names = [f"s{i}" for i in range(n)]
+    deps = {i: [j for j in range(i) if (edge_bits >> (i * (i - 1) // 2 + j)) & 1]
+            for i in range(n)}
+    steps = {
+        names[i]: Step.from_dict(names[i], {
+            "type": "worker", "topic": "job.t",
+            "depends_on": [names[j] for j in deps[i]],
+        })
+        for i in range(n)
+    }
+    wf = Workflow(id="wfP", org_id="org", steps=steps)
+    store.put_workflow(wf)
+    run = WorkflowRun(id="runP", workflow_id="wfP", org_id="org")
+    store.create_run(run)
+    engine.start_run("wfP", "runP")
+
+    failing = names[fail_step] if 0 <= fail_step < n else None
+    done = set()
+    for _ in range(n * n + 2):  # fixpoint: drain all dispatches
+        pending = [r for r in submitted if r.job_id not in done]
+        if not pending:
+            break
+        for req in pending:
+            done.add(req.job_id)
+            sid = req.labels["step_id"]
+            status = JobStatus.FAILED if sid == failing else JobStatus.SUCCEEDED
+            engine.handle_job_result(JobResult(job_id=req.job_id, status=status))
+
+    run = store.get_run("runP")
+    # downstream closure of the failing step
+    blocked = set()
+    if failing is not None:
+        idx = {nm: i for i, nm in enumerate(names)}
+        frontier = {failing}
+        while frontier:
+            cur = frontier.pop()
+            blocked.add(cur)
+            for i in range(n):
+                if names[i] not in blocked and any(names[j] in blocked for j in deps[i]):
+                    frontier.add(names[i])
+    if failing is None:
+        assert run.status == RUN_SUCCEEDED, run.status
+        assert all(run.steps[nm].status == "succeeded" for nm in names)
+    else:
+        assert run.status == RUN_FAILED, run.status
+        for nm in names:
+            if nm == failing:
+                assert run.steps[nm].status == "failed"
+            elif nm in blocked:
+                assert run.steps[nm].status in ("pending", "skipped", "blocked")
+            else:
+                # results arriving after the run went terminal are ignored
+                # (engine.handle_job_result, mirroring the reference's
+                # processed-state dedup) so in-flight siblings stay running
+                assert run.steps[nm].status in ("succeeded", "running")
