@@ -389,9 +389,12 @@ def test_workflow_random_dag_terminates_with_correct_statuses(n, edge_bits, fail
             if nm == failing:
                 assert run.steps[nm].status == "failed"
             elif nm in blocked:
-                assert run.steps[nm].status in ("pending", "skipped", "blocked")
+                sr = run.steps.get(nm)
+                assert sr is None or sr.status in ("pending", "skipped", "blocked")
             else:
                 # results arriving after the run went terminal are ignored
                 # (engine.handle_job_result, mirroring the reference's
-                # processed-state dedup) so in-flight siblings stay running
-                assert run.steps[nm].status in ("succeeded", "running")
+                # processed-state dedup) so in-flight siblings stay running;
+                # independent steps not yet dispatched have no record at all
+                sr = run.steps.get(nm)
+                assert sr is None or sr.status in ("succeeded", "running", "pending")
