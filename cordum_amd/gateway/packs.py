@@ -48,8 +48,12 @@ def extract_pack(blob: bytes) -> Dict[str, bytes]:
             for member in tf:
                 if not member.isfile():
                     continue
-                name = member.name.lstrip("./")
-                if name.startswith("/") or ".." in name.split("/"):
+                # strip only a leading "./" prefix — a charwise lstrip("./")
+                # would silently sanitize "../x" to "x" instead of rejecting
+                name = member.name
+                while name.startswith("./"):
+                    name = name[2:]
+                if name.startswith("/") or ".." in name.split("/") or not name:
                     raise PackError(f"illegal path {member.name!r}")
                 if member.size > MAX_FILE:
                     raise PackError(f"file too large: {name}")

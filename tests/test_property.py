@@ -398,3 +398,48 @@ def test_workflow_random_dag_terminates_with_correct_statuses(n, edge_bits, fail
                 # independent steps not yet dispatched have no record at all
                 sr = run.steps.get(nm)
                 assert sr is None or sr.status in ("succeeded", "running", "pending")
+
+
+# --- pack archive hardening ---------------------------------------------------
+
+import io  # noqa: E402
+import tarfile  # noqa: E402
+
+
+def _tar_with(names):
+    buf = io.BytesIO()
+    with tarfile.open(fileobj=buf, mode="w:gz") as tf:
+        for nm in names:
+            data = b"x"
+            info = tarfile.TarInfo(name=nm)
+            info.size = len(data)
+            tf.addfile(info, io.BytesIO(data))
+    return buf.getvalue()
+
+
+@settings(max_examples=200, deadline=None)
+@given(name=st.text(alphabet=st.sampled_from("ab./\\_"), min_size=1, max_size=16))
+def test_pack_extract_rejects_traversal_never_escapes(name):
+    from cordum_amd.gateway.packs import PackError, extract_pack
+
+    try:
+        blob = _tar_with([name])
+    except ValueError:
+        return  # tarfile itself refuses the name
+    try:
+        files = extract_pack(blob)
+    except PackError:
+        return  # rejected: fine
+    # anything accepted must be a clean relative path with no traversal
+    for accepted in files:
+        assert accepted
+        assert not accepted.startswith("/")
+        assert ".." not in accepted.split("/")
+
+
+def test_pack_extract_rejects_known_hostile_names():
+    from cordum_amd.gateway.packs import PackError, extract_pack
+
+    for nm in ["../evil", "a/../../evil", "/abs/path", "./../up", "..", "./.."]:
+        with pytest.raises(PackError):
+            extract_pack(_tar_with([nm, "pack.yaml"]))
