@@ -443,3 +443,49 @@ def test_pack_extract_rejects_known_hostile_names():
     for nm in ["../evil", "a/../../evil", "/abs/path", "./../up", "..", "./.."]:
         with pytest.raises(PackError):
             extract_pack(_tar_with([nm, "pack.yaml"]))
+
+
+# --- schema validator robustness ---------------------------------------------
+
+
+@settings(max_examples=150, deadline=None)
+@given(schema=json_val, value=json_val)
+def test_schema_validator_never_crashes(schema, value):
+    """validate_value on arbitrary (schema, value) pairs returns a bool list
+    outcome or raises nothing — hostile schemas must not crash the gateway."""
+    from cordum_amd.store.schema_registry import validate_value
+
+    errs = validate_value(schema, value)
+    assert isinstance(errs, list)
+
+
+# --- bus NAK redelivery -------------------------------------------------------
+
+
+@settings(max_examples=80, deadline=None)
+@given(n_msgs=st.integers(min_value=1, max_value=6),
+       nak_rounds=st.integers(min_value=0, max_value=3))
+def test_bus_nak_redelivers_until_ack(n_msgs, nak_rounds):
+    """A handler that NAKs k times then succeeds must see each message
+    exactly k+1 times (JetStream redelivery semantics, bus/nats.go:146-168),
+    pinned to the NAKing consumer."""
+    from cordum_amd.bus import LoopbackBus, RetryAfter
+    from cordum_amd.protocol.capv2 import BusPacket, JobRequest
+
+    clock = ManualClock()
+    bus = LoopbackBus(clock=clock)
+    seen = {}
+
+    def handler(subject, pkt):
+        jid = pkt.job_request.job_id
+        seen[jid] = seen.get(jid, 0) + 1
+        if seen[jid] <= nak_rounds:
+            raise RetryAfter(1.0, "busy")
+
+    bus.subscribe("sys.job.submit", handler, queue_group="g")
+    for i in range(n_msgs):
+        bus.publish("sys.job.submit", BusPacket(job_request=JobRequest(job_id=f"j{i}")))
+    for _ in range(nak_rounds + 2):
+        clock.advance(2)
+        bus.pump()
+    assert seen == {f"j{i}": nak_rounds + 1 for i in range(n_msgs)}
