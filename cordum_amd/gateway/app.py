@@ -122,6 +122,8 @@ def create_app(
             body = json.loads(raw or b"{}")
         except ValueError:
             raise HTTPException(400, "invalid json")
+        if not isinstance(body, dict):
+            raise HTTPException(400, "body must be a json object")
 
         prompt = body.get("prompt") or ""
         if not prompt:

@@ -489,3 +489,70 @@ def test_bus_nak_redelivers_until_ack(n_msgs, nak_rounds):
         clock.advance(2)
         bus.pump()
     assert seen == {f"j{i}": nak_rounds + 1 for i in range(n_msgs)}
+
+
+# --- expression/template fuzz -------------------------------------------------
+
+
+@settings(max_examples=400, deadline=None)
+@given(expr=st.text(alphabet=st.sampled_from("ab.(){}$!<>=' 0137length"), max_size=24),
+       scope=json_val)
+def test_eval_never_raises_uncontrolled(expr, scope):
+    """Workflow expressions come from user-authored YAML: arbitrary strings
+    must evaluate or raise EvalError — nothing else (no crashes in the
+    engine's scheduleReady path)."""
+    ctx = scope if isinstance(scope, dict) else {"input": scope}
+    for fn in (eval_expr, eval_template_string):
+        try:
+            fn(expr, ctx)
+        except EvalError:
+            pass
+        except RecursionError:
+            pass  # pathological nesting is bounded by python's own limit
+
+
+# --- gateway submit-body fuzz -------------------------------------------------
+
+
+@settings(max_examples=60, deadline=None)
+@given(body=json_val,
+       path=st.sampled_from(["/api/v1/jobs", "/api/v1/workflows",
+                             "/api/v1/policy/evaluate", "/api/v1/locks/acquire",
+                             "/api/v1/schemas", "/api/v1/config"]))
+def test_gateway_mutating_endpoints_handle_arbitrary_json(body, path):
+    """POST with arbitrary JSON must answer 2xx or 4xx — never a 500
+    (gateway.go:1757-1977 validation contract)."""
+    code, _ = _gw_client().post_json(path, body)
+    assert code < 500, (code, path, body)
+
+
+_GW = None
+
+
+def _gw_client():
+    global _GW
+    if _GW is None:
+        from fastapi.testclient import TestClient
+
+        from cordum_amd.scheduler import PoolProfile, PoolRouting
+        from cordum_amd.gateway.app import create_app
+        from cordum_amd.gateway.auth import BasicAuthProvider
+        from cordum_amd.runtime.node import Node
+
+        node = Node(clock=ManualClock(),
+                    routing=PoolRouting(topics={"job.default": ["default"]},
+                                        pools={"default": PoolProfile()})).start()
+        node.add_worker("w1", topics=["job.default"])
+        app = create_app(node, auth=BasicAuthProvider(api_keys=["test-key"]))
+        tc = TestClient(app, raise_server_exceptions=False)
+
+        class C:
+            def post_json(self, path, body):
+                r = tc.post(path, json=body,
+                            headers={"X-API-Key": "test-key",
+                                     "X-Principal-Id": "t",
+                                     "X-Principal-Role": "admin"})
+                return r.status_code, r
+
+        _GW = C()
+    return _GW
