@@ -74,7 +74,10 @@ def cmd_init(args):
 def cmd_workflow(args):
     c = make_client(args)
     if args.action == "create":
-        doc = json.loads(Path(args.file).read_text())
+        import yaml
+
+        path = Path(args.file or args.id)
+        doc = yaml.safe_load(path.read_text())  # YAML or JSON, like the Go CLI
         out(c.create_workflow(doc))
     elif args.action == "list":
         out(c.list_workflows())
@@ -88,7 +91,8 @@ def cmd_run(args):
     c = make_client(args)
     if args.action == "start":
         inp = json.loads(args.input) if args.input else {}
-        out(c.start_run(args.workflow, inp, dry_run=args.dry_run))
+        wf_id = args.workflow or args.id  # `run start <wf>` or -w <wf>
+        out(c.start_run(wf_id, inp, dry_run=args.dry_run))
     elif args.action == "get":
         out(c.get_run(args.id))
     elif args.action == "list":

@@ -1,0 +1,80 @@
+"""cordumctl CLI: parse + client-path tests against a live in-process node
+(mirrors the reference's cordumctl_smoke.sh, but as a unit suite)."""
+import json
+import threading
+
+import pytest
+
+from cordum_amd.cli.cordumctl import main as ctl_main
+
+
+@pytest.fixture(scope="module")
+def server():
+    import uvicorn
+
+    from cordum_amd.gateway.app import create_app
+    from cordum_amd.gateway.auth import BasicAuthProvider
+    from cordum_amd.runtime.node import Node
+    from cordum_amd.scheduler import PoolProfile, PoolRouting
+
+    node = Node(routing=PoolRouting(topics={"job.default": ["default"],
+                                            "job.echo": ["default"]},
+                                    pools={"default": PoolProfile()})).start()
+    node.add_worker("w1", topics=["job.default", "job.echo"])
+    app = create_app(node, auth=BasicAuthProvider(api_keys=["k"]))
+    cfg = uvicorn.Config(app, host="127.0.0.1", port=18231, log_level="error")
+    srv = uvicorn.Server(cfg)
+    t = threading.Thread(target=srv.run, daemon=True)
+    t.start()
+    import time
+    for _ in range(100):
+        if srv.started:
+            break
+        time.sleep(0.05)
+    yield "http://127.0.0.1:18231"
+    srv.should_exit = True
+    t.join(timeout=5)
+
+
+def run_ctl(server, *argv, capsys=None):
+    rc = ctl_main(["--server", server, "--api-key", "k", *argv])
+    out = capsys.readouterr().out if capsys else ""
+    return rc, out
+
+
+def test_status(server, capsys):
+    rc, out = run_ctl(server, "status", capsys=capsys)
+    assert rc in (0, None)
+    assert "ok" in out or "uptime" in out
+
+
+def test_job_submit_and_status(server, capsys):
+    rc, out = run_ctl(server, "job", "submit", "--topic", "job.echo",
+                      "--prompt", "hello", capsys=capsys)
+    assert rc in (0, None)
+    doc = json.loads(out)
+    jid = doc.get("job_id") or doc.get("jobId")
+    assert jid
+    rc, out = run_ctl(server, "job", "status", jid, capsys=capsys)
+    assert rc in (0, None)
+    assert jid in out
+
+
+def test_workflow_create_and_run(server, capsys):
+    wf = {"id": "cliwf", "steps": {"a": {"type": "worker", "topic": "job.echo",
+                                         "input": {"prompt": "x"}}}}
+    import tempfile, yaml, os
+    with tempfile.NamedTemporaryFile("w", suffix=".yaml", delete=False) as f:
+        yaml.safe_dump(wf, f)
+        path = f.name
+    try:
+        rc, out = run_ctl(server, "workflow", "create", "--file", path, capsys=capsys)
+        assert rc in (0, None)
+        rc, out = run_ctl(server, "run", "start", "cliwf", capsys=capsys)
+        assert rc in (0, None)
+        run_id = json.loads(out).get("run_id") or json.loads(out).get("id")
+        assert run_id
+        rc, out = run_ctl(server, "run", "timeline", run_id, capsys=capsys)
+        assert rc in (0, None)
+    finally:
+        os.unlink(path)
