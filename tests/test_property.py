@@ -571,3 +571,18 @@ def test_gateway_list_endpoints_handle_junk_query_params(q, path):
     r = gw.tc.get(f"{path}?cursor={q}&limit={q}&state={q}&topic={q}",
                   headers=gw.headers)
     assert r.status_code < 500, (r.status_code, path, q)
+
+
+@settings(max_examples=80, deadline=None)
+@given(junk=st.text(alphabet=st.sampled_from("ab:/~.%-_0"), max_size=20))
+def test_gateway_pointer_and_bundle_ids_handle_junk(junk):
+    """Pointer params and `~`-escaped bundle ids are attacker-controlled path
+    material: junk must never 500 (memory reader gateway.go:1206+, bundle id
+    unescape policy_bundles.go:24-122)."""
+    gw = _gw_client()
+    r = gw.tc.get(f"/api/v1/memory?ptr={junk}", headers=gw.headers)
+    assert r.status_code < 500, ("memory", r.status_code, junk)
+    r = gw.tc.get(f"/api/v1/policy/bundles/{junk or 'x'}", headers=gw.headers)
+    assert r.status_code < 500, ("bundle", r.status_code, junk)
+    r = gw.tc.get(f"/api/v1/artifacts/{junk or 'x'}", headers=gw.headers)
+    assert r.status_code < 500, ("artifact", r.status_code, junk)
