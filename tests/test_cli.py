@@ -78,3 +78,42 @@ def test_workflow_create_and_run(server, capsys):
         assert rc in (0, None)
     finally:
         os.unlink(path)
+
+
+# --- SDK client against the live server --------------------------------------
+
+
+@pytest.fixture(scope="module")
+def sdk(server):
+    from cordum_amd.sdk.client import Client
+
+    return Client(base_url=server, api_key="k",
+                  principal_id="tester", role="admin")
+
+
+def test_sdk_job_lifecycle(sdk):
+    doc = sdk.submit_job("hello sdk", topic="job.echo")
+    jid = doc.get("job_id") or doc.get("jobId")
+    assert jid
+    job = sdk.get_job(jid)
+    assert job.get("id") == jid  # detail envelope uses `id` (gateway.go:1011)
+    listing = sdk.list_jobs(limit=5)
+    assert "items" in listing
+
+
+def test_sdk_artifacts_and_memory(sdk):
+    ptr = sdk.artifacts_put(b"sdk-bytes")["ptr"]
+    assert ptr.startswith("redis://art:")
+    got = sdk.memory(ptr)
+    assert got is not None
+
+
+def test_sdk_policy_evaluate(sdk):
+    res = sdk.policy_evaluate(tenant="default", topic="job.echo")
+    dec = res.get("decision") or res.get("result", {}).get("decision")
+    assert dec is not None
+
+
+def test_sdk_status_and_workers(sdk):
+    assert sdk.status().get("status") in ("ok", "degraded")
+    assert "workers" in sdk.workers() or "items" in sdk.workers()
