@@ -8,7 +8,7 @@ semantics for durable subjects (msg-id dedup with a 2-minute window
 heartbeats/cancel/progress/workflow events.
 
 Cross-GPU traffic does NOT go through this bus object: batched job
-descriptors travel over RCCL all-to-all (parallel/fabric.py, SURVEY.md §2.5);
+descriptors travel over RCCL all-to-all (ops/pipeline.py, SURVEY.md §2.5);
 this bus carries the host-side control flow (gateway ↔ scheduler ↔ workflow
 engine ↔ in-process workers) and is the seam the loopback tests use
 (reference test seam: scheduler/integration_test.go:18-45).

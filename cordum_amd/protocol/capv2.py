@@ -13,8 +13,8 @@ stable across processes and hosts.
 
 Everything here is plain Python + bytes; the scheduler's hot path never
 round-trips through this codec (device job descriptors are packed tensors —
-see cordum_amd/ops/descriptors.py). This codec serves the API boundary, the
-WAL, and cross-process workers.
+see cordum_amd/ops/policy_compile.py JobEncoder and ops/pipeline.py). This
+codec serves the API boundary, the WAL, and cross-process workers.
 """
 from __future__ import annotations
 
