@@ -817,7 +817,11 @@ class DevicePipeline:
             self.acc_counts.zero_()
             torch.cuda.synchronize(self.device)
             self._pad_graphs = (g1s, g2s, g3, g4)
-        except Exception:
+        except Exception as e:
+            import sys
+
+            print(f"[cordum] pad-graph capture failed, running eager segments: {e!r}",
+                  file=sys.stderr)
             self._pad_graphs = ()  # capture unsupported: stay eager
 
     def _tick_padded(self) -> TickStats:
