@@ -586,3 +586,32 @@ def test_gateway_pointer_and_bundle_ids_handle_junk(junk):
     assert r.status_code < 500, ("bundle", r.status_code, junk)
     r = gw.tc.get(f"/api/v1/artifacts/{junk or 'x'}", headers=gw.headers)
     assert r.status_code < 500, ("artifact", r.status_code, junk)
+
+
+@settings(max_examples=200, deadline=None)
+@given(job_id=st.text(min_size=1, max_size=10), topic=texts,
+       cfg=texts, other_env=st.dictionaries(label_keys, texts, max_size=3))
+def test_job_hash_strips_exactly_the_approval_binding_exclusions(job_id, topic, cfg, other_env):
+    """job_hash ignores approval_* labels, the bus msg-id label and
+    CORDUM_EFFECTIVE_CONFIG env (job_hash.go:15-48) — and ONLY those:
+    any other label/env/field change must change the hash."""
+    base = JobRequest(job_id=job_id, topic=topic, env=dict(other_env))
+    h0 = job_hash(base)
+
+    excluded = JobRequest.decode(base.encode())
+    excluded.env = dict(excluded.env)
+    excluded.env["CORDUM_EFFECTIVE_CONFIG"] = cfg
+    excluded.labels = {"approval_granted": "true", "approval_reason": "r",
+                       "cordum.bus_msg_id": "m1"}
+    assert job_hash(excluded) == h0
+
+    changed = JobRequest.decode(base.encode())
+    changed.env = dict(changed.env)
+    changed.env["OTHER"] = "x"
+    assert job_hash(changed) != h0
+    changed2 = JobRequest.decode(base.encode())
+    changed2.labels = {"team": "x"}
+    assert job_hash(changed2) != h0
+    changed3 = JobRequest.decode(base.encode())
+    changed3.topic = topic + "!"
+    assert job_hash(changed3) != h0
