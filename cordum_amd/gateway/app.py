@@ -289,7 +289,7 @@ def create_app(
             clone.meta.capability = rem.replacement_capability
         labels = {
             k: v for k, v in clone.labels.items()
-            if not k.startswith("approval_") and k != BUS_MSG_ID_LABEL
+            if not k.lower().startswith("approval_") and k != BUS_MSG_ID_LABEL
         }
         for k, v in rem.add_labels.items():
             labels[k] = v
@@ -510,7 +510,7 @@ def create_app(
 
     @api.post("/workflows/{wf_id}/runs/{run_id}/steps/{step_id}/approve")
     async def approve_wf_step(wf_id: str, run_id: str, step_id: str, request: Request,
-                              p: Principal = Depends(principal)):
+                              p: Principal = Depends(admin)):
         body = await _json_body(request)
         approved = bool(body.get("approved", True))
         try:

@@ -12,7 +12,8 @@ configsvc watch callback (no polling needed in-process).
 In the reference this is a separate gRPC microservice; here it is fused into
 the scheduler's process (the gRPC hop becomes a function
 call). The BATCHED evaluation path for the device data plane lives in
-ops/policy_eval.py and is fed by the same compiled policy.
+ops/pipeline.py (K1 in ops/hip/cordum_kernels.hip) and is fed by the same
+compiled policy (ops/policy_compile.py).
 """
 from __future__ import annotations
 

@@ -10,9 +10,10 @@ empty value never matches), topic globs via path.Match :347-363,
 MCP allow/deny :365-416.
 
 This host evaluator is the oracle for the batched HIP policy kernel
-(ops/policy_eval.py + ops/hip/policy_kernels.hip): the compiler in
-ops/policy_compile.py lowers these rules to int tensors and the kernel must
-agree with `Policy.evaluate` on every input (tests/test_policy_kernel.py).
+(ops/hip/cordum_kernels.hip K1): the compiler in ops/policy_compile.py
+lowers these rules to int tensors and the kernel must agree with
+`Policy.evaluate` on every input (tests/test_policy_compile.py,
+tests/test_gpu_kernels.py).
 """
 from __future__ import annotations
 

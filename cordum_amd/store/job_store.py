@@ -10,7 +10,7 @@ per-job locks :203-217).
 Key naming is kept Redis-compatible (`job:meta:<id>`, `job:index:<STATE>` …)
 so the WAL/checkpoint layout (store/wal.py) matches the reference's Redis
 key scheme. On the GPU data plane the HOT copy of {state, deadline, attempts,
-tenant, topic} lives in the HBM job table (ops/job_table.py); this host store
+tenant, topic} lives in the HBM job table (ops/pipeline.py); this host store
 is the system of record for everything the API can read back.
 """
 from __future__ import annotations

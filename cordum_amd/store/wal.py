@@ -17,6 +17,7 @@ import base64
 import json
 import os
 import tempfile
+import threading
 from pathlib import Path
 
 from ..protocol.capv2 import JobRequest
@@ -29,10 +30,27 @@ class Checkpointer:
         self.dir.mkdir(parents=True, exist_ok=True)
         self._wal_path = self.dir / "wal.jsonl"
         self._wal_file = None
+        # Serializes wal_append against checkpoint: without it a submission
+        # WAL-appended after the snapshot was built would be deleted by the
+        # truncate and lost on crash-restore (gateway handlers run on a
+        # threadpool, so appends are concurrent with checkpoints).
+        self._lock = threading.Lock()
 
     # -- checkpoint -----------------------------------------------------------
     def checkpoint(self) -> str:
-        """Write a full snapshot atomically; truncates the WAL."""
+        """Write a full snapshot atomically; truncates the WAL.
+
+        Holds the WAL lock across quiesce+snapshot+truncate so no append can
+        land between the snapshot build and the truncate."""
+        with self._lock:
+            return self._checkpoint_locked()
+
+    def _checkpoint_locked(self) -> str:
+        # quiesce: every WAL-appended submission is pumped into the stores
+        # before the snapshot is built, so truncating the WAL loses nothing
+        drain = getattr(self.node, "drain", None)
+        if callable(drain):
+            drain()
         snap = {
             "version": 1,
             "jobs": self.node.job_store.snapshot(),
@@ -77,12 +95,13 @@ class Checkpointer:
 
     # -- WAL -------------------------------------------------------------------
     def wal_append(self, req: JobRequest, trace_id: str = "") -> None:
-        if self._wal_file is None:
-            self._wal_file = open(self._wal_path, "a")
-        rec = {"trace_id": trace_id, "req": base64.b64encode(req.encode()).decode()}
-        self._wal_file.write(json.dumps(rec) + "\n")
-        self._wal_file.flush()
-        os.fsync(self._wal_file.fileno())
+        with self._lock:
+            if self._wal_file is None:
+                self._wal_file = open(self._wal_path, "a")
+            rec = {"trace_id": trace_id, "req": base64.b64encode(req.encode()).decode()}
+            self._wal_file.write(json.dumps(rec) + "\n")
+            self._wal_file.flush()
+            os.fsync(self._wal_file.fileno())
 
     def _truncate_wal(self) -> None:
         if self._wal_file is not None:

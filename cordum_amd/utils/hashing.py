@@ -23,10 +23,12 @@ def sha256_hex(data: bytes) -> str:
 
 def job_hash(req: capv2.JobRequest) -> str:
     clone = req.copy()
+    # job_hash.go:27 lowercases the key before the prefix check, so mixed-case
+    # Approval_* labels are stripped too (cross-implementation hash parity)
     clone.labels = {
         k: v
         for k, v in clone.labels.items()
-        if not k.startswith("approval_") and k != BUS_MSG_ID_LABEL
+        if not k.lower().startswith("approval_") and k != BUS_MSG_ID_LABEL
     }
     clone.env = {k: v for k, v in clone.env.items() if k != EFFECTIVE_CONFIG_ENV}
     return sha256_hex(clone.encode())

@@ -154,6 +154,17 @@ def test_approval_requires_admin(node):
     assert c.post(f"/api/v1/approvals/{job_id}/approve", json={}).status_code == 403
 
 
+def test_workflow_step_approval_requires_admin(node):
+    """Workflow step approval is admin-gated like job approvals
+    (gateway.go:3558 requireRole "admin")."""
+    auth = BasicAuthProvider(api_keys=["k"], tenant="default")
+    app = create_app(node, auth=auth)
+    c = TestClient(app)
+    c.headers.update({"X-API-Key": "k", "X-Principal-Role": "user"})
+    r = c.post("/api/v1/workflows/any/runs/any/steps/any/approve", json={})
+    assert r.status_code == 403
+
+
 def test_approval_reject(client, node):
     r = client.post("/api/v1/jobs", json={"prompt": "x", "topic": "job.echo", "risk_tags": ["risky"]})
     job_id = r.json()["job_id"]

@@ -602,6 +602,7 @@ def test_job_hash_strips_exactly_the_approval_binding_exclusions(job_id, topic, 
     excluded.env = dict(excluded.env)
     excluded.env["CORDUM_EFFECTIVE_CONFIG"] = cfg
     excluded.labels = {"approval_granted": "true", "approval_reason": "r",
+                       "Approval_Note": "mixed-case stripped too (job_hash.go:27 lowercases)",
                        "cordum.bus_msg_id": "m1"}
     assert job_hash(excluded) == h0
 
