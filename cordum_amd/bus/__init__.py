@@ -1,1 +1,9 @@
-from .bus import Bus, LoopbackBus, RetryAfter, Subscription, compute_msg_id, subject_matches
+from .bus import (
+    DEFAULT_MAX_DELIVER,
+    Bus,
+    LoopbackBus,
+    RetryAfter,
+    Subscription,
+    compute_msg_id,
+    subject_matches,
+)

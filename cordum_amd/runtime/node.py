@@ -54,7 +54,14 @@ class Node:
         routing: Optional[PoolRouting] = None,
         policy_yaml: str = "",
         safety_cache_ttl_s: float = 30.0,
+        dispatch: str = "host",
+        device=None,
+        backend: Optional[str] = None,
     ):
+        """dispatch="device" runs the safety gate and worker scoring through
+        the K1/K2 kernels (runtime/device_dispatch.py): on a GPU host with the
+        HIP extension (backend="ext", the default there), on CPU via the torch
+        reference ops (backend="ref") — same code path, same semantics."""
         self.clock = clock
         self.bus = LoopbackBus(clock=clock)
         self.job_store = JobStore(clock=clock)
@@ -72,16 +79,56 @@ class Node:
         )
         self.strategy = LeastLoadedStrategy(routing or DEFAULT_ROUTING)
         self.metrics_exporter = PromMetrics()
-        self.scheduler = SchedulerEngine(
-            self.bus,
-            self.job_store,
-            SafetyChecker(self.safety_kernel, clock=clock),
-            self.registry,
-            self.strategy,
-            configsvc=self.configsvc,
-            metrics=self.metrics_exporter,
-            clock=clock,
-        )
+        self.dispatch_mode = dispatch
+        self.device = None
+        self._ext = None
+        self.device_gate = None
+        self.device_worker_table = None
+        self.device_pools: List = []
+        if dispatch == "device":
+            import torch
+
+            from ..ops.worker_table import DeviceWorkerTable
+            from .device_dispatch import DeviceDispatchEngine
+            from .device_gate import DeviceBatchGate
+
+            if device is None:
+                device = "cuda:0" if torch.cuda.is_available() else "cpu"
+            self.device = torch.device(device)
+            backend = backend or ("ext" if self.device.type == "cuda" else "ref")
+            if backend == "ref":
+                from ..ops.pipeline import _RefOps
+
+                self._ext = _RefOps()
+            else:
+                from ..ops import get_ext
+
+                self._ext = get_ext(required=True)
+            self.device_gate = DeviceBatchGate(self.safety_kernel, self.device, self._ext)
+            self.device_worker_table = DeviceWorkerTable(self.device, self._ext)
+            self.scheduler = DeviceDispatchEngine(
+                self.bus,
+                self.job_store,
+                SafetyChecker(self.safety_kernel, clock=clock),
+                self.registry,
+                self.strategy,
+                configsvc=self.configsvc,
+                metrics=self.metrics_exporter,
+                clock=clock,
+                gate=self.device_gate,
+                worker_table=self.device_worker_table,
+            )
+        else:
+            self.scheduler = SchedulerEngine(
+                self.bus,
+                self.job_store,
+                SafetyChecker(self.safety_kernel, clock=clock),
+                self.registry,
+                self.strategy,
+                configsvc=self.configsvc,
+                metrics=self.metrics_exporter,
+                clock=clock,
+            )
         self.workflow_store = WorkflowStore(clock=clock)
         self.workflow = WorkflowEngine(
             self.workflow_store,
@@ -132,22 +179,49 @@ class Node:
         self.workers.append(w)
         return w
 
+    def add_device_worker_pool(self, n_workers: int = 4, pool: str = "default",
+                               topics: Optional[List[str]] = None, **kw):
+        """GPU-resident echo worker pool (ops/worker_pool.py): jobs routed to
+        its worker subjects execute as one batched device kernel."""
+        assert self.dispatch_mode == "device", "device pool needs dispatch='device'"
+        from ..ops.worker_pool import DeviceWorkerPool
+
+        p = DeviceWorkerPool(
+            self.bus, self.memory, self.device, self._ext,
+            n_workers=n_workers, pool=pool, topics=topics or ["job.default"],
+            clock=self.clock, **kw,
+        ).start()
+        self.device_pools.append(p)
+        return p
+
     # -- ticking -----------------------------------------------------------------
+    def _pump_dispatch(self) -> int:
+        """One bus pump + device flush + device-pool batch execution round."""
+        n = self.bus.pump()
+        flush = getattr(self.scheduler, "flush", None)
+        if flush is not None:
+            n += flush()
+        for p in self.device_pools:
+            n += p.execute()
+        return n
+
     def tick(self) -> None:
         """One control-loop iteration: drain bus queues, fire timers,
         heartbeats, reconcilers."""
         for w in self.workers:
             w.send_heartbeat()
-        self.bus.pump()
+        for p in self.device_pools:
+            p.send_heartbeats()
+        self._pump_dispatch()
         self.workflow.pump_timers()
-        self.bus.pump()
+        self._pump_dispatch()
 
     def reconcile(self) -> None:
         self.scheduler_reconciler.tick()
         self.pending_replayer.tick()
         self.run_reconciler.tick()
         self.write_worker_snapshot()
-        self.bus.pump()
+        self._pump_dispatch()
 
     def write_worker_snapshot(self) -> None:
         """Periodic cluster snapshot under `sys:workers:snapshot`
@@ -158,7 +232,7 @@ class Node:
     def drain(self, max_iters: int = 64) -> None:
         """Pump until quiescent at the current clock (tests / sync callers)."""
         for _ in range(max_iters):
-            n = self.bus.pump()
+            n = self._pump_dispatch()
             n += self.workflow.pump_timers()
             if n == 0:
                 return

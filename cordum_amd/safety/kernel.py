@@ -211,12 +211,12 @@ class SafetyKernel:
         return self.evaluate(req)
 
     def explain_rows(self, req: PolicyCheckRequest) -> List[Dict[str, Any]]:
-        inp = self._input_from_request(req)
+        inp = self.input_from_request(req)
         with self._mu:
             policy = self._policy
         return policy.explain(inp) if policy else []
 
-    def _input_from_request(self, req: PolicyCheckRequest) -> pol.PolicyInput:
+    def input_from_request(self, req: PolicyCheckRequest) -> pol.PolicyInput:
         meta = req.meta
         tenant = (req.tenant or "").strip()
         with self._mu:
@@ -262,11 +262,33 @@ class SafetyKernel:
         if not topic.startswith("job."):
             return PolicyCheckResponse(decision=DecisionType.DENY, reason="unsupported topic")
 
-        inp = self._input_from_request(req)
+        inp = self.input_from_request(req)
 
         pd = pol.PolicyDecision(decision=pol.DECISION_ALLOW)
         if policy is not None:
             pd = policy.evaluate(inp)
+        resp = self.finish_response(pd, inp, topic, req, policy, snapshot)
+        if cache_key:
+            cached = PolicyCheckResponse.decode(resp.encode())
+            cached.approval_ref = ""
+            self._cache_put(cache_key, cached)
+        return resp
+
+    def finish_response(
+        self,
+        pd: pol.PolicyDecision,
+        inp: pol.PolicyInput,
+        topic: str,
+        req: PolicyCheckRequest,
+        policy: Optional[pol.SafetyPolicy],
+        snapshot: str,
+    ) -> PolicyCheckResponse:
+        """The post-first-match half of evaluate(): tenant MCP gate, decision
+        normalization, effective-config restrictions, response assembly.
+        Shared verbatim with the batched device path (runtime/device_gate.py)
+        so kernel==device decisions are the same code for everything after
+        the rule match (which is what the K1 kernel computes)."""
+        if policy is not None:
             tp = policy.tenants.get(inp.tenant)
             if tp is not None:
                 ok, mcp_reason = pol.mcp_allowed(tp.mcp, inp.mcp)
@@ -303,7 +325,7 @@ class SafetyKernel:
                 decision, reason = DecisionType.DENY, mcp_reason
 
         approval_required = pd.approval_required or decision == DecisionType.REQUIRE_HUMAN
-        resp = PolicyCheckResponse(
+        return PolicyCheckResponse(
             decision=decision,
             reason=reason,
             policy_snapshot=snapshot,
@@ -313,11 +335,6 @@ class SafetyKernel:
             approval_ref=req.job_id if approval_required else "",
             remediations=list(pd.remediations),
         )
-        if cache_key:
-            cached = PolicyCheckResponse.decode(resp.encode())
-            cached.approval_ref = ""
-            self._cache_put(cache_key, cached)
-        return resp
 
     # -- decision cache (kernel.go:259-303) --------------------------------------
     def _cache_key(self, req: PolicyCheckRequest, snapshot: str) -> str:

@@ -232,15 +232,7 @@ class SafetyPolicy:
     def evaluate(self, inp: PolicyInput) -> PolicyDecision:
         for rule in self.effective_rules():
             if match_rule(rule.match, inp):
-                decision = normalize_decision(rule.decision)
-                return PolicyDecision(
-                    decision=decision,
-                    reason=rule.reason,
-                    rule_id=rule.id,
-                    constraints=rule.constraints,
-                    approval_required=decision == DECISION_REQUIRE_APPROVAL,
-                    remediations=list(rule.remediations),
-                )
+                return decision_from_rule(rule)
         return PolicyDecision(decision=DECISION_ALLOW)
 
     def explain(self, inp: PolicyInput) -> List[Dict[str, Any]]:
@@ -249,6 +241,21 @@ class SafetyPolicy:
         for rule in self.effective_rules():
             rows.append({"rule_id": rule.id, "matched": match_rule(rule.match, inp), "decision": normalize_decision(rule.decision)})
         return rows
+
+
+def decision_from_rule(rule: PolicyRule) -> PolicyDecision:
+    """PolicyDecision for a matched rule — the evaluate() match arm, shared
+    with the batched device path so a K1 first-match index yields the exact
+    same decision record (safety_policy.go:187-206)."""
+    decision = normalize_decision(rule.decision)
+    return PolicyDecision(
+        decision=decision,
+        reason=rule.reason,
+        rule_id=rule.id,
+        constraints=rule.constraints,
+        approval_required=decision == DECISION_REQUIRE_APPROVAL,
+        remediations=list(rule.remediations),
+    )
 
 
 def legacy_rules(policy: SafetyPolicy) -> List[PolicyRule]:

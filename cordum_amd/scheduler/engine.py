@@ -168,49 +168,10 @@ class Engine:
 
         self.attach_effective_config(req)
         record = self.check_safety_decision(req)
-        decision = record.decision
-
-        if decision in ("allow", "allow_with_constraints"):
-            if record.constraints is not None:
-                apply_constraints(req, record.constraints)
-        elif decision == "throttle":
-            raise RetryAfter(SAFETY_THROTTLE_DELAY_S, f"safety throttle: {record.reason}")
-        elif decision == "require_approval":
-            self._set_state_quiet(job_id, JobState.APPROVAL_REQUIRED)
+        if not self.apply_decision(req, record):
             return
-        elif decision == "deny":
-            self._set_state_quiet(job_id, JobState.DENIED)
-            self.metrics.inc_safety_denied(topic)
-            self.emit_dlq(job_id, topic, JobStatus.DENIED, record.reason, errs.REASON_SAFETY_DENIED)
+        if not self.pre_dispatch_checks(req, record):
             return
-        else:
-            self._set_state_quiet(job_id, JobState.DENIED)
-            self.metrics.inc_safety_denied(topic)
-            self.emit_dlq(job_id, topic, JobStatus.DENIED, record.reason, errs.REASON_SAFETY_UNKNOWN)
-            return
-
-        # max retries (engine.go:349-365)
-        max_retries = _max_retries(record.constraints)
-        if max_retries > 0:
-            attempts = int(self.job_store.get_job_meta(job_id).get("attempts", 0) or 0)
-            if attempts >= max_retries + 1:
-                reason = f"max retries exceeded (attempts={attempts}, max_retries={max_retries})"
-                self._set_state_quiet(job_id, JobState.FAILED)
-                self.emit_dlq(job_id, topic, JobStatus.FAILED, reason, errs.REASON_MAX_RETRIES)
-                return
-
-        # tenant concurrency (engine.go:367-381)
-        max_concurrent = _max_concurrent(record.constraints)
-        if max_concurrent > 0:
-            tenant = _tenant_of(req)
-            if self.job_store.tenant_active_count(tenant) > max_concurrent:
-                # the job itself is PENDING and counted; limit is on others + self
-                raise RetryAfter(RETRY_DELAY_NO_WORKERS_S, "tenant limit")
-
-        # deadline registration (engine.go:383-390)
-        if req.budget is not None and req.budget.deadline_ms > 0:
-            at = self.clock.now_micros() + req.budget.deadline_ms * 1000
-            self.job_store.set_deadline(job_id, at)
 
         # routing (engine.go:392-407)
         workers = self.registry.snapshot()
@@ -224,36 +185,101 @@ class Engine:
             self.emit_dlq(job_id, topic, JobStatus.FAILED, str(e), errs.reason_code_for(e))
             return
 
+        self.dispatch(req, trace_id, subject)
+
+    def apply_decision(self, req: JobRequest, record: SafetyDecisionRecord) -> bool:
+        """Decision branches (engine.go:298-347): returns True to continue to
+        dispatch, False when the job reached a decision-terminal state; raises
+        RetryAfter for throttle."""
+        job_id = (req.job_id or "").strip()
+        topic = (req.topic or "").strip()
+        decision = record.decision
+        if decision in ("allow", "allow_with_constraints"):
+            if record.constraints is not None:
+                apply_constraints(req, record.constraints)
+            return True
+        if decision == "throttle":
+            raise RetryAfter(SAFETY_THROTTLE_DELAY_S, f"safety throttle: {record.reason}")
+        if decision == "require_approval":
+            self._set_state_quiet(job_id, JobState.APPROVAL_REQUIRED)
+            return False
+        if decision == "deny":
+            self._set_state_quiet(job_id, JobState.DENIED)
+            self.metrics.inc_safety_denied(topic)
+            self.emit_dlq(job_id, topic, JobStatus.DENIED, record.reason, errs.REASON_SAFETY_DENIED)
+            return False
+        self._set_state_quiet(job_id, JobState.DENIED)
+        self.metrics.inc_safety_denied(topic)
+        self.emit_dlq(job_id, topic, JobStatus.DENIED, record.reason, errs.REASON_SAFETY_UNKNOWN)
+        return False
+
+    def pre_dispatch_checks(self, req: JobRequest, record: SafetyDecisionRecord) -> bool:
+        """Max retries + tenant concurrency + deadline registration
+        (engine.go:349-390); raises RetryAfter on tenant limit."""
+        job_id = (req.job_id or "").strip()
+        topic = (req.topic or "").strip()
+        max_retries = _max_retries(record.constraints)
+        if max_retries > 0:
+            attempts = int(self.job_store.get_job_meta(job_id).get("attempts", 0) or 0)
+            if attempts >= max_retries + 1:
+                reason = f"max retries exceeded (attempts={attempts}, max_retries={max_retries})"
+                self._set_state_quiet(job_id, JobState.FAILED)
+                self.emit_dlq(job_id, topic, JobStatus.FAILED, reason, errs.REASON_MAX_RETRIES)
+                return False
+
+        max_concurrent = _max_concurrent(record.constraints)
+        if max_concurrent > 0:
+            tenant = _tenant_of(req)
+            if self.job_store.tenant_active_count(tenant) > max_concurrent:
+                # the job itself is PENDING and counted; limit is on others + self
+                raise RetryAfter(RETRY_DELAY_NO_WORKERS_S, "tenant limit")
+
+        if req.budget is not None and req.budget.deadline_ms > 0:
+            at = self.clock.now_micros() + req.budget.deadline_ms * 1000
+            self.job_store.set_deadline(job_id, at)
+        return True
+
+    def dispatch(self, req: JobRequest, trace_id: str, subject: str) -> None:
+        """SCHEDULED -> publish -> DISPATCHED -> RUNNING (engine.go:409-441)."""
+        job_id = (req.job_id or "").strip()
         self.job_store.set_state(job_id, JobState.SCHEDULED)
         packet = BusPacket(trace_id=trace_id, protocol_version=1, job_request=req)
         self.bus.publish(subject, packet)
-        self.metrics.inc_dispatched(topic)
+        self.metrics.inc_dispatched((req.topic or "").strip())
         self.job_store.set_state(job_id, JobState.DISPATCHED)
         self.job_store.set_state(job_id, JobState.RUNNING)
 
     # -- safety ---------------------------------------------------------------
-    def check_safety_decision(self, req: JobRequest) -> SafetyDecisionRecord:
+    def approval_resume_record(self, req: JobRequest) -> Optional[SafetyDecisionRecord]:
+        """engine.go:484-522: the approval label is honored only when the
+        stored decision required approval AND the job hash matches. Returns
+        the allow record, or None (proceed to a fresh safety check)."""
         job_id = (req.job_id or "").strip()
         approved = (req.labels or {}).get("approval_granted", "").strip().lower() == "true"
-        if approved:
-            prev = self.job_store.get_safety_decision(job_id)
-            if prev is not None and (prev.approval_required or prev.decision == "require_approval") and prev.job_hash:
-                h = job_hash(req)
-                if h == prev.job_hash:
-                    record = SafetyDecisionRecord(
-                        decision="allow",
-                        reason="approval granted",
-                        checked_at=self.clock.now_micros(),
-                        constraints=prev.constraints,
-                        policy_snapshot=prev.policy_snapshot,
-                        rule_id=prev.rule_id,
-                        job_hash=prev.job_hash,
-                    )
-                    self.job_store.set_safety_decision(job_id, record)
-                    return record
-                # hash mismatch: approval label ignored
+        if not approved:
+            return None
+        prev = self.job_store.get_safety_decision(job_id)
+        if prev is not None and (prev.approval_required or prev.decision == "require_approval") and prev.job_hash:
+            h = job_hash(req)
+            if h == prev.job_hash:
+                record = SafetyDecisionRecord(
+                    decision="allow",
+                    reason="approval granted",
+                    checked_at=self.clock.now_micros(),
+                    constraints=prev.constraints,
+                    policy_snapshot=prev.policy_snapshot,
+                    rule_id=prev.rule_id,
+                    job_hash=prev.job_hash,
+                )
+                self.job_store.set_safety_decision(job_id, record)
+                return record
+            # hash mismatch: approval label ignored
+        return None
 
-        record = self.safety.check(req)
+    def finalize_safety_record(self, req: JobRequest, record: SafetyDecisionRecord) -> SafetyDecisionRecord:
+        """Approval-binding fixups + persistence, shared by the per-job host
+        check and the batched device gate (runtime/device_dispatch.py)."""
+        job_id = (req.job_id or "").strip()
         if not record.checked_at:
             record.checked_at = self.clock.now_micros()
         if record.approval_required and record.decision in ("allow", "allow_with_constraints"):
@@ -262,6 +288,12 @@ class Engine:
             record.job_hash = job_hash(req)
         self.job_store.set_safety_decision(job_id, record)
         return record
+
+    def check_safety_decision(self, req: JobRequest) -> SafetyDecisionRecord:
+        resumed = self.approval_resume_record(req)
+        if resumed is not None:
+            return resumed
+        return self.finalize_safety_record(req, self.safety.check(req))
 
     # -- results ---------------------------------------------------------------
     def handle_job_result(self, res: JobResult) -> None:

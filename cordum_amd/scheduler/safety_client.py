@@ -67,6 +67,21 @@ def build_check_request(req: JobRequest) -> PolicyCheckRequest:
     return check
 
 
+def record_from_response(resp: PolicyCheckResponse) -> SafetyDecisionRecord:
+    """PolicyCheckResponse -> stored decision record (shared with the batched
+    device gate so both paths persist identical records)."""
+    return SafetyDecisionRecord(
+        decision=DECISION_NAMES.get(resp.decision, ""),
+        reason=resp.reason,
+        rule_id=resp.rule_id,
+        policy_snapshot=resp.policy_snapshot,
+        constraints=resp.constraints,
+        approval_required=resp.approval_required,
+        approval_ref=resp.approval_ref,
+        remediations=list(resp.remediations),
+    )
+
+
 class SafetyChecker:
     """Wraps a kernel-like object exposing check(PolicyCheckRequest)."""
 
@@ -133,16 +148,7 @@ class SafetyChecker:
             self._record_failure()
             return SafetyDecisionRecord(decision="deny", reason=f"safety kernel error: {e}")
         self._record_success()
-        return SafetyDecisionRecord(
-            decision=DECISION_NAMES.get(resp.decision, ""),
-            reason=resp.reason,
-            rule_id=resp.rule_id,
-            policy_snapshot=resp.policy_snapshot,
-            constraints=resp.constraints,
-            approval_required=resp.approval_required,
-            approval_ref=resp.approval_ref,
-            remediations=list(resp.remediations),
-        )
+        return record_from_response(resp)
 
     def list_snapshots(self):
         return self._kernel.list_snapshots()

@@ -129,7 +129,13 @@ class LeastLoadedStrategy(Strategy):
         with self._mu:
             return self._routing.clone()
 
-    def pick_subject(self, req: JobRequest, workers: Dict[str, Heartbeat]) -> str:
+    def resolve(self, req: JobRequest) -> Tuple[List[str], Dict[str, str], str]:
+        """The host-cheap half of pick_subject: topic->pool resolution,
+        preferred_pool hint, `requires` filtering, placement-label extraction.
+        Returns (eligible_pools, required_labels, preferred_worker_id); raises
+        NoPoolMapping exactly where pick_subject would. The scoring half runs
+        either in the host loop below or in the K2 device kernel
+        (ops/worker_table.py) over the same resolution."""
         if req is None or not req.topic:
             raise errs.NoPoolMapping("missing topic")
         routing = self.current_routing()
@@ -155,9 +161,12 @@ class LeastLoadedStrategy(Strategy):
         ]
         if not eligible:
             raise errs.NoPoolMapping("no pool satisfies requires")
+        return eligible, required_labels, labels.get("preferred_worker_id", "")
+
+    def pick_subject(self, req: JobRequest, workers: Dict[str, Heartbeat]) -> str:
+        eligible, required_labels, preferred = self.resolve(req)
         pool_set = set(eligible)
 
-        preferred = labels.get("preferred_worker_id", "")
         if preferred:
             hb = workers.get(preferred)
             if hb is not None and hb.pool in pool_set and matches_labels(hb, required_labels) and not is_overloaded(hb):

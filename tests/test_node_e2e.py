@@ -14,14 +14,18 @@ from cordum_amd.utils.clock import ManualClock
 from cordum_amd.workflow import RUN_SUCCEEDED, RUN_WAITING, Step, Workflow, WorkflowRun
 
 
-@pytest.fixture
-def node():
+@pytest.fixture(params=["host", "device"])
+def node(request):
+    """Every e2e flow runs twice: host dispatch engine and the device
+    dispatch engine (K1/K2 batched path, CPU reference backend in CI —
+    the identical code path the GPU runs with the HIP extension)."""
     clock = ManualClock()
     routing = PoolRouting(
         topics={"job.default": ["default"], "job.echo": ["default"]},
         pools={"default": PoolProfile()},
     )
-    n = Node(clock=clock, routing=routing).start()
+    n = Node(clock=clock, routing=routing, dispatch=request.param,
+             backend="ref" if request.param == "device" else None).start()
     n.add_worker("w1", topics=["job.default", "job.echo"])
     return n
 
