@@ -2,13 +2,23 @@
 """Flagship benchmark: jobs dispatched/sec (whole node) + p50 dispatch->result
 latency — the BASELINE.json metric on synthetic echo jobs.
 
-Each timed step runs ONE full control-plane tick on every rank
-(cordum_amd/ops/pipeline.py): submit -> K1 policy gate -> heartbeat
-all-gather -> K2 least-loaded routing -> state transitions -> RCCL
-all-to-all dispatch -> device echo workers -> result return -> SUCCEEDED.
-Work is synthetic (random job descriptors + random 256 B payloads), nothing
-is cached across steps (a ring of distinct pre-encoded batches), and every
-job traverses the full state machine.
+Two timed windows, both reported in the ONE output line:
+
+1. END-TO-END INGEST (the headline `value`): each step encodes a FRESH random
+   job batch on the host (vectorized JobEncoder semantics, new content every
+   step), H2D-copies descriptors + payload bytes into the device staging
+   tensors, runs the full control-plane tick (K1 policy gate -> heartbeat
+   fan-in -> K2 least-loaded routing -> state transitions -> dispatch ->
+   device echo workers -> SUCCEEDED), and reads the per-job results
+   (checksums + decisions) back to the host. Encode, H2D, tick and D2H are
+   all inside the timed region.
+2. IN-HBM TICK (`config.in_hbm_tick_jobs_per_s`): the steady-state dispatch
+   tick over batches resident in HBM (a ring of pre-staged distinct batches,
+   hipGraph-captured) — the rate of the dispatch engine itself once jobs are
+   in device memory.
+
+Counting: `value` counts DISPATCHED+COMPLETED jobs only; DENIED jobs are
+reported separately in config.denied_jobs (they are decided, not dispatched).
 
 Usage (the driver contract):
   python bench.py [--gpus N] [--steps K] [--warmup W]
@@ -71,6 +81,7 @@ def main() -> int:
 
     from cordum_amd.ops.pipeline import DevicePipeline
 
+    backend = "ext" if use_gpu else "ref"
     pipe = DevicePipeline(
         device=device,
         batch_size=args.batch,
@@ -79,7 +90,21 @@ def main() -> int:
         payload_words=max(1, args.payload_bytes // 4),
         world_size=world_size,
         rank=rank,
-        backend="ext" if use_gpu else "ref",
+        backend=backend,
+    )
+    # e2e window pipeline: bitset K1 so the captured graph reads the staging
+    # tensors the per-step H2D copies write (smaller ring: only slot 0 is used)
+    pipe_e2e = DevicePipeline(
+        device=device,
+        batch_size=args.batch,
+        n_local_workers=args.workers,
+        n_rules=args.rules,
+        payload_words=max(1, args.payload_bytes // 4),
+        world_size=world_size,
+        rank=rank,
+        n_batches=2,
+        backend=backend,
+        use_mfma=False,
     )
 
     def barrier():
@@ -88,16 +113,31 @@ def main() -> int:
         if use_gpu:
             torch.cuda.synchronize(device)
 
-    # warmup
+    # warmup both windows (graph capture, allocator steady state, RCCL init)
     for _ in range(args.warmup):
         pipe.tick()
+    for _ in range(max(2, args.warmup)):
+        pipe_e2e.tick_e2e()
     barrier()
 
+    # ---- window 1: end-to-end ingest (headline) --------------------------------
+    e2e_steps = []
+    e2e_completed = 0
+    e2e_denied = 0
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        st = pipe_e2e.tick_e2e()
+        e2e_steps.append(st.wall_s)
+        e2e_completed += st.completed
+        e2e_denied += st.denied
+    barrier()
+    e2e_elapsed = time.perf_counter() - t0
+
+    # ---- window 2: in-HBM steady-state tick -------------------------------------
     if use_gpu:
         # sync-free timed window: stats fold into a device accumulator inside
-        # the captured tick (read ONCE after the closing barrier), per-step
-        # latency from hipEvents — the contract brackets the WHOLE window
-        # with barrier+synchronize, not every step
+        # the captured tick (read ONCE after the closing barrier); per-step
+        # latency from hipEvents
         pipe.reset_stats()
         events = [torch.cuda.Event(enable_timing=True) for _ in range(args.steps + 1)]
         t0 = time.perf_counter()
@@ -106,42 +146,46 @@ def main() -> int:
             pipe.tick_async()
             events[s + 1].record()
         barrier()
-        elapsed = time.perf_counter() - t0
-        completed, denied = pipe.collect_stats()
-        completed += denied  # denied jobs are also fully decided
-        step_times = [events[s].elapsed_time(events[s + 1]) / 1000.0
+        tick_elapsed = time.perf_counter() - t0
+        tick_completed, tick_denied = pipe.collect_stats()
+        tick_steps = [events[s].elapsed_time(events[s + 1]) / 1000.0
                       for s in range(args.steps)]
     else:
         t0 = time.perf_counter()
-        step_times = []
-        completed = 0
-        denied = 0
+        tick_steps = []
+        tick_completed = 0
+        tick_denied = 0
         for _ in range(args.steps):
             st = pipe.tick()
-            step_times.append(st.wall_s)
-            completed += st.completed + st.denied  # denied jobs also fully decided
-            denied += st.denied
+            tick_steps.append(st.wall_s)
+            tick_completed += st.completed
+            tick_denied += st.denied
         barrier()
-        elapsed = time.perf_counter() - t0
+        tick_elapsed = time.perf_counter() - t0
 
     # MAX elapsed over ranks; SUM of completed jobs over ranks
-    if world_size > 1:
-        t = torch.tensor([elapsed], device=device)
+    def reduce_window(elapsed, completed):
+        if world_size == 1:
+            return elapsed, completed
+        t = torch.tensor([elapsed], device=device, dtype=torch.float64)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
-        elapsed = float(t.item())
         c = torch.tensor([completed], device=device, dtype=torch.float64)
         dist.all_reduce(c, op=dist.ReduceOp.SUM)
-        completed = int(c.item())
+        return float(t.item()), int(c.item())
 
-    jobs_per_sec = completed / elapsed
-    ms_per_step = elapsed / args.steps * 1000.0
-    p50_ms = statistics.median(step_times) * 1000.0
+    e2e_elapsed, e2e_completed = reduce_window(e2e_elapsed, e2e_completed)
+    tick_elapsed, tick_completed = reduce_window(tick_elapsed, tick_completed)
+
+    e2e_jobs_per_sec = e2e_completed / e2e_elapsed
+    tick_jobs_per_sec = tick_completed / tick_elapsed
+    ms_per_step = e2e_elapsed / args.steps * 1000.0
+    p50_ms = statistics.median(e2e_steps) * 1000.0
 
     baseline_sustained = 38234.0  # BASELINE.md peak-stress sustained jobs/s
     if rank == 0:
         out = {
             "metric": "jobs dispatched/sec (whole node)",
-            "value": round(jobs_per_sec, 1),
+            "value": round(e2e_jobs_per_sec, 1),
             "unit": "jobs/s",
             "n_gpus": world_size,
             "steps": args.steps,
@@ -149,9 +193,11 @@ def main() -> int:
             "ms_per_step": round(ms_per_step, 3),
             "higher_is_better": True,
             "scaling": "weak",
-            "vs_baseline": round(jobs_per_sec / baseline_sustained, 2),
+            "vs_baseline": round(e2e_jobs_per_sec / baseline_sustained, 2),
             "dtype": "int64-bitset/int32 (control-plane integer path)",
-            "data": "synthetic (random job descriptors + random 256B payloads, ring of distinct batches)",
+            "data": "synthetic (FRESH random job batch encoded+H2D per step; "
+                    "payload bytes from a pre-generated pinned ring, stamped "
+                    "per step and H2D-copied in the timed window)",
             "config": {
                 "model": "cordum control-plane dispatch pipeline (echo workers)",
                 "global_batch": args.batch * world_size,
@@ -159,8 +205,13 @@ def main() -> int:
                 "parallelism": f"shard{world_size}",
                 "rules": args.rules,
                 "workers_per_rank": args.workers,
+                "value_is": "end-to-end ingest rate (host encode + H2D + full "
+                            "tick + result D2H all timed)",
+                "in_hbm_tick_jobs_per_s": round(tick_jobs_per_sec, 1),
+                "in_hbm_tick_ms_per_step": round(tick_elapsed / args.steps * 1000.0, 3),
                 "p50_dispatch_result_ms": round(p50_ms, 3),
-                "denied_jobs": denied,
+                "denied_jobs": e2e_denied + tick_denied,
+                "denied_excluded_from_value": True,
             },
         }
         print(json.dumps(out))
