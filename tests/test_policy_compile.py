@@ -122,3 +122,42 @@ def test_empty_policy():
     enc = JobEncoder(compiled)
     batch = enc.encode([pol.PolicyInput(tenant="x", topic="job.y")])
     assert int(first_match_reference(compiled, batch)[0]) == -1
+
+
+def test_synthetic_encoder_rows_are_exact_encodings():
+    """SyntheticEncoder (the e2e bench's vectorized encoder) must emit rows
+    that are exact JobEncoder encodings: every dimension of every job equals
+    the LUT row built through the real encoder (or zero = absent)."""
+    import torch
+
+    from cordum_amd.ops.pipeline import SyntheticEncoder, make_synthetic_policy
+    from cordum_amd.ops.policy_compile import (
+        ALL_REQUIRES, DIM_RISK, DIM_TENANT, DIM_TOPIC, JobBatch, compile_policy,
+    )
+
+    compiled = compile_policy(make_synthetic_policy(256), words=1)
+    assert compiled.exact
+    enc = SyntheticEncoder(compiled, seed=3)
+    J, W = 512, compiled.words
+    out = JobBatch(
+        torch.zeros((J, 7, W), dtype=torch.int64),
+        torch.zeros((J, 2, W), dtype=torch.int64),
+        torch.zeros((J,), dtype=torch.uint8),
+        torch.zeros((J, 4, W), dtype=torch.int64),
+        torch.zeros((J,), dtype=torch.uint8),
+    )
+    enc.fresh(out)
+    tenant_rows = {int(r[0]) for r in enc.tenant_lut}
+    topic_rows = {int(r[0]) for r in enc.topic_lut}
+    risk_rows = {int(r[0]) for r in enc.risk_lut} | {0}
+    req_rows = {int(r[0]) for r in enc.req_lut} | {0}
+    for j in range(J):
+        assert int(out.any_bits[j, DIM_TENANT, 0]) in tenant_rows
+        assert int(out.any_bits[j, DIM_TOPIC, 0]) in topic_rows
+        assert int(out.any_bits[j, DIM_RISK, 0]) in risk_rows
+        assert int(out.all_bits[j, ALL_REQUIRES, 0]) in req_rows
+    # fresh batches differ step to step
+    out2 = JobBatch(out.any_bits.clone(), out.all_bits.clone(), out.secrets.clone(),
+                    out.mcp_bits.clone(), out.mcp_used.clone())
+    enc.fresh(out2)
+    assert not torch.equal(out.any_bits, out2.any_bits)
