@@ -52,8 +52,6 @@ def main() -> int:
     import torch
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
-    rank = int(os.environ.get("RANK", "0"))
-    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
     if world_size == 1 and args.gpus > 1:
         print(
             "error: for --gpus N>1 launch via torch.distributed.run "
@@ -69,15 +67,9 @@ def main() -> int:
 
     import torch.distributed as dist
 
-    if world_size > 1:
-        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-        dist.init_process_group(backend="nccl" if use_gpu else "gloo",
-                                rank=rank, world_size=world_size)
-    if use_gpu:
-        torch.cuda.set_device(local_rank)
-        device = torch.device(f"cuda:{local_rank}")
-    else:
-        device = torch.device("cpu")
+    from cordum_amd.parallel import init_fabric
+
+    rank, world_size, device = init_fabric()
 
     from cordum_amd.ops.pipeline import DevicePipeline
 

@@ -11,9 +11,11 @@ torch.distributed (backend "nccl" == RCCL on ROCm):
                         splits sync)
  - result return     -> mirrored all_to_all_single
 
-The implementation lives in ops/pipeline.py (DevicePipeline._tick_padded and
-the pack/echo/load-feedback kernels in ops/hip/cordum_kernels.hip); this
-package exposes the fabric helpers.
+The collective call sites live in ops/pipeline.py (DevicePipeline's
+heartbeat fan-in at :905-908/:1056-1058 and the padded exchanges at
+:917-927/:1093-1136, backed by the pack/echo/load-feedback kernels in
+ops/hip/cordum_kernels.hip); this package owns process-group bring-up
+(init_fabric), used by bench.py and the serve launcher.
 """
 from __future__ import annotations
 
@@ -44,19 +46,3 @@ def init_fabric(backend: str | None = None) -> tuple[int, int, torch.device]:
             world_size=world,
         )
     return rank, world, device
-
-
-def heartbeat_all_gather(global_buf: torch.Tensor, local: torch.Tensor) -> None:
-    """sys.heartbeat -> all-gather of per-rank worker load vectors."""
-    if dist.is_initialized() and dist.get_world_size() > 1:
-        dist.all_gather_into_tensor(global_buf, local)
-    else:
-        global_buf.copy_(local)
-
-
-def padded_all_to_all(recv: torch.Tensor, send: torch.Tensor) -> None:
-    """Equal-split all_to_all (the padded dispatch exchange)."""
-    if dist.is_initialized() and dist.get_world_size() > 1:
-        dist.all_to_all_single(recv, send)
-    else:
-        recv.copy_(send)
