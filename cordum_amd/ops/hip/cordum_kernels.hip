@@ -965,7 +965,13 @@ __global__ __launch_bounds__(BLOCK) void pack_by_dest_kernel(
     int* __restrict__ send_slots,             // [world*B]
     int* __restrict__ send_widx,              // [world*B] local worker idx
     int* __restrict__ send_cnt,               // [world] pre-zeroed
-    int nwl, int cap)
+    int nwl, int cap,
+    int* __restrict__ rq_src,                 // [rq_cap] requeue ring: slot (this batch)
+    int* __restrict__ rq_widx,                // [rq_cap] global widx
+    int* __restrict__ rq_attempts,            // [rq_cap] delivery attempts
+    int* __restrict__ rq_count,               // [1]
+    unsigned long long* __restrict__ rq_dead, // [1] dropped after max attempts / ring full
+    int rq_cap)
 {
     const int i = blockIdx.x * BLOCK + threadIdx.x;
     const int lane = threadIdx.x % WAVE;
@@ -992,15 +998,101 @@ __global__ __launch_bounds__(BLOCK) void pack_by_dest_kernel(
         }
         pending &= ~same;
     }
-    if (my_pos >= 0 && my_pos < cap) {
+    if (my_pos < 0) return;
+    if (my_pos < cap) {
         send_slots[(size_t)dest * cap + my_pos] = slot;
         send_widx[(size_t)dest * cap + my_pos] = widx % nwl;
+    } else {
+        // destination segment full: NAK-with-redelivery (bus/nats.go:146-168
+        // analog) — park in the requeue ring, re-admitted next tick
+        const int p = atomicAdd(rq_count, 1);
+        if (p < rq_cap) {
+            rq_src[p] = slot;      // payload row in this batch's home arena
+            rq_widx[p] = widx;
+            rq_attempts[p] = 1;
+        } else {
+            atomicAdd(rq_dead, 1ull);  // ring full -> DLQ-bound, counted
+        }
     }
 }
 
-// gather payload rows into the padded send arena, region-valid
+// Redelivery pack: drain LAST tick's requeue ring into the padded send
+// segments ahead of the fresh batch (redelivered jobs have priority, like
+// a NAK'd message re-surfacing before new publishes). Entries that find
+// their destination full again go back to the (new) requeue ring with
+// attempts+1; past RQ_MAX_DELIVER they are dropped to rq_dead (the DLQ
+// analog of JetStream max-deliver).
+#define RQ_MAX_DELIVER 5
+__global__ __launch_bounds__(BLOCK) void pack_requeue_kernel(
+    const int* __restrict__ rq_prev_widx,     // [rq_cap]
+    const int* __restrict__ rq_prev_attempts, // [rq_cap]
+    const int* __restrict__ rq_prev_count,    // [1]
+    int* __restrict__ send_slots,             // [world*cap]
+    int* __restrict__ send_widx,              // [world*cap]
+    int* __restrict__ send_cnt,               // [world] pre-zeroed
+    int nwl, int cap,
+    int* __restrict__ rq_src,
+    int* __restrict__ rq_widx,
+    int* __restrict__ rq_attempts,
+    int* __restrict__ rq_count,
+    unsigned long long* __restrict__ rq_dead,
+    int rq_cap)
+{
+    const int j = blockIdx.x * BLOCK + threadIdx.x;
+    const int n = min(*rq_prev_count, rq_cap);
+    if (j >= n) return;
+    const int widx = rq_prev_widx[j];
+    const int dest = widx / nwl;
+    const int pos = atomicAdd(&send_cnt[dest], 1);
+    if (pos < cap) {
+        // flagged slot: payload comes from rq_prev_payload row j
+        send_slots[(size_t)dest * cap + pos] = -1 - j;
+        send_widx[(size_t)dest * cap + pos] = widx % nwl;
+    } else {
+        const int att = rq_prev_attempts[j] + 1;
+        if (att > RQ_MAX_DELIVER) {
+            atomicAdd(rq_dead, 1ull);
+            return;
+        }
+        const int p = atomicAdd(rq_count, 1);
+        if (p < rq_cap) {
+            rq_src[p] = -1 - j;   // payload row in rq_prev_payload
+            rq_widx[p] = widx;
+            rq_attempts[p] = att;
+        } else {
+            atomicAdd(rq_dead, 1ull);
+        }
+    }
+}
+
+// Materialize this tick's requeue-ring payloads into their own arena so the
+// entries survive home-arena reuse (ring slots rotate every tick): source is
+// the current batch (src >= 0) or last tick's rq arena (flagged).
+__global__ __launch_bounds__(BLOCK) void materialize_rq_payload_kernel(
+    const unsigned int* __restrict__ payload,         // [B*stride] this batch
+    const unsigned int* __restrict__ rq_prev_payload, // [rq_cap*stride]
+    const int* __restrict__ rq_src,                   // [rq_cap]
+    const int* __restrict__ rq_count,                 // [1]
+    unsigned int* __restrict__ rq_payload,            // [rq_cap*stride]
+    int stride, int rq_cap)
+{
+    const int j = (blockIdx.x * BLOCK + threadIdx.x) / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const int n = min(*rq_count, rq_cap);
+    if (j >= n) return;
+    const int src = rq_src[j];
+    const unsigned int* row = src >= 0 ? payload + (size_t)src * stride
+                                       : rq_prev_payload + (size_t)(-1 - src) * stride;
+    unsigned int* dst = rq_payload + (size_t)j * stride;
+    for (int k = lane; k < stride; k += WAVE)
+        dst[k] = row[k];
+}
+
+// gather payload rows into the padded send arena, region-valid; flagged
+// (negative) slots are redeliveries whose payload lives in rq_prev_payload
 __global__ __launch_bounds__(BLOCK) void gather_payload_padded_kernel(
     const unsigned int* __restrict__ payload,  // [B*stride] home arena
+    const unsigned int* __restrict__ rq_prev_payload, // [rq_cap*stride]
     const int* __restrict__ send_slots,        // [world*cap]
     const int* __restrict__ send_cnt,          // [world]
     unsigned int* __restrict__ send_payload,   // [world*cap*stride]
@@ -1011,10 +1103,13 @@ __global__ __launch_bounds__(BLOCK) void gather_payload_padded_kernel(
     if (w >= world * cap) return;
     const int region = w / cap, e = w % cap;
     if (e >= send_cnt[region]) return;
-    const size_t src = (size_t)send_slots[w] * stride;
+    const int sslot = send_slots[w];
+    const unsigned int* src = sslot >= 0
+        ? payload + (size_t)sslot * stride
+        : rq_prev_payload + (size_t)(-1 - sslot) * stride;
     const size_t dst = (size_t)w * stride;
     for (int k = lane; k < stride; k += WAVE)
-        send_payload[dst + k] = payload[src + k];
+        send_payload[dst + k] = src[k];
 }
 
 // region-valid echo over the padded receive arena
@@ -1056,6 +1151,7 @@ __global__ __launch_bounds__(BLOCK) void apply_transitions_padded_kernel(
     if (i >= world * cap) return;
     if ((i % cap) >= cnt[i / cap]) return;
     const int slot = slots_pad[i];
+    if (slot < 0) return;  // redelivered entry: its batch slot was recycled
     const unsigned char from = states[slot];
     if (!d_transition_lut[from * N_STATES + to]) return;
     states[slot] = to;
@@ -1544,23 +1640,65 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor> run_readi
 void pack_by_dest(torch::Tensor routable_slots, torch::Tensor routable_widx,
                   torch::Tensor routable_count, torch::Tensor send_slots,
                   torch::Tensor send_widx, torch::Tensor send_cnt,
-                  int64_t nwl, int64_t cap, int64_t capacity)
+                  int64_t nwl, int64_t cap, int64_t capacity,
+                  torch::Tensor rq_src, torch::Tensor rq_widx,
+                  torch::Tensor rq_attempts, torch::Tensor rq_count,
+                  torch::Tensor rq_dead)
 {
     const int blocks = ((int)capacity + BLOCK - 1) / BLOCK;
     hipLaunchKernelGGL(pack_by_dest_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
         routable_slots.data_ptr<int>(), routable_widx.data_ptr<int>(),
         routable_count.data_ptr<int>(), send_slots.data_ptr<int>(),
-        send_widx.data_ptr<int>(), send_cnt.data_ptr<int>(), (int)nwl, (int)cap);
+        send_widx.data_ptr<int>(), send_cnt.data_ptr<int>(), (int)nwl, (int)cap,
+        rq_src.data_ptr<int>(), rq_widx.data_ptr<int>(),
+        rq_attempts.data_ptr<int>(), rq_count.data_ptr<int>(),
+        (unsigned long long*)rq_dead.data_ptr<int64_t>(),
+        (int)rq_src.size(0));
 }
 
-void gather_payload_padded(torch::Tensor payload, torch::Tensor send_slots,
+void pack_requeue(torch::Tensor rq_prev_widx, torch::Tensor rq_prev_attempts,
+                  torch::Tensor rq_prev_count, torch::Tensor send_slots,
+                  torch::Tensor send_widx, torch::Tensor send_cnt,
+                  int64_t nwl, int64_t cap,
+                  torch::Tensor rq_src, torch::Tensor rq_widx,
+                  torch::Tensor rq_attempts, torch::Tensor rq_count,
+                  torch::Tensor rq_dead)
+{
+    const int rq_cap = (int)rq_src.size(0);
+    const int blocks = (rq_cap + BLOCK - 1) / BLOCK;
+    hipLaunchKernelGGL(pack_requeue_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        rq_prev_widx.data_ptr<int>(), rq_prev_attempts.data_ptr<int>(),
+        rq_prev_count.data_ptr<int>(), send_slots.data_ptr<int>(),
+        send_widx.data_ptr<int>(), send_cnt.data_ptr<int>(), (int)nwl, (int)cap,
+        rq_src.data_ptr<int>(), rq_widx.data_ptr<int>(),
+        rq_attempts.data_ptr<int>(), rq_count.data_ptr<int>(),
+        (unsigned long long*)rq_dead.data_ptr<int64_t>(), rq_cap);
+}
+
+void materialize_rq_payload(torch::Tensor payload, torch::Tensor rq_prev_payload,
+                            torch::Tensor rq_src, torch::Tensor rq_count,
+                            torch::Tensor rq_payload, int64_t stride)
+{
+    const int rq_cap = (int)rq_src.size(0);
+    const int blocks = (rq_cap + (BLOCK / WAVE) - 1) / (BLOCK / WAVE);
+    hipLaunchKernelGGL(materialize_rq_payload_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        (const unsigned int*)payload.data_ptr<int32_t>(),
+        (const unsigned int*)rq_prev_payload.data_ptr<int32_t>(),
+        rq_src.data_ptr<int>(), rq_count.data_ptr<int>(),
+        (unsigned int*)rq_payload.data_ptr<int32_t>(), (int)stride, rq_cap);
+}
+
+void gather_payload_padded(torch::Tensor payload, torch::Tensor rq_prev_payload,
+                           torch::Tensor send_slots,
                            torch::Tensor send_cnt, torch::Tensor send_payload,
                            int64_t stride, int64_t cap, int64_t world)
 {
     const int waves = (int)(world * cap);
     const int blocks = (waves + (BLOCK / WAVE) - 1) / (BLOCK / WAVE);
     hipLaunchKernelGGL(gather_payload_padded_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
-        (const unsigned int*)payload.data_ptr<int32_t>(), send_slots.data_ptr<int>(),
+        (const unsigned int*)payload.data_ptr<int32_t>(),
+        (const unsigned int*)rq_prev_payload.data_ptr<int32_t>(),
+        send_slots.data_ptr<int>(),
         send_cnt.data_ptr<int>(), (unsigned int*)send_payload.data_ptr<int32_t>(),
         (int)stride, (int)cap, (int)world);
 }
@@ -1630,6 +1768,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("load_feedback", &load_feedback, "per-worker active-count histogram");
     m.def("run_readiness", &run_readiness, "K3 run/step readiness sweep");
     m.def("pack_by_dest", &pack_by_dest, "padded per-destination dispatch pack");
+    m.def("pack_requeue", &pack_requeue, "redeliver last tick's requeue ring into the send segments");
+    m.def("materialize_rq_payload", &materialize_rq_payload, "copy requeued payload rows into the rq arena");
     m.def("gather_payload_padded", &gather_payload_padded, "payload gather into padded send arena");
     m.def("echo_padded", &echo_padded, "region-valid echo over padded recv arena");
     m.def("apply_transitions_padded", &apply_transitions_padded, "K5 over padded slot list");

@@ -222,15 +222,32 @@ class _RefOps:
         to = torch.full((n,), int(to_state), dtype=torch.uint8)
         apply_transitions_ref(states, attempts, deadlines, slots[:n], to)
 
-    def pack_by_dest(self, rs, rw, rc, ss, sw, sc, nwl, cap, capacity):
+    def pack_by_dest(self, rs, rw, rc, ss, sw, sc, nwl, cap, capacity,
+                     rq_src, rq_widx, rq_attempts, rq_count, rq_dead):
         from .reference import pack_by_dest_ref
 
-        pack_by_dest_ref(rs, rw, rc, ss, sw, sc, int(nwl), int(cap))
+        pack_by_dest_ref(rs, rw, rc, ss, sw, sc, int(nwl), int(cap),
+                         rq_src, rq_widx, rq_attempts, rq_count, rq_dead)
 
-    def gather_payload_padded(self, payload, ss, sc, sp, stride, cap, world):
+    def pack_requeue(self, pw, pa, pc, ss, sw, sc, nwl, cap,
+                     rq_src, rq_widx, rq_attempts, rq_count, rq_dead):
+        from .reference import pack_requeue_ref
+
+        pack_requeue_ref(pw, pa, pc, ss, sw, sc, int(nwl), int(cap),
+                         rq_src, rq_widx, rq_attempts, rq_count, rq_dead)
+
+    def materialize_rq_payload(self, payload, prev_payload, rq_src, rq_count,
+                               rq_payload, stride):
+        from .reference import materialize_rq_payload_ref
+
+        materialize_rq_payload_ref(payload, prev_payload, rq_src, rq_count,
+                                   rq_payload, int(stride))
+
+    def gather_payload_padded(self, payload, prev_payload, ss, sc, sp, stride, cap, world):
         from .reference import gather_payload_padded_ref
 
-        gather_payload_padded_ref(payload, ss, sc, sp, int(stride), int(cap), int(world))
+        gather_payload_padded_ref(payload, prev_payload, ss, sc, sp,
+                                  int(stride), int(cap), int(world))
 
     def echo_padded(self, rp, rc, ra, rsums, stride, cap, world):
         from .reference import echo_padded_ref
@@ -392,7 +409,9 @@ class DevicePipeline:
         seed: int = 7,
         backend: str = "ext",
         use_mfma: Optional[bool] = None,
+        pad_cap: Optional[int] = None,
     ):
+        self._pad_cap_override = pad_cap
         device = torch.device(device)
         self.ext = _RefOps() if backend == "ref" else get_ext(required=True)
         self.device = device
@@ -811,9 +830,9 @@ class DevicePipeline:
             return
         B, W, world, dev = self.B, self.payload_words, self.world, self.device
         # per-destination capacity: spreading makes the expected share B/world;
-        # 4x headroom covers imbalance, and overflow (counted as unrouted
-        # backpressure, like a worker-at-capacity NAK) is accounted below
-        self.pad_cap = min(B, max(64, (4 * B) // max(1, world)))
+        # 4x headroom covers imbalance, and overflow is parked in the device
+        # requeue ring below and redelivered next tick (NAK redelivery)
+        self.pad_cap = self._pad_cap_override or min(B, max(64, (4 * B) // max(1, world)))
         cap = self.pad_cap
 
         def zi(n):
@@ -829,6 +848,20 @@ class DevicePipeline:
         self.pad_res = torch.zeros(world * cap * W, dtype=torch.int32, device=dev)
         self.pad_sums = zi(world * cap)
         self.pad_sums_back = zi(world * cap)
+        # device requeue ring: capacity-overflowed dispatches are parked here
+        # (payload copied out of the rotating home arena) and redelivered
+        # ahead of the next tick's fresh batch — the NAK/redelivery analog of
+        # bus/nats.go:146-168; rq_dead counts max-deliver/ring-full drops
+        self.rq_src = zi(B)
+        self.rq_widx = zi(B)
+        self.rq_attempts = zi(B)
+        self.rq_count = zi(1)
+        self.rq_payload = torch.zeros(B * W, dtype=torch.int32, device=dev)
+        self.rq_prev_widx = zi(B)
+        self.rq_prev_attempts = zi(B)
+        self.rq_prev_count = zi(1)
+        self.rq_prev_payload = torch.zeros(B * W, dtype=torch.int32, device=dev)
+        self.rq_dead = torch.zeros(1, dtype=torch.int64, device=dev)
         if not hasattr(self, "_pend_states"):
             self._pend_states = torch.full((B,), PENDING, dtype=torch.uint8, device=dev)
 
@@ -868,16 +901,34 @@ class DevicePipeline:
             ext.apply_transitions_dyn(self.states, self.attempts, self.deadlines,
                                       self.routable_slots, self.routable_count, st, B)
         cap = self.pad_cap
+        # rotate the requeue ring: last tick's parked entries become this
+        # tick's redeliveries (they pack FIRST — redelivery has priority)
+        self.rq_prev_widx.copy_(self.rq_widx)
+        self.rq_prev_attempts.copy_(self.rq_attempts)
+        self.rq_prev_count.copy_(self.rq_count)
+        self.rq_prev_payload.copy_(self.rq_payload)
+        self.rq_count.zero_()
         self.pad_send_cnt.zero_()
+        ext.pack_requeue(self.rq_prev_widx, self.rq_prev_attempts, self.rq_prev_count,
+                         self.pad_send_slots, self.pad_send_widx, self.pad_send_cnt,
+                         self.NWL, cap,
+                         self.rq_src, self.rq_widx, self.rq_attempts,
+                         self.rq_count, self.rq_dead)
         ext.pack_by_dest(self.routable_slots, self.routable_widx, self.routable_count,
                          self.pad_send_slots, self.pad_send_widx, self.pad_send_cnt,
-                         self.NWL, cap, B)
-        # clamp: entries beyond capacity were not written (overflow = dropped
-        # for retry, the at-capacity NAK analog); dispatched = sum of clamped
+                         self.NWL, cap, B,
+                         self.rq_src, self.rq_widx, self.rq_attempts,
+                         self.rq_count, self.rq_dead)
+        # clamp: entries beyond capacity were parked in the requeue ring;
+        # dispatched = sum of clamped counts (fresh + redelivered this tick)
         self.pad_send_cnt.clamp_(max=cap)
         self._counts[3:4].copy_(self.pad_send_cnt.sum().reshape(1))
-        ext.gather_payload_padded(self.payloads[slot], self.pad_send_slots, self.pad_send_cnt,
+        ext.gather_payload_padded(self.payloads[slot], self.rq_prev_payload,
+                                  self.pad_send_slots, self.pad_send_cnt,
                                   self.pad_send_payload, self.payload_words, cap, world)
+        ext.materialize_rq_payload(self.payloads[slot], self.rq_prev_payload,
+                                   self.rq_src, self.rq_count,
+                                   self.rq_payload, self.payload_words)
 
     def _pad_g3(self) -> None:
         ext, world, cap = self.ext, self.world, self.pad_cap
@@ -961,11 +1012,14 @@ class DevicePipeline:
             g4 = torch.cuda.CUDAGraph()
             with torch.cuda.graph(g4):
                 self._pad_g4()
-            # reset the durable DLQ ring + stats accumulator: the eager
-            # warmup ran for real
+            # reset the durable DLQ ring + stats accumulator + requeue ring:
+            # the eager warmup ran for real
             self.dlq_head.zero_()
             self.dlq_ring.fill_(-1)
             self.acc_counts.zero_()
+            self.rq_count.zero_()
+            self.rq_prev_count.zero_()
+            self.rq_dead.zero_()
             torch.cuda.synchronize(self.device)
             self._pad_graphs = (g1s, g2s, g3, g4)
         except Exception as e:
@@ -1010,10 +1064,13 @@ class DevicePipeline:
             self._refresh_order()
         counts = self._counts.cpu()  # the tick's only host sync
         denied = int(counts[0])
-        dispatched = int(counts[3])  # routable minus capacity overflow
+        dispatched = int(counts[3])  # fresh + redelivered packed this tick
+        # parked for next-tick redelivery (raw count may exceed ring capacity:
+        # the excess was counted into rq_dead at park time)
+        backlog = min(int(self.rq_count.cpu()[0]), int(self.rq_src.shape[0]))
         dt = time.perf_counter() - t0
         stats = TickStats(completed=dispatched, denied=denied,
-                          unrouted=B - dispatched - denied, wall_s=dt)
+                          unrouted=backlog, wall_s=dt)
         self.total_completed += stats.completed
         self.total_denied += stats.denied
         return stats

@@ -194,28 +194,81 @@ def run_readiness_ref(step_state, deps_mask, n_steps, run_active):
 # -- padded cross-rank dispatch references ------------------------------------
 
 
+RQ_MAX_DELIVER = 5
+
+
+def _rq_park(rq_src, rq_widx, rq_attempts, rq_count, rq_dead,
+             src: int, widx: int, attempts: int) -> None:
+    p = int(rq_count[0])
+    rq_count[0] = p + 1
+    if p < rq_src.shape[0]:
+        rq_src[p] = src
+        rq_widx[p] = widx
+        rq_attempts[p] = attempts
+    else:
+        rq_dead[0] += 1
+
+
 def pack_by_dest_ref(routable_slots, routable_widx, routable_count,
-                     send_slots, send_widx, send_cnt, nwl: int, cap: int):
+                     send_slots, send_widx, send_cnt, nwl: int, cap: int,
+                     rq_src, rq_widx, rq_attempts, rq_count, rq_dead):
     n = int(routable_count[0])
     for i in range(n):
         slot = int(routable_slots[i])
         widx = int(routable_widx[i])
         dest = widx // nwl
         pos = int(send_cnt[dest])
+        send_cnt[dest] += 1
         if pos < cap:
             send_slots[dest * cap + pos] = slot
             send_widx[dest * cap + pos] = widx % nwl
+        else:
+            _rq_park(rq_src, rq_widx, rq_attempts, rq_count, rq_dead,
+                     slot, widx, 1)
+
+
+def pack_requeue_ref(rq_prev_widx, rq_prev_attempts, rq_prev_count,
+                     send_slots, send_widx, send_cnt, nwl: int, cap: int,
+                     rq_src, rq_widx, rq_attempts, rq_count, rq_dead):
+    n = min(int(rq_prev_count[0]), int(rq_prev_widx.shape[0]))
+    for j in range(n):
+        widx = int(rq_prev_widx[j])
+        dest = widx // nwl
+        pos = int(send_cnt[dest])
         send_cnt[dest] += 1
+        if pos < cap:
+            send_slots[dest * cap + pos] = -1 - j
+            send_widx[dest * cap + pos] = widx % nwl
+        else:
+            att = int(rq_prev_attempts[j]) + 1
+            if att > RQ_MAX_DELIVER:
+                rq_dead[0] += 1
+            else:
+                _rq_park(rq_src, rq_widx, rq_attempts, rq_count, rq_dead,
+                         -1 - j, widx, att)
 
 
-def gather_payload_padded_ref(payload, send_slots, send_cnt, send_payload,
-                              stride: int, cap: int, world: int):
+def materialize_rq_payload_ref(payload, rq_prev_payload, rq_src, rq_count,
+                               rq_payload, stride: int):
     pl = payload.view(-1, stride)
+    prev = rq_prev_payload.view(-1, stride)
+    dst = rq_payload.view(-1, stride)
+    n = min(int(rq_count[0]), int(rq_src.shape[0]))
+    for j in range(n):
+        s = int(rq_src[j])
+        dst[j] = pl[s] if s >= 0 else prev[-1 - s]
+
+
+def gather_payload_padded_ref(payload, rq_prev_payload, send_slots, send_cnt,
+                              send_payload, stride: int, cap: int, world: int):
+    pl = payload.view(-1, stride)
+    prev = rq_prev_payload.view(-1, stride)
     sp = send_payload.view(-1, stride)
     for r in range(world):
         for e in range(int(send_cnt[r])):
             i = r * cap + e
-            sp[i] = pl[int(send_slots[i])]
+            s = int(send_slots[i])
+            sp[i] = pl[s] if s >= 0 else prev[-1 - s]
 
 
 def echo_padded_ref(recv_payload, recv_cnt, res_arena, res_sums,
@@ -239,6 +292,8 @@ def apply_transitions_padded_ref(states, attempts, deadlines, slots_pad, cnt,
     for r in range(world):
         for e in range(int(cnt[r])):
             slot = int(slots_pad[r * cap + e])
+            if slot < 0:  # redelivered entry: its batch slot was recycled
+                continue
             frm = int(states[slot])
             if lut[frm][to_state]:
                 states[slot] = to_state

@@ -228,13 +228,24 @@ def test_padded_dispatch_kernels_match_reference(ext):
     routable_widx = torch.randint(0, world * nwl, (n,), dtype=torch.int32, generator=g)
     payload = torch.randint(-(1 << 31), (1 << 31) - 1, (B * stride,), dtype=torch.int32, generator=g)
 
+    def rq_bufs(dev_=None):
+        kw = {"device": dev_} if dev_ is not None else {}
+        return (torch.zeros(B, dtype=torch.int32, **kw),
+                torch.zeros(B, dtype=torch.int32, **kw),
+                torch.zeros(B, dtype=torch.int32, **kw),
+                torch.zeros(1, dtype=torch.int32, **kw),
+                torch.zeros(1, dtype=torch.int64, **kw))
+
     # references
     ss_r = torch.zeros(world * B, dtype=torch.int32)
     sw_r = torch.zeros(world * B, dtype=torch.int32)
     sc_r = torch.zeros(world, dtype=torch.int32)
-    ref.pack_by_dest_ref(routable_slots, routable_widx, routable_count, ss_r, sw_r, sc_r, nwl, B)
+    rq_r = rq_bufs()
+    ref.pack_by_dest_ref(routable_slots, routable_widx, routable_count, ss_r, sw_r, sc_r, nwl, B,
+                         *rq_r)
+    rq_prev_payload = torch.zeros(B * stride, dtype=torch.int32)
     sp_r = torch.zeros(world * B * stride, dtype=torch.int32)
-    ref.gather_payload_padded_ref(payload, ss_r, sc_r, sp_r, stride, B, world)
+    ref.gather_payload_padded_ref(payload, rq_prev_payload, ss_r, sc_r, sp_r, stride, B, world)
     ra_r = torch.zeros_like(sp_r)
     sums_r = torch.zeros(world * B, dtype=torch.int32)
     ref.echo_padded_ref(sp_r, sc_r, ra_r, sums_r, stride, B, world)
@@ -250,9 +261,11 @@ def test_padded_dispatch_kernels_match_reference(ext):
     ss = torch.zeros(world * B, dtype=torch.int32, device=d)
     sw = torch.zeros(world * B, dtype=torch.int32, device=d)
     sc = torch.zeros(world, dtype=torch.int32, device=d)
+    rq_d = rq_bufs(d)
     ext.pack_by_dest(routable_slots.to(d), routable_widx.to(d), routable_count.to(d),
-                     ss, sw, sc, nwl, B, n)
+                     ss, sw, sc, nwl, B, n, *rq_d)
     assert torch.equal(sc.cpu(), sc_r)
+    assert int(rq_d[3].cpu()[0]) == 0 and int(rq_d[4].cpu()[0]) == 0  # cap=B: no overflow
     # per-dest sets must match (order within a segment may differ)
     for r in range(world):
         k = int(sc_r[r])
@@ -262,7 +275,7 @@ def test_padded_dispatch_kernels_match_reference(ext):
     # use the REFERENCE packing on device for the order-dependent stages
     ss_d, sw_d, sc_d = ss_r.to(d), sw_r.to(d), sc_r.to(d)
     sp = torch.zeros(world * B * stride, dtype=torch.int32, device=d)
-    ext.gather_payload_padded(payload.to(d), ss_d, sc_d, sp, stride, B, world)
+    ext.gather_payload_padded(payload.to(d), rq_prev_payload.to(d), ss_d, sc_d, sp, stride, B, world)
     assert torch.equal(sp.cpu(), sp_r)
     ra = torch.zeros_like(sp)
     sums = torch.zeros(world * B, dtype=torch.int32, device=d)
@@ -278,6 +291,93 @@ def test_padded_dispatch_kernels_match_reference(ext):
     wal = torch.zeros(nwl, dtype=torch.int32, device=d)
     ext.load_feedback_padded(sw_d, sc_d, wal, B, world)
     assert torch.equal(wal.cpu(), wal_r)
+
+
+def test_requeue_kernels_match_reference(ext):
+    """pack_by_dest overflow parking + pack_requeue redelivery +
+    materialize_rq_payload vs the CPU oracles, with a forced-small cap."""
+    from cordum_amd.ops import reference as ref
+
+    g = torch.Generator().manual_seed(99)
+    B, world, nwl, stride, cap = 512, 4, 8, 8, 16
+    n = 400
+    routable_count = torch.tensor([n], dtype=torch.int32)
+    routable_slots = torch.randperm(B, generator=g)[:n].to(torch.int32)
+    routable_widx = torch.randint(0, world * nwl, (n,), dtype=torch.int32, generator=g)
+    payload = torch.randint(-(1 << 31), (1 << 31) - 1, (B * stride,),
+                            dtype=torch.int32, generator=g)
+    prev_payload = torch.randint(-(1 << 31), (1 << 31) - 1, (B * stride,),
+                                 dtype=torch.int32, generator=g)
+    prev_widx = torch.randint(0, world * nwl, (B,), dtype=torch.int32, generator=g)
+    prev_att = torch.randint(1, 5, (B,), dtype=torch.int32, generator=g)
+    prev_count = torch.tensor([60], dtype=torch.int32)
+
+    def mk(dev_=None):
+        kw = {"device": dev_} if dev_ is not None else {}
+        return {
+            "ss": torch.zeros(world * cap, dtype=torch.int32, **kw),
+            "sw": torch.zeros(world * cap, dtype=torch.int32, **kw),
+            "sc": torch.zeros(world, dtype=torch.int32, **kw),
+            "rq_src": torch.zeros(B, dtype=torch.int32, **kw),
+            "rq_widx": torch.zeros(B, dtype=torch.int32, **kw),
+            "rq_att": torch.zeros(B, dtype=torch.int32, **kw),
+            "rq_count": torch.zeros(1, dtype=torch.int32, **kw),
+            "rq_dead": torch.zeros(1, dtype=torch.int64, **kw),
+            "rq_payload": torch.zeros(B * stride, dtype=torch.int32, **kw),
+        }
+
+    r = mk()
+    ref.pack_requeue_ref(prev_widx, prev_att, prev_count, r["ss"], r["sw"], r["sc"],
+                         nwl, cap, r["rq_src"], r["rq_widx"], r["rq_att"],
+                         r["rq_count"], r["rq_dead"])
+    ref.pack_by_dest_ref(routable_slots, routable_widx, routable_count,
+                         r["ss"], r["sw"], r["sc"], nwl, cap,
+                         r["rq_src"], r["rq_widx"], r["rq_att"],
+                         r["rq_count"], r["rq_dead"])
+    ref.materialize_rq_payload_ref(payload, prev_payload, r["rq_src"],
+                                   r["rq_count"], r["rq_payload"], stride)
+
+    d = dev()
+    v = mk(d)
+    ext.pack_requeue(prev_widx.to(d), prev_att.to(d), prev_count.to(d),
+                     v["ss"], v["sw"], v["sc"], nwl, cap,
+                     v["rq_src"], v["rq_widx"], v["rq_att"],
+                     v["rq_count"], v["rq_dead"])
+    ext.pack_by_dest(routable_slots.to(d), routable_widx.to(d), routable_count.to(d),
+                     v["ss"], v["sw"], v["sc"], nwl, cap, n,
+                     v["rq_src"], v["rq_widx"], v["rq_att"],
+                     v["rq_count"], v["rq_dead"])
+    ext.materialize_rq_payload(payload.to(d), prev_payload.to(d), v["rq_src"],
+                               v["rq_count"], v["rq_payload"], stride)
+
+    # counts agree exactly; ring contents agree as SETS (atomic order differs)
+    assert torch.equal(v["sc"].cpu().clamp(max=cap), r["sc"].clamp(max=cap))
+    assert int(v["rq_count"].cpu()[0]) == int(r["rq_count"][0])
+    assert int(v["rq_dead"].cpu()[0]) == int(r["rq_dead"][0])
+    nq = min(int(r["rq_count"][0]), B)
+    got = sorted(zip(v["rq_src"][:nq].cpu().tolist(), v["rq_widx"][:nq].cpu().tolist(),
+                     v["rq_att"][:nq].cpu().tolist()))
+    want = sorted(zip(r["rq_src"][:nq].tolist(), r["rq_widx"][:nq].tolist(),
+                      r["rq_att"][:nq].tolist()))
+    assert got == want
+    # per-dest packed sets agree (flagged + fresh together)
+    for q in range(world):
+        k = min(int(r["sc"][q]), cap)
+        a = sorted(zip(v["ss"][q * cap:q * cap + k].cpu().tolist(),
+                       v["sw"][q * cap:q * cap + k].cpu().tolist()))
+        b = sorted(zip(r["ss"][q * cap:q * cap + k].tolist(),
+                       r["sw"][q * cap:q * cap + k].tolist()))
+        assert a == b
+    # materialized payload rows: each device ring entry's payload matches the
+    # row its (device-order) src points at
+    pl = payload.view(B, stride)
+    prev = prev_payload.view(B, stride)
+    vp = v["rq_payload"].cpu().view(B, stride)
+    vsrc = v["rq_src"].cpu()
+    for j in range(nq):
+        s = int(vsrc[j])
+        want_row = pl[s] if s >= 0 else prev[-1 - s]
+        assert torch.equal(vp[j], want_row)
 
 
 def test_fused_tick_kernels_match_refops(ext):
