@@ -458,6 +458,297 @@ __global__ __launch_bounds__(BLOCK) void run_readiness_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// K3-WF: device-resident workflow engine tick (config #3)
+// ---------------------------------------------------------------------------
+// Batched equivalent of the per-run scheduleReady walk + HandleJobResult +
+// updateRunStatus loop (workflow/engine.go:453-827, :1524-1560, :1623-1699)
+// over HBM-resident run tables. Step kinds: WORKER (1 child job), FOR_EACH
+// (fanout_n children, aggregated like engine.go:1623-1645), APPROVAL
+// (WAITING hold until a host grant — the human-in-the-loop hop stays on the
+// host, like gateway.go:3553), CONDITION (pre-evaluated bit -> SUCCEEDED or
+// SKIPPED), DELAY (next_ready tick gate). Failed children retry with
+// exponential tick backoff up to max_retries (computeBackoff analog), then
+// the step FAILs and the run FAILs (failure blocks downstream,
+// depsSatisfied :1231-1242 — SKIPPED satisfies, FAILED never does).
+#define WFS_PENDING 0
+#define WFS_DISPATCHED 1
+#define WFS_WAITING 2
+#define WFS_SUCCEEDED 3
+#define WFS_FAILED 4
+#define WFS_SKIPPED 5
+#define WFK_WORKER 0
+#define WFK_FOR_EACH 1
+#define WFK_APPROVAL 2
+#define WFK_CONDITION 3
+#define WFK_DELAY 4
+
+__global__ __launch_bounds__(BLOCK) void wf_sweep_kernel(
+    unsigned char* __restrict__ step_state,   // [NR*64]
+    const long long* __restrict__ deps_mask,  // [NR*64]
+    const unsigned char* __restrict__ n_steps,// [NR]
+    const unsigned char* __restrict__ run_active, // [NR]
+    const unsigned char* __restrict__ step_kind,  // [NR*64]
+    const long long* __restrict__ cond_bits,  // [NR] bit s = condition value
+    const int* __restrict__ next_ready,       // [NR*64] tick gate
+    int tick,
+    int* __restrict__ disp_runs,              // [cap] WORKER/FOR_EACH to expand
+    int* __restrict__ disp_steps,             // [cap]
+    int* __restrict__ disp_count,             // [1]
+    int* __restrict__ appr_runs,              // [acap] fresh WAITING approvals
+    int* __restrict__ appr_steps,             // [acap]
+    int* __restrict__ appr_count,             // [1]
+    int NR, int cap, int acap)
+{
+    const int run = blockIdx.x * BLOCK + threadIdx.x;
+    const int lane = threadIdx.x % WAVE;
+    long long disp = 0, appr = 0;
+    if (run < NR && run_active[run]) {
+        const int ns = n_steps[run];
+        unsigned char* st = &step_state[(size_t)run * 64];
+        unsigned long long satisfied = 0;
+        for (int t = 0; t < ns; ++t) {
+            const unsigned char s = st[t];
+            satisfied |= (unsigned long long)(s == WFS_SUCCEEDED || s == WFS_SKIPPED) << t;
+        }
+        const long long* dm = &deps_mask[(size_t)run * 64];
+        const long long cb = cond_bits[run];
+        for (int s = 0; s < ns; ++s) {
+            if (st[s] != WFS_PENDING) continue;
+            if (((unsigned long long)dm[s] & ~satisfied) != 0) continue;
+            if (next_ready[(size_t)run * 64 + s] > tick) continue;
+            switch (step_kind[(size_t)run * 64 + s]) {
+            case WFK_WORKER:
+            case WFK_FOR_EACH:
+                disp |= 1ll << s;
+                break;
+            case WFK_APPROVAL:
+                st[s] = WFS_WAITING;   // held for the host-side grant
+                appr |= 1ll << s;
+                break;
+            case WFK_CONDITION:
+                st[s] = ((cb >> s) & 1) ? WFS_SUCCEEDED : WFS_SKIPPED;
+                break;
+            case WFK_DELAY:            // gate already passed
+                st[s] = WFS_SUCCEEDED;
+                break;
+            }
+        }
+    }
+    while (true) {   // wave-aggregated appends, one ready step per pass
+        if (!__any(disp != 0)) break;
+        int step = -1;
+        if (disp != 0) { step = __ffsll(disp) - 1; disp &= disp - 1; }
+        const int pos = wave_append_slot(step >= 0, disp_count, lane);
+        if (pos >= 0 && pos < cap) { disp_runs[pos] = run; disp_steps[pos] = step; }
+    }
+    while (true) {
+        if (!__any(appr != 0)) break;
+        int step = -1;
+        if (appr != 0) { step = __ffsll(appr) - 1; appr &= appr - 1; }
+        const int pos = wave_append_slot(step >= 0, appr_count, lane);
+        if (pos >= 0 && pos < acap) { appr_runs[pos] = run; appr_steps[pos] = step; }
+    }
+}
+
+// for_each expansion: one wave per dispatch entry; all-or-nothing child-slot
+// reservation in the child arena (full arena -> step stays PENDING and the
+// next sweep retries, natural backpressure like max_parallel windowing)
+__global__ __launch_bounds__(BLOCK) void wf_expand_kernel(
+    const int* __restrict__ disp_runs,
+    const int* __restrict__ disp_steps,
+    const int* __restrict__ disp_count,
+    unsigned char* __restrict__ step_state,   // [NR*64]
+    int* __restrict__ children_todo,          // [NR*64] children left to emit
+    int* __restrict__ children_out,           // [NR*64] in flight
+    int* __restrict__ child_tag,              // [CB] run*64+step
+    int* __restrict__ child_seq,              // [CB] per-step emission ordinal
+    int* __restrict__ child_widx,             // [CB] global worker pick
+    int* __restrict__ child_count,            // [1]
+    int* __restrict__ children_emitted,       // [NR*64] monotone ordinal source
+    const int* __restrict__ order,            // [NWG] spread order (K2c)
+    const int* __restrict__ valid_count,      // [1]
+    int disp_cap, int CB)
+{
+    const int e = (blockIdx.x * BLOCK + threadIdx.x) / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    if (e >= min(*disp_count, disp_cap)) return;
+    const int run = disp_runs[e];
+    const int step = disp_steps[e];
+    const size_t rs = (size_t)run * 64 + step;
+    int base = 0, todo = 0, seq0 = 0;
+    if (lane == 0) {
+        todo = children_todo[rs];
+        // monotone per-step emission counter: a deterministic per-child
+        // ordinal stream identical on every backend regardless of atomic
+        // packing order (retried children get FRESH ordinals)
+        seq0 = children_emitted[rs];
+        base = atomicAdd(child_count, todo);
+        if (base + todo > CB) {          // arena full: roll back, retry later
+            atomicSub(child_count, todo);
+            todo = -1;
+        } else {
+            step_state[rs] = WFS_DISPATCHED;
+            children_out[rs] += todo;
+            children_emitted[rs] += todo;
+            children_todo[rs] = 0;
+        }
+    }
+    base = __shfl(base, 0, WAVE);
+    todo = __shfl(todo, 0, WAVE);
+    seq0 = __shfl(seq0, 0, WAVE);
+    if (todo <= 0) return;
+    const int tag = run * 64 + step;
+    const int V = max(1, *valid_count);
+    for (int k = lane; k < todo; k += WAVE) {
+        child_tag[base + k] = tag;
+        child_seq[base + k] = seq0 + k;
+        // spread: children are unconstrained -> round-robin the least-loaded
+        // order (K2c semantics, deterministic)
+        child_widx[base + k] = order[(base + k) % min(V, 1024)];
+    }
+}
+
+// deterministic per-entry failure injection (config #5 retry waves): a
+// splitmix-style hash of (tag, tick, arena position) -> fail if < fail_ppt
+// per mille. The CPU oracle mirrors this bit-for-bit.
+__device__ __forceinline__ unsigned int wf_mix(unsigned int x) {
+    x ^= x >> 16; x *= 0x7feb352du; x ^= x >> 15; x *= 0x846ca68bu; x ^= x >> 16;
+    return x;
+}
+
+// apply child results over the packed send segments (the owner rank's view
+// of what it dispatched this tick; echo workers are deterministic-success,
+// failures are injected by a hash of the child's (tag, emission ordinal) —
+// independent of packing order, so every backend agrees). Flagged entries
+// are redeliveries whose tag/seq live in the rq_prev carries.
+__global__ __launch_bounds__(BLOCK) void wf_apply_kernel(
+    const int* __restrict__ send_slots,       // [world*cap]
+    const int* __restrict__ send_cnt,         // [world]
+    const int* __restrict__ child_tag,        // [CB]
+    const int* __restrict__ child_seq,        // [CB]
+    const int* __restrict__ rq_prev_tag,      // [rq_cap]
+    const int* __restrict__ rq_prev_seq,      // [rq_cap]
+    int* __restrict__ children_done,          // [NR*64]
+    int* __restrict__ children_fail,          // [NR*64]
+    int* __restrict__ children_out,           // [NR*64]
+    int fail_ppt, int cap, int world)
+{
+    const int i = blockIdx.x * BLOCK + threadIdx.x;
+    if (i >= world * cap) return;
+    if ((i % cap) >= min(send_cnt[i / cap], cap)) return;
+    const int sslot = send_slots[i];
+    const int tag = sslot >= 0 ? child_tag[sslot] : rq_prev_tag[-1 - sslot];
+    const int seq = sslot >= 0 ? child_seq[sslot] : rq_prev_seq[-1 - sslot];
+    const unsigned int h = wf_mix((unsigned int)tag * 2654435761u
+                                  ^ (unsigned int)seq * 40503u);
+    if ((int)(h % 1000u) < fail_ppt) atomicAdd(&children_fail[tag], 1);
+    else atomicAdd(&children_done[tag], 1);
+    atomicSub(&children_out[tag], 1);
+}
+
+// dead-letter application: entries dropped by the requeue ring (max-deliver
+// exceeded / ring full) count as failed children so their steps can retry or
+// fail instead of hanging forever
+__global__ __launch_bounds__(BLOCK) void wf_apply_dead_kernel(
+    const int* __restrict__ dead_src,         // [dcap] slot or -1-j flags
+    const int* __restrict__ dead_count,       // [1]
+    const int* __restrict__ child_tag,
+    const int* __restrict__ rq_prev_tag,
+    int* __restrict__ children_fail,
+    int* __restrict__ children_out,
+    int dcap)
+{
+    const int i = blockIdx.x * BLOCK + threadIdx.x;
+    if (i >= min(*dead_count, dcap)) return;
+    const int s = dead_src[i];
+    const int tag = s >= 0 ? child_tag[s] : rq_prev_tag[-1 - s];
+    atomicAdd(&children_fail[tag], 1);
+    atomicSub(&children_out[tag], 1);
+}
+
+// step commit: aggregate children (engine.go:1623-1645) + retry/backoff
+// (computeBackoff :1573-1595, attempts capped by max_retries)
+__global__ __launch_bounds__(BLOCK) void wf_commit_kernel(
+    unsigned char* __restrict__ step_state,   // [NR*64]
+    int* __restrict__ step_attempts,          // [NR*64]
+    int* __restrict__ children_todo,
+    int* __restrict__ children_out,
+    int* __restrict__ children_done,          // kept: successes accumulate
+    int* __restrict__ children_fail,
+    int* __restrict__ next_ready,
+    int tick, int max_retries, int NRS)
+{
+    const int i = blockIdx.x * BLOCK + threadIdx.x;
+    if (i >= NRS) return;
+    if (step_state[i] != WFS_DISPATCHED) return;
+    if (children_out[i] != 0) return;
+    const int fail = children_fail[i];
+    if (fail > 0) {
+        if (step_attempts[i] < max_retries) {
+            step_attempts[i] += 1;
+            children_todo[i] += fail;      // only failed children re-run
+            children_fail[i] = 0;
+            next_ready[i] = tick + min(1 << step_attempts[i], 16);
+            step_state[i] = WFS_PENDING;
+        } else {
+            step_state[i] = WFS_FAILED;
+        }
+    } else if (children_todo[i] > 0) {
+        step_state[i] = WFS_PENDING;       // partial emission resumes
+    } else {
+        step_state[i] = WFS_SUCCEEDED;
+    }
+}
+
+// run status roll-up (updateRunStatus :1647-1699): all steps SUCCEEDED or
+// SKIPPED -> run SUCCEEDED; any FAILED step -> run FAILED (retries already
+// exhausted at step level). counts[0] += succeeded, counts[1] += failed.
+__global__ __launch_bounds__(BLOCK) void wf_status_kernel(
+    const unsigned char* __restrict__ step_state,
+    const unsigned char* __restrict__ n_steps,
+    unsigned char* __restrict__ run_active,
+    unsigned char* __restrict__ run_state,    // [NR] WFS_* of the run
+    unsigned long long* __restrict__ counts,  // [2]
+    int NR)
+{
+    const int run = blockIdx.x * BLOCK + threadIdx.x;
+    if (run >= NR || !run_active[run]) return;
+    const int ns = n_steps[run];
+    const unsigned char* st = &step_state[(size_t)run * 64];
+    bool all_ok = true, any_fail = false;
+    for (int s = 0; s < ns; ++s) {
+        const unsigned char v = st[s];
+        any_fail |= (v == WFS_FAILED);
+        all_ok &= (v == WFS_SUCCEEDED || v == WFS_SKIPPED);
+    }
+    if (any_fail) {
+        run_active[run] = 0;
+        run_state[run] = WFS_FAILED;
+        atomicAdd(&counts[1], 1ull);
+    } else if (all_ok) {
+        run_active[run] = 0;
+        run_state[run] = WFS_SUCCEEDED;
+        atomicAdd(&counts[0], 1ull);
+    }
+}
+
+// host-driven approval grants (the admin hop): verdict 1 -> SUCCEEDED,
+// 0 -> FAILED (rejection fails the step, which fails the run)
+__global__ __launch_bounds__(BLOCK) void wf_grant_kernel(
+    const int* __restrict__ grant_runs,
+    const int* __restrict__ grant_steps,
+    const unsigned char* __restrict__ verdicts,
+    int n,
+    unsigned char* __restrict__ step_state)
+{
+    const int i = blockIdx.x * BLOCK + threadIdx.x;
+    if (i >= n) return;
+    const size_t rs = (size_t)grant_runs[i] * 64 + grant_steps[i];
+    if (step_state[rs] == WFS_WAITING)
+        step_state[rs] = verdicts[i] ? WFS_SUCCEEDED : WFS_FAILED;
+}
+
+// ---------------------------------------------------------------------------
 // K4: deadline / staleness scan -> TIMEOUT candidates (compacted list)
 // ---------------------------------------------------------------------------
 __global__ __launch_bounds__(BLOCK) void deadline_scan_kernel(
@@ -971,7 +1262,10 @@ __global__ __launch_bounds__(BLOCK) void pack_by_dest_kernel(
     int* __restrict__ rq_attempts,            // [rq_cap] delivery attempts
     int* __restrict__ rq_count,               // [1]
     unsigned long long* __restrict__ rq_dead, // [1] dropped after max attempts / ring full
-    int rq_cap)
+    int rq_cap,
+    int* __restrict__ dead_src,               // [dead_cap] per-tick dead list (DLQ-bound)
+    int* __restrict__ dead_count,             // [1] pre-zeroed per tick
+    int dead_cap)
 {
     const int i = blockIdx.x * BLOCK + threadIdx.x;
     const int lane = threadIdx.x % WAVE;
@@ -1012,6 +1306,8 @@ __global__ __launch_bounds__(BLOCK) void pack_by_dest_kernel(
             rq_attempts[p] = 1;
         } else {
             atomicAdd(rq_dead, 1ull);  // ring full -> DLQ-bound, counted
+            const int dp = atomicAdd(dead_count, 1);
+            if (dp < dead_cap) dead_src[dp] = slot;
         }
     }
 }
@@ -1036,7 +1332,10 @@ __global__ __launch_bounds__(BLOCK) void pack_requeue_kernel(
     int* __restrict__ rq_attempts,
     int* __restrict__ rq_count,
     unsigned long long* __restrict__ rq_dead,
-    int rq_cap)
+    int rq_cap,
+    int* __restrict__ dead_src,               // [dead_cap]
+    int* __restrict__ dead_count,             // [1] pre-zeroed per tick
+    int dead_cap)
 {
     const int j = blockIdx.x * BLOCK + threadIdx.x;
     const int n = min(*rq_prev_count, rq_cap);
@@ -1050,17 +1349,21 @@ __global__ __launch_bounds__(BLOCK) void pack_requeue_kernel(
         send_widx[(size_t)dest * cap + pos] = widx % nwl;
     } else {
         const int att = rq_prev_attempts[j] + 1;
-        if (att > RQ_MAX_DELIVER) {
-            atomicAdd(rq_dead, 1ull);
-            return;
+        bool drop = att > RQ_MAX_DELIVER;
+        if (!drop) {
+            const int p = atomicAdd(rq_count, 1);
+            if (p < rq_cap) {
+                rq_src[p] = -1 - j;   // payload row in rq_prev_payload
+                rq_widx[p] = widx;
+                rq_attempts[p] = att;
+            } else {
+                drop = true;
+            }
         }
-        const int p = atomicAdd(rq_count, 1);
-        if (p < rq_cap) {
-            rq_src[p] = -1 - j;   // payload row in rq_prev_payload
-            rq_widx[p] = widx;
-            rq_attempts[p] = att;
-        } else {
+        if (drop) {
             atomicAdd(rq_dead, 1ull);
+            const int dp = atomicAdd(dead_count, 1);
+            if (dp < dead_cap) dead_src[dp] = -1 - j;
         }
     }
 }
@@ -1643,7 +1946,8 @@ void pack_by_dest(torch::Tensor routable_slots, torch::Tensor routable_widx,
                   int64_t nwl, int64_t cap, int64_t capacity,
                   torch::Tensor rq_src, torch::Tensor rq_widx,
                   torch::Tensor rq_attempts, torch::Tensor rq_count,
-                  torch::Tensor rq_dead)
+                  torch::Tensor rq_dead,
+                  torch::Tensor dead_src, torch::Tensor dead_count)
 {
     const int blocks = ((int)capacity + BLOCK - 1) / BLOCK;
     hipLaunchKernelGGL(pack_by_dest_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
@@ -1653,7 +1957,9 @@ void pack_by_dest(torch::Tensor routable_slots, torch::Tensor routable_widx,
         rq_src.data_ptr<int>(), rq_widx.data_ptr<int>(),
         rq_attempts.data_ptr<int>(), rq_count.data_ptr<int>(),
         (unsigned long long*)rq_dead.data_ptr<int64_t>(),
-        (int)rq_src.size(0));
+        (int)rq_src.size(0),
+        dead_src.data_ptr<int>(), dead_count.data_ptr<int>(),
+        (int)dead_src.size(0));
 }
 
 void pack_requeue(torch::Tensor rq_prev_widx, torch::Tensor rq_prev_attempts,
@@ -1662,7 +1968,8 @@ void pack_requeue(torch::Tensor rq_prev_widx, torch::Tensor rq_prev_attempts,
                   int64_t nwl, int64_t cap,
                   torch::Tensor rq_src, torch::Tensor rq_widx,
                   torch::Tensor rq_attempts, torch::Tensor rq_count,
-                  torch::Tensor rq_dead)
+                  torch::Tensor rq_dead,
+                  torch::Tensor dead_src, torch::Tensor dead_count)
 {
     const int rq_cap = (int)rq_src.size(0);
     const int blocks = (rq_cap + BLOCK - 1) / BLOCK;
@@ -1672,7 +1979,108 @@ void pack_requeue(torch::Tensor rq_prev_widx, torch::Tensor rq_prev_attempts,
         send_widx.data_ptr<int>(), send_cnt.data_ptr<int>(), (int)nwl, (int)cap,
         rq_src.data_ptr<int>(), rq_widx.data_ptr<int>(),
         rq_attempts.data_ptr<int>(), rq_count.data_ptr<int>(),
-        (unsigned long long*)rq_dead.data_ptr<int64_t>(), rq_cap);
+        (unsigned long long*)rq_dead.data_ptr<int64_t>(), rq_cap,
+        dead_src.data_ptr<int>(), dead_count.data_ptr<int>(),
+        (int)dead_src.size(0));
+}
+
+void wf_sweep(torch::Tensor step_state, torch::Tensor deps_mask, torch::Tensor n_steps,
+              torch::Tensor run_active, torch::Tensor step_kind, torch::Tensor cond_bits,
+              torch::Tensor next_ready, int64_t tick,
+              torch::Tensor disp_runs, torch::Tensor disp_steps, torch::Tensor disp_count,
+              torch::Tensor appr_runs, torch::Tensor appr_steps, torch::Tensor appr_count)
+{
+    const int NR = (int)n_steps.size(0);
+    const int blocks = (NR + BLOCK - 1) / BLOCK;
+    hipLaunchKernelGGL(wf_sweep_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        step_state.data_ptr<uint8_t>(), (const long long*)deps_mask.data_ptr<int64_t>(),
+        n_steps.data_ptr<uint8_t>(), run_active.data_ptr<uint8_t>(),
+        step_kind.data_ptr<uint8_t>(), (const long long*)cond_bits.data_ptr<int64_t>(),
+        next_ready.data_ptr<int>(), (int)tick,
+        disp_runs.data_ptr<int>(), disp_steps.data_ptr<int>(), disp_count.data_ptr<int>(),
+        appr_runs.data_ptr<int>(), appr_steps.data_ptr<int>(), appr_count.data_ptr<int>(),
+        NR, (int)disp_runs.size(0), (int)appr_runs.size(0));
+}
+
+void wf_expand(torch::Tensor disp_runs, torch::Tensor disp_steps, torch::Tensor disp_count,
+               torch::Tensor step_state, torch::Tensor children_todo, torch::Tensor children_out,
+               torch::Tensor child_tag, torch::Tensor child_seq, torch::Tensor child_widx,
+               torch::Tensor child_count, torch::Tensor children_emitted,
+               torch::Tensor order, torch::Tensor valid_count)
+{
+    const int cap = (int)disp_runs.size(0);
+    const int blocks = (cap + (BLOCK / WAVE) - 1) / (BLOCK / WAVE);
+    hipLaunchKernelGGL(wf_expand_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        disp_runs.data_ptr<int>(), disp_steps.data_ptr<int>(), disp_count.data_ptr<int>(),
+        step_state.data_ptr<uint8_t>(), children_todo.data_ptr<int>(),
+        children_out.data_ptr<int>(), child_tag.data_ptr<int>(),
+        child_seq.data_ptr<int>(), child_widx.data_ptr<int>(), child_count.data_ptr<int>(),
+        children_emitted.data_ptr<int>(),
+        order.data_ptr<int>(), valid_count.data_ptr<int>(),
+        cap, (int)child_tag.size(0));
+}
+
+void wf_apply(torch::Tensor send_slots, torch::Tensor send_cnt, torch::Tensor child_tag,
+              torch::Tensor child_seq, torch::Tensor rq_prev_tag, torch::Tensor rq_prev_seq,
+              torch::Tensor children_done,
+              torch::Tensor children_fail, torch::Tensor children_out,
+              int64_t fail_ppt, int64_t cap, int64_t world)
+{
+    const int n = (int)(world * cap);
+    const int blocks = (n + BLOCK - 1) / BLOCK;
+    hipLaunchKernelGGL(wf_apply_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        send_slots.data_ptr<int>(), send_cnt.data_ptr<int>(), child_tag.data_ptr<int>(),
+        child_seq.data_ptr<int>(), rq_prev_tag.data_ptr<int>(), rq_prev_seq.data_ptr<int>(),
+        children_done.data_ptr<int>(),
+        children_fail.data_ptr<int>(), children_out.data_ptr<int>(),
+        (int)fail_ppt, (int)cap, (int)world);
+}
+
+void wf_apply_dead(torch::Tensor dead_src, torch::Tensor dead_count, torch::Tensor child_tag,
+                   torch::Tensor rq_prev_tag, torch::Tensor children_fail,
+                   torch::Tensor children_out)
+{
+    const int dcap = (int)dead_src.size(0);
+    const int blocks = (dcap + BLOCK - 1) / BLOCK;
+    hipLaunchKernelGGL(wf_apply_dead_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        dead_src.data_ptr<int>(), dead_count.data_ptr<int>(), child_tag.data_ptr<int>(),
+        rq_prev_tag.data_ptr<int>(), children_fail.data_ptr<int>(),
+        children_out.data_ptr<int>(), dcap);
+}
+
+void wf_commit(torch::Tensor step_state, torch::Tensor step_attempts,
+               torch::Tensor children_todo, torch::Tensor children_out,
+               torch::Tensor children_done, torch::Tensor children_fail,
+               torch::Tensor next_ready, int64_t tick, int64_t max_retries)
+{
+    const int NRS = (int)step_state.numel();
+    const int blocks = (NRS + BLOCK - 1) / BLOCK;
+    hipLaunchKernelGGL(wf_commit_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        step_state.data_ptr<uint8_t>(), step_attempts.data_ptr<int>(),
+        children_todo.data_ptr<int>(), children_out.data_ptr<int>(),
+        children_done.data_ptr<int>(), children_fail.data_ptr<int>(),
+        next_ready.data_ptr<int>(), (int)tick, (int)max_retries, NRS);
+}
+
+void wf_status(torch::Tensor step_state, torch::Tensor n_steps, torch::Tensor run_active,
+               torch::Tensor run_state, torch::Tensor counts)
+{
+    const int NR = (int)n_steps.size(0);
+    const int blocks = (NR + BLOCK - 1) / BLOCK;
+    hipLaunchKernelGGL(wf_status_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        step_state.data_ptr<uint8_t>(), n_steps.data_ptr<uint8_t>(),
+        run_active.data_ptr<uint8_t>(), run_state.data_ptr<uint8_t>(),
+        (unsigned long long*)counts.data_ptr<int64_t>(), NR);
+}
+
+void wf_grant(torch::Tensor grant_runs, torch::Tensor grant_steps, torch::Tensor verdicts,
+              int64_t n, torch::Tensor step_state)
+{
+    if (n <= 0) return;
+    const int blocks = ((int)n + BLOCK - 1) / BLOCK;
+    hipLaunchKernelGGL(wf_grant_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        grant_runs.data_ptr<int>(), grant_steps.data_ptr<int>(),
+        verdicts.data_ptr<uint8_t>(), (int)n, step_state.data_ptr<uint8_t>());
 }
 
 void materialize_rq_payload(torch::Tensor payload, torch::Tensor rq_prev_payload,
@@ -1769,6 +2177,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("run_readiness", &run_readiness, "K3 run/step readiness sweep");
     m.def("pack_by_dest", &pack_by_dest, "padded per-destination dispatch pack");
     m.def("pack_requeue", &pack_requeue, "redeliver last tick's requeue ring into the send segments");
+    m.def("wf_sweep", &wf_sweep, "K3-WF readiness sweep + approval holds + condition/delay commits");
+    m.def("wf_expand", &wf_expand, "K3-WF for_each/worker child expansion into the child arena");
+    m.def("wf_apply", &wf_apply, "K3-WF child result application with failure injection");
+    m.def("wf_apply_dead", &wf_apply_dead, "K3-WF dead-letter child application");
+    m.def("wf_commit", &wf_commit, "K3-WF step aggregation + retry/backoff commit");
+    m.def("wf_status", &wf_status, "K3-WF run status roll-up");
+    m.def("wf_grant", &wf_grant, "K3-WF host approval grants");
     m.def("materialize_rq_payload", &materialize_rq_payload, "copy requeued payload rows into the rq arena");
     m.def("gather_payload_padded", &gather_payload_padded, "payload gather into padded send arena");
     m.def("echo_padded", &echo_padded, "region-valid echo over padded recv arena");

@@ -223,18 +223,60 @@ class _RefOps:
         apply_transitions_ref(states, attempts, deadlines, slots[:n], to)
 
     def pack_by_dest(self, rs, rw, rc, ss, sw, sc, nwl, cap, capacity,
-                     rq_src, rq_widx, rq_attempts, rq_count, rq_dead):
+                     rq_src, rq_widx, rq_attempts, rq_count, rq_dead,
+                     dead_src, dead_count):
         from .reference import pack_by_dest_ref
 
         pack_by_dest_ref(rs, rw, rc, ss, sw, sc, int(nwl), int(cap),
-                         rq_src, rq_widx, rq_attempts, rq_count, rq_dead)
+                         rq_src, rq_widx, rq_attempts, rq_count, rq_dead,
+                         dead_src, dead_count)
 
     def pack_requeue(self, pw, pa, pc, ss, sw, sc, nwl, cap,
-                     rq_src, rq_widx, rq_attempts, rq_count, rq_dead):
+                     rq_src, rq_widx, rq_attempts, rq_count, rq_dead,
+                     dead_src, dead_count):
         from .reference import pack_requeue_ref
 
         pack_requeue_ref(pw, pa, pc, ss, sw, sc, int(nwl), int(cap),
-                         rq_src, rq_widx, rq_attempts, rq_count, rq_dead)
+                         rq_src, rq_widx, rq_attempts, rq_count, rq_dead,
+                         dead_src, dead_count)
+
+    def wf_sweep(self, *a):
+        from .reference import wf_sweep_ref
+
+        args = list(a)
+        args[7] = int(args[7])
+        wf_sweep_ref(*args)
+
+    def wf_expand(self, *a):
+        from .reference import wf_expand_ref
+
+        wf_expand_ref(*a)
+
+    def wf_apply(self, ss, sc, ct, cs, rpt, rps, cd, cf, co, fail_ppt, cap, world):
+        from .reference import wf_apply_ref
+
+        wf_apply_ref(ss, sc, ct, cs, rpt, rps, cd, cf, co, int(fail_ppt),
+                     int(cap), int(world))
+
+    def wf_apply_dead(self, *a):
+        from .reference import wf_apply_dead_ref
+
+        wf_apply_dead_ref(*a)
+
+    def wf_commit(self, st, sa, ctd, co, cd, cf, nr, tick, max_retries):
+        from .reference import wf_commit_ref
+
+        wf_commit_ref(st, sa, ctd, co, cd, cf, nr, int(tick), int(max_retries))
+
+    def wf_status(self, *a):
+        from .reference import wf_status_ref
+
+        wf_status_ref(*a)
+
+    def wf_grant(self, gr, gs, v, n, st):
+        from .reference import wf_grant_ref
+
+        wf_grant_ref(gr, gs, v, int(n), st)
 
     def materialize_rq_payload(self, payload, prev_payload, rq_src, rq_count,
                                rq_payload, stride):
@@ -862,6 +904,10 @@ class DevicePipeline:
         self.rq_prev_count = zi(1)
         self.rq_prev_payload = torch.zeros(B * W, dtype=torch.int32, device=dev)
         self.rq_dead = torch.zeros(1, dtype=torch.int64, device=dev)
+        # per-tick dead list (consumers that track per-entry identity, e.g.
+        # the workflow tick, resolve these; the job bench only counts them)
+        self.dead_src = zi(64)
+        self.dead_count = zi(1)
         if not hasattr(self, "_pend_states"):
             self._pend_states = torch.full((B,), PENDING, dtype=torch.uint8, device=dev)
 
@@ -908,17 +954,20 @@ class DevicePipeline:
         self.rq_prev_count.copy_(self.rq_count)
         self.rq_prev_payload.copy_(self.rq_payload)
         self.rq_count.zero_()
+        self.dead_count.zero_()
         self.pad_send_cnt.zero_()
         ext.pack_requeue(self.rq_prev_widx, self.rq_prev_attempts, self.rq_prev_count,
                          self.pad_send_slots, self.pad_send_widx, self.pad_send_cnt,
                          self.NWL, cap,
                          self.rq_src, self.rq_widx, self.rq_attempts,
-                         self.rq_count, self.rq_dead)
+                         self.rq_count, self.rq_dead,
+                         self.dead_src, self.dead_count)
         ext.pack_by_dest(self.routable_slots, self.routable_widx, self.routable_count,
                          self.pad_send_slots, self.pad_send_widx, self.pad_send_cnt,
                          self.NWL, cap, B,
                          self.rq_src, self.rq_widx, self.rq_attempts,
-                         self.rq_count, self.rq_dead)
+                         self.rq_count, self.rq_dead,
+                         self.dead_src, self.dead_count)
         # clamp: entries beyond capacity were parked in the requeue ring;
         # dispatched = sum of clamped counts (fresh + redelivered this tick)
         self.pad_send_cnt.clamp_(max=cap)

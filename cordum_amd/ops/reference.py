@@ -197,8 +197,18 @@ def run_readiness_ref(step_state, deps_mask, n_steps, run_active):
 RQ_MAX_DELIVER = 5
 
 
+def _rq_dead_mark(rq_dead, dead_src, dead_count, src: int) -> None:
+    rq_dead[0] += 1
+    if dead_src is not None:
+        dp = int(dead_count[0])
+        dead_count[0] = dp + 1
+        if dp < dead_src.shape[0]:
+            dead_src[dp] = src
+
+
 def _rq_park(rq_src, rq_widx, rq_attempts, rq_count, rq_dead,
-             src: int, widx: int, attempts: int) -> None:
+             src: int, widx: int, attempts: int,
+             dead_src=None, dead_count=None) -> None:
     p = int(rq_count[0])
     rq_count[0] = p + 1
     if p < rq_src.shape[0]:
@@ -206,12 +216,13 @@ def _rq_park(rq_src, rq_widx, rq_attempts, rq_count, rq_dead,
         rq_widx[p] = widx
         rq_attempts[p] = attempts
     else:
-        rq_dead[0] += 1
+        _rq_dead_mark(rq_dead, dead_src, dead_count, src)
 
 
 def pack_by_dest_ref(routable_slots, routable_widx, routable_count,
                      send_slots, send_widx, send_cnt, nwl: int, cap: int,
-                     rq_src, rq_widx, rq_attempts, rq_count, rq_dead):
+                     rq_src, rq_widx, rq_attempts, rq_count, rq_dead,
+                     dead_src=None, dead_count=None):
     n = int(routable_count[0])
     for i in range(n):
         slot = int(routable_slots[i])
@@ -224,12 +235,13 @@ def pack_by_dest_ref(routable_slots, routable_widx, routable_count,
             send_widx[dest * cap + pos] = widx % nwl
         else:
             _rq_park(rq_src, rq_widx, rq_attempts, rq_count, rq_dead,
-                     slot, widx, 1)
+                     slot, widx, 1, dead_src, dead_count)
 
 
 def pack_requeue_ref(rq_prev_widx, rq_prev_attempts, rq_prev_count,
                      send_slots, send_widx, send_cnt, nwl: int, cap: int,
-                     rq_src, rq_widx, rq_attempts, rq_count, rq_dead):
+                     rq_src, rq_widx, rq_attempts, rq_count, rq_dead,
+                     dead_src=None, dead_count=None):
     n = min(int(rq_prev_count[0]), int(rq_prev_widx.shape[0]))
     for j in range(n):
         widx = int(rq_prev_widx[j])
@@ -242,10 +254,10 @@ def pack_requeue_ref(rq_prev_widx, rq_prev_attempts, rq_prev_count,
         else:
             att = int(rq_prev_attempts[j]) + 1
             if att > RQ_MAX_DELIVER:
-                rq_dead[0] += 1
+                _rq_dead_mark(rq_dead, dead_src, dead_count, -1 - j)
             else:
                 _rq_park(rq_src, rq_widx, rq_attempts, rq_count, rq_dead,
-                         -1 - j, widx, att)
+                         -1 - j, widx, att, dead_src, dead_count)
 
 
 def materialize_rq_payload_ref(payload, rq_prev_payload, rq_src, rq_count,
@@ -307,3 +319,170 @@ def load_feedback_padded_ref(recv_widx, recv_cnt, w_active_local, cap: int, worl
     for r in range(world):
         for e in range(int(recv_cnt[r])):
             w_active_local[int(recv_widx[r * cap + e])] += 1
+
+
+# -- K3-WF device workflow engine references ----------------------------------
+# States/kinds mirror the kernel constants (cordum_kernels.hip K3-WF section).
+
+WFS_PENDING, WFS_DISPATCHED, WFS_WAITING = 0, 1, 2
+WFS_SUCCEEDED, WFS_FAILED, WFS_SKIPPED = 3, 4, 5
+WFK_WORKER, WFK_FOR_EACH, WFK_APPROVAL, WFK_CONDITION, WFK_DELAY = 0, 1, 2, 3, 4
+
+
+def wf_sweep_ref(step_state, deps_mask, n_steps, run_active, step_kind,
+                 cond_bits, next_ready, tick: int,
+                 disp_runs, disp_steps, disp_count,
+                 appr_runs, appr_steps, appr_count):
+    NR = int(n_steps.shape[0])
+    cap, acap = int(disp_runs.shape[0]), int(appr_runs.shape[0])
+    for run in range(NR):
+        if not int(run_active[run]):
+            continue
+        ns = int(n_steps[run])
+        base = run * 64
+        satisfied = 0
+        for t in range(ns):
+            if int(step_state[base + t]) in (WFS_SUCCEEDED, WFS_SKIPPED):
+                satisfied |= 1 << t
+        for s in range(ns):
+            if int(step_state[base + s]) != WFS_PENDING:
+                continue
+            need = int(deps_mask[base + s]) & ((1 << 64) - 1)
+            if need & ~satisfied & ((1 << 64) - 1):
+                continue
+            if int(next_ready[base + s]) > tick:
+                continue
+            kind = int(step_kind[base + s])
+            if kind in (WFK_WORKER, WFK_FOR_EACH):
+                pos = int(disp_count[0])
+                disp_count[0] = pos + 1
+                if pos < cap:
+                    disp_runs[pos] = run
+                    disp_steps[pos] = s
+            elif kind == WFK_APPROVAL:
+                step_state[base + s] = WFS_WAITING
+                pos = int(appr_count[0])
+                appr_count[0] = pos + 1
+                if pos < acap:
+                    appr_runs[pos] = run
+                    appr_steps[pos] = s
+            elif kind == WFK_CONDITION:
+                step_state[base + s] = (
+                    WFS_SUCCEEDED if (int(cond_bits[run]) >> s) & 1 else WFS_SKIPPED)
+            elif kind == WFK_DELAY:
+                step_state[base + s] = WFS_SUCCEEDED
+
+
+def wf_expand_ref(disp_runs, disp_steps, disp_count, step_state,
+                  children_todo, children_out, child_tag, child_seq,
+                  child_widx, child_count, children_emitted,
+                  order, valid_count):
+    CB = int(child_tag.shape[0])
+    n = min(int(disp_count[0]), int(disp_runs.shape[0]))
+    V = min(max(1, int(valid_count[0])), 1024)
+    for e in range(n):
+        run, step = int(disp_runs[e]), int(disp_steps[e])
+        rs = run * 64 + step
+        todo = int(children_todo[rs])
+        # monotone per-step emission counter -> deterministic per-child
+        # ordinal stream (matches the kernel; retries get fresh ordinals)
+        seq0 = int(children_emitted[rs])
+        base = int(child_count[0])
+        if base + todo > CB:
+            continue  # arena full: retry next sweep
+        child_count[0] = base + todo
+        step_state[rs] = WFS_DISPATCHED
+        children_out[rs] += todo
+        children_emitted[rs] += todo
+        children_todo[rs] = 0
+        tag = run * 64 + step
+        for k in range(todo):
+            child_tag[base + k] = tag
+            child_seq[base + k] = seq0 + k
+            child_widx[base + k] = int(order[(base + k) % V])
+
+
+def _wf_mix(x: int) -> int:
+    x &= 0xFFFFFFFF
+    x ^= x >> 16
+    x = (x * 0x7FEB352D) & 0xFFFFFFFF
+    x ^= x >> 15
+    x = (x * 0x846CA68B) & 0xFFFFFFFF
+    x ^= x >> 16
+    return x
+
+
+def wf_fail_hash(tag: int, seq: int) -> int:
+    return _wf_mix((tag * 2654435761 ^ seq * 40503) & 0xFFFFFFFF)
+
+
+def wf_apply_ref(send_slots, send_cnt, child_tag, child_seq, rq_prev_tag,
+                 rq_prev_seq, children_done, children_fail, children_out,
+                 fail_ppt: int, cap: int, world: int):
+    for r in range(world):
+        for e in range(min(int(send_cnt[r]), cap)):
+            i = r * cap + e
+            s = int(send_slots[i])
+            tag = int(child_tag[s]) if s >= 0 else int(rq_prev_tag[-1 - s])
+            seq = int(child_seq[s]) if s >= 0 else int(rq_prev_seq[-1 - s])
+            if wf_fail_hash(tag, seq) % 1000 < fail_ppt:
+                children_fail[tag] += 1
+            else:
+                children_done[tag] += 1
+            children_out[tag] -= 1
+
+
+def wf_apply_dead_ref(dead_src, dead_count, child_tag, rq_prev_tag,
+                      children_fail, children_out):
+    n = min(int(dead_count[0]), int(dead_src.shape[0]))
+    for i in range(n):
+        s = int(dead_src[i])
+        tag = int(child_tag[s]) if s >= 0 else int(rq_prev_tag[-1 - s])
+        children_fail[tag] += 1
+        children_out[tag] -= 1
+
+
+def wf_commit_ref(step_state, step_attempts, children_todo, children_out,
+                  children_done, children_fail, next_ready,
+                  tick: int, max_retries: int):
+    for i in range(int(step_state.numel())):
+        if int(step_state[i]) != WFS_DISPATCHED or int(children_out[i]) != 0:
+            continue
+        fail = int(children_fail[i])
+        if fail > 0:
+            if int(step_attempts[i]) < max_retries:
+                step_attempts[i] += 1
+                children_todo[i] += fail
+                children_fail[i] = 0
+                next_ready[i] = tick + min(1 << int(step_attempts[i]), 16)
+                step_state[i] = WFS_PENDING
+            else:
+                step_state[i] = WFS_FAILED
+        elif int(children_todo[i]) > 0:
+            step_state[i] = WFS_PENDING
+        else:
+            step_state[i] = WFS_SUCCEEDED
+
+
+def wf_status_ref(step_state, n_steps, run_active, run_state, counts):
+    NR = int(n_steps.shape[0])
+    for run in range(NR):
+        if not int(run_active[run]):
+            continue
+        ns = int(n_steps[run])
+        vals = [int(step_state[run * 64 + s]) for s in range(ns)]
+        if any(v == WFS_FAILED for v in vals):
+            run_active[run] = 0
+            run_state[run] = WFS_FAILED
+            counts[1] += 1
+        elif all(v in (WFS_SUCCEEDED, WFS_SKIPPED) for v in vals):
+            run_active[run] = 0
+            run_state[run] = WFS_SUCCEEDED
+            counts[0] += 1
+
+
+def wf_grant_ref(grant_runs, grant_steps, verdicts, n: int, step_state):
+    for i in range(n):
+        rs = int(grant_runs[i]) * 64 + int(grant_steps[i])
+        if int(step_state[rs]) == WFS_WAITING:
+            step_state[rs] = WFS_SUCCEEDED if int(verdicts[i]) else WFS_FAILED

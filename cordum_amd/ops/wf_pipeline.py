@@ -1,0 +1,460 @@
+"""Device-resident workflow engine tick (config #3: 1->256 fan-out +
+approval gate over 8 pools, RCCL all-to-all over xGMI).
+
+This is the batched HBM counterpart of the host workflow engine
+(cordum_amd/workflow/engine.py; semantics oracle core/workflow/engine.go
+:453-827 scheduleReady, :1524-1560 applyResult, :1623-1645
+aggregateChildren, :1647-1699 updateRunStatus). Run/step state lives in
+per-rank HBM tables; each tick is a fixed kernel sequence:
+
+  wf_sweep    readiness gates (deps, backoff, delay), approval holds,
+              condition commits          -> dispatch list of (run, step)
+  wf_expand   for_each/worker expansion  -> child-job arena (all-or-nothing
+              reservation; arena-full steps retry next sweep)
+  K2 routing  children spread over the least-loaded global worker order,
+              packed per destination rank (pack_requeue + pack_by_dest with
+              the NAK requeue ring), payload gathered from the child arena
+  exchange    all_to_all_single over RCCL/xGMI (world > 1)
+  echo        device worker pools execute on the receiving rank
+  wf_apply    owner-rank result application with deterministic failure
+              injection (retry waves), dead-letter application for
+              max-deliver drops
+  wf_commit   child aggregation, retry + exponential tick backoff, FAILED
+              after max_retries
+  wf_status   run roll-up -> SUCCEEDED/FAILED counts
+  host hop    fresh approval holds drained D2H; the approver (the admin
+              analog, gateway.go:3553) grants them next tick H2D
+
+Step kinds: WORKER (one child job), FOR_EACH (fanout children, per-child
+retry), APPROVAL (host-granted), CONDITION (pre-evaluated bit ->
+SUCCEEDED/SKIPPED), DELAY (tick gate). Dependencies are 64-bit step masks;
+SKIPPED satisfies a dependency, FAILED permanently blocks (the run fails).
+"""
+from __future__ import annotations
+
+import time
+from dataclasses import dataclass, field
+from typing import List, Optional, Sequence
+
+import torch
+import torch.distributed as dist
+
+from . import get_ext
+from .pipeline import _RefOps
+
+WFS_PENDING, WFS_DISPATCHED, WFS_WAITING = 0, 1, 2
+WFS_SUCCEEDED, WFS_FAILED, WFS_SKIPPED = 3, 4, 5
+WFK_WORKER, WFK_FOR_EACH, WFK_APPROVAL, WFK_CONDITION, WFK_DELAY = 0, 1, 2, 3, 4
+
+
+@dataclass
+class StepSpec:
+    kind: int
+    deps: Sequence[int] = ()
+    fanout: int = 1          # children for FOR_EACH (WORKER always 1)
+    delay_ticks: int = 0     # DELAY gate
+    cond: bool = True        # CONDITION value (pre-evaluated)
+
+
+@dataclass
+class DagSpec:
+    """One workflow DAG (<= 64 steps)."""
+    steps: List[StepSpec] = field(default_factory=list)
+
+    @classmethod
+    def fanout_approval(cls, fanout: int = 256) -> "DagSpec":
+        """Config #3 shape: seed -> 1->N fan-out -> approval gate -> final."""
+        return cls(steps=[
+            StepSpec(WFK_WORKER),
+            StepSpec(WFK_FOR_EACH, deps=[0], fanout=fanout),
+            StepSpec(WFK_APPROVAL, deps=[1]),
+            StepSpec(WFK_WORKER, deps=[2]),
+        ])
+
+
+@dataclass
+class WfStats:
+    runs_succeeded: int = 0
+    runs_failed: int = 0
+    ticks: int = 0
+    wall_s: float = 0.0
+    children_dispatched: int = 0
+
+
+class WorkflowPipeline:
+    """Per-rank device workflow engine over a set of runs (same or mixed
+    DAGs). World > 1 exchanges children over the padded all-to-all."""
+
+    def __init__(
+        self,
+        device: torch.device,
+        dags: Sequence[DagSpec],
+        n_local_workers: int = 256,
+        payload_words: int = 64,
+        world_size: int = 1,
+        rank: int = 0,
+        backend: str = "ext",
+        fail_ppt: int = 0,
+        max_retries: int = 2,
+        approval_verdict: int = 1,
+        child_cap: Optional[int] = None,
+        pad_cap: Optional[int] = None,
+        seed: int = 7,
+    ):
+        device = torch.device(device)
+        self.ext = _RefOps() if backend == "ref" else get_ext(required=True)
+        self.device = device
+        self.world = world_size
+        self.rank = rank
+        self.NWL = n_local_workers
+        self.W = payload_words
+        self.fail_ppt = fail_ppt
+        self.max_retries = max_retries
+        self.approval_verdict = approval_verdict
+        NR = len(dags)
+        self.NR = NR
+        assert all(len(d.steps) <= 64 for d in dags), "device DAGs cap at 64 steps"
+
+        # ---- host-side template tables (the run-creation image) -------------
+        st = torch.zeros(NR * 64, dtype=torch.uint8)
+        deps = torch.zeros(NR * 64, dtype=torch.int64)
+        kinds = torch.zeros(NR * 64, dtype=torch.uint8)
+        todo = torch.zeros(NR * 64, dtype=torch.int32)
+        nready = torch.zeros(NR * 64, dtype=torch.int32)
+        nsteps = torch.zeros(NR, dtype=torch.uint8)
+        cond = torch.zeros(NR, dtype=torch.int64)
+        total_children = 0
+        for r, dag in enumerate(dags):
+            nsteps[r] = len(dag.steps)
+            cbits = 0
+            for s, spec in enumerate(dag.steps):
+                i = r * 64 + s
+                kinds[i] = spec.kind
+                m = 0
+                for d in spec.deps:
+                    m |= 1 << d
+                deps[i] = m - (1 << 64) if m >= (1 << 63) else m
+                if spec.kind == WFK_FOR_EACH:
+                    todo[i] = spec.fanout
+                    total_children += spec.fanout
+                elif spec.kind == WFK_WORKER:
+                    todo[i] = 1
+                    total_children += 1
+                if spec.kind == WFK_DELAY:
+                    nready[i] = spec.delay_ticks
+                if spec.cond:
+                    cbits |= 1 << s
+            cond[r] = cbits - (1 << 64) if cbits >= (1 << 63) else cbits
+        self.total_children = total_children
+        self._tmpl = {
+            "step_state": st, "deps_mask": deps, "step_kind": kinds,
+            "children_todo": todo, "next_ready": nready, "n_steps": nsteps,
+            "cond_bits": cond,
+        }
+
+        d = device
+        self.step_state = st.to(d)
+        self.deps_mask = deps.to(d)
+        self.step_kind = kinds.to(d)
+        self.children_todo = todo.to(d)
+        self.next_ready = nready.to(d)
+        self.n_steps = nsteps.to(d)
+        self.cond_bits = cond.to(d)
+        self.run_active = torch.ones(NR, dtype=torch.uint8, device=d)
+        self.run_state = torch.zeros(NR, dtype=torch.uint8, device=d)
+        self.step_attempts = torch.zeros(NR * 64, dtype=torch.int32, device=d)
+        self.children_out = torch.zeros(NR * 64, dtype=torch.int32, device=d)
+        self.children_emitted = torch.zeros(NR * 64, dtype=torch.int32, device=d)
+        self.children_done = torch.zeros(NR * 64, dtype=torch.int32, device=d)
+        self.children_fail = torch.zeros(NR * 64, dtype=torch.int32, device=d)
+        self.wf_counts = torch.zeros(2, dtype=torch.int64, device=d)
+
+        # dispatch + approval lists
+        self.disp_cap = NR * 64
+        self.disp_runs = torch.zeros(self.disp_cap, dtype=torch.int32, device=d)
+        self.disp_steps = torch.zeros(self.disp_cap, dtype=torch.int32, device=d)
+        self.disp_count = torch.zeros(1, dtype=torch.int32, device=d)
+        self.appr_runs = torch.zeros(NR, dtype=torch.int32, device=d)
+        self.appr_steps = torch.zeros(NR, dtype=torch.int32, device=d)
+        self.appr_count = torch.zeros(1, dtype=torch.int32, device=d)
+
+        # ---- child-job arena -------------------------------------------------
+        CB = child_cap or max(1024, min(total_children, 1 << 18))
+        self.CB = CB
+        self.child_tag = torch.zeros(CB, dtype=torch.int32, device=d)
+        self.child_seq = torch.zeros(CB, dtype=torch.int32, device=d)
+        self.child_widx = torch.zeros(CB, dtype=torch.int32, device=d)
+        self.child_count = torch.zeros(1, dtype=torch.int32, device=d)
+        g = torch.Generator().manual_seed(seed * 17 + rank)
+        self.child_payload = torch.randint(
+            -(1 << 31), (1 << 31) - 1, (CB * payload_words,),
+            dtype=torch.int32, generator=g).to(d)
+
+        # ---- worker table (global view over all ranks' pools) ---------------
+        NWG = self.NWL * world_size
+        self.w_pool = (torch.arange(NWG, dtype=torch.int32) // self.NWL).to(d)
+        maxp = max(64, (4 * CB) // max(1, self.NWL))
+        self.w_maxp = torch.full((NWG,), maxp, dtype=torch.int32, device=d)
+        self.w_labels = torch.zeros(NWG, dtype=torch.int64, device=d)
+        self.w_active_local = torch.zeros(self.NWL, dtype=torch.int32, device=d)
+        self.w_active = torch.zeros(NWG, dtype=torch.int32, device=d)
+        self.w_cpu_local = (torch.rand(self.NWL, generator=g) * 50).to(d)
+        self.w_gpu_local = (torch.rand(self.NWL, generator=g) * 50).to(d)
+        self.w_cpu = torch.zeros(NWG, dtype=torch.float32, device=d)
+        self.w_gpu = torch.zeros(NWG, dtype=torch.float32, device=d)
+        self.w_keys = torch.zeros(NWG, dtype=torch.int64, device=d)
+        self.order_buf = torch.arange(NWG, dtype=torch.int32, device=d)
+        self.valid_buf = torch.tensor([NWG], dtype=torch.int32, device=d)
+
+        # ---- padded exchange arenas (child dispatch across ranks) -----------
+        self.pad_cap = pad_cap or min(CB, max(64, (4 * CB) // max(1, world_size)))
+        cap, Wd, world = self.pad_cap, payload_words, world_size
+
+        def zi(n):
+            return torch.zeros(n, dtype=torch.int32, device=d)
+
+        self.pad_send_slots = zi(world * cap)
+        self.pad_send_widx = zi(world * cap)
+        self.pad_send_cnt = zi(world)
+        self.pad_recv_cnt = zi(world)
+        self.pad_recv_widx = zi(world * cap)
+        self.pad_send_payload = torch.zeros(world * cap * Wd, dtype=torch.int32, device=d)
+        self.pad_recv_payload = torch.zeros_like(self.pad_send_payload)
+        self.pad_res = torch.zeros_like(self.pad_send_payload)
+        self.pad_sums = zi(world * cap)
+        self.pad_sums_back = zi(world * cap)
+        # requeue ring (children parked on destination overflow) + tag carry
+        RQ = CB
+        self.rq_src = zi(RQ)
+        self.rq_widx = zi(RQ)
+        self.rq_attempts = zi(RQ)
+        self.rq_count = zi(1)
+        self.rq_dead = torch.zeros(1, dtype=torch.int64, device=d)
+        self.rq_payload = torch.zeros(RQ * Wd, dtype=torch.int32, device=d)
+        self.rq_tag = zi(RQ)
+        self.rq_seq = zi(RQ)
+        self.rq_prev_widx = zi(RQ)
+        self.rq_prev_attempts = zi(RQ)
+        self.rq_prev_count = zi(1)
+        self.rq_prev_payload = torch.zeros_like(self.rq_payload)
+        self.rq_prev_tag = zi(RQ)
+        self.rq_prev_seq = zi(RQ)
+        self.dead_src = zi(RQ)
+        self.dead_count = zi(1)
+
+        # host-side pinned staging for the approval hop
+        pin = d.type == "cuda"
+        self._appr_host_runs = torch.zeros(NR, dtype=torch.int32)
+        self._appr_host_steps = torch.zeros(NR, dtype=torch.int32)
+        if pin:
+            self._appr_host_runs = self._appr_host_runs.pin_memory()
+            self._appr_host_steps = self._appr_host_steps.pin_memory()
+        self._pending_grants: List[tuple] = []
+
+        self._tick = 0
+        self.children_dispatched_total = 0
+
+    # ---- run admission -------------------------------------------------------
+    def reset_runs(self) -> None:
+        """Re-admit the whole run wave: copy the creation image back into the
+        device tables (this is the honest run-creation H2D cost when the
+        template lives on the host; templates are staged once)."""
+        if not hasattr(self, "_tmpl_dev"):
+            self._tmpl_dev = {k: v.to(self.device) for k, v in self._tmpl.items()}
+        t = self._tmpl_dev
+        self.step_state.copy_(t["step_state"])
+        self.children_todo.copy_(t["children_todo"])
+        self.next_ready.copy_(t["next_ready"])
+        self.run_active.fill_(1)
+        self.run_state.zero_()
+        self.step_attempts.zero_()
+        self.children_out.zero_()
+        self.children_emitted.zero_()
+        self.children_done.zero_()
+        self.children_fail.zero_()
+        self.rq_count.zero_()
+        self.rq_prev_count.zero_()
+        self.dead_count.zero_()
+        self._pending_grants.clear()
+        self._tick = 0  # delay gates + backoff are wave-relative ticks
+
+    # ---- collectives ---------------------------------------------------------
+    def _heartbeats(self) -> None:
+        if self.world > 1:
+            dist.all_gather_into_tensor(self.w_active, self.w_active_local)
+            if not getattr(self, "_hb_static_done", False):
+                dist.all_gather_into_tensor(self.w_cpu, self.w_cpu_local)
+                dist.all_gather_into_tensor(self.w_gpu, self.w_gpu_local)
+                self._hb_static_done = True
+        else:
+            self.w_active.copy_(self.w_active_local)
+            self.w_cpu.copy_(self.w_cpu_local)
+            self.w_gpu.copy_(self.w_gpu_local)
+
+    def _exchange_out(self) -> None:
+        if self.world > 1:
+            dist.all_to_all_single(self.pad_recv_cnt, self.pad_send_cnt)
+            dist.all_to_all_single(self.pad_recv_widx, self.pad_send_widx)
+            dist.all_to_all_single(self.pad_recv_payload, self.pad_send_payload)
+        else:
+            self.pad_recv_cnt.copy_(self.pad_send_cnt)
+            self.pad_recv_widx.copy_(self.pad_send_widx)
+            self.pad_recv_payload.copy_(self.pad_send_payload)
+
+    def _exchange_back(self) -> None:
+        if self.world > 1:
+            dist.all_to_all_single(self.pad_sums_back, self.pad_sums)
+        else:
+            self.pad_sums_back.copy_(self.pad_sums)
+
+    def _refresh_order(self) -> None:
+        if self._tick % 8 != 1:
+            return
+        self.order_buf.copy_(torch.argsort(self.w_keys).to(torch.int32))
+        valid = ((self.w_keys >> 32) & 0xFFFFFFFF).ne(0xFFFFFFFE)
+        self.valid_buf.copy_(valid.sum().to(torch.int32).reshape(1))
+
+    # ---- one workflow tick ---------------------------------------------------
+    def tick(self) -> None:
+        ext = self.ext
+        tick = self._tick
+        self._tick += 1
+        cap, world, Wd = self.pad_cap, self.world, self.W
+
+        # host hop: approvals drained LAST tick are granted now (1-tick
+        # admin latency, like the reference's async approve endpoint)
+        if self._pending_grants:
+            runs, steps = self._pending_grants.pop()
+            n = runs.shape[0]
+            v = torch.full((n,), self.approval_verdict, dtype=torch.uint8,
+                           device=self.device)
+            ext.wf_grant(runs.to(self.device), steps.to(self.device), v, n,
+                         self.step_state)
+
+        # sweep: readiness -> dispatch list + fresh approval holds
+        self.disp_count.zero_()
+        self.appr_count.zero_()
+        ext.wf_sweep(self.step_state, self.deps_mask, self.n_steps,
+                     self.run_active, self.step_kind, self.cond_bits,
+                     self.next_ready, tick,
+                     self.disp_runs, self.disp_steps, self.disp_count,
+                     self.appr_runs, self.appr_steps, self.appr_count)
+
+        # expand into the child arena, spread-routed over the global order
+        self.child_count.zero_()
+        ext.wf_expand(self.disp_runs, self.disp_steps, self.disp_count,
+                      self.step_state, self.children_todo, self.children_out,
+                      self.child_tag, self.child_seq, self.child_widx,
+                      self.child_count, self.children_emitted,
+                      self.order_buf, self.valid_buf)
+
+        # K2 load view + routing order refresh
+        ext.worker_precompute_into(self.w_pool, self.w_active, self.w_maxp,
+                                   self.w_cpu, self.w_gpu, self.w_keys)
+        self._heartbeats()
+
+        # pack per destination: redeliveries first, then fresh children
+        self.rq_prev_widx.copy_(self.rq_widx)
+        self.rq_prev_attempts.copy_(self.rq_attempts)
+        self.rq_prev_count.copy_(self.rq_count)
+        self.rq_prev_payload.copy_(self.rq_payload)
+        self.rq_prev_tag.copy_(self.rq_tag)
+        self.rq_prev_seq.copy_(self.rq_seq)
+        self.rq_count.zero_()
+        self.dead_count.zero_()
+        self.pad_send_cnt.zero_()
+        ext.pack_requeue(self.rq_prev_widx, self.rq_prev_attempts, self.rq_prev_count,
+                         self.pad_send_slots, self.pad_send_widx, self.pad_send_cnt,
+                         self.NWL, cap,
+                         self.rq_src, self.rq_widx, self.rq_attempts,
+                         self.rq_count, self.rq_dead,
+                         self.dead_src, self.dead_count)
+        ext.pack_by_dest(self._child_slots(), self.child_widx, self.child_count,
+                         self.pad_send_slots, self.pad_send_widx, self.pad_send_cnt,
+                         self.NWL, cap, self.CB,
+                         self.rq_src, self.rq_widx, self.rq_attempts,
+                         self.rq_count, self.rq_dead,
+                         self.dead_src, self.dead_count)
+        self.pad_send_cnt.clamp_(max=cap)
+        ext.gather_payload_padded(self.child_payload, self.rq_prev_payload,
+                                  self.pad_send_slots, self.pad_send_cnt,
+                                  self.pad_send_payload, Wd, cap, world)
+        # park-survival: payload + tag of newly parked children
+        ext.materialize_rq_payload(self.child_payload, self.rq_prev_payload,
+                                   self.rq_src, self.rq_count,
+                                   self.rq_payload, Wd)
+        ext.materialize_rq_payload(self.child_tag, self.rq_prev_tag,
+                                   self.rq_src, self.rq_count,
+                                   self.rq_tag, 1)
+        ext.materialize_rq_payload(self.child_seq, self.rq_prev_seq,
+                                   self.rq_src, self.rq_count,
+                                   self.rq_seq, 1)
+
+        # dispatch across ranks + device worker execution + results home
+        self._exchange_out()
+        ext.echo_padded(self.pad_recv_payload, self.pad_recv_cnt, self.pad_res,
+                        self.pad_sums, Wd, cap, world)
+        self.w_active_local.zero_()
+        ext.load_feedback_padded(self.pad_recv_widx, self.pad_recv_cnt,
+                                 self.w_active_local, cap, world)
+        self._exchange_back()
+
+        # owner-rank application + aggregation + roll-up
+        ext.wf_apply(self.pad_send_slots, self.pad_send_cnt, self.child_tag,
+                     self.child_seq, self.rq_prev_tag, self.rq_prev_seq,
+                     self.children_done, self.children_fail,
+                     self.children_out, self.fail_ppt, cap, world)
+        ext.wf_apply_dead(self.dead_src, self.dead_count, self.child_tag,
+                          self.rq_prev_tag, self.children_fail, self.children_out)
+        ext.wf_commit(self.step_state, self.step_attempts, self.children_todo,
+                      self.children_out, self.children_done, self.children_fail,
+                      self.next_ready, tick, self.max_retries)
+        ext.wf_status(self.step_state, self.n_steps, self.run_active,
+                      self.run_state, self.wf_counts)
+        self._refresh_order()
+
+        # drain fresh approval holds (D2H) for next tick's grant
+        na = int(self.appr_count.cpu()[0])
+        if na > 0:
+            na = min(na, self.NR)
+            self._pending_grants.append(
+                (self.appr_runs[:na].cpu().clone(), self.appr_steps[:na].cpu().clone()))
+
+    def _child_slots(self):
+        """Routable slot list for pack_by_dest = child arena indices 0..C-1."""
+        if not hasattr(self, "_iota"):
+            self._iota = torch.arange(self.CB, dtype=torch.int32, device=self.device)
+        return self._iota
+
+    # ---- wave driver ---------------------------------------------------------
+    def counts(self) -> tuple:
+        c = self.wf_counts.cpu()
+        return int(c[0]), int(c[1])
+
+    def active(self) -> int:
+        return int(self.run_active.sum().cpu())
+
+    def run_wave(self, max_ticks: int = 256) -> WfStats:
+        """Admit the full run wave and tick until every run is terminal."""
+        t0 = time.perf_counter()
+        self.reset_runs()
+        start_ok, start_fail = self.counts()
+        ticks = 0
+        # world>1: all ranks must tick in lockstep (collectives); keep
+        # ticking until every rank's runs are done
+        while ticks < max_ticks:
+            self.tick()
+            ticks += 1
+            if ticks % 4 == 0:
+                done = self.active() == 0
+                if self.world > 1:
+                    t = torch.tensor([0 if done else 1],
+                                     dtype=torch.int64, device=self.device
+                                     if self.device.type == "cuda" else "cpu")
+                    dist.all_reduce(t)
+                    done = int(t.item()) == 0
+                if done:
+                    break
+        ok, fail = self.counts()
+        return WfStats(runs_succeeded=ok - start_ok, runs_failed=fail - start_fail,
+                       ticks=ticks, wall_s=time.perf_counter() - t0)

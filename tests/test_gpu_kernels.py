@@ -234,7 +234,9 @@ def test_padded_dispatch_kernels_match_reference(ext):
                 torch.zeros(B, dtype=torch.int32, **kw),
                 torch.zeros(B, dtype=torch.int32, **kw),
                 torch.zeros(1, dtype=torch.int32, **kw),
-                torch.zeros(1, dtype=torch.int64, **kw))
+                torch.zeros(1, dtype=torch.int64, **kw),
+                torch.zeros(8, dtype=torch.int32, **kw),
+                torch.zeros(1, dtype=torch.int32, **kw))
 
     # references
     ss_r = torch.zeros(world * B, dtype=torch.int32)
@@ -324,16 +326,18 @@ def test_requeue_kernels_match_reference(ext):
             "rq_count": torch.zeros(1, dtype=torch.int32, **kw),
             "rq_dead": torch.zeros(1, dtype=torch.int64, **kw),
             "rq_payload": torch.zeros(B * stride, dtype=torch.int32, **kw),
+            "dead_src": torch.zeros(B, dtype=torch.int32, **kw),
+            "dead_count": torch.zeros(1, dtype=torch.int32, **kw),
         }
 
     r = mk()
     ref.pack_requeue_ref(prev_widx, prev_att, prev_count, r["ss"], r["sw"], r["sc"],
                          nwl, cap, r["rq_src"], r["rq_widx"], r["rq_att"],
-                         r["rq_count"], r["rq_dead"])
+                         r["rq_count"], r["rq_dead"], r["dead_src"], r["dead_count"])
     ref.pack_by_dest_ref(routable_slots, routable_widx, routable_count,
                          r["ss"], r["sw"], r["sc"], nwl, cap,
                          r["rq_src"], r["rq_widx"], r["rq_att"],
-                         r["rq_count"], r["rq_dead"])
+                         r["rq_count"], r["rq_dead"], r["dead_src"], r["dead_count"])
     ref.materialize_rq_payload_ref(payload, prev_payload, r["rq_src"],
                                    r["rq_count"], r["rq_payload"], stride)
 
@@ -342,11 +346,11 @@ def test_requeue_kernels_match_reference(ext):
     ext.pack_requeue(prev_widx.to(d), prev_att.to(d), prev_count.to(d),
                      v["ss"], v["sw"], v["sc"], nwl, cap,
                      v["rq_src"], v["rq_widx"], v["rq_att"],
-                     v["rq_count"], v["rq_dead"])
+                     v["rq_count"], v["rq_dead"], v["dead_src"], v["dead_count"])
     ext.pack_by_dest(routable_slots.to(d), routable_widx.to(d), routable_count.to(d),
                      v["ss"], v["sw"], v["sc"], nwl, cap, n,
                      v["rq_src"], v["rq_widx"], v["rq_att"],
-                     v["rq_count"], v["rq_dead"])
+                     v["rq_count"], v["rq_dead"], v["dead_src"], v["dead_count"])
     ext.materialize_rq_payload(payload.to(d), prev_payload.to(d), v["rq_src"],
                                v["rq_count"], v["rq_payload"], stride)
 
@@ -354,6 +358,7 @@ def test_requeue_kernels_match_reference(ext):
     assert torch.equal(v["sc"].cpu().clamp(max=cap), r["sc"].clamp(max=cap))
     assert int(v["rq_count"].cpu()[0]) == int(r["rq_count"][0])
     assert int(v["rq_dead"].cpu()[0]) == int(r["rq_dead"][0])
+    assert int(v["dead_count"].cpu()[0]) == int(r["dead_count"][0])
     nq = min(int(r["rq_count"][0]), B)
     got = sorted(zip(v["rq_src"][:nq].cpu().tolist(), v["rq_widx"][:nq].cpu().tolist(),
                      v["rq_att"][:nq].cpu().tolist()))
