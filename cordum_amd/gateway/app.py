@@ -203,6 +203,7 @@ def create_app(
         node.job_store.set_state(job_id, JobState.PENDING)
         node.job_store.set_job_meta(
             job_id, topic=topic, tenant=org, trace_id=trace_id,
+            team=req.env.get("team_id", ""),
             actor_id=req.meta.actor_id, actor_type=actor_type_raw,
             idempotency_key=idem, capability=req.meta.capability,
             pack_id=req.meta.pack_id, risk_tags=risk_tags,
@@ -218,18 +219,47 @@ def create_app(
         state: str = "",
         topic: str = "",
         tenant: str = "",
+        team: str = "",
+        trace_id: str = "",
+        updated_after: int = 0,
+        updated_before: int = 0,
         limit: int = Query(50, le=500),
         cursor: Optional[int] = None,
     ):
-        ids, next_cursor = node.job_store.list_recent(limit=limit * 4, cursor=cursor)
+        """gateway.go:918-1009 — filters state/topic/tenant/team/trace +
+        updated-window, micros cursor. A state filter pages the per-state
+        index directly (complete at any scale); trace_id lists the trace
+        set; otherwise the recent index pages by score."""
+        updated_after = _normalize_micros(updated_after)
+        updated_before = _normalize_micros(updated_before)
+        cursor = _normalize_micros(cursor) if cursor else None
+        state_u = state.upper() if state else ""
+        if trace_id:
+            ids, next_cursor = node.job_store.get_trace(trace_id), None
+        elif state_u:
+            try:
+                st = JobState[state_u]
+            except KeyError:
+                return {"items": [], "next_cursor": None}
+            ids, next_cursor = node.job_store.list_jobs_by_state_page(
+                st, cursor=cursor, limit=limit)
+        else:
+            ids, next_cursor = node.job_store.list_recent(limit=limit, cursor=cursor)
         items = []
         for jid in ids:
             meta = node.job_store.get_job_meta(jid)
-            if state and meta.get("state", "") != state.upper():
+            if state_u and meta.get("state", "") != state_u:
                 continue
             if topic and meta.get("topic", "") != topic:
                 continue
             if tenant and meta.get("tenant", "") != tenant:
+                continue
+            if team and meta.get("team", "") != team:
+                continue
+            up = meta.get("updated_at", 0)
+            if updated_after and up < updated_after:
+                continue
+            if updated_before and up > updated_before:
                 continue
             items.append(_job_summary(jid, meta))
             if len(items) >= limit:
@@ -827,12 +857,28 @@ async def _json_body(request: Request) -> Dict[str, Any]:
     return body
 
 
+def _normalize_micros(ts: Optional[int]) -> int:
+    """gateway.go:2407-2421 — accept seconds/millis/micros/nanos, return
+    micros (upper-bound variant without the sub-unit rounding: callers here
+    compare strictly so the +999_999 fill is immaterial for paging)."""
+    if not ts or ts <= 0:
+        return 0
+    if ts < 1_000_000_000_000:          # seconds
+        return ts * 1_000_000
+    if ts < 1_000_000_000_000_000:      # millis
+        return ts * 1_000
+    if ts < 1_000_000_000_000_000_000:  # micros
+        return ts
+    return ts // 1_000                  # nanos
+
+
 def _job_summary(job_id: str, meta: Dict[str, Any]) -> Dict[str, Any]:
     return {
         "id": job_id,
         "state": meta.get("state", ""),
         "topic": meta.get("topic", ""),
         "tenant": meta.get("tenant", ""),
+        "team": meta.get("team", ""),
         "trace_id": meta.get("trace_id", ""),
         "updated_at": meta.get("updated_at", 0),
         "attempts": meta.get("attempts", 0),

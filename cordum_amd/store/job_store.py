@@ -243,6 +243,21 @@ class JobStore:
         with self._mu:
             return len(self._state_index[state])
 
+    def list_jobs_by_state_page(
+        self, state: JobState, cursor: Optional[int] = None, limit: int = 50
+    ) -> Tuple[List[str], Optional[int]]:
+        """Newest-first page over the per-state ZSET analog with a micros
+        cursor (job_store.go:676+ / gateway.go:918-1009): state-filtered
+        listings page the state index directly instead of post-filtering the
+        recent index, so results stay complete at any scale."""
+        with self._mu:
+            items = sorted(self._state_index[state].items(), key=lambda kv: -kv[1])
+            if cursor is not None:
+                items = [(j, t) for j, t in items if t < cursor]
+            page = items[:limit]
+            next_cursor = page[-1][1] if len(items) > limit and page else None
+            return [j for j, _ in page], next_cursor
+
     def list_recent(self, limit: int = 100, cursor: Optional[int] = None) -> Tuple[List[str], Optional[int]]:
         """Recent jobs newest-first with micros cursor (gateway.go:918-1009)."""
         with self._mu:
