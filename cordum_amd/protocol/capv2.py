@@ -272,9 +272,11 @@ def _encode_field(spec: F, val: Any) -> bytes:
             kv = bytearray()
             kb = mk.encode("utf-8")
             vb = val[mk].encode("utf-8")
+            # map entries serialize BOTH fields even when empty (protobuf
+            # MapEntry semantics — verified byte-for-byte against the
+            # google.protobuf runtime in tests/test_capv2_interop.py)
             kv += _tag(1, _WT_LEN) + _enc_varint(len(kb)) + kb
-            if vb:
-                kv += _tag(2, _WT_LEN) + _enc_varint(len(vb)) + vb
+            kv += _tag(2, _WT_LEN) + _enc_varint(len(vb)) + vb
             out += _tag(spec.num, _WT_LEN) + _enc_varint(len(kv)) + bytes(kv)
         return bytes(out)
     raise ValueError(f"unknown kind {k}")
@@ -734,7 +736,11 @@ class BusPacket(Message):
 
     trace_id: str = ""
     tenant_id: str = ""
-    protocol_version: int = 1
+    # proto3 implicit presence: the wire default MUST be 0 (a 1 default
+    # would re-materialize on decode(encode(x)) of a foreign packet and
+    # break byte round-trips — caught by tests/test_capv2_interop.py).
+    # Senders set protocol_version=1 explicitly (capsdk constants).
+    protocol_version: int = 0
     labels: Dict[str, str] = dc_field(default_factory=dict)
     job_request: Optional[JobRequest] = None
     job_result: Optional[JobResult] = None

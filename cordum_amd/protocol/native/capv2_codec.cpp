@@ -165,14 +165,14 @@ static void encode_field(const FieldSpec& f, py::handle val, std::string& out) {
       std::sort(items.begin(), items.end());  // deterministic, like capv2.py
       for (auto& kv : items) {
         std::string entry;
+        // map entries serialize BOTH fields even when empty (protobuf
+        // MapEntry semantics; byte-verified in tests/test_capv2_interop.py)
         put_tag(entry, 1, 2);
         put_varint(entry, kv.first.size());
         entry += kv.first;
-        if (!kv.second.empty()) {
-          put_tag(entry, 2, 2);
-          put_varint(entry, kv.second.size());
-          entry += kv.second;
-        }
+        put_tag(entry, 2, 2);
+        put_varint(entry, kv.second.size());
+        entry += kv.second;
         put_tag(out, f.num, 2);
         put_varint(out, entry.size());
         out += entry;

@@ -1,0 +1,225 @@
+"""CAP v2 wire interop proof (VERDICT round-1 item #6).
+
+Parses the vendored cordum_amd/protocol/capv2.proto with a minimal proto3
+parser, builds REAL google.protobuf message classes from it at runtime
+(FileDescriptorProto -> descriptor pool -> message_factory — functionally
+what protoc would generate), and property-tests byte equality both ways
+against the hand-rolled deterministic codec in cordum_amd/protocol/capv2.py:
+
+    ours.encode()  == protobuf.SerializeToString(deterministic=True)
+    ours.decode(protobuf bytes) round-trips to identical bytes
+
+This proves the codec is genuine protobuf wire format under the declared
+numbering, so any protoc-generated binding of capv2.proto (Go, Python, C++)
+interoperates with this node byte-for-byte (pb.go:1-76 shapes)."""
+import random
+import re
+from pathlib import Path
+
+import pytest
+
+from cordum_amd.protocol import capv2
+
+PROTO_PATH = Path(__file__).resolve().parent.parent / "cordum_amd" / "protocol" / "capv2.proto"
+
+pb = pytest.importorskip("google.protobuf")
+from google.protobuf import descriptor_pb2, descriptor_pool, message_factory  # noqa: E402
+
+TYPE = descriptor_pb2.FieldDescriptorProto
+
+
+def parse_proto(text: str):
+    """Minimal proto3 parser for the subset capv2.proto uses: enums,
+    messages with scalar/enum/message/repeated/map fields."""
+    text = re.sub(r"//[^\n]*", "", text)
+    enums, messages = {}, {}
+    for m in re.finditer(r"enum\s+(\w+)\s*\{([^}]*)\}", text):
+        vals = re.findall(r"(\w+)\s*=\s*(\d+)\s*;", m.group(2))
+        enums[m.group(1)] = [(n, int(v)) for n, v in vals]
+    for m in re.finditer(r"message\s+(\w+)\s*\{([^}]*)\}", text):
+        fields = []
+        for f in re.finditer(
+            r"(repeated\s+)?(map\s*<\s*string\s*,\s*string\s*>|[\w.]+)\s+(\w+)\s*=\s*(\d+)\s*;",
+            m.group(2),
+        ):
+            rep, typ, name, num = f.group(1), f.group(2), f.group(3), int(f.group(4))
+            fields.append((name, num, typ.replace(" ", ""), bool(rep)))
+        messages[m.group(1)] = fields
+    return enums, messages
+
+
+SCALARS = {
+    "int64": TYPE.TYPE_INT64,
+    "int32": TYPE.TYPE_INT32,
+    "string": TYPE.TYPE_STRING,
+    "bytes": TYPE.TYPE_BYTES,
+    "double": TYPE.TYPE_DOUBLE,
+    "bool": TYPE.TYPE_BOOL,
+}
+
+
+@pytest.fixture(scope="module")
+def classes():
+    enums, messages = parse_proto(PROTO_PATH.read_text())
+    fdp = descriptor_pb2.FileDescriptorProto()
+    fdp.name = "capv2_interop.proto"
+    fdp.package = "cap.v2.interop"
+    fdp.syntax = "proto3"
+    for ename, vals in enums.items():
+        e = fdp.enum_type.add()
+        e.name = ename
+        for n, v in vals:
+            ev = e.value.add()
+            ev.name = n
+            ev.number = v
+    for mname, fields in messages.items():
+        msg = fdp.message_type.add()
+        msg.name = mname
+        for fname, num, typ, rep in fields:
+            fld = msg.field.add()
+            fld.name = fname
+            fld.number = num
+            if typ == "map<string,string>":
+                entry = msg.nested_type.add()
+                entry.name = "".join(p.capitalize() for p in fname.split("_")) + "Entry"
+                entry.options.map_entry = True
+                for kn, kno in (("key", 1), ("value", 2)):
+                    kf = entry.field.add()
+                    kf.name = kn
+                    kf.number = kno
+                    kf.type = TYPE.TYPE_STRING
+                    kf.label = TYPE.LABEL_OPTIONAL
+                fld.type = TYPE.TYPE_MESSAGE
+                fld.type_name = f".cap.v2.interop.{mname}.{entry.name}"
+                fld.label = TYPE.LABEL_REPEATED
+            elif typ in SCALARS:
+                fld.type = SCALARS[typ]
+                fld.label = TYPE.LABEL_REPEATED if rep else TYPE.LABEL_OPTIONAL
+            elif typ in enums:
+                fld.type = TYPE.TYPE_ENUM
+                fld.type_name = f".cap.v2.interop.{typ}"
+                fld.label = TYPE.LABEL_OPTIONAL
+            else:  # message type
+                fld.type = TYPE.TYPE_MESSAGE
+                fld.type_name = f".cap.v2.interop.{typ}"
+                fld.label = TYPE.LABEL_REPEATED if rep else TYPE.LABEL_OPTIONAL
+    pool = descriptor_pool.DescriptorPool()
+    fd = pool.Add(fdp)
+    return {name: message_factory.GetMessageClass(fd.message_types_by_name[name])
+            for name in messages}
+
+
+def test_proto_numbering_matches_codec(classes):
+    """Every FIELDS entry in capv2.py must match the vendored .proto:
+    same field names, numbers, and cardinality — the .proto is the
+    authoritative contract."""
+    enums, messages = parse_proto(PROTO_PATH.read_text())
+    for cls_name, fields in messages.items():
+        cls = getattr(capv2, cls_name)
+        declared = {name: num for name, num, _, _ in fields}
+        ours = {name: spec.num for name, spec in cls.FIELDS.items()}
+        assert ours == declared, cls_name
+    prefixes = {"JobPriority": "JOB_PRIORITY_", "JobStatus": "JOB_STATUS_",
+                "ActorType": "ACTOR_TYPE_", "DecisionType": "DECISION_TYPE_"}
+    for ename, vals in enums.items():
+        cls = getattr(capv2, ename)
+        declared = {n.replace(prefixes[ename], ""): v for n, v in vals}
+        ours = {m.name: int(m.value) for m in cls}
+        assert ours == declared, ename
+
+
+def _rand_str(rng):
+    return "".join(rng.choice("abcdefghijklmnop-_.:/") for _ in range(rng.randint(0, 12)))
+
+
+def _rand_map(rng):
+    return {("k%d" % i) + _rand_str(rng): _rand_str(rng)
+            for i in range(rng.randint(0, 4))}
+
+
+def _fill_both(rng, ours, theirs, classes, depth=0):
+    """Randomly populate the same values into our message and the protobuf
+    message (recursing into submessages)."""
+    for name, spec in type(ours).FIELDS.items():
+        if spec.kind in ("int",):
+            v = rng.choice([0, 1, 7, 1 << 20, (1 << 53) - 1])
+            setattr(ours, name, v)
+            setattr(theirs, name, v)
+        elif spec.kind == "enum":
+            v = rng.randint(0, 4)
+            setattr(ours, name, v)
+            setattr(theirs, name, v)
+        elif spec.kind == "bool":
+            v = rng.random() < 0.5
+            setattr(ours, name, v)
+            setattr(theirs, name, v)
+        elif spec.kind == "double":
+            v = rng.choice([0.0, 1.5, -3.25, 0.875])
+            setattr(ours, name, v)
+            setattr(theirs, name, v)
+        elif spec.kind == "str":
+            v = _rand_str(rng)
+            setattr(ours, name, v)
+            setattr(theirs, name, v)
+        elif spec.kind == "bytes":
+            v = bytes(rng.randrange(256) for _ in range(rng.randint(0, 8)))
+            setattr(ours, name, v)
+            setattr(theirs, name, v)
+        elif spec.kind == "rep_str":
+            v = [_rand_str(rng) for _ in range(rng.randint(0, 3))]
+            setattr(ours, name, list(v))
+            getattr(theirs, name).extend(v)
+        elif spec.kind == "map_ss":
+            v = _rand_map(rng)
+            setattr(ours, name, dict(v))
+            getattr(theirs, name).update(v)
+        elif spec.kind == "msg" and depth < 2 and rng.random() < 0.7:
+            sub_ours = spec.sub()
+            sub_theirs = getattr(theirs, name)
+            _fill_both(rng, sub_ours, sub_theirs, classes, depth + 1)
+            setattr(ours, name, sub_ours)
+        elif spec.kind == "rep_msg":
+            for _ in range(rng.randint(0, 2)):
+                sub_ours = capv2.PolicyRemediation()
+                sub_theirs = getattr(theirs, name).add()
+                _fill_both(rng, sub_ours, sub_theirs, classes, depth + 1)
+                ours.remediations.append(sub_ours)
+
+
+ALL_TYPES = [
+    "Budget", "ContextHints", "JobMetadata", "JobRequest", "JobResult",
+    "JobProgress", "JobCancel", "Heartbeat", "SystemAlert",
+    "BudgetConstraints", "SandboxProfile", "ToolchainConstraints",
+    "DiffConstraints", "PolicyConstraints", "PolicyRemediation",
+    "PolicyCheckRequest", "PolicyCheckResponse", "BusPacket",
+]
+
+
+@pytest.mark.parametrize("type_name", ALL_TYPES)
+def test_encode_bytes_equal_protobuf(classes, type_name):
+    rng = random.Random(hash(type_name) & 0xFFFF)
+    for _ in range(50):
+        ours = getattr(capv2, type_name)()
+        theirs = classes[type_name]()
+        _fill_both(rng, ours, theirs, classes)
+        got = ours.encode()
+        want = theirs.SerializeToString(deterministic=True)
+        assert got == want, f"{type_name}: codec bytes != protobuf bytes"
+
+
+@pytest.mark.parametrize("type_name", ALL_TYPES)
+def test_decode_protobuf_bytes(classes, type_name):
+    """decode(protobuf.Serialize(x)) must re-encode to identical bytes
+    (full information preservation both directions)."""
+    rng = random.Random(hash(type_name) & 0xFFF)
+    for _ in range(50):
+        ours = getattr(capv2, type_name)()
+        theirs = classes[type_name]()
+        _fill_both(rng, ours, theirs, classes)
+        wire = theirs.SerializeToString(deterministic=True)
+        rt = getattr(capv2, type_name).decode(wire)
+        assert rt.encode() == wire
+        # and the protobuf side parses OUR bytes to the same message
+        back = classes[type_name]()
+        back.ParseFromString(ours.encode())
+        assert back.SerializeToString(deterministic=True) == wire
