@@ -294,14 +294,19 @@ def make_packs_router(node, principal_dep, admin_dep) -> APIRouter:
         except PackError as e:
             raise HTTPException(404, str(e))
 
-    # marketplace: catalog registry (no egress in this environment — catalog
-    # entries are served from the config doc; URL installs require a pinned
-    # sha256 AND an enabled catalog entry, enforced here; the fetch itself is
-    # delegated to the caller-supplied blob)
+    # -- marketplace (packs.go:454-607, :933-1200) ---------------------------
+    # Catalog registry doc cfg:system:pack_catalogs {"catalogs":[{id,title,
+    # url,enabled}]}; each enabled catalog's file ({updated_at, packs:[...]})
+    # is fetched through node.marketplace_fetcher(url) -> bytes (pluggable:
+    # there is no egress in this deployment, so the default fetcher refuses;
+    # tests and real deployments inject one — marketplace_test.go uses a stub
+    # HTTP server the same way). Snapshot cached 30 s; URL installs must
+    # match an enabled catalog entry AND its pinned sha256.
+    mkt = Marketplace(node, installer)
+
     @router.get("/marketplace/packs")
-    def marketplace(p=Depends(principal_dep)):
-        catalogs = node.configsvc.get(*CATALOGS_DOC) or {}
-        return {"items": list((catalogs.get("entries") or {}).values())}
+    def marketplace(p=Depends(admin_dep)):
+        return mkt.snapshot()
 
     @router.post("/marketplace/install")
     async def marketplace_install(request: Request, p=Depends(admin_dep)):
@@ -310,23 +315,153 @@ def make_packs_router(node, principal_dep, admin_dep) -> APIRouter:
             body = json.loads(body_raw or b"{}")
         except ValueError:
             raise HTTPException(400, "invalid json")
-        name = (body.get("name") or "").strip()
-        catalogs = node.configsvc.get(*CATALOGS_DOC) or {}
-        entry = (catalogs.get("entries") or {}).get(name)
-        if entry is None or not entry.get("enabled", True):
-            raise HTTPException(403, "pack not in an enabled catalog")
-        blob_b64 = body.get("archive_base64", "")
-        if not blob_b64:
-            raise HTTPException(400, "archive_base64 required (no egress in this deployment)")
-        import base64
-
-        blob = base64.b64decode(blob_b64)
-        want = entry.get("sha256", "")
-        if want and _digest(blob) != want:
-            raise HTTPException(403, "archive digest does not match catalog entry")
-        try:
-            return installer.install(blob, source=f"marketplace:{name}")
-        except PackError as e:
-            raise HTTPException(400, str(e))
+        if not isinstance(body, dict):
+            raise HTTPException(400, "invalid json payload")
+        return mkt.install(body)
 
     return router
+
+
+DEFAULT_CATALOG_URL = "https://packs.cordum.io/catalog.json"
+MARKETPLACE_CACHE_TTL = 30.0
+
+
+class Marketplace:
+    """Catalog registry + 30 s snapshot cache + sha256-pinned installs
+    (packs.go:454-607 handleMarketplaceInstall, :933 marketplaceSnapshot,
+    :1128/:1168 findMarketplaceEntry/ByURL)."""
+
+    def __init__(self, node, installer):
+        self.node = node
+        self.installer = installer
+        self._cache = None
+        self._cache_at = 0.0
+
+    def _fetch(self, url: str) -> bytes:
+        fetcher = getattr(self.node, "marketplace_fetcher", None)
+        if fetcher is None:
+            raise PackError("marketplace fetch unavailable (no egress; "
+                            "configure a marketplace fetcher)")
+        return fetcher(url)
+
+    def _catalogs(self) -> List[Dict[str, Any]]:
+        doc = self.node.configsvc.get(*CATALOGS_DOC) or {}
+        cats = doc.get("catalogs")
+        if cats is None:
+            cats = [{"id": "cordum", "title": "Cordum Packs",
+                     "url": DEFAULT_CATALOG_URL, "enabled": True}]
+        return [c for c in cats if isinstance(c, dict)]
+
+    def _load_entries(self):
+        statuses, entries = [], []
+        for idx, cat in enumerate(self._catalogs()):
+            cid = (cat.get("id") or "").strip() or f"catalog-{idx + 1}"
+            status = {
+                "id": cid,
+                "title": (cat.get("title") or "").strip(),
+                "url": (cat.get("url") or "").strip(),
+                "enabled": bool(cat.get("enabled", True)),
+            }
+            if not status["enabled"]:
+                statuses.append(status)
+                continue
+            try:
+                raw = self._fetch(status["url"])
+                doc = json.loads(raw)
+            except Exception as e:  # catalog errors are per-catalog, not fatal
+                status["error"] = str(e)
+                statuses.append(status)
+                continue
+            status["updated_at"] = doc.get("updated_at", "")
+            statuses.append(status)
+            for pack in doc.get("packs", []) or []:
+                if isinstance(pack, dict):
+                    entries.append({"pack": pack, "catalog_id": cid,
+                                    "catalog_title": status["title"]})
+        return statuses, entries
+
+    def snapshot(self, refresh: bool = False) -> Dict[str, Any]:
+        now = self.node.clock.now() if hasattr(self.node, "clock") else 0.0
+        if not refresh and self._cache is not None and \
+                now - self._cache_at < MARKETPLACE_CACHE_TTL:
+            resp = dict(self._cache)
+            resp["cached"] = True
+            return resp
+        statuses, entries = self._load_entries()
+        items = []
+        for e in entries:
+            p = e["pack"]
+            items.append({
+                "id": p.get("id", ""), "version": p.get("version", ""),
+                "title": p.get("title", ""), "description": p.get("description", ""),
+                "author": p.get("author", ""), "homepage": p.get("homepage", ""),
+                "source": p.get("source", ""), "license": p.get("license", ""),
+                "url": p.get("url", ""), "sha256": p.get("sha256", ""),
+                "capabilities": p.get("capabilities", []) or [],
+                "requires": p.get("requires", []) or [],
+                "risk_tags": p.get("risk_tags", []) or [],
+                "catalog_id": e["catalog_id"],
+                "catalog_title": e["catalog_title"],
+            })
+        resp = {"catalogs": statuses, "items": items, "fetched_at": now}
+        self._cache, self._cache_at = resp, now
+        return dict(resp)
+
+    def _find_entry(self, catalog_id: str, pack_id: str, version: str):
+        _, entries = self._load_entries()
+        for e in entries:
+            p = e["pack"]
+            if e["catalog_id"] == catalog_id and p.get("id") == pack_id:
+                if version and p.get("version") != version:
+                    continue
+                return e
+        raise HTTPException(404, "marketplace entry not found")
+
+    def _find_entry_by_url(self, url: str):
+        _, entries = self._load_entries()
+        for e in entries:
+            if (e["pack"].get("url") or "").strip() == url:
+                return e
+        raise HTTPException(404, "pack url not found in any enabled catalog")
+
+    def install(self, body: Dict[str, Any]) -> Dict[str, Any]:
+        url = (body.get("url") or "").strip()
+        sha = (body.get("sha256") or "").strip()
+        if url:
+            if not sha:
+                raise HTTPException(400, "sha256 required")
+            entry = self._find_entry_by_url(url)
+            p = entry["pack"]
+            entry_url = (p.get("url") or "").strip()
+            entry_sha = (p.get("sha256") or "").strip()
+            if not entry_url or not entry_sha:
+                raise HTTPException(400, "marketplace entry missing url or sha256")
+            if sha.lower() != entry_sha.lower():
+                raise HTTPException(400, "sha256 mismatch")
+            url, sha = entry_url, entry_sha
+        else:
+            catalog_id = (body.get("catalog_id") or "").strip()
+            pack_id = (body.get("pack_id") or "").strip()
+            if not catalog_id or not pack_id:
+                raise HTTPException(400, "catalog_id and pack_id required")
+            entry = self._find_entry(catalog_id, pack_id,
+                                     (body.get("version") or "").strip())
+            p = entry["pack"]
+            url = (p.get("url") or "").strip()
+            sha = (p.get("sha256") or "").strip()
+        if not url:
+            raise HTTPException(400, "download url required")
+        if not sha:
+            raise HTTPException(400, "sha256 required")
+        try:
+            blob = self._fetch(url)
+        except PackError as e:
+            raise HTTPException(400, str(e))
+        except Exception as e:
+            raise HTTPException(400, f"download failed: {e}")
+        if _digest(blob).lower() != sha.lower():
+            raise HTTPException(400, "sha256 mismatch")
+        try:
+            return self.installer.install(blob, source=f"marketplace:{url}")
+        except PackError as e:
+            raise HTTPException(400, str(e))

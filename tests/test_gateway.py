@@ -364,3 +364,94 @@ def test_dashboard_and_health(client):
     r = client.get("/dashboard")
     assert r.status_code == 200 and "cordum-mi355x" in r.text
     assert client.get("/metrics").status_code == 200
+
+
+def test_marketplace_catalog_flow(client, node):
+    """packs.go:454-607: catalog registry + 30s cache + sha256-pinned
+    install, with a stub fetcher standing in for the catalog HTTP server
+    (like marketplace_test.go's httptest servers)."""
+    import hashlib
+
+    blob = make_pack_tgz(HELLO_PACK, HELLO_FILES)
+    sha = hashlib.sha256(blob).hexdigest()
+    catalog = {
+        "updated_at": "2026-01-01T00:00:00Z",
+        "packs": [{
+            "id": "hello-pack", "version": "1.0.0", "title": "Hello",
+            "url": "https://packs.example.com/hello.tgz", "sha256": sha,
+        }],
+    }
+    fetches = []
+
+    def fetcher(url):
+        fetches.append(url)
+        if url.endswith("catalog.json"):
+            return json.dumps(catalog).encode()
+        if url.endswith("hello.tgz"):
+            return blob
+        raise ValueError("unknown url " + url)
+
+    node.marketplace_fetcher = fetcher
+    node.configsvc.set("system", "pack_catalogs", {"catalogs": [
+        {"id": "main", "title": "Main", "url": "https://packs.example.com/catalog.json",
+         "enabled": True},
+        {"id": "off", "url": "https://off.example.com/catalog.json", "enabled": False},
+    ]})
+
+    r = client.get("/api/v1/marketplace/packs")
+    assert r.status_code == 200, r.text
+    snap = r.json()
+    assert [c["id"] for c in snap["catalogs"]] == ["main", "off"]
+    assert snap["catalogs"][1]["enabled"] is False
+    assert snap["items"][0]["id"] == "hello-pack"
+    assert snap["items"][0]["sha256"] == sha
+    # 30s cache: the second snapshot does not re-fetch
+    n_fetches = len(fetches)
+    r = client.get("/api/v1/marketplace/packs")
+    assert r.json()["cached"] is True
+    assert len(fetches) == n_fetches
+
+    # catalog_id + pack_id install path
+    r = client.post("/api/v1/marketplace/install",
+                    json={"catalog_id": "main", "pack_id": "hello-pack"})
+    assert r.status_code == 200, r.text
+    assert r.json()["status"] == "ACTIVE"
+
+    # URL install requires the pinned sha256 and catalog membership
+    r = client.post("/api/v1/marketplace/install",
+                    json={"url": "https://packs.example.com/hello.tgz"})
+    assert r.status_code == 400  # sha256 required
+    r = client.post("/api/v1/marketplace/install",
+                    json={"url": "https://packs.example.com/hello.tgz",
+                          "sha256": "deadbeef"})
+    assert r.status_code == 400  # sha mismatch vs catalog pin
+    r = client.post("/api/v1/marketplace/install",
+                    json={"url": "https://rogue.example.com/evil.tgz",
+                          "sha256": sha})
+    assert r.status_code == 404  # not in any enabled catalog
+    r = client.post("/api/v1/marketplace/install",
+                    json={"url": "https://packs.example.com/hello.tgz",
+                          "sha256": sha.upper()})
+    assert r.status_code == 200  # case-insensitive digest compare
+
+    # tampered payload: catalog pin protects the install
+    def bad_fetcher(url):
+        if url.endswith("catalog.json"):
+            return json.dumps(catalog).encode()
+        return b"tampered"
+
+    node.marketplace_fetcher = bad_fetcher
+    r = client.post("/api/v1/marketplace/install",
+                    json={"catalog_id": "main", "pack_id": "hello-pack"})
+    assert r.status_code == 400
+    assert "sha256 mismatch" in r.text
+
+
+def test_marketplace_no_fetcher_refuses(client, node):
+    if hasattr(node, "marketplace_fetcher"):
+        del node.marketplace_fetcher
+    node.configsvc.set("system", "pack_catalogs", {"catalogs": [
+        {"id": "main", "url": "https://packs.example.com/catalog.json"}]})
+    snap = client.get("/api/v1/marketplace/packs").json()
+    assert "error" in snap["catalogs"][0]
+    assert snap["items"] == []
