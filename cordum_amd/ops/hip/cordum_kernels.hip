@@ -565,6 +565,8 @@ __global__ __launch_bounds__(BLOCK) void wf_expand_kernel(
     int* __restrict__ child_widx,             // [CB] global worker pick
     int* __restrict__ child_count,            // [1]
     int* __restrict__ children_emitted,       // [NR*64] monotone ordinal source
+    int* __restrict__ dispatch_tick,          // [NR*64] stamped for K4-WF
+    int tick,
     const int* __restrict__ order,            // [NWG] spread order (K2c)
     const int* __restrict__ valid_count,      // [1]
     int disp_cap, int CB)
@@ -591,6 +593,7 @@ __global__ __launch_bounds__(BLOCK) void wf_expand_kernel(
             children_out[rs] += todo;
             children_emitted[rs] += todo;
             children_todo[rs] = 0;
+            dispatch_tick[rs] = tick;
         }
     }
     base = __shfl(base, 0, WAVE);
@@ -631,7 +634,7 @@ __global__ __launch_bounds__(BLOCK) void wf_apply_kernel(
     int* __restrict__ children_done,          // [NR*64]
     int* __restrict__ children_fail,          // [NR*64]
     int* __restrict__ children_out,           // [NR*64]
-    int fail_ppt, int cap, int world)
+    int fail_ppt, int drop_ppt, int cap, int world)
 {
     const int i = blockIdx.x * BLOCK + threadIdx.x;
     if (i >= world * cap) return;
@@ -639,6 +642,11 @@ __global__ __launch_bounds__(BLOCK) void wf_apply_kernel(
     const int sslot = send_slots[i];
     const int tag = sslot >= 0 ? child_tag[sslot] : rq_prev_tag[-1 - sslot];
     const int seq = sslot >= 0 ? child_seq[sslot] : rq_prev_seq[-1 - sslot];
+    // lost-result injection (crashed worker): the child simply vanishes;
+    // children_out stays up and the K4-WF timeout scan recovers it later
+    const unsigned int hd = wf_mix((unsigned int)tag * 0x85EBCA6Bu
+                                   ^ (unsigned int)seq * 0xC2B2AE35u);
+    if ((int)(hd % 1000u) < drop_ppt) return;
     const unsigned int h = wf_mix((unsigned int)tag * 2654435761u
                                   ^ (unsigned int)seq * 40503u);
     if ((int)(h % 1000u) < fail_ppt) atomicAdd(&children_fail[tag], 1);
@@ -666,6 +674,77 @@ __global__ __launch_bounds__(BLOCK) void wf_apply_dead_kernel(
     atomicSub(&children_out[tag], 1);
 }
 
+// K4-WF: stale-step timeout scan (reconciler.go:88-144 analog over the
+// run/step table). A step DISPATCHED for > cutoff ticks with children still
+// outstanding has lost them (worker crash / dropped result): the remainder
+// is declared TIMEOUT — counted, converted to failed children — so the
+// commit pass retries them with backoff instead of hanging forever.
+__global__ __launch_bounds__(BLOCK) void wf_timeout_scan_kernel(
+    const unsigned char* __restrict__ step_state,
+    int* __restrict__ children_out,
+    int* __restrict__ children_fail,
+    const int* __restrict__ dispatch_tick,    // [NR*64] last expansion tick
+    int tick, int cutoff,
+    unsigned long long* __restrict__ timeout_count,
+    int NRS)
+{
+    const int i = blockIdx.x * BLOCK + threadIdx.x;
+    if (i >= NRS) return;
+    if (step_state[i] != WFS_DISPATCHED) return;
+    const int lost = children_out[i];
+    if (lost <= 0) return;
+    if (dispatch_tick[i] > tick - cutoff) return;
+    children_fail[i] += lost;
+    children_out[i] = 0;
+    atomicAdd(timeout_count, (unsigned long long)lost);
+}
+
+// continuous re-admission (config #5 "hold 1M concurrent runs"): terminal
+// runs in `filter_state` are reset from the creation template and re-enter
+// the system. SUCCEEDED runs re-admit on device every tick; FAILED runs
+// wait for the host DLQ drain (dlq_store.go analog) which re-admits them
+// through the same kernel after recording the entries.
+__global__ __launch_bounds__(BLOCK) void wf_readmit_kernel(
+    unsigned char* __restrict__ run_active,
+    unsigned char* __restrict__ run_state,
+    const unsigned char* __restrict__ n_steps,
+    unsigned char* __restrict__ step_state,
+    int* __restrict__ step_attempts,
+    int* __restrict__ children_todo,
+    int* __restrict__ children_out,
+    int* __restrict__ children_done,
+    int* __restrict__ children_fail,
+    int* __restrict__ children_emitted,
+    int* __restrict__ next_ready,
+    int* __restrict__ dispatch_tick,
+    const int* __restrict__ todo_tmpl,        // [NR*64]
+    const int* __restrict__ nready_tmpl,      // [NR*64]
+    int filter_state,
+    unsigned long long* __restrict__ admit_count,
+    int NR)
+{
+    const int run = blockIdx.x * BLOCK + threadIdx.x;
+    if (run >= NR) return;
+    if (run_active[run] || run_state[run] != filter_state) return;
+    const int ns = n_steps[run];
+    const size_t base = (size_t)run * 64;
+    for (int s = 0; s < ns; ++s) {
+        const size_t i = base + s;
+        step_state[i] = WFS_PENDING;
+        step_attempts[i] = 0;
+        children_todo[i] = todo_tmpl[i];
+        children_out[i] = 0;
+        children_done[i] = 0;
+        children_fail[i] = 0;
+        children_emitted[i] = 0;
+        next_ready[i] = nready_tmpl[i];
+        dispatch_tick[i] = 0;
+    }
+    run_state[run] = 0;
+    run_active[run] = 1;
+    atomicAdd(admit_count, 1ull);
+}
+
 // step commit: aggregate children (engine.go:1623-1645) + retry/backoff
 // (computeBackoff :1573-1595, attempts capped by max_retries)
 __global__ __launch_bounds__(BLOCK) void wf_commit_kernel(
@@ -676,7 +755,9 @@ __global__ __launch_bounds__(BLOCK) void wf_commit_kernel(
     int* __restrict__ children_done,          // kept: successes accumulate
     int* __restrict__ children_fail,
     int* __restrict__ next_ready,
-    int tick, int max_retries, int NRS)
+    int tick, int max_retries,
+    unsigned long long* __restrict__ retry_count,  // cumulative children retried
+    int NRS)
 {
     const int i = blockIdx.x * BLOCK + threadIdx.x;
     if (i >= NRS) return;
@@ -690,6 +771,7 @@ __global__ __launch_bounds__(BLOCK) void wf_commit_kernel(
             children_fail[i] = 0;
             next_ready[i] = tick + min(1 << step_attempts[i], 16);
             step_state[i] = WFS_PENDING;
+            atomicAdd(retry_count, (unsigned long long)fail);
         } else {
             step_state[i] = WFS_FAILED;
         }
@@ -2006,6 +2088,7 @@ void wf_expand(torch::Tensor disp_runs, torch::Tensor disp_steps, torch::Tensor 
                torch::Tensor step_state, torch::Tensor children_todo, torch::Tensor children_out,
                torch::Tensor child_tag, torch::Tensor child_seq, torch::Tensor child_widx,
                torch::Tensor child_count, torch::Tensor children_emitted,
+               torch::Tensor dispatch_tick, int64_t tick,
                torch::Tensor order, torch::Tensor valid_count)
 {
     const int cap = (int)disp_runs.size(0);
@@ -2016,15 +2099,52 @@ void wf_expand(torch::Tensor disp_runs, torch::Tensor disp_steps, torch::Tensor 
         children_out.data_ptr<int>(), child_tag.data_ptr<int>(),
         child_seq.data_ptr<int>(), child_widx.data_ptr<int>(), child_count.data_ptr<int>(),
         children_emitted.data_ptr<int>(),
+        dispatch_tick.data_ptr<int>(), (int)tick,
         order.data_ptr<int>(), valid_count.data_ptr<int>(),
         cap, (int)child_tag.size(0));
+}
+
+void wf_timeout_scan(torch::Tensor step_state, torch::Tensor children_out,
+                     torch::Tensor children_fail, torch::Tensor dispatch_tick,
+                     int64_t tick, int64_t cutoff, torch::Tensor timeout_count)
+{
+    const int NRS = (int)step_state.numel();
+    const int blocks = (NRS + BLOCK - 1) / BLOCK;
+    hipLaunchKernelGGL(wf_timeout_scan_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        step_state.data_ptr<uint8_t>(), children_out.data_ptr<int>(),
+        children_fail.data_ptr<int>(), dispatch_tick.data_ptr<int>(),
+        (int)tick, (int)cutoff,
+        (unsigned long long*)timeout_count.data_ptr<int64_t>(), NRS);
+}
+
+void wf_readmit(torch::Tensor run_active, torch::Tensor run_state, torch::Tensor n_steps,
+                torch::Tensor step_state, torch::Tensor step_attempts,
+                torch::Tensor children_todo, torch::Tensor children_out,
+                torch::Tensor children_done, torch::Tensor children_fail,
+                torch::Tensor children_emitted, torch::Tensor next_ready,
+                torch::Tensor dispatch_tick, torch::Tensor todo_tmpl,
+                torch::Tensor nready_tmpl, int64_t filter_state,
+                torch::Tensor admit_count)
+{
+    const int NR = (int)n_steps.size(0);
+    const int blocks = (NR + BLOCK - 1) / BLOCK;
+    hipLaunchKernelGGL(wf_readmit_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        run_active.data_ptr<uint8_t>(), run_state.data_ptr<uint8_t>(),
+        n_steps.data_ptr<uint8_t>(), step_state.data_ptr<uint8_t>(),
+        step_attempts.data_ptr<int>(), children_todo.data_ptr<int>(),
+        children_out.data_ptr<int>(), children_done.data_ptr<int>(),
+        children_fail.data_ptr<int>(), children_emitted.data_ptr<int>(),
+        next_ready.data_ptr<int>(), dispatch_tick.data_ptr<int>(),
+        todo_tmpl.data_ptr<int>(), nready_tmpl.data_ptr<int>(),
+        (int)filter_state,
+        (unsigned long long*)admit_count.data_ptr<int64_t>(), NR);
 }
 
 void wf_apply(torch::Tensor send_slots, torch::Tensor send_cnt, torch::Tensor child_tag,
               torch::Tensor child_seq, torch::Tensor rq_prev_tag, torch::Tensor rq_prev_seq,
               torch::Tensor children_done,
               torch::Tensor children_fail, torch::Tensor children_out,
-              int64_t fail_ppt, int64_t cap, int64_t world)
+              int64_t fail_ppt, int64_t drop_ppt, int64_t cap, int64_t world)
 {
     const int n = (int)(world * cap);
     const int blocks = (n + BLOCK - 1) / BLOCK;
@@ -2033,7 +2153,7 @@ void wf_apply(torch::Tensor send_slots, torch::Tensor send_cnt, torch::Tensor ch
         child_seq.data_ptr<int>(), rq_prev_tag.data_ptr<int>(), rq_prev_seq.data_ptr<int>(),
         children_done.data_ptr<int>(),
         children_fail.data_ptr<int>(), children_out.data_ptr<int>(),
-        (int)fail_ppt, (int)cap, (int)world);
+        (int)fail_ppt, (int)drop_ppt, (int)cap, (int)world);
 }
 
 void wf_apply_dead(torch::Tensor dead_src, torch::Tensor dead_count, torch::Tensor child_tag,
@@ -2051,7 +2171,8 @@ void wf_apply_dead(torch::Tensor dead_src, torch::Tensor dead_count, torch::Tens
 void wf_commit(torch::Tensor step_state, torch::Tensor step_attempts,
                torch::Tensor children_todo, torch::Tensor children_out,
                torch::Tensor children_done, torch::Tensor children_fail,
-               torch::Tensor next_ready, int64_t tick, int64_t max_retries)
+               torch::Tensor next_ready, int64_t tick, int64_t max_retries,
+               torch::Tensor retry_count)
 {
     const int NRS = (int)step_state.numel();
     const int blocks = (NRS + BLOCK - 1) / BLOCK;
@@ -2059,7 +2180,8 @@ void wf_commit(torch::Tensor step_state, torch::Tensor step_attempts,
         step_state.data_ptr<uint8_t>(), step_attempts.data_ptr<int>(),
         children_todo.data_ptr<int>(), children_out.data_ptr<int>(),
         children_done.data_ptr<int>(), children_fail.data_ptr<int>(),
-        next_ready.data_ptr<int>(), (int)tick, (int)max_retries, NRS);
+        next_ready.data_ptr<int>(), (int)tick, (int)max_retries,
+        (unsigned long long*)retry_count.data_ptr<int64_t>(), NRS);
 }
 
 void wf_status(torch::Tensor step_state, torch::Tensor n_steps, torch::Tensor run_active,
@@ -2182,6 +2304,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("wf_apply", &wf_apply, "K3-WF child result application with failure injection");
     m.def("wf_apply_dead", &wf_apply_dead, "K3-WF dead-letter child application");
     m.def("wf_commit", &wf_commit, "K3-WF step aggregation + retry/backoff commit");
+    m.def("wf_timeout_scan", &wf_timeout_scan, "K4-WF stale-step timeout scan");
+    m.def("wf_readmit", &wf_readmit, "K3-WF continuous run re-admission from template");
     m.def("wf_status", &wf_status, "K3-WF run status roll-up");
     m.def("wf_grant", &wf_grant, "K3-WF host approval grants");
     m.def("materialize_rq_payload", &materialize_rq_payload, "copy requeued payload rows into the rq arena");

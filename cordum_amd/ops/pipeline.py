@@ -250,23 +250,26 @@ class _RefOps:
     def wf_expand(self, *a):
         from .reference import wf_expand_ref
 
-        wf_expand_ref(*a)
+        args = list(a)
+        args[12] = int(args[12])  # tick
+        wf_expand_ref(*args)
 
-    def wf_apply(self, ss, sc, ct, cs, rpt, rps, cd, cf, co, fail_ppt, cap, world):
+    def wf_apply(self, ss, sc, ct, cs, rpt, rps, cd, cf, co, fail_ppt,
+                 drop_ppt, cap, world):
         from .reference import wf_apply_ref
 
         wf_apply_ref(ss, sc, ct, cs, rpt, rps, cd, cf, co, int(fail_ppt),
-                     int(cap), int(world))
+                     int(drop_ppt), int(cap), int(world))
 
     def wf_apply_dead(self, *a):
         from .reference import wf_apply_dead_ref
 
         wf_apply_dead_ref(*a)
 
-    def wf_commit(self, st, sa, ctd, co, cd, cf, nr, tick, max_retries):
+    def wf_commit(self, st, sa, ctd, co, cd, cf, nr, tick, max_retries, rc):
         from .reference import wf_commit_ref
 
-        wf_commit_ref(st, sa, ctd, co, cd, cf, nr, int(tick), int(max_retries))
+        wf_commit_ref(st, sa, ctd, co, cd, cf, nr, int(tick), int(max_retries), rc)
 
     def wf_status(self, *a):
         from .reference import wf_status_ref
@@ -277,6 +280,18 @@ class _RefOps:
         from .reference import wf_grant_ref
 
         wf_grant_ref(gr, gs, v, int(n), st)
+
+    def wf_timeout_scan(self, st, co, cf, dt, tick, cutoff, tc):
+        from .reference import wf_timeout_scan_ref
+
+        wf_timeout_scan_ref(st, co, cf, dt, int(tick), int(cutoff), tc)
+
+    def wf_readmit(self, *a):
+        from .reference import wf_readmit_ref
+
+        args = list(a)
+        args[14] = int(args[14])  # filter_state
+        wf_readmit_ref(*args)
 
     def materialize_rq_payload(self, payload, prev_payload, rq_src, rq_count,
                                rq_payload, stride):

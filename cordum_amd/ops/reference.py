@@ -376,7 +376,7 @@ def wf_sweep_ref(step_state, deps_mask, n_steps, run_active, step_kind,
 def wf_expand_ref(disp_runs, disp_steps, disp_count, step_state,
                   children_todo, children_out, child_tag, child_seq,
                   child_widx, child_count, children_emitted,
-                  order, valid_count):
+                  dispatch_tick, tick, order, valid_count):
     CB = int(child_tag.shape[0])
     n = min(int(disp_count[0]), int(disp_runs.shape[0]))
     V = min(max(1, int(valid_count[0])), 1024)
@@ -395,6 +395,7 @@ def wf_expand_ref(disp_runs, disp_steps, disp_count, step_state,
         children_out[rs] += todo
         children_emitted[rs] += todo
         children_todo[rs] = 0
+        dispatch_tick[rs] = tick
         tag = run * 64 + step
         for k in range(todo):
             child_tag[base + k] = tag
@@ -416,15 +417,21 @@ def wf_fail_hash(tag: int, seq: int) -> int:
     return _wf_mix((tag * 2654435761 ^ seq * 40503) & 0xFFFFFFFF)
 
 
+def wf_drop_hash(tag: int, seq: int) -> int:
+    return _wf_mix((tag * 0x85EBCA6B ^ seq * 0xC2B2AE35) & 0xFFFFFFFF)
+
+
 def wf_apply_ref(send_slots, send_cnt, child_tag, child_seq, rq_prev_tag,
                  rq_prev_seq, children_done, children_fail, children_out,
-                 fail_ppt: int, cap: int, world: int):
+                 fail_ppt: int, drop_ppt: int, cap: int, world: int):
     for r in range(world):
         for e in range(min(int(send_cnt[r]), cap)):
             i = r * cap + e
             s = int(send_slots[i])
             tag = int(child_tag[s]) if s >= 0 else int(rq_prev_tag[-1 - s])
             seq = int(child_seq[s]) if s >= 0 else int(rq_prev_seq[-1 - s])
+            if wf_drop_hash(tag, seq) % 1000 < drop_ppt:
+                continue  # lost result: K4-WF timeout scan recovers it
             if wf_fail_hash(tag, seq) % 1000 < fail_ppt:
                 children_fail[tag] += 1
             else:
@@ -444,7 +451,7 @@ def wf_apply_dead_ref(dead_src, dead_count, child_tag, rq_prev_tag,
 
 def wf_commit_ref(step_state, step_attempts, children_todo, children_out,
                   children_done, children_fail, next_ready,
-                  tick: int, max_retries: int):
+                  tick: int, max_retries: int, retry_count=None):
     for i in range(int(step_state.numel())):
         if int(step_state[i]) != WFS_DISPATCHED or int(children_out[i]) != 0:
             continue
@@ -456,6 +463,8 @@ def wf_commit_ref(step_state, step_attempts, children_todo, children_out,
                 children_fail[i] = 0
                 next_ready[i] = tick + min(1 << int(step_attempts[i]), 16)
                 step_state[i] = WFS_PENDING
+                if retry_count is not None:
+                    retry_count[0] += fail
             else:
                 step_state[i] = WFS_FAILED
         elif int(children_todo[i]) > 0:
@@ -486,3 +495,40 @@ def wf_grant_ref(grant_runs, grant_steps, verdicts, n: int, step_state):
         rs = int(grant_runs[i]) * 64 + int(grant_steps[i])
         if int(step_state[rs]) == WFS_WAITING:
             step_state[rs] = WFS_SUCCEEDED if int(verdicts[i]) else WFS_FAILED
+
+
+def wf_timeout_scan_ref(step_state, children_out, children_fail,
+                        dispatch_tick, tick: int, cutoff: int, timeout_count):
+    for i in range(int(step_state.numel())):
+        if int(step_state[i]) != WFS_DISPATCHED:
+            continue
+        lost = int(children_out[i])
+        if lost <= 0 or int(dispatch_tick[i]) > tick - cutoff:
+            continue
+        children_fail[i] += lost
+        children_out[i] = 0
+        timeout_count[0] += lost
+
+
+def wf_readmit_ref(run_active, run_state, n_steps, step_state, step_attempts,
+                   children_todo, children_out, children_done, children_fail,
+                   children_emitted, next_ready, dispatch_tick,
+                   todo_tmpl, nready_tmpl, filter_state: int, admit_count):
+    NR = int(n_steps.shape[0])
+    for run in range(NR):
+        if int(run_active[run]) or int(run_state[run]) != filter_state:
+            continue
+        for s in range(int(n_steps[run])):
+            i = run * 64 + s
+            step_state[i] = WFS_PENDING
+            step_attempts[i] = 0
+            children_todo[i] = int(todo_tmpl[i])
+            children_out[i] = 0
+            children_done[i] = 0
+            children_fail[i] = 0
+            children_emitted[i] = 0
+            next_ready[i] = int(nready_tmpl[i])
+            dispatch_tick[i] = 0
+        run_state[run] = 0
+        run_active[run] = 1
+        admit_count[0] += 1

@@ -95,10 +95,16 @@ class WorkflowPipeline:
         rank: int = 0,
         backend: str = "ext",
         fail_ppt: int = 0,
+        drop_ppt: int = 0,
         max_retries: int = 2,
+        # must exceed the max requeue-ring lifetime (RQ_MAX_DELIVER + 1
+        # ticks) so a parked-but-live child is never declared lost — a
+        # timeout firing on a child that later applies would double-account
+        timeout_cutoff: int = 8,
         approval_verdict: int = 1,
         child_cap: Optional[int] = None,
         pad_cap: Optional[int] = None,
+        replicate: Optional[int] = None,
         seed: int = 7,
     ):
         device = torch.device(device)
@@ -109,20 +115,25 @@ class WorkflowPipeline:
         self.NWL = n_local_workers
         self.W = payload_words
         self.fail_ppt = fail_ppt
+        self.drop_ppt = drop_ppt
         self.max_retries = max_retries
+        self.timeout_cutoff = timeout_cutoff
         self.approval_verdict = approval_verdict
-        NR = len(dags)
+        if replicate is not None:
+            assert len(dags) == 1
+        NR = replicate if replicate is not None else len(dags)
         self.NR = NR
         assert all(len(d.steps) <= 64 for d in dags), "device DAGs cap at 64 steps"
 
         # ---- host-side template tables (the run-creation image) -------------
-        st = torch.zeros(NR * 64, dtype=torch.uint8)
-        deps = torch.zeros(NR * 64, dtype=torch.int64)
-        kinds = torch.zeros(NR * 64, dtype=torch.uint8)
-        todo = torch.zeros(NR * 64, dtype=torch.int32)
-        nready = torch.zeros(NR * 64, dtype=torch.int32)
-        nsteps = torch.zeros(NR, dtype=torch.uint8)
-        cond = torch.zeros(NR, dtype=torch.int64)
+        TN = 1 if replicate is not None else NR
+        st = torch.zeros(TN * 64, dtype=torch.uint8)
+        deps = torch.zeros(TN * 64, dtype=torch.int64)
+        kinds = torch.zeros(TN * 64, dtype=torch.uint8)
+        todo = torch.zeros(TN * 64, dtype=torch.int32)
+        nready = torch.zeros(TN * 64, dtype=torch.int32)
+        nsteps = torch.zeros(TN, dtype=torch.uint8)
+        cond = torch.zeros(TN, dtype=torch.int64)
         total_children = 0
         for r, dag in enumerate(dags):
             nsteps[r] = len(dag.steps)
@@ -145,6 +156,15 @@ class WorkflowPipeline:
                 if spec.cond:
                     cbits |= 1 << s
             cond[r] = cbits - (1 << 64) if cbits >= (1 << 63) else cbits
+        if replicate is not None:
+            st = st.repeat(NR)
+            deps = deps.repeat(NR)
+            kinds = kinds.repeat(NR)
+            todo = todo.repeat(NR)
+            nready = nready.repeat(NR)
+            nsteps = nsteps.repeat(NR)
+            cond = cond.repeat(NR)
+            total_children *= NR
         self.total_children = total_children
         self._tmpl = {
             "step_state": st, "deps_mask": deps, "step_kind": kinds,
@@ -168,9 +188,14 @@ class WorkflowPipeline:
         self.children_done = torch.zeros(NR * 64, dtype=torch.int32, device=d)
         self.children_fail = torch.zeros(NR * 64, dtype=torch.int32, device=d)
         self.wf_counts = torch.zeros(2, dtype=torch.int64, device=d)
+        self.dispatch_tick = torch.zeros(NR * 64, dtype=torch.int32, device=d)
+        self.timeout_count = torch.zeros(1, dtype=torch.int64, device=d)
+        self.retry_count = torch.zeros(1, dtype=torch.int64, device=d)
+        self.admit_count = torch.zeros(1, dtype=torch.int64, device=d)
 
-        # dispatch + approval lists
-        self.disp_cap = NR * 64
+        # dispatch + approval lists (overflowing entries simply retry on the
+        # next sweep, so the cap bounds memory, not correctness)
+        self.disp_cap = min(NR * 64, 1 << 22)
         self.disp_runs = torch.zeros(self.disp_cap, dtype=torch.int32, device=d)
         self.disp_steps = torch.zeros(self.disp_cap, dtype=torch.int32, device=d)
         self.disp_count = torch.zeros(1, dtype=torch.int32, device=d)
@@ -275,6 +300,7 @@ class WorkflowPipeline:
         self.rq_count.zero_()
         self.rq_prev_count.zero_()
         self.dead_count.zero_()
+        self.dispatch_tick.zero_()
         self._pending_grants.clear()
         self._tick = 0  # delay gates + backoff are wave-relative ticks
 
@@ -346,6 +372,7 @@ class WorkflowPipeline:
                       self.step_state, self.children_todo, self.children_out,
                       self.child_tag, self.child_seq, self.child_widx,
                       self.child_count, self.children_emitted,
+                      self.dispatch_tick, tick,
                       self.order_buf, self.valid_buf)
 
         # K2 load view + routing order refresh
@@ -403,12 +430,17 @@ class WorkflowPipeline:
         ext.wf_apply(self.pad_send_slots, self.pad_send_cnt, self.child_tag,
                      self.child_seq, self.rq_prev_tag, self.rq_prev_seq,
                      self.children_done, self.children_fail,
-                     self.children_out, self.fail_ppt, cap, world)
+                     self.children_out, self.fail_ppt, self.drop_ppt, cap, world)
         ext.wf_apply_dead(self.dead_src, self.dead_count, self.child_tag,
                           self.rq_prev_tag, self.children_fail, self.children_out)
+        # K4-WF: steps whose children were lost (worker crash) time out and
+        # the remainder retries with backoff (reconciler.go:88-144)
+        ext.wf_timeout_scan(self.step_state, self.children_out,
+                            self.children_fail, self.dispatch_tick,
+                            tick, self.timeout_cutoff, self.timeout_count)
         ext.wf_commit(self.step_state, self.step_attempts, self.children_todo,
                       self.children_out, self.children_done, self.children_fail,
-                      self.next_ready, tick, self.max_retries)
+                      self.next_ready, tick, self.max_retries, self.retry_count)
         ext.wf_status(self.step_state, self.n_steps, self.run_active,
                       self.run_state, self.wf_counts)
         self._refresh_order()
@@ -425,6 +457,54 @@ class WorkflowPipeline:
         if not hasattr(self, "_iota"):
             self._iota = torch.arange(self.CB, dtype=torch.int32, device=self.device)
         return self._iota
+
+    # ---- continuous mode (config #5) ----------------------------------------
+    def _ensure_tmpl_dev(self):
+        if not hasattr(self, "_tmpl_dev"):
+            self._tmpl_dev = {k: v.to(self.device) for k, v in self._tmpl.items()}
+        return self._tmpl_dev
+
+    def readmit_succeeded(self) -> None:
+        """Terminal SUCCEEDED runs re-enter from the creation template so the
+        table stays full (config #5 'hold 1M concurrent runs')."""
+        t = self._ensure_tmpl_dev()
+        self.ext.wf_readmit(self.run_active, self.run_state, self.n_steps,
+                            self.step_state, self.step_attempts,
+                            self.children_todo, self.children_out,
+                            self.children_done, self.children_fail,
+                            self.children_emitted, self.next_ready,
+                            self.dispatch_tick, t["children_todo"],
+                            t["next_ready"], WFS_SUCCEEDED, self.admit_count)
+
+    def drain_failed_to_dlq(self, dlq=None) -> int:
+        """Host DLQ drain (dlq_store.go analog): record every FAILED run,
+        then re-admit those rows through the same template reset. Returns
+        the number drained; the caller's DLQ bookkeeping must reconcile
+        exactly with the device failed counter at the end of a soak."""
+        failed = (self.run_state == WFS_FAILED) & (self.run_active == 0)
+        ids = torch.nonzero(failed).flatten()
+        n = int(ids.numel())
+        if n and dlq is not None:
+            from ..store.dlq_store import DLQEntry
+
+            gen = int(self.admit_count.cpu()[0])
+            for r in ids.cpu().tolist():
+                dlq.add(DLQEntry(job_id=f"run-{r}-g{gen}", topic="job.soak",
+                                 status="JOB_STATUS_FAILED",
+                                 reason="max retries exceeded (soak)",
+                                 reason_code="max_retries_exceeded",
+                                 last_state="FAILED",
+                                 attempts=self.max_retries))
+        if n:
+            t = self._ensure_tmpl_dev()
+            self.ext.wf_readmit(self.run_active, self.run_state, self.n_steps,
+                                self.step_state, self.step_attempts,
+                                self.children_todo, self.children_out,
+                                self.children_done, self.children_fail,
+                                self.children_emitted, self.next_ready,
+                                self.dispatch_tick, t["children_todo"],
+                                t["next_ready"], WFS_FAILED, self.admit_count)
+        return n
 
     # ---- wave driver ---------------------------------------------------------
     def counts(self) -> tuple:

@@ -261,3 +261,39 @@ def test_two_rank_workflow_fanout(tmp_path):
         out = json.load(open(tmp_path / f"rank{r}.json"))
         assert out["ok"] == 4 and out["fail"] == 0
         assert out["remote_work"] > 0
+
+
+def test_continuous_soak_reconciles():
+    """Config #5 mechanics in miniature: continuous re-admission holds the
+    table full; injected failures retry with backoff; injected LOST results
+    are recovered by the K4-WF timeout scan; failed runs drain to the host
+    DLQ; device counters reconcile exactly against host bookkeeping."""
+    from cordum_amd.store.dlq_store import DLQStore
+
+    NR = 32
+    dag = DagSpec(steps=[
+        StepSpec(WFK_WORKER),
+        StepSpec(WFK_FOR_EACH, deps=[0], fanout=4),
+        StepSpec(WFK_WORKER, deps=[1]),
+    ])
+    pipe = mk_pipe([dag], replicate=NR, fail_ppt=250, drop_ppt=50,
+                   max_retries=1)
+    dlq = DLQStore()
+    pipe.reset_runs()
+    drained = 0
+    for t in range(400):
+        pipe.tick()
+        pipe.readmit_succeeded()
+        if t % 25 == 24:
+            drained += pipe.drain_failed_to_dlq(dlq)
+    drained += pipe.drain_failed_to_dlq(dlq)
+    ok, fail = pipe.counts()
+    active = pipe.active()
+    admissions = NR + int(pipe.admit_count[0])
+    assert ok > NR  # table cycled many times
+    assert fail > 0  # max-retries exhaustion happened
+    assert fail == drained  # every failure reached the host DLQ
+    assert admissions == ok + fail + active  # nothing lost, nothing invented
+    assert int(pipe.timeout_count[0]) > 0  # lost children were recovered
+    assert int(pipe.retry_count[0]) > 0  # retry waves happened
+    assert active == NR  # the table is still full (held concurrency)
