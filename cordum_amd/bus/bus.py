@@ -18,6 +18,7 @@ from __future__ import annotations
 import heapq
 import itertools
 import threading
+from functools import lru_cache
 from dataclasses import dataclass, field
 from typing import Callable, Dict, List, Optional, Tuple
 
@@ -41,12 +42,17 @@ class RetryAfter(Exception):
         self.cause = cause
 
 
+@lru_cache(maxsize=8192)
+def _tokens(s: str):
+    return s.split(".")
+
+
 def subject_matches(pattern: str, subject: str) -> bool:
     """NATS subject matching: tokens split on '.', '*' = one token, '>' = tail."""
     if pattern == subject:
         return True
-    pt = pattern.split(".")
-    st = subject.split(".")
+    pt = _tokens(pattern)
+    st = _tokens(subject)
     for i, p in enumerate(pt):
         if p == ">":
             return len(st) > i  # NATS '>': one or MORE remaining tokens

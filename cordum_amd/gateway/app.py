@@ -124,7 +124,41 @@ def create_app(
             raise HTTPException(400, "invalid json")
         if not isinstance(body, dict):
             raise HTTPException(400, "body must be a json object")
+        out = _submit_one(body, p)
+        node.drain()
+        return out
 
+    @api.post("/jobs/batch")
+    async def submit_jobs_batch(request: Request, p: Principal = Depends(principal)):
+        """Batched ingest (MI355X-native addition): one HTTP request admits
+        up to 4096 jobs; per-job semantics are identical to POST /jobs (same
+        validation, idempotency, secrets scan, persistence) and the whole
+        batch is flushed through the dispatch engine once — this is the
+        HTTP-side amortization that feeds the batched device tick."""
+        raw = await request.body()
+        if len(raw) > 16 * MAX_BODY_BYTES:
+            raise HTTPException(413, "body too large")
+        try:
+            body = json.loads(raw or b"{}")
+        except ValueError:
+            raise HTTPException(400, "invalid json")
+        jobs = body.get("jobs") if isinstance(body, dict) else None
+        if not isinstance(jobs, list) or not jobs:
+            raise HTTPException(400, "jobs array required")
+        if len(jobs) > 4096:
+            raise HTTPException(400, "too many jobs (max 4096)")
+        items = []
+        for jb in jobs:
+            if not isinstance(jb, dict):
+                raise HTTPException(400, "every job must be a json object")
+            try:
+                items.append(_submit_one(jb, p))
+            except HTTPException as e:
+                items.append({"error": e.detail, "status": e.status_code})
+        node.drain()
+        return {"items": items}
+
+    def _submit_one(body: Dict[str, Any], p: Principal) -> Dict[str, Any]:
         prompt = body.get("prompt") or ""
         if not prompt:
             raise HTTPException(400, "prompt is required")
@@ -210,7 +244,6 @@ def create_app(
             requires=list(req.meta.requires), principal=req.principal_id,
         )
         node.submit_job(req, trace_id=trace_id, context=json.dumps(context_payload).encode())
-        node.drain()
         return {"job_id": job_id, "trace_id": trace_id}
 
     @api.get("/jobs")
