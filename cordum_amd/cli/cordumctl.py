@@ -73,7 +73,9 @@ def cmd_init(args):
 
 def cmd_workflow(args):
     c = make_client(args)
-    if args.action == "create":
+    if args.action == "scaffold":
+        cmd_pack_scaffold(Path(args.dir), args.id or "my-pack", force=args.force)
+    elif args.action == "create":
         import yaml
 
         path = Path(args.file or args.id)
@@ -149,9 +151,92 @@ def build_pack_archive(directory: Path) -> bytes:
     return buf.getvalue()
 
 
+
+PACK_SCAFFOLD_MANIFEST = """apiVersion: cordum.io/v1alpha1
+kind: Pack
+
+metadata:
+  id: {pid}
+  version: 0.1.0
+  title: {pid} pack
+  description: {pid} workflow pack (scaffolded).
+
+compatibility:
+  protocolVersion: 1
+
+topics:
+  - topic: job.{pid}.echo
+    capability: {pid}.echo
+
+resources:
+  schemas:
+    - schemas/echo-input.json
+  workflows:
+    - workflows/echo.json
+
+overlays:
+  config:
+    - scope: system
+      key: default
+      json_merge_patch:
+        pools:
+          topics:
+            job.{pid}.echo: [default]
+          pools:
+            default: {{}}
+  policy:
+    - id: {pid}/policy
+      bundle_fragment: |
+        version: {pid}-v1
+        rules:
+          - id: {pid}-allow-echo
+            decision: allow
+            match:
+              topics: ["job.{pid}.echo"]
+"""
+
+
+def cmd_pack_scaffold(root: Path, pack_id: str, force: bool = False) -> None:
+    """`cordumctl pack scaffold` — generate a working pack skeleton
+    (reference: cmd/cordumctl/pack_create.go scaffoldPack templates)."""
+
+    def write(rel: str, content: str):
+        path = root / rel
+        if path.exists() and not force:
+            raise SystemExit(f"file exists: {path} (use --force)")
+        path.parent.mkdir(parents=True, exist_ok=True)
+        path.write_text(content)
+
+    write("pack.yaml", PACK_SCAFFOLD_MANIFEST.format(pid=pack_id))
+    write("schemas/echo-input.json", json.dumps({
+        "$id": f"{pack_id}/echo-input",
+        "type": "object",
+        "properties": {"message": {"type": "string"},
+                       "author": {"type": "string"}},
+        "required": ["message"],
+        "additionalProperties": False,
+    }, indent=2))
+    write("workflows/echo.json", json.dumps({
+        "id": f"{pack_id}.echo",
+        "name": f"{pack_id} echo",
+        "org_id": "default",
+        "steps": {"echo": {"type": "worker", "topic": f"job.{pack_id}.echo",
+                           "input_schema_id": f"{pack_id}/echo-input",
+                           "input": {"message": "${input.message}",
+                                     "author": "${input.author}"},
+                           "meta": {"pack_id": pack_id,
+                                    "capability": f"{pack_id}.echo"}}},
+    }, indent=2))
+    write("README.md", f"# {pack_id}\n\nScaffolded pack. Install with:\n\n"
+                       f"    cordumctl pack install --file .\n")
+    print(f"scaffolded pack {pack_id!r} in {root}")
+
+
 def cmd_pack(args):
     c = make_client(args)
-    if args.action == "create":
+    if args.action == "scaffold":
+        cmd_pack_scaffold(Path(args.dir), args.id or "my-pack", force=args.force)
+    elif args.action == "create":
         blob = build_pack_archive(Path(args.dir))
         Path(args.output).write_bytes(blob)
         print(f"wrote {args.output} ({len(blob)} bytes)")
@@ -168,6 +253,49 @@ def cmd_pack(args):
         out(c.uninstall_pack(args.id))
     elif args.action == "verify":
         out(c.verify_pack(args.id))
+
+
+
+def cmd_dev(args):
+    """`cordumctl dev` — local dev stack: the single-process node with echo
+    workers and the hello-pack example installed (the reference's dev mode
+    brings up the compose stack; one process replaces it here)."""
+    import threading
+    import time as _time
+
+    from ..cli.serve import serve
+
+    def install_example():
+        _time.sleep(1.5)
+        try:
+            repo = Path(__file__).resolve().parent.parent.parent
+            pack_dir = repo / "examples" / "hello-pack"
+            if pack_dir.is_dir():
+                blob = build_pack_archive(pack_dir)
+                from ..sdk.client import Client
+
+                Client(base_url=f"http://127.0.0.1:{args.port}",
+                       role="admin", principal_id="dev").install_pack(blob)
+                print("hello-pack installed")
+        except Exception as e:
+            print(f"(dev) example pack install skipped: {e}", file=sys.stderr)
+
+    threading.Thread(target=install_example, daemon=True).start()
+    print("Cordum stack started (dev mode).")
+    print(f"Gateway: http://127.0.0.1:{args.port}")
+    serve(port=args.port, workers=2)
+
+
+def cmd_up(args):
+    """`cordumctl up` — the production-shaped single-node launch: durable
+    WAL/checkpoints under --state-dir, echo workers on configured pools."""
+    from ..cli.serve import serve
+
+    Path(args.state_dir).mkdir(parents=True, exist_ok=True)
+    print("Cordum stack started.")
+    print(f"Gateway: http://127.0.0.1:{args.port}")
+    print(f"State: {args.state_dir}")
+    serve(port=args.port, workers=2, checkpoint_dir=args.state_dir)
 
 
 def main(argv=None) -> int:
@@ -226,12 +354,23 @@ def main(argv=None) -> int:
     p.set_defaults(fn=cmd_job)
 
     p = sub.add_parser("pack")
-    p.add_argument("action", choices=["create", "install", "uninstall", "list", "show", "verify"])
+    p.add_argument("action", choices=["scaffold", "create", "install", "uninstall", "list", "show", "verify"])
     p.add_argument("id", nargs="?")
     p.add_argument("--dir", default=".")
     p.add_argument("--file", "-f")
     p.add_argument("--output", "-o", default="pack.tgz")
+    p.add_argument("--force", action="store_true")
     p.set_defaults(fn=cmd_pack)
+
+    p = sub.add_parser("dev", help="run a local dev stack (serve + example pack)")
+    p.add_argument("--port", type=int, default=8081)
+    p.set_defaults(fn=cmd_dev)
+
+    p = sub.add_parser("up", help="run the production-shaped single-node stack "
+                                  "(serve + durable checkpoints)")
+    p.add_argument("--port", type=int, default=8081)
+    p.add_argument("--state-dir", default=str(Path.home() / ".cordum" / "state"))
+    p.set_defaults(fn=cmd_up)
 
     args = ap.parse_args(argv)
     try:

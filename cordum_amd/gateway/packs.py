@@ -82,13 +82,32 @@ def load_manifest(files: Dict[str, bytes]) -> Dict[str, Any]:
     if manifest.get("apiVersion") != API_VERSION:
         raise PackError(f"unsupported apiVersion {manifest.get('apiVersion')!r}")
     meta = manifest.get("metadata") or {}
-    if not meta.get("name"):
-        raise PackError("metadata.name required")
+    # the reference manifest uses metadata.id (packs.go:140-146,
+    # examples/hello-pack/pack.yaml); accept `name` as an alias
+    pack_id = meta.get("id") or meta.get("name")
+    if not pack_id:
+        raise PackError("metadata.id required")
+    meta["name"] = pack_id
+    meta["id"] = pack_id
     compat = manifest.get("compatibility") or {}
     pv = compat.get("protocolVersion")
     if pv is not None and int(pv) != 1:
         raise PackError(f"incompatible protocolVersion {pv}")
     return manifest
+
+
+def _resource_paths(entries) -> list:
+    """resources.{schemas,workflows} entries are plain paths OR the
+    reference's {id, path} objects (packs.go packResource)."""
+    out = []
+    for e in entries or []:
+        if isinstance(e, str):
+            out.append((None, e))
+        elif isinstance(e, dict) and e.get("path"):
+            out.append((e.get("id") or None, e["path"]))
+        else:
+            raise PackError(f"invalid resource entry: {e!r}")
+    return out
 
 
 def _digest(data: bytes) -> str:
@@ -115,12 +134,12 @@ class PackInstaller:
         try:
             resources = manifest.get("resources") or {}
             # schemas
-            for path in resources.get("schemas") or []:
+            for rid, path in _resource_paths(resources.get("schemas")):
                 raw = files.get(path)
                 if raw is None:
                     raise PackError(f"schema file missing: {path}")
                 doc = _parse_doc(raw, path)
-                sid = doc.get("$id") or doc.get("id") or path.rsplit("/", 1)[-1].rsplit(".", 1)[0]
+                sid = rid or doc.get("$id") or doc.get("id") or path.rsplit("/", 1)[-1].rsplit(".", 1)[0]
                 prev = self.node.schemas.get(sid)
                 if prev is not None and _digest(json.dumps(prev, sort_keys=True).encode()) == _digest(
                     json.dumps(doc, sort_keys=True).encode()
@@ -135,7 +154,7 @@ class PackInstaller:
             # workflows
             from ..workflow import Workflow
 
-            for path in resources.get("workflows") or []:
+            for _wf_rid, path in _resource_paths(resources.get("workflows")):
                 raw = files.get(path)
                 if raw is None:
                     raise PackError(f"workflow file missing: {path}")
@@ -158,7 +177,15 @@ class PackInstaller:
             for ov in overlays.get("config") or []:
                 scope = ov.get("scope", "system")
                 key = ov.get("key", "default")
-                patch = ov.get("json_merge_patch") or {}
+                patch = ov.get("json_merge_patch")
+                if patch is None and ov.get("path"):
+                    # reference overlay form: strategy + patch content in a
+                    # bundled file (packs.go packConfigOverlay)
+                    raw = files.get(ov["path"])
+                    if raw is None:
+                        raise PackError(f"overlay file missing: {ov['path']}")
+                    patch = _parse_doc(raw, ov["path"])
+                patch = patch or {}
                 prev = self.node.configsvc.get(scope, key)
                 self.node.configsvc.patch(scope, key, patch)
                 plan["config_overlays"] += 1
@@ -166,7 +193,14 @@ class PackInstaller:
                                 self.node.configsvc.set(scope, key, prev or {}))
             for i, ov in enumerate(overlays.get("policy") or []):
                 frag = ov.get("bundle_fragment", "")
-                bundle_id = ov.get("id") or f"{pack_id}/{i}"
+                if not frag and ov.get("path"):
+                    raw = files.get(ov["path"])
+                    if raw is None:
+                        raise PackError(f"overlay file missing: {ov['path']}")
+                    frag = raw.decode("utf-8")
+                bundle_id = ov.get("id") or ov.get("name") or f"{pack_id}/{i}"
+                if "/" not in bundle_id:
+                    bundle_id = f"{pack_id}/{bundle_id}"
                 prev_doc = self.node.configsvc.get(*POLICY_DOC) or {}
                 self.node.configsvc.patch(*POLICY_DOC, {
                     "bundles": {bundle_id: {"enabled": True, "content": frag, "pack_id": pack_id}},

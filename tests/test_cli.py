@@ -117,3 +117,119 @@ def test_sdk_policy_evaluate(sdk):
 def test_sdk_status_and_workers(sdk):
     assert sdk.status().get("status") in ("ok", "degraded")
     assert "workers" in sdk.workers() or "items" in sdk.workers()
+
+
+def test_pack_scaffold_installs(tmp_path):
+    """`cordumctl pack scaffold` output must install cleanly end-to-end."""
+    from fastapi.testclient import TestClient
+
+    from cordum_amd.cli.cordumctl import build_pack_archive, cmd_pack_scaffold
+    from cordum_amd.gateway import create_app
+    from cordum_amd.runtime.node import Node
+
+    cmd_pack_scaffold(tmp_path / "p", "scafpack")
+    node = Node().start()
+    client = TestClient(create_app(node))
+    blob = build_pack_archive(tmp_path / "p")
+    r = client.post("/api/v1/packs/install", content=blob,
+                    headers={"X-Principal-Id": "p", "X-Principal-Role": "admin"})
+    assert r.status_code == 200, r.text
+    assert r.json()["status"] == "ACTIVE"
+    r = client.get("/api/v1/workflows/scafpack.echo", headers={"X-Principal-Id": "p"})
+    assert r.status_code == 200
+
+
+def test_reference_format_pack_installs(tmp_path):
+    """The REFERENCE's manifest dialect (metadata.id, topics[].name,
+    {id,path} resources, file-based overlays with strategy keys — the exact
+    shape of examples/hello-pack/pack.yaml + pack_create.go templates) must
+    install unmodified."""
+    import io
+    import tarfile
+
+    from fastapi.testclient import TestClient
+
+    from cordum_amd.gateway import create_app
+    from cordum_amd.runtime.node import Node
+
+    files = {
+        "pack.yaml": """apiVersion: cordum.io/v1alpha1
+kind: Pack
+metadata:
+  id: ref-pack
+  version: 0.1.0
+  title: Ref Pack
+  description: reference-dialect pack.
+compatibility:
+  protocolVersion: 1
+  minCoreVersion: 0.6.0
+topics:
+  - name: job.ref.echo
+    capability: ref.echo
+resources:
+  schemas:
+    - id: ref-pack/EchoInput
+      path: schemas/EchoInput.json
+  workflows:
+    - id: ref-pack.echo
+      path: workflows/echo.yaml
+overlays:
+  config:
+    - name: pools
+      scope: system
+      key: default
+      strategy: json_merge_patch
+      path: overlays/pools.patch.yaml
+  policy:
+    - name: safety
+      strategy: bundle_fragment
+      path: overlays/policy.fragment.yaml
+""",
+        "schemas/EchoInput.json": '{"type": "object", "required": ["message"]}',
+        "workflows/echo.yaml": """id: ref-pack.echo
+name: Ref Echo
+org_id: default
+steps:
+  echo:
+    type: worker
+    topic: job.ref.echo
+    input:
+      message: "${input.message}"
+""",
+        "overlays/pools.patch.yaml": """pools:
+  topics:
+    job.ref.echo: [default]
+  pools:
+    default: {}
+""",
+        "overlays/policy.fragment.yaml": """version: ref-v1
+rules:
+  - id: ref-allow
+    decision: allow
+    match:
+      topics: ["job.ref.echo"]
+""",
+    }
+    buf = io.BytesIO()
+    with tarfile.open(fileobj=buf, mode="w:gz") as tf:
+        for name, content in files.items():
+            info = tarfile.TarInfo(name)
+            data = content.encode()
+            info.size = len(data)
+            tf.addfile(info, io.BytesIO(data))
+    node = Node().start()
+    client = TestClient(create_app(node))
+    r = client.post("/api/v1/packs/install", content=buf.getvalue(),
+                    headers={"X-Principal-Id": "p", "X-Principal-Role": "admin"})
+    assert r.status_code == 200, r.text
+    plan = r.json()["plan"]
+    assert plan["schemas"][0]["id"] == "ref-pack/EchoInput"
+    assert plan["workflows"][0]["id"] == "ref-pack.echo"
+    assert plan["config_overlays"] == 1 and plan["policy_overlays"] == 1
+    # routing overlay landed: job.ref.echo routes to the default pool
+    r = client.get("/api/v1/workflows/ref-pack.echo", headers={"X-Principal-Id": "p"})
+    assert r.status_code == 200
+    # policy fragment landed under the pack-scoped bundle id
+    bundles = client.get("/api/v1/policy/bundles",
+                         headers={"X-Principal-Id": "p"}).json()["items"]
+    assert any(b["id"] == "ref-pack/safety" for b in bundles)
