@@ -46,7 +46,8 @@ def cmd_serve(args):
 
     serve(host=args.host, port=args.port, config_dir=args.config_dir,
           workers=args.workers, checkpoint_dir=args.checkpoint_dir,
-          checkpoint_interval_s=args.checkpoint_interval)
+          checkpoint_interval_s=args.checkpoint_interval,
+          dashboard_dir=args.dashboard_dir)
 
 
 def cmd_init(args):
@@ -307,6 +308,8 @@ def main(argv=None) -> int:
     sub.add_parser("status").set_defaults(fn=cmd_status)
 
     p = sub.add_parser("serve", help="run the single-process control-plane node")
+    p.add_argument("--dashboard-dir", default="",
+                   help="serve a dashboard build statically at /")
     p.add_argument("--host", default="127.0.0.1")
     p.add_argument("--port", type=int, default=8080)
     p.add_argument("--config-dir", default="")

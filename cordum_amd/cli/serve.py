@@ -39,7 +39,8 @@ def build_node(config_dir: str = "", checkpoint_dir: str = ""):
 
 
 def serve(host: str = "127.0.0.1", port: int = 8080, config_dir: str = "",
-          workers: int = 2, checkpoint_dir: str = "", checkpoint_interval_s: float = 30.0):
+          workers: int = 2, checkpoint_dir: str = "", checkpoint_interval_s: float = 30.0,
+          dashboard_dir: str = ""):
     import uvicorn
 
     from ..gateway import create_app
@@ -51,7 +52,8 @@ def serve(host: str = "127.0.0.1", port: int = 8080, config_dir: str = "",
                         topics=sorted(node.strategy.current_routing().topics))
 
     app = create_app(node, rate_limit_rps=cfg.api_rate_limit_rps,
-                     rate_limit_burst=cfg.api_rate_limit_burst)
+                     rate_limit_burst=cfg.api_rate_limit_burst,
+                     dashboard_dir=dashboard_dir)
 
     stop = threading.Event()
 

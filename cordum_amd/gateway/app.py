@@ -63,6 +63,7 @@ def create_app(
     auth: Optional[BasicAuthProvider] = None,
     rate_limit_rps: float = 0.0,
     rate_limit_burst: int = 100,
+    dashboard_dir: str = "",
 ) -> FastAPI:
     app = FastAPI(title="cordum-mi355x gateway", version="0.1.0")
     # CORS for the dashboard (gateway.go:2051-2143)
