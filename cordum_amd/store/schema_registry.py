@@ -6,8 +6,12 @@ payloads and for config-file validation.
 
 The validator is a self-contained JSON-Schema (draft-07 subset) implementation
 covering the keywords the reference's workflows/packs use: type, properties,
-required, items, enum, const, minimum/maximum, minLength/maxLength,
-pattern, additionalProperties, anyOf/allOf/oneOf, format (opaque).
+required, items (incl. tuple form + additionalItems), enum, const,
+minimum/maximum (+exclusive), multipleOf, minLength/maxLength, pattern,
+additionalProperties, min/maxProperties, min/maxItems, uniqueItems,
+anyOf/allOf/oneOf/not, local $ref (#/definitions, #/$defs), format (opaque).
+Conformance corpus: tests/test_schema_conformance.py (JSON-Schema-Test-Suite
+style triples).
 """
 from __future__ import annotations
 
@@ -42,10 +46,24 @@ def _type_ok(t: str, v: Any) -> bool:
     return True
 
 
-def validate_value(schema: Dict[str, Any], value: Any, path: str = "$") -> List[str]:
+def validate_value(schema: Dict[str, Any], value: Any, path: str = "$",
+                   root: Optional[Dict[str, Any]] = None) -> List[str]:
     errs: List[str] = []
     if not isinstance(schema, dict):
         return errs
+    if root is None:
+        root = schema
+    # local $ref resolution (#/definitions/... or #/$defs/...), draft-07
+    ref = schema.get("$ref")
+    if isinstance(ref, str) and ref.startswith("#/"):
+        target: Any = root
+        for part in ref[2:].split("/"):
+            part = part.replace("~1", "/").replace("~0", "~")
+            if isinstance(target, dict) and part in target:
+                target = target[part]
+            else:
+                return [f"{path}: unresolvable $ref {ref}"]
+        return validate_value(target, value, path, root)
     t = schema.get("type")
     if t is not None:
         types = t if isinstance(t, list) else [t]
@@ -61,6 +79,15 @@ def validate_value(schema: Dict[str, Any], value: Any, path: str = "$") -> List[
             errs.append(f"{path}: {value} < minimum {schema['minimum']}")
         if "maximum" in schema and value > schema["maximum"]:
             errs.append(f"{path}: {value} > maximum {schema['maximum']}")
+        if "exclusiveMinimum" in schema and value <= schema["exclusiveMinimum"]:
+            errs.append(f"{path}: {value} <= exclusiveMinimum {schema['exclusiveMinimum']}")
+        if "exclusiveMaximum" in schema and value >= schema["exclusiveMaximum"]:
+            errs.append(f"{path}: {value} >= exclusiveMaximum {schema['exclusiveMaximum']}")
+        if "multipleOf" in schema:
+            m = schema["multipleOf"]
+            q = value / m
+            if abs(q - round(q)) > 1e-9:
+                errs.append(f"{path}: {value} not a multiple of {m}")
     if isinstance(value, str):
         if "minLength" in schema and len(value) < schema["minLength"]:
             errs.append(f"{path}: shorter than minLength {schema['minLength']}")
@@ -77,23 +104,49 @@ def validate_value(schema: Dict[str, Any], value: Any, path: str = "$") -> List[
         for req in schema.get("required", []):
             if req not in value:
                 errs.append(f"{path}: missing required property {req!r}")
+        if "minProperties" in schema and len(value) < schema["minProperties"]:
+            errs.append(f"{path}: fewer than minProperties {schema['minProperties']}")
+        if "maxProperties" in schema and len(value) > schema["maxProperties"]:
+            errs.append(f"{path}: more than maxProperties {schema['maxProperties']}")
         for k, v in value.items():
             if k in props:
-                errs.extend(validate_value(props[k], v, f"{path}.{k}"))
+                errs.extend(validate_value(props[k], v, f"{path}.{k}", root))
             elif schema.get("additionalProperties") is False:
                 errs.append(f"{path}: unexpected property {k!r}")
             elif isinstance(schema.get("additionalProperties"), dict):
-                errs.extend(validate_value(schema["additionalProperties"], v, f"{path}.{k}"))
-    if isinstance(value, list) and "items" in schema:
-        for i, item in enumerate(value):
-            errs.extend(validate_value(schema["items"], item, f"{path}[{i}]"))
+                errs.extend(validate_value(schema["additionalProperties"], v, f"{path}.{k}", root))
+    if isinstance(value, list):
+        items = schema.get("items")
+        if isinstance(items, dict):
+            for i, item in enumerate(value):
+                errs.extend(validate_value(items, item, f"{path}[{i}]", root))
+        elif isinstance(items, list):  # draft-07 tuple validation
+            for i, sub in enumerate(items):
+                if i < len(value):
+                    errs.extend(validate_value(sub, value[i], f"{path}[{i}]", root))
+            extra = schema.get("additionalItems")
+            if extra is False and len(value) > len(items):
+                errs.append(f"{path}: more items than tuple schema allows")
+            elif isinstance(extra, dict):
+                for i in range(len(items), len(value)):
+                    errs.extend(validate_value(extra, value[i], f"{path}[{i}]", root))
         if "minItems" in schema and len(value) < schema["minItems"]:
             errs.append(f"{path}: fewer than minItems {schema['minItems']}")
         if "maxItems" in schema and len(value) > schema["maxItems"]:
             errs.append(f"{path}: more than maxItems {schema['maxItems']}")
+        if schema.get("uniqueItems"):
+            seen = []
+            for item in value:
+                if item in seen:
+                    errs.append(f"{path}: items not unique")
+                    break
+                seen.append(item)
+    if "not" in schema:
+        if not validate_value(schema["not"], value, path, root):
+            errs.append(f"{path}: matches schema in 'not'")
     for comb, mode in (("anyOf", "any"), ("allOf", "all"), ("oneOf", "one")):
         if comb in schema:
-            sub_errs = [validate_value(s, value, path) for s in schema[comb]]
+            sub_errs = [validate_value(s, value, path, root) for s in schema[comb]]
             ok = [not e for e in sub_errs]
             if mode == "any" and not any(ok):
                 errs.append(f"{path}: no {comb} branch matched")
