@@ -490,7 +490,7 @@ __global__ __launch_bounds__(BLOCK) void wf_sweep_kernel(
     const unsigned char* __restrict__ step_kind,  // [NR*64]
     const long long* __restrict__ cond_bits,  // [NR] bit s = condition value
     const int* __restrict__ next_ready,       // [NR*64] tick gate
-    int tick,
+    const int* __restrict__ tick_p,           // [1] device tick counter
     int* __restrict__ disp_runs,              // [cap] WORKER/FOR_EACH to expand
     int* __restrict__ disp_steps,             // [cap]
     int* __restrict__ disp_count,             // [1]
@@ -501,6 +501,7 @@ __global__ __launch_bounds__(BLOCK) void wf_sweep_kernel(
 {
     const int run = blockIdx.x * BLOCK + threadIdx.x;
     const int lane = threadIdx.x % WAVE;
+    const int tick = *tick_p;
     long long disp = 0, appr = 0;
     if (run < NR && run_active[run]) {
         const int ns = n_steps[run];
@@ -566,7 +567,7 @@ __global__ __launch_bounds__(BLOCK) void wf_expand_kernel(
     int* __restrict__ child_count,            // [1]
     int* __restrict__ children_emitted,       // [NR*64] monotone ordinal source
     int* __restrict__ dispatch_tick,          // [NR*64] stamped for K4-WF
-    int tick,
+    const int* __restrict__ tick_p,           // [1]
     const int* __restrict__ order,            // [NWG] spread order (K2c)
     const int* __restrict__ valid_count,      // [1]
     int disp_cap, int CB)
@@ -593,7 +594,7 @@ __global__ __launch_bounds__(BLOCK) void wf_expand_kernel(
             children_out[rs] += todo;
             children_emitted[rs] += todo;
             children_todo[rs] = 0;
-            dispatch_tick[rs] = tick;
+            dispatch_tick[rs] = *tick_p;
         }
     }
     base = __shfl(base, 0, WAVE);
@@ -684,7 +685,7 @@ __global__ __launch_bounds__(BLOCK) void wf_timeout_scan_kernel(
     int* __restrict__ children_out,
     int* __restrict__ children_fail,
     const int* __restrict__ dispatch_tick,    // [NR*64] last expansion tick
-    int tick, int cutoff,
+    const int* __restrict__ tick_p, int cutoff,
     unsigned long long* __restrict__ timeout_count,
     int NRS)
 {
@@ -693,7 +694,7 @@ __global__ __launch_bounds__(BLOCK) void wf_timeout_scan_kernel(
     if (step_state[i] != WFS_DISPATCHED) return;
     const int lost = children_out[i];
     if (lost <= 0) return;
-    if (dispatch_tick[i] > tick - cutoff) return;
+    if (dispatch_tick[i] > *tick_p - cutoff) return;
     children_fail[i] += lost;
     children_out[i] = 0;
     atomicAdd(timeout_count, (unsigned long long)lost);
@@ -755,7 +756,7 @@ __global__ __launch_bounds__(BLOCK) void wf_commit_kernel(
     int* __restrict__ children_done,          // kept: successes accumulate
     int* __restrict__ children_fail,
     int* __restrict__ next_ready,
-    int tick, int max_retries,
+    const int* __restrict__ tick_p, int max_retries,
     unsigned long long* __restrict__ retry_count,  // cumulative children retried
     int NRS)
 {
@@ -769,7 +770,7 @@ __global__ __launch_bounds__(BLOCK) void wf_commit_kernel(
             step_attempts[i] += 1;
             children_todo[i] += fail;      // only failed children re-run
             children_fail[i] = 0;
-            next_ready[i] = tick + min(1 << step_attempts[i], 16);
+            next_ready[i] = *tick_p + min(1 << step_attempts[i], 16);
             step_state[i] = WFS_PENDING;
             atomicAdd(retry_count, (unsigned long long)fail);
         } else {
@@ -2068,7 +2069,7 @@ void pack_requeue(torch::Tensor rq_prev_widx, torch::Tensor rq_prev_attempts,
 
 void wf_sweep(torch::Tensor step_state, torch::Tensor deps_mask, torch::Tensor n_steps,
               torch::Tensor run_active, torch::Tensor step_kind, torch::Tensor cond_bits,
-              torch::Tensor next_ready, int64_t tick,
+              torch::Tensor next_ready, torch::Tensor tick,
               torch::Tensor disp_runs, torch::Tensor disp_steps, torch::Tensor disp_count,
               torch::Tensor appr_runs, torch::Tensor appr_steps, torch::Tensor appr_count)
 {
@@ -2078,7 +2079,7 @@ void wf_sweep(torch::Tensor step_state, torch::Tensor deps_mask, torch::Tensor n
         step_state.data_ptr<uint8_t>(), (const long long*)deps_mask.data_ptr<int64_t>(),
         n_steps.data_ptr<uint8_t>(), run_active.data_ptr<uint8_t>(),
         step_kind.data_ptr<uint8_t>(), (const long long*)cond_bits.data_ptr<int64_t>(),
-        next_ready.data_ptr<int>(), (int)tick,
+        next_ready.data_ptr<int>(), tick.data_ptr<int>(),
         disp_runs.data_ptr<int>(), disp_steps.data_ptr<int>(), disp_count.data_ptr<int>(),
         appr_runs.data_ptr<int>(), appr_steps.data_ptr<int>(), appr_count.data_ptr<int>(),
         NR, (int)disp_runs.size(0), (int)appr_runs.size(0));
@@ -2088,7 +2089,7 @@ void wf_expand(torch::Tensor disp_runs, torch::Tensor disp_steps, torch::Tensor 
                torch::Tensor step_state, torch::Tensor children_todo, torch::Tensor children_out,
                torch::Tensor child_tag, torch::Tensor child_seq, torch::Tensor child_widx,
                torch::Tensor child_count, torch::Tensor children_emitted,
-               torch::Tensor dispatch_tick, int64_t tick,
+               torch::Tensor dispatch_tick, torch::Tensor tick,
                torch::Tensor order, torch::Tensor valid_count)
 {
     const int cap = (int)disp_runs.size(0);
@@ -2099,21 +2100,21 @@ void wf_expand(torch::Tensor disp_runs, torch::Tensor disp_steps, torch::Tensor 
         children_out.data_ptr<int>(), child_tag.data_ptr<int>(),
         child_seq.data_ptr<int>(), child_widx.data_ptr<int>(), child_count.data_ptr<int>(),
         children_emitted.data_ptr<int>(),
-        dispatch_tick.data_ptr<int>(), (int)tick,
+        dispatch_tick.data_ptr<int>(), tick.data_ptr<int>(),
         order.data_ptr<int>(), valid_count.data_ptr<int>(),
         cap, (int)child_tag.size(0));
 }
 
 void wf_timeout_scan(torch::Tensor step_state, torch::Tensor children_out,
                      torch::Tensor children_fail, torch::Tensor dispatch_tick,
-                     int64_t tick, int64_t cutoff, torch::Tensor timeout_count)
+                     torch::Tensor tick, int64_t cutoff, torch::Tensor timeout_count)
 {
     const int NRS = (int)step_state.numel();
     const int blocks = (NRS + BLOCK - 1) / BLOCK;
     hipLaunchKernelGGL(wf_timeout_scan_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
         step_state.data_ptr<uint8_t>(), children_out.data_ptr<int>(),
         children_fail.data_ptr<int>(), dispatch_tick.data_ptr<int>(),
-        (int)tick, (int)cutoff,
+        tick.data_ptr<int>(), (int)cutoff,
         (unsigned long long*)timeout_count.data_ptr<int64_t>(), NRS);
 }
 
@@ -2171,7 +2172,7 @@ void wf_apply_dead(torch::Tensor dead_src, torch::Tensor dead_count, torch::Tens
 void wf_commit(torch::Tensor step_state, torch::Tensor step_attempts,
                torch::Tensor children_todo, torch::Tensor children_out,
                torch::Tensor children_done, torch::Tensor children_fail,
-               torch::Tensor next_ready, int64_t tick, int64_t max_retries,
+               torch::Tensor next_ready, torch::Tensor tick, int64_t max_retries,
                torch::Tensor retry_count)
 {
     const int NRS = (int)step_state.numel();
@@ -2180,7 +2181,7 @@ void wf_commit(torch::Tensor step_state, torch::Tensor step_attempts,
         step_state.data_ptr<uint8_t>(), step_attempts.data_ptr<int>(),
         children_todo.data_ptr<int>(), children_out.data_ptr<int>(),
         children_done.data_ptr<int>(), children_fail.data_ptr<int>(),
-        next_ready.data_ptr<int>(), (int)tick, (int)max_retries,
+        next_ready.data_ptr<int>(), tick.data_ptr<int>(), (int)max_retries,
         (unsigned long long*)retry_count.data_ptr<int64_t>(), NRS);
 }
 

@@ -243,16 +243,12 @@ class _RefOps:
     def wf_sweep(self, *a):
         from .reference import wf_sweep_ref
 
-        args = list(a)
-        args[7] = int(args[7])
-        wf_sweep_ref(*args)
+        wf_sweep_ref(*a)
 
     def wf_expand(self, *a):
         from .reference import wf_expand_ref
 
-        args = list(a)
-        args[12] = int(args[12])  # tick
-        wf_expand_ref(*args)
+        wf_expand_ref(*a)
 
     def wf_apply(self, ss, sc, ct, cs, rpt, rps, cd, cf, co, fail_ppt,
                  drop_ppt, cap, world):
@@ -269,7 +265,7 @@ class _RefOps:
     def wf_commit(self, st, sa, ctd, co, cd, cf, nr, tick, max_retries, rc):
         from .reference import wf_commit_ref
 
-        wf_commit_ref(st, sa, ctd, co, cd, cf, nr, int(tick), int(max_retries), rc)
+        wf_commit_ref(st, sa, ctd, co, cd, cf, nr, tick, int(max_retries), rc)
 
     def wf_status(self, *a):
         from .reference import wf_status_ref
@@ -284,7 +280,7 @@ class _RefOps:
     def wf_timeout_scan(self, st, co, cf, dt, tick, cutoff, tc):
         from .reference import wf_timeout_scan_ref
 
-        wf_timeout_scan_ref(st, co, cf, dt, int(tick), int(cutoff), tc)
+        wf_timeout_scan_ref(st, co, cf, dt, tick, int(cutoff), tc)
 
     def wf_readmit(self, *a):
         from .reference import wf_readmit_ref

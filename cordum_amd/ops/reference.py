@@ -330,9 +330,10 @@ WFK_WORKER, WFK_FOR_EACH, WFK_APPROVAL, WFK_CONDITION, WFK_DELAY = 0, 1, 2, 3, 4
 
 
 def wf_sweep_ref(step_state, deps_mask, n_steps, run_active, step_kind,
-                 cond_bits, next_ready, tick: int,
+                 cond_bits, next_ready, tick,
                  disp_runs, disp_steps, disp_count,
                  appr_runs, appr_steps, appr_count):
+    tick = int(tick[0]) if hasattr(tick, "__getitem__") else int(tick)
     NR = int(n_steps.shape[0])
     cap, acap = int(disp_runs.shape[0]), int(appr_runs.shape[0])
     for run in range(NR):
@@ -377,6 +378,7 @@ def wf_expand_ref(disp_runs, disp_steps, disp_count, step_state,
                   children_todo, children_out, child_tag, child_seq,
                   child_widx, child_count, children_emitted,
                   dispatch_tick, tick, order, valid_count):
+    tick = int(tick[0]) if hasattr(tick, "__getitem__") else int(tick)
     CB = int(child_tag.shape[0])
     n = min(int(disp_count[0]), int(disp_runs.shape[0]))
     V = min(max(1, int(valid_count[0])), 1024)
@@ -451,7 +453,8 @@ def wf_apply_dead_ref(dead_src, dead_count, child_tag, rq_prev_tag,
 
 def wf_commit_ref(step_state, step_attempts, children_todo, children_out,
                   children_done, children_fail, next_ready,
-                  tick: int, max_retries: int, retry_count=None):
+                  tick, max_retries: int, retry_count=None):
+    tick = int(tick[0]) if hasattr(tick, "__getitem__") else int(tick)
     for i in range(int(step_state.numel())):
         if int(step_state[i]) != WFS_DISPATCHED or int(children_out[i]) != 0:
             continue
@@ -498,7 +501,8 @@ def wf_grant_ref(grant_runs, grant_steps, verdicts, n: int, step_state):
 
 
 def wf_timeout_scan_ref(step_state, children_out, children_fail,
-                        dispatch_tick, tick: int, cutoff: int, timeout_count):
+                        dispatch_tick, tick, cutoff: int, timeout_count):
+    tick = int(tick[0]) if hasattr(tick, "__getitem__") else int(tick)
     for i in range(int(step_state.numel())):
         if int(step_state[i]) != WFS_DISPATCHED:
             continue

@@ -190,6 +190,9 @@ class WorkflowPipeline:
         self.wf_counts = torch.zeros(2, dtype=torch.int64, device=d)
         self.dispatch_tick = torch.zeros(NR * 64, dtype=torch.int32, device=d)
         self.timeout_count = torch.zeros(1, dtype=torch.int64, device=d)
+        # device tick counter: kernels read it from memory so the whole tick
+        # body is hipGraph-capturable (a scalar arg would freeze in capture)
+        self.tick_buf = torch.full((1,), -1, dtype=torch.int32, device=d)
         self.retry_count = torch.zeros(1, dtype=torch.int64, device=d)
         self.admit_count = torch.zeros(1, dtype=torch.int64, device=d)
 
@@ -275,6 +278,7 @@ class WorkflowPipeline:
             self._appr_host_runs = self._appr_host_runs.pin_memory()
             self._appr_host_steps = self._appr_host_steps.pin_memory()
         self._pending_grants: List[tuple] = []
+        self._wf_graph = None
 
         self._tick = 0
         self.children_dispatched_total = 0
@@ -303,6 +307,7 @@ class WorkflowPipeline:
         self.dispatch_tick.zero_()
         self._pending_grants.clear()
         self._tick = 0  # delay gates + backoff are wave-relative ticks
+        self.tick_buf.fill_(-1)
 
     # ---- collectives ---------------------------------------------------------
     def _heartbeats(self) -> None:
@@ -341,28 +346,20 @@ class WorkflowPipeline:
         self.valid_buf.copy_(valid.sum().to(torch.int32).reshape(1))
 
     # ---- one workflow tick ---------------------------------------------------
-    def tick(self) -> None:
+    def _tick_device_body(self) -> None:
+        """The fixed kernel sequence of one tick (no host decisions): this is
+        what gets hipGraph-captured on GPU (world == 1). The device tick
+        counter increments in-graph; collectives stay outside (world > 1
+        runs the same body eagerly with the exchanges inline)."""
         ext = self.ext
-        tick = self._tick
-        self._tick += 1
         cap, world, Wd = self.pad_cap, self.world, self.W
-
-        # host hop: approvals drained LAST tick are granted now (1-tick
-        # admin latency, like the reference's async approve endpoint)
-        if self._pending_grants:
-            runs, steps = self._pending_grants.pop()
-            n = runs.shape[0]
-            v = torch.full((n,), self.approval_verdict, dtype=torch.uint8,
-                           device=self.device)
-            ext.wf_grant(runs.to(self.device), steps.to(self.device), v, n,
-                         self.step_state)
-
+        self.tick_buf += 1
         # sweep: readiness -> dispatch list + fresh approval holds
         self.disp_count.zero_()
         self.appr_count.zero_()
         ext.wf_sweep(self.step_state, self.deps_mask, self.n_steps,
                      self.run_active, self.step_kind, self.cond_bits,
-                     self.next_ready, tick,
+                     self.next_ready, self.tick_buf,
                      self.disp_runs, self.disp_steps, self.disp_count,
                      self.appr_runs, self.appr_steps, self.appr_count)
 
@@ -372,7 +369,7 @@ class WorkflowPipeline:
                       self.step_state, self.children_todo, self.children_out,
                       self.child_tag, self.child_seq, self.child_widx,
                       self.child_count, self.children_emitted,
-                      self.dispatch_tick, tick,
+                      self.dispatch_tick, self.tick_buf,
                       self.order_buf, self.valid_buf)
 
         # K2 load view + routing order refresh
@@ -437,12 +434,63 @@ class WorkflowPipeline:
         # the remainder retries with backoff (reconciler.go:88-144)
         ext.wf_timeout_scan(self.step_state, self.children_out,
                             self.children_fail, self.dispatch_tick,
-                            tick, self.timeout_cutoff, self.timeout_count)
+                            self.tick_buf, self.timeout_cutoff, self.timeout_count)
         ext.wf_commit(self.step_state, self.step_attempts, self.children_todo,
                       self.children_out, self.children_done, self.children_fail,
-                      self.next_ready, tick, self.max_retries, self.retry_count)
+                      self.next_ready, self.tick_buf, self.max_retries, self.retry_count)
         ext.wf_status(self.step_state, self.n_steps, self.run_active,
                       self.run_state, self.wf_counts)
+
+    def _ensure_wf_graph(self) -> bool:
+        if self.device.type != "cuda" or self.world > 1:
+            return False
+        if self._wf_graph is not None:
+            return True
+        # eager warmup (allocator steady state), then capture; the warmup
+        # ran for real, so restore the run tables (reset) and the cumulative
+        # counters (snapshot) afterwards — the triggering tick then replays
+        # as tick 0 of a fresh wave
+        saved = [(t, t.clone()) for t in
+                 (self.wf_counts, self.timeout_count, self.retry_count,
+                  self.admit_count, self.rq_dead)]
+        for _ in range(2):
+            self._tick_device_body()
+        torch.cuda.synchronize(self.device)
+        try:
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                self._tick_device_body()
+            self._wf_graph = g
+        except Exception as e:
+            import sys
+
+            print(f"[cordum] wf-graph capture failed, running eager: {e!r}",
+                  file=sys.stderr)
+            self._wf_graph = ()
+        self.reset_runs()
+        for t, v in saved:
+            t.copy_(v)
+        torch.cuda.synchronize(self.device)
+        return bool(self._wf_graph)
+
+    def tick(self) -> None:
+        ext = self.ext
+        self._tick += 1
+
+        # host hop: approvals drained LAST tick are granted now (1-tick
+        # admin latency, like the reference's async approve endpoint)
+        if self._pending_grants:
+            runs, steps = self._pending_grants.pop()
+            n = runs.shape[0]
+            v = torch.full((n,), self.approval_verdict, dtype=torch.uint8,
+                           device=self.device)
+            ext.wf_grant(runs.to(self.device), steps.to(self.device), v, n,
+                         self.step_state)
+
+        if self.device.type == "cuda" and self.world == 1 and self._ensure_wf_graph():
+            self._wf_graph.replay()
+        else:
+            self._tick_device_body()
         self._refresh_order()
 
         # drain fresh approval holds (D2H) for next tick's grant
