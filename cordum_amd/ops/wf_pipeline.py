@@ -442,7 +442,10 @@ class WorkflowPipeline:
                       self.run_state, self.wf_counts)
 
     def _ensure_wf_graph(self) -> bool:
-        if self.device.type != "cuda" or self.world > 1:
+        import os
+
+        if self.device.type != "cuda" or self.world > 1 \
+                or os.environ.get("CORDUM_WF_NO_GRAPH"):
             return False
         if self._wf_graph is not None:
             return True
