@@ -586,9 +586,13 @@ def create_app(
 
     # -------------------------------------------------------------- approvals
     @api.get("/approvals")
-    def list_approvals(p: Principal = Depends(principal), limit: int = 50):
-        """ListJobsByState(APPROVAL_REQUIRED) + safety summary (gateway.go:3644-3698)."""
-        ids = node.job_store.list_jobs_by_state(JobState.APPROVAL_REQUIRED, limit=limit)
+    def list_approvals(p: Principal = Depends(principal), limit: int = 50,
+                       cursor: Optional[int] = None):
+        """ListJobsByState(APPROVAL_REQUIRED) + safety summary, micros-cursor
+        pagination (gateway.go:3644-3698)."""
+        cursor = _normalize_micros(cursor) if cursor else None
+        ids, next_cursor = node.job_store.list_jobs_by_state_page(
+            JobState.APPROVAL_REQUIRED, cursor=cursor, limit=limit)
         items = []
         for jid in ids:
             meta = node.job_store.get_job_meta(jid)
@@ -597,7 +601,7 @@ def create_app(
             if rec is not None:
                 item["safety"] = _safety_record_json(rec)
             items.append(item)
-        return {"items": items}
+        return {"items": items, "next_cursor": next_cursor}
 
     @api.post("/approvals/{job_id}/approve")
     async def approve_job(job_id: str, request: Request, p: Principal = Depends(admin)):
