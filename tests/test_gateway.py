@@ -455,3 +455,30 @@ def test_marketplace_no_fetcher_refuses(client, node):
     snap = client.get("/api/v1/marketplace/packs").json()
     assert "error" in snap["catalogs"][0]
     assert snap["items"] == []
+
+
+def test_job_detail_field_parity(client, node):
+    """GET /jobs/{id} must carry the reference's full detail surface
+    (gateway.go:1011-1173, ~35 fields incl. inlined context/result JSON,
+    safety decision, approval audit, workflow ids)."""
+    r = client.post("/api/v1/jobs", json={
+        "topic": "job.default", "prompt": "detail", "capability": "echo",
+        "actor_id": "a1", "actor_type": "service", "pack_id": "p1",
+        "risk_tags": ["t"], "requires": ["r"], "idempotency_key": "detail-1",
+        "labels": {"workflow_id": "wf9", "run_id": "r9", "step_id": "s9"},
+    })
+    assert r.status_code == 200, r.text
+    jid = r.json()["job_id"]
+    d = client.get(f"/api/v1/jobs/{jid}").json()
+    for f in ("id", "state", "trace_id", "context_ptr", "context",
+              "result_ptr", "topic", "tenant", "actor_id", "actor_type",
+              "idempotency_key", "capability", "pack_id", "risk_tags",
+              "requires", "attempts", "safety_decision", "safety_reason",
+              "safety_rule_id", "safety_snapshot", "safety_constraints",
+              "safety_remediations", "safety_job_hash", "approval_required",
+              "approval_ref", "labels", "workflow_id", "run_id", "step_id"):
+        assert f in d, f"missing job-detail field {f!r}"
+    assert d["context"]["prompt"] == "detail"
+    assert d["workflow_id"] == "wf9" and d["run_id"] == "r9" and d["step_id"] == "s9"
+    if d["state"] == "SUCCEEDED":
+        assert d["result"] is not None
