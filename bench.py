@@ -56,6 +56,9 @@ def run_workflow_bench(args, device, world_size, rank, backend, use_gpu) -> int:
         world_size=world_size,
         rank=rank,
         backend=backend,
+        # arena sized for the whole wave: partial-emission backpressure is a
+        # correctness feature, not a bench configuration
+        child_cap=args.runs * (args.fanout + 2),
     )
 
     def barrier():
@@ -147,7 +150,7 @@ def main() -> int:
     ap.add_argument("--workflow", action="store_true",
                     help="config #3 mode: device workflow engine waves "
                          "(1->fanout fan-out + approval gate per run)")
-    ap.add_argument("--runs", type=int, default=512, help="workflow runs per rank per wave")
+    ap.add_argument("--runs", type=int, default=2048, help="workflow runs per rank per wave")
     ap.add_argument("--fanout", type=int, default=256, help="for_each children per run")
     args = ap.parse_args()
 
