@@ -156,6 +156,10 @@ def main() -> int:
 
     import torch
 
+    from cordum_amd.utils.threads import cap_torch_threads
+
+    cap_torch_threads()
+
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
     if world_size == 1 and args.gpus > 1:
         print(

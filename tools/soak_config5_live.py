@@ -43,6 +43,10 @@ def main() -> int:
 
     import torch
 
+    from cordum_amd.utils.threads import cap_torch_threads
+
+    cap_torch_threads()
+
     from cordum_amd.ops.wf_pipeline import (
         DagSpec, StepSpec, WFK_APPROVAL, WFK_FOR_EACH, WFK_WORKER,
         WorkflowPipeline,
