@@ -220,8 +220,7 @@ def main() -> int:
     # warmup both windows (graph capture, allocator steady state, RCCL init)
     for _ in range(args.warmup):
         pipe.tick()
-    for _ in range(max(2, args.warmup)):
-        pipe_e2e.tick_e2e()
+    pipe_e2e.e2e_run(max(2, args.warmup))
     barrier()
 
     # GC discipline for the timed windows: the captured graphs + pipelines are
@@ -236,15 +235,11 @@ def main() -> int:
     gc.disable()
 
     # ---- window 1: end-to-end ingest (headline) --------------------------------
-    e2e_steps = []
-    e2e_completed = 0
-    e2e_denied = 0
+    # depth-2 pipelined on 1 GPU (encode t+1 on host while the device runs
+    # tick t); the multi-rank and CPU paths run the sequential loop inside
+    # e2e_run. Latencies are per-batch encode-start -> results-on-host.
     t0 = time.perf_counter()
-    for _ in range(args.steps):
-        st = pipe_e2e.tick_e2e()
-        e2e_steps.append(st.wall_s)
-        e2e_completed += st.completed
-        e2e_denied += st.denied
+    e2e_completed, e2e_denied, e2e_steps = pipe_e2e.e2e_run(args.steps)
     barrier()
     e2e_elapsed = time.perf_counter() - t0
 

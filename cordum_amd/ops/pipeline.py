@@ -809,6 +809,97 @@ class DevicePipeline:
         st.wall_s = time.perf_counter() - t0
         return st
 
+    def e2e_run(self, steps: int):
+        """Pipelined end-to-end ingest window: encode batch t+1 on the host
+        WHILE the device executes tick t (depth-2 software pipeline over the
+        two staging ring slots). Every cost — encode, H2D, full tick, result
+        D2H — still happens inside the timed window; it is overlapped, not
+        skipped (production ingestion pipelines the same way). All device
+        work is stream-ordered: the result D2H for tick t is enqueued before
+        tick t+1's graph, so the shared output buffers cannot race.
+
+        Returns (completed, denied, batch_latencies_s): latency = encode
+        start -> results for that batch readable on host."""
+        self.ensure_e2e()
+        if self.device.type != "cuda" or self.world > 1 or not self._fused_capable:
+            completed = denied = 0
+            lats = []
+            for _ in range(steps):
+                st = self.tick_e2e()
+                completed += st.completed
+                denied += st.denied
+                lats.append(st.wall_s)
+            return completed, denied, lats
+
+        self._ensure_graphs()
+        B, W = self.B, self.payload_words
+        nslots = 2
+        if not hasattr(self, "_e2e_hosts2"):
+            # per-slot host staging + pinned result buffers
+            def host(shape, dtype):
+                return torch.zeros(shape, dtype=dtype).pin_memory()
+
+            Wc = self.compiled.words
+            self._e2e_hosts2 = [
+                JobBatch(host((B, 7, Wc), torch.int64), host((B, 2, Wc), torch.int64),
+                         host((B,), torch.uint8), host((B, 4, Wc), torch.int64),
+                         host((B,), torch.uint8))
+                for _ in range(nslots)
+            ]
+            self._e2e_out_sums = [torch.zeros(B, dtype=torch.int32).pin_memory()
+                                  for _ in range(nslots)]
+            self._e2e_out_dec = [torch.zeros(B, dtype=torch.int8).pin_memory()
+                                 for _ in range(nslots)]
+            self._e2e_out_counts = [torch.zeros(4, dtype=torch.int32).pin_memory()
+                                    for _ in range(nslots)]
+            self._e2e_ev = [torch.cuda.Event() for _ in range(nslots)]
+
+        completed = denied = 0
+        lats = [0.0] * steps
+        t_enc = [0.0] * nslots
+
+        def harvest(s_prev: int) -> None:
+            nonlocal completed, denied
+            slot = s_prev % nslots
+            self._e2e_ev[slot].synchronize()
+            c = self._e2e_out_counts[slot]
+            denied_n = int(c[0])
+            routable = int(c[2])
+            completed += routable
+            denied += denied_n
+            lats[s_prev] = time.perf_counter() - t_enc[slot]
+            # results are in the pinned buffers (checksums + decisions);
+            # keep the last batch visible for assertions
+            self._e2e_sums = self._e2e_out_sums[slot]
+            self._e2e_decisions = self._e2e_out_dec[slot]
+
+        for s in range(steps):
+            slot = s % nslots
+            # 1) host encode (overlaps the device executing step s-1)
+            t_enc[slot] = time.perf_counter()
+            hb = self._e2e_hosts2[slot]
+            self._e2e_enc.fresh(hb)
+            payload = self._e2e_payloads[s % len(self._e2e_payloads)]
+            payload.view(B, W)[:, 0] = s
+            # 2) now require step s-1's results (bounds pipeline depth at 2)
+            if s >= 1:
+                harvest(s - 1)
+            # 3) stage + launch + result egress, all stream-ordered
+            jb = self.batches[slot]
+            jb.any_bits.copy_(hb.any_bits, non_blocking=True)
+            jb.all_bits.copy_(hb.all_bits, non_blocking=True)
+            jb.secrets.copy_(hb.secrets, non_blocking=True)
+            jb.mcp_bits.copy_(hb.mcp_bits, non_blocking=True)
+            jb.mcp_used.copy_(hb.mcp_used, non_blocking=True)
+            self.payloads[slot].copy_(payload, non_blocking=True)
+            self._graphs[slot].replay()
+            self._e2e_out_sums[slot].copy_(self.res_sums, non_blocking=True)
+            self._e2e_out_dec[slot].copy_(self.out_decision, non_blocking=True)
+            self._e2e_out_counts[slot].copy_(self._counts, non_blocking=True)
+            self._e2e_ev[slot].record()
+        harvest(steps - 1)
+        return completed, denied, lats
+
     # -- one control-plane tick -------------------------------------------------
     def tick(self) -> TickStats:
         if getattr(self, "_fused_capable", False):
