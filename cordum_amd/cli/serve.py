@@ -8,6 +8,15 @@ import time
 from pathlib import Path
 
 
+def _cuda() -> bool:
+    try:
+        import torch
+
+        return torch.cuda.is_available()
+    except ImportError:
+        return False
+
+
 def build_node(config_dir: str = "", checkpoint_dir: str = ""):
     from ..config import (
         NodeConfig,
@@ -22,8 +31,26 @@ def build_node(config_dir: str = "", checkpoint_dir: str = ""):
     routing = load_pools(cfg.pool_config_path)
     timeouts = load_timeouts(cfg.timeout_config_path)
     policy_yaml = load_safety_yaml(cfg.safety_policy_path)
+    # dispatch plane: on a GPU host the batched K1/K2 device engine is the
+    # default (the north star); CORDUM_DISPATCH=host|device overrides
+    import os as _os
+
+    dispatch = _os.environ.get("CORDUM_DISPATCH", "")
+    if not dispatch:
+        try:
+            import torch
+
+            from ..utils.threads import cap_torch_threads
+
+            cap_torch_threads()
+            dispatch = "device" if torch.cuda.is_available() else "host"
+        except ImportError:
+            dispatch = "host"
     node = Node(routing=routing, policy_yaml=policy_yaml,
-                safety_cache_ttl_s=cfg.safety_decision_cache_ttl_s).start()
+                safety_cache_ttl_s=cfg.safety_decision_cache_ttl_s,
+                dispatch=dispatch,
+                device="cuda:0" if dispatch == "device" and _cuda() else "cpu",
+                backend="ext" if dispatch == "device" and _cuda() else "ref").start()
     node.scheduler_reconciler.update_timeouts(timeouts.dispatch_timeout_s, timeouts.running_timeout_s)
     seed_system_config(node.configsvc, cfg.system_config_path)
 
@@ -47,9 +74,12 @@ def serve(host: str = "127.0.0.1", port: int = 8080, config_dir: str = "",
     from ..runtime.worker import echo_handler
 
     node, cfg, checkpointer = build_node(config_dir, checkpoint_dir)
-    for i in range(workers):
-        node.add_worker(f"worker-{i}", handler=echo_handler,
-                        topics=sorted(node.strategy.current_routing().topics))
+    topics = sorted(node.strategy.current_routing().topics)
+    if node.dispatch_mode == "device":
+        node.add_device_worker_pool(n_workers=max(4, workers), topics=topics)
+    else:
+        for i in range(workers):
+            node.add_worker(f"worker-{i}", handler=echo_handler, topics=topics)
 
     app = create_app(node, rate_limit_rps=cfg.api_rate_limit_rps,
                      rate_limit_burst=cfg.api_rate_limit_burst,

@@ -146,3 +146,45 @@ def test_crash_recovery_with_checkpoint(tmp_path):
             proc2.wait(timeout=10)
         except subprocess.TimeoutExpired:
             proc2.kill()
+
+
+def test_serve_device_dispatch_mode():
+    """`cordumctl serve` with CORDUM_DISPATCH=device runs the batched K1/K2
+    engine behind the gateway (ref backend on CPU; the GPU box runs the same
+    path on the HIP extension — tests/test_gpu_serve.py)."""
+    port = free_port()
+    env = dict(os.environ, PYTHONPATH=str(REPO), CORDUM_DISPATCH="device")
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "cordum_amd.cli.cordumctl", "serve",
+         "--port", str(port), "--workers", "2"],
+        cwd=str(REPO), env=env,
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
+    )
+    try:
+        from cordum_amd.sdk.client import Client
+
+        client = Client(base_url=f"http://127.0.0.1:{port}", role="admin",
+                        principal_id="e2e")
+        for _ in range(100):
+            try:
+                client.status()
+                break
+            except Exception:
+                if proc.poll() is not None:
+                    raise RuntimeError(proc.stdout.read().decode()[-2000:])
+                time.sleep(0.2)
+        out = client.submit_job(topic="job.default", prompt="via device engine")
+        jid = out["job_id"]
+        for _ in range(100):
+            d = client.get_job(jid)
+            if d["state"] in ("SUCCEEDED", "FAILED", "DENIED"):
+                break
+            time.sleep(0.1)
+        assert d["state"] == "SUCCEEDED", d
+        assert d["result"] is not None
+    finally:
+        proc.send_signal(signal.SIGTERM)
+        try:
+            proc.wait(timeout=10)
+        except subprocess.TimeoutExpired:
+            proc.kill()
