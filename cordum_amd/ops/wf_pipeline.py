@@ -78,7 +78,6 @@ class WfStats:
     runs_failed: int = 0
     ticks: int = 0
     wall_s: float = 0.0
-    children_dispatched: int = 0
 
 
 class WorkflowPipeline:
@@ -281,7 +280,6 @@ class WorkflowPipeline:
         self._wf_graph = None
 
         self._tick = 0
-        self.children_dispatched_total = 0
 
     # ---- run admission -------------------------------------------------------
     def reset_runs(self) -> None:

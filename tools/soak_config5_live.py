@@ -80,7 +80,6 @@ def main() -> int:
     t0 = time.perf_counter()
     ticks = 0
     drained_total = 0
-    drains = 0
     samples = []
     last_ok = 0
     last_t = t0
@@ -90,7 +89,6 @@ def main() -> int:
         ticks += 1
         if ticks % args.drain_every == 0:
             drained_total += pipe.drain_failed_to_dlq(dlq)
-            drains += 1
             now = time.perf_counter()
             ok, fail = pipe.counts()
             samples.append({
