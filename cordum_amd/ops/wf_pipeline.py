@@ -224,11 +224,17 @@ class WorkflowPipeline:
         self.w_maxp = torch.full((NWG,), maxp, dtype=torch.int32, device=d)
         self.w_labels = torch.zeros(NWG, dtype=torch.int64, device=d)
         self.w_active_local = torch.zeros(self.NWL, dtype=torch.int32, device=d)
-        self.w_active = torch.zeros(NWG, dtype=torch.int32, device=d)
         self.w_cpu_local = (torch.rand(self.NWL, generator=g) * 50).to(d)
         self.w_gpu_local = (torch.rand(self.NWL, generator=g) * 50).to(d)
-        self.w_cpu = torch.zeros(NWG, dtype=torch.float32, device=d)
-        self.w_gpu = torch.zeros(NWG, dtype=torch.float32, device=d)
+        if world_size == 1:
+            # the local view IS the global view: no per-tick D2D copies
+            self.w_active = self.w_active_local
+            self.w_cpu = self.w_cpu_local
+            self.w_gpu = self.w_gpu_local
+        else:
+            self.w_active = torch.zeros(NWG, dtype=torch.int32, device=d)
+            self.w_cpu = torch.zeros(NWG, dtype=torch.float32, device=d)
+            self.w_gpu = torch.zeros(NWG, dtype=torch.float32, device=d)
         self.w_keys = torch.zeros(NWG, dtype=torch.int64, device=d)
         self.order_buf = torch.arange(NWG, dtype=torch.int32, device=d)
         self.valid_buf = torch.tensor([NWG], dtype=torch.int32, device=d)
@@ -250,8 +256,14 @@ class WorkflowPipeline:
         self.pad_res = torch.zeros_like(self.pad_send_payload)
         self.pad_sums = zi(world * cap)
         self.pad_sums_back = zi(world * cap)
-        # requeue ring (children parked on destination overflow) + tag carry
-        RQ = CB
+        # requeue ring (children parked on destination overflow) + tag carry.
+        # Sized well below the child arena: overflow is the exception, and
+        # the ring's payload arena is ping-pong COPIED every tick inside the
+        # captured graph — a CB-sized ring made that copy 28% of the tick
+        # (profiles/r29_wf_tick_ktrace). Ring-full drops land in the dead
+        # list -> wf_apply_dead -> the children retry, so a small ring is
+        # safe backpressure, not loss.
+        RQ = min(CB, max(4096, CB // 16))
         self.rq_src = zi(RQ)
         self.rq_widx = zi(RQ)
         self.rq_attempts = zi(RQ)
@@ -315,10 +327,7 @@ class WorkflowPipeline:
                 dist.all_gather_into_tensor(self.w_cpu, self.w_cpu_local)
                 dist.all_gather_into_tensor(self.w_gpu, self.w_gpu_local)
                 self._hb_static_done = True
-        else:
-            self.w_active.copy_(self.w_active_local)
-            self.w_cpu.copy_(self.w_cpu_local)
-            self.w_gpu.copy_(self.w_gpu_local)
+        # world == 1: the global tensors alias the local ones — nothing to do
 
     def _exchange_out(self) -> None:
         if self.world > 1:
