@@ -34,8 +34,11 @@ def init_fabric(backend: str | None = None) -> tuple[int, int, torch.device]:
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
     if torch.cuda.is_available():
-        torch.cuda.set_device(local_rank)
-        device = torch.device(f"cuda:{local_rank}")
+        # oversubscription fallback: more ranks than visible GPUs shares
+        # devices round-robin (lets a multi-rank job validate on one GPU)
+        idx = local_rank % max(1, torch.cuda.device_count())
+        torch.cuda.set_device(idx)
+        device = torch.device(f"cuda:{idx}")
     else:
         device = torch.device("cpu")
     if world > 1 and not dist.is_initialized():
