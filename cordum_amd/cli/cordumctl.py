@@ -47,7 +47,7 @@ def cmd_serve(args):
     serve(host=args.host, port=args.port, config_dir=args.config_dir,
           workers=args.workers, checkpoint_dir=args.checkpoint_dir,
           checkpoint_interval_s=args.checkpoint_interval,
-          dashboard_dir=args.dashboard_dir)
+          dashboard_dir=args.dashboard_dir, bridge_port=args.bridge_port)
 
 
 def cmd_init(args):
@@ -310,6 +310,9 @@ def main(argv=None) -> int:
     p = sub.add_parser("serve", help="run the single-process control-plane node")
     p.add_argument("--dashboard-dir", default="",
                    help="serve a dashboard build statically at /")
+    p.add_argument("--bridge-port", type=int, default=-1,
+                   help="TCP bus-bridge port for external CAP workers "
+                        "(0 = auto, -1 = disabled)")
     p.add_argument("--host", default="127.0.0.1")
     p.add_argument("--port", type=int, default=8080)
     p.add_argument("--config-dir", default="")

@@ -67,7 +67,7 @@ def build_node(config_dir: str = "", checkpoint_dir: str = ""):
 
 def serve(host: str = "127.0.0.1", port: int = 8080, config_dir: str = "",
           workers: int = 2, checkpoint_dir: str = "", checkpoint_interval_s: float = 30.0,
-          dashboard_dir: str = ""):
+          dashboard_dir: str = "", bridge_port: int = -1):
     import uvicorn
 
     from ..gateway import create_app
@@ -84,6 +84,19 @@ def serve(host: str = "127.0.0.1", port: int = 8080, config_dir: str = "",
     app = create_app(node, rate_limit_rps=cfg.api_rate_limit_rps,
                      rate_limit_burst=cfg.api_rate_limit_burst,
                      dashboard_dir=dashboard_dir)
+
+    # TCP bus bridge for external CAP workers (the NATS attach seam;
+    # --bridge-port 0 picks a free port, -1 disables)
+    import os as _os2
+
+    if bridge_port < 0:
+        bridge_port = int(_os2.environ.get("CORDUM_BRIDGE_PORT", "-1"))
+    bridge = None
+    if bridge_port >= 0:
+        from ..bus.tcp_bridge import BusBridgeServer
+
+        bridge = BusBridgeServer(node, host=host, port=bridge_port).start()
+        print(f"bus bridge listening on {host}:{bridge.port}", flush=True)
 
     stop = threading.Event()
 
@@ -107,3 +120,5 @@ def serve(host: str = "127.0.0.1", port: int = 8080, config_dir: str = "",
         uvicorn.run(app, host=host, port=port, log_level="warning")
     finally:
         stop.set()
+        if bridge is not None:
+            bridge.stop()
