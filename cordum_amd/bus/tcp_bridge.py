@@ -172,6 +172,7 @@ class BusBridgeServer:
             except OSError:
                 return
             conn = _BridgeConn(self, sock)
+            self._conns = [c for c in self._conns if c.alive]
             self._conns.append(conn)
             threading.Thread(target=conn.serve, daemon=True).start()
 

@@ -48,7 +48,10 @@ class RemoteWorker:
         self.bus = BridgeClient(bridge_host, bridge_port)
         self.api = Client(base_url=api_base, api_key=api_key,
                           principal_id=worker_id)
+        from concurrent.futures import ThreadPoolExecutor
+
         self._sem = threading.Semaphore(max_parallel)
+        self._pool = ThreadPoolExecutor(max_workers=max_parallel)
         self._active: Dict[str, bool] = {}
         self._cancelled: Dict[str, bool] = {}
         self._mu = threading.Lock()
@@ -110,8 +113,7 @@ class RemoteWorker:
                 continue
             req = pkt.job_request
             trace = pkt.trace_id
-            threading.Thread(target=self._run_job, args=(req, trace),
-                             daemon=True).start()
+            self._pool.submit(self._run_job, req, trace)
 
     def _run_job(self, req: JobRequest, trace_id: str) -> None:
         with self._sem:
