@@ -160,3 +160,54 @@ rules:
     n.submit_job(stored, context=None)
     n.drain()
     assert n.job_store.get_state("dep-1") == JobState.SUCCEEDED
+
+
+def test_worker_crash_timeout_dlq_and_retry():
+    """Failure detection end-to-end (reconciler.go:88-144 + dlq retry
+    gateway.go:3452): a worker receives a job and never reports back; the
+    staleness reconciler times it out, the DLQ records it, and the retry
+    path re-runs it to success once a healthy worker serves the topic."""
+    from cordum_amd.protocol.capv2 import BusPacket, Heartbeat
+    from cordum_amd.protocol import subjects as subj
+
+    clock = ManualClock()
+    routing = PoolRouting(topics={"job.default": ["default"]},
+                          pools={"default": PoolProfile()})
+    n = Node(clock=clock, routing=routing).start()
+
+    # a "worker" that heartbeats (so routing picks it) but swallows jobs
+    swallowed = []
+    n.bus.subscribe(subj.worker_subject("zombie"),
+                    lambda s, p: swallowed.append(p.job_request.job_id))
+    n.bus.publish(subj.SUBJECT_HEARTBEAT, BusPacket(
+        protocol_version=1,
+        heartbeat=Heartbeat(worker_id="zombie", pool="default",
+                            max_parallel_jobs=4)))
+    n.drain()
+
+    n.submit_job(JobRequest(job_id="lost-1", topic="job.default",
+                            tenant_id="default"), context=b'{"p": 1}')
+    n.drain()
+    assert swallowed == ["lost-1"]
+    from cordum_amd.protocol import JobState as JS
+
+    assert n.job_store.get_state("lost-1") in (JS.DISPATCHED, JS.RUNNING)
+
+    # past the dispatch/running cutoffs the reconciler declares TIMEOUT
+    # (the reference reconciler only transitions state — no DLQ entry,
+    # reconciler.go:88-129 — clients observe TIMEOUT and resubmit/remediate)
+    clock.advance(10_000)
+    n.reconcile()
+    n.drain()
+    assert n.job_store.get_state("lost-1") == JS.TIMEOUT
+    events = n.job_store.get_events("lost-1")
+    assert any(e.endswith("|TIMEOUT") for e in events)
+
+    # the zombie's heartbeat has expired with the clock advance; a healthy
+    # worker joins and the resubmission runs to completion
+    n.add_worker("healthy", topics=["job.default"])
+    n.submit_job(JobRequest(job_id="lost-1-resubmit", topic="job.default",
+                            tenant_id="default"), context=b'{"p": 1}')
+    n.drain()
+    assert n.job_store.get_state("lost-1-resubmit") == JS.SUCCEEDED
+    assert swallowed == ["lost-1"]  # the zombie never got the resubmission
