@@ -290,6 +290,11 @@ class WorkflowPipeline:
             self._appr_host_steps = self._appr_host_steps.pin_memory()
         self._pending_grants: List[tuple] = []
         self._wf_graph = None
+        if device.type == "cuda":
+            self._appr_h_count = torch.zeros(1, dtype=torch.int32).pin_memory()
+            self._appr_ev = torch.cuda.Event()
+        else:
+            self._appr_h_count = None
 
         self._tick = 0
 
@@ -316,6 +321,7 @@ class WorkflowPipeline:
         self.dead_count.zero_()
         self.dispatch_tick.zero_()
         self._pending_grants.clear()
+        self._appr_deferred = False
         self._tick = 0  # delay gates + backoff are wave-relative ticks
         self.tick_buf.fill_(-1)
 
@@ -447,6 +453,12 @@ class WorkflowPipeline:
                       self.next_ready, self.tick_buf, self.max_retries, self.retry_count)
         ext.wf_status(self.step_state, self.n_steps, self.run_active,
                       self.run_state, self.wf_counts)
+        if self._appr_h_count is not None:
+            # stream-ordered async D2H of the approval count into pinned
+            # memory: the host polls it with event.query() instead of a
+            # blocking sync every tick (the grant hop already tolerates a
+            # tick of lag)
+            self._appr_h_count.copy_(self.appr_count, non_blocking=True)
 
     def _ensure_wf_graph(self) -> bool:
         import os
@@ -483,9 +495,24 @@ class WorkflowPipeline:
         torch.cuda.synchronize(self.device)
         return bool(self._wf_graph)
 
+    def _drain_approvals(self) -> None:
+        na = int(self._appr_h_count[0]) if self._appr_h_count is not None             else int(self.appr_count.cpu()[0])
+        if na > 0:
+            na = min(na, self.NR)
+            self._pending_grants.append(
+                (self.appr_runs[:na].cpu().clone(), self.appr_steps[:na].cpu().clone()))
+
     def tick(self) -> None:
         ext = self.ext
         self._tick += 1
+
+        # a drain deferred from last tick (its D2H event was still in
+        # flight) MUST land before this tick's body overwrites the approval
+        # ring — by now the event has long completed, so this rarely blocks
+        if getattr(self, "_appr_deferred", False):
+            self._appr_ev.synchronize()
+            self._drain_approvals()
+            self._appr_deferred = False
 
         # host hop: approvals drained LAST tick are granted now (1-tick
         # admin latency, like the reference's async approve endpoint)
@@ -503,12 +530,18 @@ class WorkflowPipeline:
             self._tick_device_body()
         self._refresh_order()
 
-        # drain fresh approval holds (D2H) for next tick's grant
-        na = int(self.appr_count.cpu()[0])
-        if na > 0:
-            na = min(na, self.NR)
-            self._pending_grants.append(
-                (self.appr_runs[:na].cpu().clone(), self.appr_steps[:na].cpu().clone()))
+        # drain fresh approval holds for next tick's grant. On GPU the count
+        # arrives via the in-body pinned D2H; poll non-blockingly — if the
+        # tick is still in flight, defer the drain to the top of the next
+        # tick (before the ring is overwritten) instead of stalling here
+        if self._appr_h_count is not None:
+            self._appr_ev.record()
+            if not self._appr_ev.query():
+                self._appr_deferred = True
+                return
+            self._drain_approvals()
+        else:
+            self._drain_approvals()
 
     def _child_slots(self):
         """Routable slot list for pack_by_dest = child arena indices 0..C-1."""
