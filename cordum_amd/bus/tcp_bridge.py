@@ -140,8 +140,20 @@ class _BridgeConn:
             self.close()
 
     def _make_handler(self):
+        from .bus import RetryAfter
+
         def handler(subject: str, pkt: BusPacket) -> None:
-            self.deliver(subject, pkt)
+            try:
+                self.deliver(subject, pkt)
+            except (ConnectionError, OSError) as e:
+                # a dead wire client must never fail the PUBLISHER (the
+                # scheduler's dispatch, a gateway submit). The subscription
+                # is closed by deliver(); NAK so durable subjects redeliver
+                # to a surviving queue-group member, exactly like a NATS
+                # client vanishing mid-message (bus/nats.go:146-168) — and the
+                # staleness reconciler covers a direct-subject worker that
+                # died holding a job
+                raise RetryAfter(0.05, cause=f"bridge client gone: {e}")
 
         return handler
 

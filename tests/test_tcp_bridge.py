@@ -104,3 +104,32 @@ def test_disconnected_client_unsubscribes(node, bridge):
         protocol_version=1,
         job_result=JobResult(job_id="gone", status=JobStatus.SUCCEEDED)))
     assert all(not conn.alive for conn in bridge._conns)
+
+
+def test_dead_client_does_not_fail_publisher(node, bridge):
+    """A wire client that dies mid-delivery must NAK (durable subjects
+    redeliver to survivors) instead of raising into the publisher."""
+    from cordum_amd.protocol import subjects as subj
+
+    dead = BridgeClient(port=bridge.port)
+    dead.subscribe("job.pool2", queue_group="g2")
+    live = BridgeClient(port=bridge.port)
+    live.subscribe("job.pool2", queue_group="g2")
+    time.sleep(0.1)
+    # kill the first client's socket without unsubscribing
+    dead.sock.close()
+    time.sleep(0.1)
+    got = []
+    t = collect(live, got, 4)
+    for i in range(4):
+        # durable subject: delivery to the dead member NAKs and the pump
+        # redelivers; publish must never raise
+        node.bus.publish("job.pool2", BusPacket(
+            protocol_version=1,
+            job_request=JobRequest(job_id=f"d{i}", topic="job.pool2")))
+    deadline = time.time() + 5
+    while time.time() < deadline and len(got) < 4:
+        node.bus.pump()
+        time.sleep(0.05)
+    assert len(got) == 4  # every message landed on the survivor
+    live.close()
