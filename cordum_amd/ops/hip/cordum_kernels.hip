@@ -561,6 +561,7 @@ __global__ __launch_bounds__(BLOCK) void wf_expand_kernel(
     unsigned char* __restrict__ step_state,   // [NR*64]
     int* __restrict__ children_todo,          // [NR*64] children left to emit
     int* __restrict__ children_out,           // [NR*64] in flight
+    const int* __restrict__ max_parallel,     // [NR*64] 0 = unlimited
     int* __restrict__ child_tag,              // [CB] run*64+step
     int* __restrict__ child_seq,              // [CB] per-step emission ordinal
     int* __restrict__ child_widx,             // [CB] global worker pick
@@ -581,20 +582,34 @@ __global__ __launch_bounds__(BLOCK) void wf_expand_kernel(
     int base = 0, todo = 0, seq0 = 0;
     if (lane == 0) {
         todo = children_todo[rs];
+        // for_each max_parallel windowing (dataflow_test.go:71 semantics):
+        // emit only up to the window; the commit pass re-opens the step as
+        // soon as in-flight drops below the window
+        const int maxp = max_parallel[rs];
+        if (maxp > 0) {
+            const int room = maxp - children_out[rs];
+            if (room < todo) todo = room > 0 ? room : 0;
+        }
         // monotone per-step emission counter: a deterministic per-child
         // ordinal stream identical on every backend regardless of atomic
         // packing order (retried children get FRESH ordinals)
         seq0 = children_emitted[rs];
-        base = atomicAdd(child_count, todo);
-        if (base + todo > CB) {          // arena full: roll back, retry later
-            atomicSub(child_count, todo);
-            todo = -1;
-        } else {
+        if (todo > 0) {
+            base = atomicAdd(child_count, todo);
+            if (base + todo > CB) {      // arena full: roll back, retry later
+                atomicSub(child_count, todo);
+                todo = -1;
+            } else {
+                step_state[rs] = WFS_DISPATCHED;
+                children_out[rs] += todo;
+                children_emitted[rs] += todo;
+                children_todo[rs] -= todo;
+                dispatch_tick[rs] = *tick_p;
+            }
+        } else if (children_out[rs] > 0) {
+            // window full: mark DISPATCHED so the sweep stops re-emitting
+            // until the commit pass re-opens it
             step_state[rs] = WFS_DISPATCHED;
-            children_out[rs] += todo;
-            children_emitted[rs] += todo;
-            children_todo[rs] = 0;
-            dispatch_tick[rs] = *tick_p;
         }
     }
     base = __shfl(base, 0, WAVE);
@@ -755,6 +770,7 @@ __global__ __launch_bounds__(BLOCK) void wf_commit_kernel(
     int* __restrict__ children_out,
     int* __restrict__ children_done,          // kept: successes accumulate
     int* __restrict__ children_fail,
+    const int* __restrict__ max_parallel,
     int* __restrict__ next_ready,
     const int* __restrict__ tick_p, int max_retries,
     unsigned long long* __restrict__ retry_count,  // cumulative children retried
@@ -763,7 +779,15 @@ __global__ __launch_bounds__(BLOCK) void wf_commit_kernel(
     const int i = blockIdx.x * BLOCK + threadIdx.x;
     if (i >= NRS) return;
     if (step_state[i] != WFS_DISPATCHED) return;
-    if (children_out[i] != 0) return;
+    if (children_out[i] != 0) {
+        // window slide: with children still in flight, re-open emission as
+        // soon as there is BOTH work left and room (max_parallel window, or
+        // freed child-arena space for the unlimited case)
+        if (children_todo[i] > 0 &&
+            (max_parallel[i] == 0 || children_out[i] < max_parallel[i]))
+            step_state[i] = WFS_PENDING;
+        return;
+    }
     const int fail = children_fail[i];
     if (fail > 0) {
         if (step_attempts[i] < max_retries) {
@@ -2087,6 +2111,7 @@ void wf_sweep(torch::Tensor step_state, torch::Tensor deps_mask, torch::Tensor n
 
 void wf_expand(torch::Tensor disp_runs, torch::Tensor disp_steps, torch::Tensor disp_count,
                torch::Tensor step_state, torch::Tensor children_todo, torch::Tensor children_out,
+               torch::Tensor max_parallel,
                torch::Tensor child_tag, torch::Tensor child_seq, torch::Tensor child_widx,
                torch::Tensor child_count, torch::Tensor children_emitted,
                torch::Tensor dispatch_tick, torch::Tensor tick,
@@ -2097,7 +2122,7 @@ void wf_expand(torch::Tensor disp_runs, torch::Tensor disp_steps, torch::Tensor 
     hipLaunchKernelGGL(wf_expand_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
         disp_runs.data_ptr<int>(), disp_steps.data_ptr<int>(), disp_count.data_ptr<int>(),
         step_state.data_ptr<uint8_t>(), children_todo.data_ptr<int>(),
-        children_out.data_ptr<int>(), child_tag.data_ptr<int>(),
+        children_out.data_ptr<int>(), max_parallel.data_ptr<int>(), child_tag.data_ptr<int>(),
         child_seq.data_ptr<int>(), child_widx.data_ptr<int>(), child_count.data_ptr<int>(),
         children_emitted.data_ptr<int>(),
         dispatch_tick.data_ptr<int>(), tick.data_ptr<int>(),
@@ -2172,6 +2197,7 @@ void wf_apply_dead(torch::Tensor dead_src, torch::Tensor dead_count, torch::Tens
 void wf_commit(torch::Tensor step_state, torch::Tensor step_attempts,
                torch::Tensor children_todo, torch::Tensor children_out,
                torch::Tensor children_done, torch::Tensor children_fail,
+               torch::Tensor max_parallel,
                torch::Tensor next_ready, torch::Tensor tick, int64_t max_retries,
                torch::Tensor retry_count)
 {
@@ -2181,6 +2207,7 @@ void wf_commit(torch::Tensor step_state, torch::Tensor step_attempts,
         step_state.data_ptr<uint8_t>(), step_attempts.data_ptr<int>(),
         children_todo.data_ptr<int>(), children_out.data_ptr<int>(),
         children_done.data_ptr<int>(), children_fail.data_ptr<int>(),
+        max_parallel.data_ptr<int>(),
         next_ready.data_ptr<int>(), tick.data_ptr<int>(), (int)max_retries,
         (unsigned long long*)retry_count.data_ptr<int64_t>(), NRS);
 }

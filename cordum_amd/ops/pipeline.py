@@ -262,10 +262,10 @@ class _RefOps:
 
         wf_apply_dead_ref(*a)
 
-    def wf_commit(self, st, sa, ctd, co, cd, cf, nr, tick, max_retries, rc):
+    def wf_commit(self, st, sa, ctd, co, cd, cf, mp, nr, tick, max_retries, rc):
         from .reference import wf_commit_ref
 
-        wf_commit_ref(st, sa, ctd, co, cd, cf, nr, tick, int(max_retries), rc)
+        wf_commit_ref(st, sa, ctd, co, cd, cf, mp, nr, tick, int(max_retries), rc)
 
     def wf_status(self, *a):
         from .reference import wf_status_ref

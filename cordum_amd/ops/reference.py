@@ -375,8 +375,8 @@ def wf_sweep_ref(step_state, deps_mask, n_steps, run_active, step_kind,
 
 
 def wf_expand_ref(disp_runs, disp_steps, disp_count, step_state,
-                  children_todo, children_out, child_tag, child_seq,
-                  child_widx, child_count, children_emitted,
+                  children_todo, children_out, max_parallel, child_tag,
+                  child_seq, child_widx, child_count, children_emitted,
                   dispatch_tick, tick, order, valid_count):
     tick = int(tick[0]) if hasattr(tick, "__getitem__") else int(tick)
     CB = int(child_tag.shape[0])
@@ -386,9 +386,19 @@ def wf_expand_ref(disp_runs, disp_steps, disp_count, step_state,
         run, step = int(disp_runs[e]), int(disp_steps[e])
         rs = run * 64 + step
         todo = int(children_todo[rs])
+        # for_each max_parallel windowing (dataflow_test.go:71 semantics)
+        maxp = int(max_parallel[rs])
+        if maxp > 0:
+            room = maxp - int(children_out[rs])
+            if room < todo:
+                todo = max(0, room)
         # monotone per-step emission counter -> deterministic per-child
         # ordinal stream (matches the kernel; retries get fresh ordinals)
         seq0 = int(children_emitted[rs])
+        if todo <= 0:
+            if int(children_out[rs]) > 0:
+                step_state[rs] = WFS_DISPATCHED
+            continue
         base = int(child_count[0])
         if base + todo > CB:
             continue  # arena full: retry next sweep
@@ -396,7 +406,7 @@ def wf_expand_ref(disp_runs, disp_steps, disp_count, step_state,
         step_state[rs] = WFS_DISPATCHED
         children_out[rs] += todo
         children_emitted[rs] += todo
-        children_todo[rs] = 0
+        children_todo[rs] -= todo
         dispatch_tick[rs] = tick
         tag = run * 64 + step
         for k in range(todo):
@@ -452,11 +462,19 @@ def wf_apply_dead_ref(dead_src, dead_count, child_tag, rq_prev_tag,
 
 
 def wf_commit_ref(step_state, step_attempts, children_todo, children_out,
-                  children_done, children_fail, next_ready,
+                  children_done, children_fail, max_parallel, next_ready,
                   tick, max_retries: int, retry_count=None):
     tick = int(tick[0]) if hasattr(tick, "__getitem__") else int(tick)
     for i in range(int(step_state.numel())):
-        if int(step_state[i]) != WFS_DISPATCHED or int(children_out[i]) != 0:
+        if int(step_state[i]) != WFS_DISPATCHED:
+            continue
+        if int(children_out[i]) != 0:
+            # window slide / arena resume: more work and room -> re-open
+            if int(children_todo[i]) > 0 and (
+                int(max_parallel[i]) == 0
+                or int(children_out[i]) < int(max_parallel[i])
+            ):
+                step_state[i] = WFS_PENDING
             continue
         fail = int(children_fail[i])
         if fail > 0:

@@ -54,6 +54,7 @@ class StepSpec:
     fanout: int = 1          # children for FOR_EACH (WORKER always 1)
     delay_ticks: int = 0     # DELAY gate
     cond: bool = True        # CONDITION value (pre-evaluated)
+    max_parallel: int = 0    # for_each window (0 = unlimited)
 
 
 @dataclass
@@ -130,6 +131,7 @@ class WorkflowPipeline:
         deps = torch.zeros(TN * 64, dtype=torch.int64)
         kinds = torch.zeros(TN * 64, dtype=torch.uint8)
         todo = torch.zeros(TN * 64, dtype=torch.int32)
+        maxp = torch.zeros(TN * 64, dtype=torch.int32)
         nready = torch.zeros(TN * 64, dtype=torch.int32)
         nsteps = torch.zeros(TN, dtype=torch.uint8)
         cond = torch.zeros(TN, dtype=torch.int64)
@@ -152,6 +154,7 @@ class WorkflowPipeline:
                     total_children += 1
                 if spec.kind == WFK_DELAY:
                     nready[i] = spec.delay_ticks
+                maxp[i] = spec.max_parallel
                 if spec.cond:
                     cbits |= 1 << s
             cond[r] = cbits - (1 << 64) if cbits >= (1 << 63) else cbits
@@ -160,6 +163,7 @@ class WorkflowPipeline:
             deps = deps.repeat(NR)
             kinds = kinds.repeat(NR)
             todo = todo.repeat(NR)
+            maxp = maxp.repeat(NR)
             nready = nready.repeat(NR)
             nsteps = nsteps.repeat(NR)
             cond = cond.repeat(NR)
@@ -168,17 +172,21 @@ class WorkflowPipeline:
         self._tmpl = {
             "step_state": st, "deps_mask": deps, "step_kind": kinds,
             "children_todo": todo, "next_ready": nready, "n_steps": nsteps,
-            "cond_bits": cond,
+            "cond_bits": cond, "max_parallel": maxp,
         }
 
         d = device
-        self.step_state = st.to(d)
-        self.deps_mask = deps.to(d)
-        self.step_kind = kinds.to(d)
-        self.children_todo = todo.to(d)
-        self.next_ready = nready.to(d)
-        self.n_steps = nsteps.to(d)
-        self.cond_bits = cond.to(d)
+        # copy=True is load-bearing: on CPU, .to(device) ALIASES the source,
+        # and the live tables must not share storage with the creation
+        # template (reset_runs / wf_readmit restore from it)
+        self.step_state = st.to(d, copy=True)
+        self.deps_mask = deps.to(d, copy=True)
+        self.step_kind = kinds.to(d, copy=True)
+        self.children_todo = todo.to(d, copy=True)
+        self.max_parallel = maxp.to(d, copy=True)
+        self.next_ready = nready.to(d, copy=True)
+        self.n_steps = nsteps.to(d, copy=True)
+        self.cond_bits = cond.to(d, copy=True)
         self.run_active = torch.ones(NR, dtype=torch.uint8, device=d)
         self.run_state = torch.zeros(NR, dtype=torch.uint8, device=d)
         self.step_attempts = torch.zeros(NR * 64, dtype=torch.int32, device=d)
@@ -380,6 +388,7 @@ class WorkflowPipeline:
         self.child_count.zero_()
         ext.wf_expand(self.disp_runs, self.disp_steps, self.disp_count,
                       self.step_state, self.children_todo, self.children_out,
+                      self.max_parallel,
                       self.child_tag, self.child_seq, self.child_widx,
                       self.child_count, self.children_emitted,
                       self.dispatch_tick, self.tick_buf,
@@ -450,6 +459,7 @@ class WorkflowPipeline:
                             self.tick_buf, self.timeout_cutoff, self.timeout_count)
         ext.wf_commit(self.step_state, self.step_attempts, self.children_todo,
                       self.children_out, self.children_done, self.children_fail,
+                      self.max_parallel,
                       self.next_ready, self.tick_buf, self.max_retries, self.retry_count)
         ext.wf_status(self.step_state, self.n_steps, self.run_active,
                       self.run_state, self.wf_counts)

@@ -337,3 +337,47 @@ def test_wf_property_random_dags_and_failures():
             for s in range(len(dags[r].steps)):
                 stv = int(states[r, s])
                 assert stv in (3, 4, 5, 0), (trial, r, s, stv)  # terminal or blocked-PENDING
+
+
+def test_second_wave_redoes_real_work():
+    """Regression pin for the CPU template-aliasing bug: .to(device) on CPU
+    aliases the source tensor, so the live tables shared storage with the
+    creation template and a re-admitted wave 'completed' instantly without
+    emitting any children. Two waves on one pipeline must BOTH do the full
+    child work."""
+    FAN = 16
+    pipe = mk_pipe([DagSpec(steps=[StepSpec(WFK_FOR_EACH, fanout=FAN)])] * 4)
+    st1 = pipe.run_wave()
+    assert st1.runs_succeeded == 4
+    e1 = int(pipe.children_emitted.view(-1, 64)[:, 0].sum())
+    assert e1 == 4 * FAN
+    st2 = pipe.run_wave()
+    assert st2.runs_succeeded == 4
+    assert st2.ticks >= st1.ticks  # not an instant zombie wave
+    e2 = int(pipe.children_emitted.view(-1, 64)[:, 0].sum())
+    assert e2 == 4 * FAN  # the second wave re-emitted EVERY child
+    done = pipe.children_done.view(-1, 64)
+    assert all(int(done[r, 0]) == FAN for r in range(4))
+
+
+def test_max_parallel_windows_inflight_children():
+    """dataflow_test.go:71 semantics on device: a for_each with
+    max_parallel=W never has more than W children in flight, the window
+    slides as children land, and every child still runs exactly once."""
+    FAN, W = 24, 4
+    dag = DagSpec(steps=[StepSpec(WFK_FOR_EACH, fanout=FAN, max_parallel=W)])
+    pipe = mk_pipe([dag] * 3, pad_cap=2)  # pad_cap 2: children land slowly
+    pipe.reset_runs()
+    max_seen = 0
+    for _ in range(400):
+        pipe.tick()
+        out = pipe.children_out.view(-1, 64)[:, 0]
+        max_seen = max(max_seen, int(out.max()))
+        if pipe.active() == 0:
+            break
+    ok, fail = pipe.counts()
+    assert ok == 3, (ok, fail)
+    assert max_seen <= W          # the window held
+    assert max_seen > 1           # and it actually pipelined
+    done = pipe.children_done.view(-1, 64)
+    assert all(int(done[r, 0]) == FAN for r in range(3))
