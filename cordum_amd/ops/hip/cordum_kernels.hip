@@ -2295,6 +2295,67 @@ void load_feedback_padded(torch::Tensor recv_widx, torch::Tensor recv_cnt,
         w_active_local.data_ptr<int>(), (int)cap, (int)world);
 }
 
+
+// ---------------------------------------------------------------------------
+// Host-side fused synthetic-batch encoder (the e2e ingest stand-in).
+// The torch-op version (SyntheticEncoder.fresh) is RNG+gather bound at
+// ~1 ms/16k batch; this fused pass (counter-based splitmix64 RNG + LUT row
+// gather, at::parallel_for over jobs) does the same work at memory speed.
+// Deterministic: draws depend only on (seed, step, job, stream), so every
+// backend and every re-run produces identical batches.
+// ---------------------------------------------------------------------------
+static inline uint64_t splitmix64(uint64_t x) {
+    x += 0x9E3779B97F4A7C15ull;
+    x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ull;
+    x = (x ^ (x >> 27)) * 0x94D049BB133111EBull;
+    return x ^ (x >> 31);
+}
+
+void synthetic_fresh(torch::Tensor any_bits,   // [B,7,W] int64 (host)
+                     torch::Tensor all_bits,   // [B,2,W] int64 (host)
+                     torch::Tensor tenant_lut, // [V,W] int64
+                     torch::Tensor topic_lut,  // [V,W]
+                     torch::Tensor risk_lut,   // [V,W]
+                     torch::Tensor req_lut,    // [V,W]
+                     int64_t seed, int64_t step,
+                     int64_t dim_tenant, int64_t dim_topic, int64_t dim_risk,
+                     int64_t all_requires)
+{
+    TORCH_CHECK(!any_bits.is_cuda(), "synthetic_fresh is a host encoder");
+    const int64_t B = any_bits.size(0);
+    const int64_t W = any_bits.size(2);
+    const int64_t V = tenant_lut.size(0);
+    int64_t* any = any_bits.data_ptr<int64_t>();
+    int64_t* all = all_bits.data_ptr<int64_t>();
+    const int64_t* tl = tenant_lut.data_ptr<int64_t>();
+    const int64_t* ol = topic_lut.data_ptr<int64_t>();
+    const int64_t* rl = risk_lut.data_ptr<int64_t>();
+    const int64_t* ql = req_lut.data_ptr<int64_t>();
+    const uint64_t base = splitmix64((uint64_t)seed * 0x5851F42D4C957F2Dull
+                                     ^ (uint64_t)step);
+    at::parallel_for(0, B, 512, [&](int64_t lo, int64_t hi) {
+        for (int64_t i = lo; i < hi; ++i) {
+            const uint64_t h0 = splitmix64(base ^ (uint64_t)i);
+            const uint64_t h1 = splitmix64(h0);
+            const uint64_t h2 = splitmix64(h1);
+            const int64_t t_idx = (int64_t)(h0 % (uint64_t)V);
+            const int64_t o_idx = (int64_t)((h0 >> 32) % (uint64_t)V);
+            const int64_t r_idx = (int64_t)(h1 % (uint64_t)V);
+            const int64_t q_idx = (int64_t)((h1 >> 32) % (uint64_t)V);
+            const bool has_risk = (h2 % 1000u) < 300;      // 30%
+            const bool has_req = ((h2 >> 32) % 1000u) < 100;  // 10%
+            int64_t* arow = any + (size_t)i * 7 * W;
+            int64_t* lrow = all + (size_t)i * 2 * W;
+            for (int64_t w = 0; w < W; ++w) {
+                arow[dim_tenant * W + w] = tl[t_idx * W + w];
+                arow[dim_topic * W + w] = ol[o_idx * W + w];
+                arow[dim_risk * W + w] = has_risk ? rl[r_idx * W + w] : 0;
+                lrow[all_requires * W + w] = has_req ? ql[q_idx * W + w] : 0;
+            }
+        }
+    });
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("policy_first_match", &policy_first_match, "K1 batched policy first-match");
     m.def("policy_first_match_mfma", &policy_first_match_mfma, "K1 MFMA comparison variant");
@@ -2325,6 +2386,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("echo_execute_indexed_dyn", &echo_execute_indexed_dyn, "echo worker with device count");
     m.def("load_feedback", &load_feedback, "per-worker active-count histogram");
     m.def("run_readiness", &run_readiness, "K3 run/step readiness sweep");
+    m.def("synthetic_fresh", &synthetic_fresh, "fused host synthetic-batch encoder");
     m.def("pack_by_dest", &pack_by_dest, "padded per-destination dispatch pack");
     m.def("pack_requeue", &pack_requeue, "redeliver last tick's requeue ring into the send segments");
     m.def("wf_sweep", &wf_sweep, "K3-WF readiness sweep + approval holds + condition/delay commits");

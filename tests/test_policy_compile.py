@@ -161,3 +161,54 @@ def test_synthetic_encoder_rows_are_exact_encodings():
                     out.mcp_bits.clone(), out.mcp_used.clone())
     enc.fresh(out2)
     assert not torch.equal(out.any_bits, out2.any_bits)
+
+
+def test_native_synthetic_encoder_oracle():
+    """ext.synthetic_fresh (the fused host encoder the e2e bench uses) must
+    match a pure-python splitmix64 replica row-for-row, be deterministic per
+    (seed, step), and keep the torch path's distribution."""
+    import pytest
+    import torch
+
+    from cordum_amd.ops import get_ext
+
+    ext = get_ext(required=False)
+    if ext is None or not hasattr(ext, "synthetic_fresh"):
+        pytest.skip("HIP extension not built")
+    from cordum_amd.ops.pipeline import DevicePipeline
+    from cordum_amd.ops.policy_compile import DIM_RISK, DIM_TENANT, DIM_TOPIC
+
+    pipe = DevicePipeline(device="cpu", batch_size=2048, n_local_workers=16,
+                          n_rules=128, backend="ref", use_mfma=False)
+    pipe.ensure_e2e()
+    enc, host = pipe._e2e_enc, pipe._e2e_host
+
+    enc.fresh_fast(host, 5, ext)
+    a1 = host.any_bits.clone()
+    enc.fresh_fast(host, 5, ext)
+    assert torch.equal(host.any_bits, a1)  # deterministic per (seed, step)
+    enc.fresh_fast(host, 6, ext)
+    assert not torch.equal(host.any_bits, a1)
+
+    M = (1 << 64) - 1
+
+    def sm(x):
+        x = (x + 0x9E3779B97F4A7C15) & M
+        x = ((x ^ (x >> 30)) * 0xBF58476D1CE4E5B9) & M
+        x = ((x ^ (x >> 27)) * 0x94D049BB133111EB) & M
+        return x ^ (x >> 31)
+
+    V = enc.tenant_lut.shape[0]
+    enc.fresh_fast(host, 7, ext)
+    base = sm((enc.seed * 0x5851F42D4C957F2D) & M ^ 7)
+    for i in range(128):
+        h0 = sm(base ^ i)
+        h1 = sm(h0)
+        h2 = sm(h1)
+        assert torch.equal(host.any_bits[i, DIM_TENANT], enc.tenant_lut[h0 % V])
+        assert torch.equal(host.any_bits[i, DIM_TOPIC], enc.topic_lut[(h0 >> 32) % V])
+        want = enc.risk_lut[h1 % V] if (h2 % 1000) < 300 \
+            else torch.zeros_like(enc.risk_lut[0])
+        assert torch.equal(host.any_bits[i, DIM_RISK], want)
+    frac = float((host.any_bits[:, DIM_RISK].abs().sum(dim=1) != 0).float().mean())
+    assert 0.2 < frac < 0.4
