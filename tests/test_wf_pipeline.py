@@ -159,6 +159,8 @@ def _host_oracle_run(dag: DagSpec, fanout_items: int):
         elif spec.kind == WFK_FOR_EACH:
             steps[sid] = {"type": "worker", "topic": "job.t", "depends_on": deps,
                           "for_each": "${input.items}"}
+            if spec.max_parallel:
+                steps[sid]["max_parallel"] = spec.max_parallel
         elif spec.kind == WFK_APPROVAL:
             steps[sid] = {"type": "approval", "depends_on": deps}
         elif spec.kind == WFK_DELAY:
@@ -212,6 +214,8 @@ def test_device_tick_matches_host_engine_on_random_dags():
             kind = rng.choice([WFK_WORKER, WFK_FOR_EACH, WFK_APPROVAL, WFK_DELAY])
             steps.append(StepSpec(kind, deps=deps,
                                   fanout=rng.randint(1, 9) if kind == WFK_FOR_EACH else 1,
+                                  max_parallel=rng.choice([0, 0, 2, 3])
+                                  if kind == WFK_FOR_EACH else 0,
                                   delay_ticks=rng.randint(0, 3)))
         dags.append(DagSpec(steps=steps))
     pipe = mk_pipe(dags)
