@@ -874,6 +874,8 @@ class DevicePipeline:
             self._e2e_out_counts = [torch.zeros(4, dtype=torch.int32).pin_memory()
                                     for _ in range(nslots)]
             self._e2e_ev = [torch.cuda.Event() for _ in range(nslots)]
+            self._e2e_in_ev = [torch.cuda.Event() for _ in range(nslots)]
+            self._e2e_copy_stream = torch.cuda.Stream(device=self.device)
             from concurrent.futures import ThreadPoolExecutor
 
             self._e2e_pool = ThreadPoolExecutor(max_workers=2)
@@ -926,15 +928,22 @@ class DevicePipeline:
             hb = self._e2e_hosts2[s % nhost]
             if s >= 1:
                 harvest(s - 1)
-            # stage + launch + result egress, all stream-ordered
+            # stage on the copy stream: the H2D for step s overlaps the
+            # DMA engines with step s-1 still computing on the main stream
+            # (the staging tensors of THIS slot were last read by graph s-2,
+            # whose completion harvest(s-1) just proved)
             jb = self.batches[slot]
-            jb.any_bits.copy_(hb.any_bits, non_blocking=True)
-            jb.all_bits.copy_(hb.all_bits, non_blocking=True)
-            jb.secrets.copy_(hb.secrets, non_blocking=True)
-            jb.mcp_bits.copy_(hb.mcp_bits, non_blocking=True)
-            jb.mcp_used.copy_(hb.mcp_used, non_blocking=True)
-            self.payloads[slot].copy_(self._e2e_payloads[s % len(self._e2e_payloads)],
-                                      non_blocking=True)
+            with torch.cuda.stream(self._e2e_copy_stream):
+                jb.any_bits.copy_(hb.any_bits, non_blocking=True)
+                jb.all_bits.copy_(hb.all_bits, non_blocking=True)
+                jb.secrets.copy_(hb.secrets, non_blocking=True)
+                jb.mcp_bits.copy_(hb.mcp_bits, non_blocking=True)
+                jb.mcp_used.copy_(hb.mcp_used, non_blocking=True)
+                self.payloads[slot].copy_(
+                    self._e2e_payloads[s % len(self._e2e_payloads)],
+                    non_blocking=True)
+                self._e2e_in_ev[slot].record(self._e2e_copy_stream)
+            torch.cuda.current_stream(self.device).wait_event(self._e2e_in_ev[slot])
             self._graphs[slot].replay()
             self._e2e_out_sums[slot].copy_(self.res_sums, non_blocking=True)
             self._e2e_out_dec[slot].copy_(self.out_decision, non_blocking=True)
