@@ -196,8 +196,9 @@ def main() -> int:
         rank=rank,
         backend=backend,
     )
-    # e2e window pipeline: bitset K1 so the captured graph reads the staging
-    # tensors the per-step H2D copies write (smaller ring: only slot 0 is used)
+    # e2e window pipeline: the captured graph reads the staging tensors the
+    # per-step H2D copies write — MFMA K1 included, via the device-side job
+    # pack kernel (falls back to the bitset K1 outside the MFMA envelope)
     pipe_e2e = DevicePipeline(
         device=device,
         batch_size=args.batch,
@@ -208,7 +209,7 @@ def main() -> int:
         rank=rank,
         n_batches=2,
         backend=backend,
-        use_mfma=False,
+        mfma_pack_on_device=True,
     )
 
     def barrier():

@@ -1594,6 +1594,38 @@ __global__ __launch_bounds__(BLOCK) void load_feedback_padded_kernel(
 // ---------------------------------------------------------------------------
 // Torch extension host wrappers
 // ---------------------------------------------------------------------------
+
+// device-side job packing for the MFMA K1 variant: bit words -> A-fragment
+// layout (policy_mfma.py _pack_a semantics, byte-for-byte). One thread per
+// (job-tile, dim, lane); each expands one 16-bit slice of the job's bit
+// word into 16 int8 lanes. Lets the e2e ingest window run the MFMA kernel
+// on freshly staged batches without a host packing pass.
+__global__ __launch_bounds__(BLOCK) void pack_jobs_mfma_kernel(
+    const long long* __restrict__ any_bits,   // [B,7] (1-word vocab)
+    const long long* __restrict__ all_bits,   // [B,2]
+    signed char* __restrict__ a_pack,         // [Jt,9,64,16]
+    int B, int Jt)
+{
+    const int idx = blockIdx.x * BLOCK + threadIdx.x;
+    if (idx >= Jt * 9 * 64) return;
+    const int jt = idx / (9 * 64);
+    const int rem = idx % (9 * 64);
+    const int d = rem / 64;
+    const int lane = rem % 64;
+    const int row = lane & 15;
+    const int kb = lane >> 4;
+    const int job = jt * 16 + row;
+    unsigned long long w = 0;
+    if (job < B)
+        w = (unsigned long long)(d < 7 ? any_bits[(size_t)job * 7 + d]
+                                       : all_bits[(size_t)job * 2 + (d - 7)]);
+    const unsigned int slice = (unsigned int)((w >> (kb * 16)) & 0xFFFFu);
+    signed char* out = a_pack + (size_t)idx * 16;
+    #pragma unroll
+    for (int t = 0; t < 16; ++t)
+        out[t] = (signed char)((slice >> t) & 1u);
+}
+
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 
@@ -2356,6 +2388,20 @@ void synthetic_fresh(torch::Tensor any_bits,   // [B,7,W] int64 (host)
     });
 }
 
+
+void pack_jobs_mfma_dev(torch::Tensor any_bits, torch::Tensor all_bits,
+                        torch::Tensor a_pack)
+{
+    const int B = (int)any_bits.size(0);
+    const int Jt = (int)a_pack.size(0);
+    const int n = Jt * 9 * 64;
+    const int blocks = (n + BLOCK - 1) / BLOCK;
+    hipLaunchKernelGGL(pack_jobs_mfma_kernel, dim3(blocks), dim3(BLOCK), 0, cur_stream(),
+        (const long long*)any_bits.data_ptr<int64_t>(),
+        (const long long*)all_bits.data_ptr<int64_t>(),
+        a_pack.data_ptr<int8_t>(), B, Jt);
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("policy_first_match", &policy_first_match, "K1 batched policy first-match");
     m.def("policy_first_match_mfma", &policy_first_match_mfma, "K1 MFMA comparison variant");
@@ -2387,6 +2433,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("load_feedback", &load_feedback, "per-worker active-count histogram");
     m.def("run_readiness", &run_readiness, "K3 run/step readiness sweep");
     m.def("synthetic_fresh", &synthetic_fresh, "fused host synthetic-batch encoder");
+    m.def("pack_jobs_mfma_dev", &pack_jobs_mfma_dev, "device job packing for the MFMA K1");
     m.def("pack_by_dest", &pack_by_dest, "padded per-destination dispatch pack");
     m.def("pack_requeue", &pack_requeue, "redeliver last tick's requeue ring into the send segments");
     m.def("wf_sweep", &wf_sweep, "K3-WF readiness sweep + approval holds + condition/delay commits");

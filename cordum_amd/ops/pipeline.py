@@ -478,9 +478,11 @@ class DevicePipeline:
         seed: int = 7,
         backend: str = "ext",
         use_mfma: Optional[bool] = None,
+        mfma_pack_on_device: bool = False,
         pad_cap: Optional[int] = None,
     ):
         self._pad_cap_override = pad_cap
+        self._mfma_device_pack = mfma_pack_on_device
         device = torch.device(device)
         self.ext = _RefOps() if backend == "ref" else get_ext(required=True)
         self.device = device
@@ -599,9 +601,21 @@ class DevicePipeline:
         if use_mfma is False:
             # the e2e-ingest bench refreshes the bitset descriptors in place
             # each step; forcing the bitset K1 keeps the captured graph
-            # reading the tensors the H2D copies write
+            # reading the tensors the H2D copies write. With
+            # mfma_pack_on_device the MFMA path reads the same staging
+            # tensors through the device pack kernel, so it stays usable.
             self._use_mfma = False
-        if self._use_mfma:
+        if self._use_mfma and self._mfma_device_pack:
+            from .policy_mfma import N_DIMS, pack_policy_mfma
+
+            self.mfma_policy = pack_policy_mfma(self.compiled).to(device)
+            Jt = (self.B + 15) // 16
+            # packed IN-GRAPH from the staged bit words each tick
+            self.mfma_a_packs = [
+                torch.zeros(Jt, N_DIMS, 64, 16, dtype=torch.int8, device=device)
+                for _ in range(len(self.batches))
+            ]
+        elif self._use_mfma:
             from .policy_mfma import pack_jobs_mfma, pack_policy_mfma
 
             self.mfma_policy = pack_policy_mfma(self.compiled).to(device)
@@ -657,6 +671,9 @@ class DevicePipeline:
         # as no-match so the host-side mask pass is gone too)
         if self._use_mfma:
             ext.begin_tick_first(self.states, self._counts, self.first_buf)
+            if self._mfma_device_pack:
+                ext.pack_jobs_mfma_dev(jb.any_bits, jb.all_bits,
+                                       self.mfma_a_packs[slot])
             ext.policy_first_match_mfma_into(
                 self.mfma_a_packs[slot], self.mfma_policy.b_pack, self.mfma_policy.cards,
                 self.mfma_policy.secrets, jb.secrets, self.mfma_policy.tile_dims,
@@ -765,7 +782,8 @@ class DevicePipeline:
     def ensure_e2e(self, seed: int = 11, payload_ring: int = 4) -> None:
         if hasattr(self, "_e2e_enc"):
             return
-        assert not self._use_mfma, "e2e ingest requires the bitset K1 (use_mfma=False)"
+        assert not self._use_mfma or self._mfma_device_pack, \
+            "e2e ingest needs staged descriptors: bitset K1 or device-packed MFMA"
         self._e2e_enc = SyntheticEncoder(self.compiled, seed=seed + self.rank)
         pin = self.device.type == "cuda"
         B, W = self.B, self.compiled.words

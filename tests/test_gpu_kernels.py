@@ -481,3 +481,40 @@ def test_fused_tick_kernels_match_refops(ext):
     ext.accumulate_counts(src, acc)
     ext.accumulate_counts(src, acc)
     assert acc.cpu().tolist() == [2, 4, 6, 8]
+
+
+def test_pack_jobs_mfma_device_matches_host(ext):
+    """The device A-fragment packing must equal policy_mfma.pack_jobs_mfma
+    byte-for-byte (same fragment layout the matrix cores consume)."""
+    from cordum_amd.ops.pipeline import encode_synthetic_jobs, make_synthetic_policy
+    from cordum_amd.ops.policy_compile import compile_policy
+    from cordum_amd.ops.policy_mfma import N_DIMS, pack_jobs_mfma
+
+    policy = make_synthetic_policy(300, vocab=40, seed=3)
+    compiled = compile_policy(policy, words=1)
+    assert compiled.exact
+    jb = encode_synthetic_jobs(compiled, 1000, seed=17)
+    want, _ = pack_jobs_mfma(jb)
+
+    d = dev()
+    jb_d = jb.to(d)
+    Jt = (1000 + 15) // 16
+    got = torch.zeros(Jt, N_DIMS, 64, 16, dtype=torch.int8, device=d)
+    ext.pack_jobs_mfma_dev(jb_d.any_bits, jb_d.all_bits, got)
+    assert torch.equal(got.cpu(), want)
+
+
+def test_e2e_mfma_device_pack_counts_match_bitset(ext):
+    """The e2e window with device-packed MFMA K1 must decide exactly what
+    the bitset K1 decides on identical staged batches."""
+    from cordum_amd.ops.pipeline import DevicePipeline
+
+    def run(**kw):
+        p = DevicePipeline(device="cuda:0", batch_size=4096, n_local_workers=64,
+                           n_rules=512, n_batches=2, backend="ext", **kw)
+        return p.e2e_run(8)
+
+    c1, d1, _ = run(use_mfma=False)
+    c2, d2, _ = run(mfma_pack_on_device=True)
+    assert (c1, d1) == (c2, d2)
+    assert d1 > 0
