@@ -133,3 +133,43 @@ def test_dead_client_does_not_fail_publisher(node, bridge):
         time.sleep(0.05)
     assert len(got) == 4  # every message landed on the survivor
     live.close()
+
+
+def test_frame_roundtrip_fuzz(node, bridge):
+    """Random subjects + random CAP payloads through real sockets: frames
+    must round-trip exactly (length-prefix framing, UTF-8 subjects, binary
+    payload bytes)."""
+    import random
+
+    from cordum_amd.protocol.capv2 import Heartbeat, JobRequest
+
+    rng = random.Random(9)
+    c = BridgeClient(port=bridge.port)
+    c.subscribe("fuzz.>")
+    time.sleep(0.1)
+    sent = []
+    for i in range(40):
+        depth = rng.randint(1, 4)
+        subject = "fuzz." + ".".join(
+            "".join(rng.choice("abcxyz09_-") for _ in range(rng.randint(1, 8)))
+            for _ in range(depth))
+        pkt = BusPacket(
+            trace_id="".join(rng.choice("0123456789abcdef") for _ in range(16)),
+            protocol_version=1,
+            job_request=JobRequest(
+                job_id=f"fz{i}", topic=subject,
+                labels={f"k{j}": "".join(rng.choice("µ€漢x") for _ in range(3))
+                        for j in range(rng.randint(0, 3))},
+            ) if rng.random() < 0.7 else None,
+            heartbeat=None if rng.random() < 0.7 else Heartbeat(worker_id=f"w{i}"),
+        )
+        sent.append((subject, pkt.encode()))
+        node.bus.publish(subject, pkt)
+    got = []
+    t = collect(c, got, len(sent))
+    t.join(timeout=10)
+    assert len(got) == len(sent)
+    for (subject, wire), (gs, gp) in zip(sent, got):
+        assert gs == subject
+        assert gp.encode() == wire  # byte-exact through the socket
+    c.close()
