@@ -33,7 +33,7 @@ from __future__ import annotations
 import socket
 import struct
 import threading
-from typing import Dict, List, Optional, Tuple
+from typing import List, Optional, Tuple
 
 from ..protocol.capv2 import BusPacket
 
