@@ -594,6 +594,7 @@ class DevicePipeline:
         self._use_mfma = (
             backend != "ref"
             and device.type == "cuda"
+            and world_size == 1  # the padded multi-rank tick runs the bitset K1
             and self.compiled.words == 1
             and self.compiled.n_rules <= 32768
             and int(self.compiled.mcp_any.sum()) == 0
