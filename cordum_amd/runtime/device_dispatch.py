@@ -160,6 +160,13 @@ class DeviceDispatchEngine(Engine):
         # 3) decision semantics + pre-dispatch checks (host per job)
         routable: List[int] = []
         for i, (req, trace_id) in enumerate(batch):
+            # a job can reach a terminal state between buffering and flush
+            # (cancel, a concurrent duplicate) — the buffered copy is then
+            # stale and must be dropped, like the host engine's early exit
+            # (engine.go:226-237)
+            if is_terminal(self.job_store.get_state(req.job_id)):
+                self._done(req.job_id)
+                continue
             rec = records[i]
             try:
                 if not self.apply_decision(req, rec):
@@ -229,9 +236,20 @@ class DeviceDispatchEngine(Engine):
                 self._route_error(req, trace_id, errs.NoWorkers(f"pool {pools_str!r}"))
             else:
                 _wid, subject = table.subject_for(pick)
-                self.dispatch(req, trace_id, subject)
+                self._dispatch_safe(req, trace_id, subject)
                 self._done(req.job_id)
                 self.device_routed += 1
+
+    def _dispatch_safe(self, req: JobRequest, trace_id: str, subject: str) -> None:
+        """dispatch() with the cancel race closed: a transition refused by
+        the legality table (the job went terminal under us) is a no-op, not
+        a batch-aborting error."""
+        from ..store.job_store import InvalidTransition
+
+        try:
+            self.dispatch(req, trace_id, subject)
+        except InvalidTransition:
+            pass
 
     def _host_route(self, req: JobRequest, trace_id: str) -> None:
         try:
@@ -239,7 +257,7 @@ class DeviceDispatchEngine(Engine):
         except Exception as e:
             self._route_error(req, trace_id, e)
             return
-        self.dispatch(req, trace_id, subject)
+        self._dispatch_safe(req, trace_id, subject)
         self._done(req.job_id)
         self.host_routed += 1
 

@@ -309,3 +309,25 @@ def test_gateway_submit_through_device_engine():
     assert "hello gpu" in str(d.get("result", ""))
     assert n.device_gate.jobs_evaluated >= 1
     assert n.scheduler.device_routed >= 1
+
+
+def test_cancel_before_flush_drops_buffered_job():
+    """A job cancelled while buffered in the device engine must not be
+    dispatched — and must not abort the rest of the flush batch (this
+    crashed the whole drain with InvalidTransition before the fix)."""
+    from cordum_amd.protocol import JobState
+    from cordum_amd.protocol.capv2 import JobRequest
+    from cordum_amd.runtime.node import Node
+    from cordum_amd.utils.clock import ManualClock
+
+    node = Node(clock=ManualClock(), dispatch="device", backend="ref").start()
+    node.add_device_worker_pool(n_workers=2, topics=["job.default"])
+    node.submit_job(JobRequest(job_id="cx-1", topic="job.default",
+                               tenant_id="default"), context=b"{}")
+    node.submit_job(JobRequest(job_id="cx-2", topic="job.default",
+                               tenant_id="default"), context=b"{}")
+    assert node.scheduler.pending_count() == 2
+    assert node.scheduler.cancel_job("cx-1")
+    node.drain()
+    assert node.job_store.get_state("cx-1") == JobState.CANCELLED
+    assert node.job_store.get_state("cx-2") == JobState.SUCCEEDED
