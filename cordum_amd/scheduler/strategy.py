@@ -36,6 +36,16 @@ _NON_PLACEMENT_LABELS = {
     "step_id",
     "node_id",
     "worker_id",
+    # gateway-stamped audit/traceability labels (remediation
+    # gateway.go:1594-1755, DLQ retry :3452-3553): in the reference these
+    # leak into placement and make every remediated/retried job unroutable
+    # (no worker carries them) — the same bug family as the approval_*
+    # labels below. Audit metadata must not constrain placement.
+    "remediation_of",
+    "remediation_id",
+    "retry",
+    "dlq_entry",
+    "retry_of_job",
 }
 # NOTE: the Go filter (strategy_least_loaded.go:196-224) drops only
 # "approval_granted"; the gateway also stamps approval_reason/approval_note on

@@ -211,3 +211,21 @@ def test_worker_crash_timeout_dlq_and_retry():
     n.drain()
     assert n.job_store.get_state("lost-1-resubmit") == JS.SUCCEEDED
     assert swallowed == ["lost-1"]  # the zombie never got the resubmission
+
+
+def test_remediation_reroutes_and_dispatches(node):
+    """Remediation must produce a job that actually ROUTES: the audit
+    labels the gateway stamps (remediation_of/remediation_id) are not
+    placement constraints (reference bug: strategy_least_loaded.go's
+    filter keeps them, so remediated jobs are unroutable there; documented
+    deviation like the approval_* family)."""
+    from cordum_amd.protocol.capv2 import PolicyRemediation
+    from cordum_amd.store.job_store import SafetyDecisionRecord
+
+    req = JobRequest(job_id="rm-1", topic="job.echo", tenant_id="default",
+                     labels={"remediation_of": "old-1", "remediation_id": "r1",
+                             "retry": "true", "dlq_entry": "old-1",
+                             "retry_of_job": "old-1"})
+    node.submit_job(req, context=b'{"m": 1}')
+    node.drain()
+    assert node.job_store.get_state("rm-1") == JobState.SUCCEEDED
