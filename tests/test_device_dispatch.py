@@ -331,3 +331,37 @@ def test_cancel_before_flush_drops_buffered_job():
     node.drain()
     assert node.job_store.get_state("cx-1") == JobState.CANCELLED
     assert node.job_store.get_state("cx-2") == JobState.SUCCEEDED
+
+
+def test_pending_replayer_redrives_device_engine():
+    """A job that exhausts its redelivery budget (no workers) parks
+    PENDING; when capacity appears, the pending replayer re-drives it
+    through the batched device path (pending_replayer.go semantics)."""
+    from cordum_amd.protocol import JobState
+    from cordum_amd.protocol.capv2 import JobRequest
+    from cordum_amd.runtime.node import Node
+    from cordum_amd.scheduler import PoolProfile, PoolRouting
+    from cordum_amd.utils.clock import ManualClock
+
+    clock = ManualClock()
+    routing = PoolRouting(topics={"job.default": ["default"]},
+                          pools={"default": PoolProfile()})
+    node = Node(clock=clock, routing=routing, dispatch="device",
+                backend="ref").start()
+    node.submit_job(JobRequest(job_id="p1", topic="job.default",
+                               tenant_id="default"), context=b"{}")
+    node.drain()
+    for _ in range(10):  # exhaust the NAK budget against an empty fleet
+        clock.advance(5)
+        node.tick()
+    assert node.job_store.get_state("p1") == JobState.PENDING
+    assert not node.scheduler._redeliveries  # budget spent, not looping
+
+    node.add_device_worker_pool(n_workers=2, topics=["job.default"])
+    clock.advance(400)  # past the replayer age threshold
+    node.reconcile()
+    node.drain()
+    for _ in range(6):
+        clock.advance(5)
+        node.tick()
+    assert node.job_store.get_state("p1") == JobState.SUCCEEDED
