@@ -288,7 +288,9 @@ def test_crash_restart_recovers_state(tmp_path):
     handlers)."""
     port = free_port()
     state = str(tmp_path / "state")
-    env = dict(os.environ, PYTHONPATH=str(REPO))
+    # device dispatch engine (ref backend on CPU): the WAL replay must
+    # re-drive submissions through the batched path too
+    env = dict(os.environ, PYTHONPATH=str(REPO), CORDUM_DISPATCH="device")
 
     def start():
         return subprocess.Popen(
