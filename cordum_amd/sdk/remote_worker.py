@@ -60,9 +60,11 @@ class RemoteWorker:
     # -- wiring ---------------------------------------------------------------
     def start(self) -> "RemoteWorker":
         for t in self.topics:
-            # queue group per topic: external workers load-balance with each
-            # other AND with in-process pools (worker.go:96-111)
-            self.bus.subscribe(t, queue_group=f"workers:{t}")
+            # SAME queue-group convention as the in-process worker runtime
+            # (pool.<pool>, worker.go:96-111): external and in-process
+            # members of a pool load-balance together — a different group
+            # name would DOUBLE-deliver pool-topic fallback publishes
+            self.bus.subscribe(t, queue_group=f"pool.{self.pool}")
         self.bus.subscribe(subj.worker_subject(self.worker_id))
         self.bus.subscribe(subj.SUBJECT_CANCEL)
         self.send_heartbeat()

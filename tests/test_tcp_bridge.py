@@ -173,3 +173,58 @@ def test_frame_roundtrip_fuzz(node, bridge):
         assert gs == subject
         assert gp.encode() == wire  # byte-exact through the socket
     c.close()
+
+
+def test_remote_and_inprocess_workers_share_pool_group(node, bridge):
+    """A pool-topic publish must reach exactly ONE member of the pool even
+    when the pool mixes an in-process worker and a wire-attached worker
+    (same queue-group convention; a divergent group name double-delivered)."""
+    from cordum_amd.sdk.remote_worker import RemoteWorker
+
+    hits = []
+    node.add_worker("inproc-1", handler=lambda req, ctx: hits.append(req.job_id) or b"{}",
+                    topics=["job.mixed"])
+
+    ext_hits = []
+
+    class _NoApi:
+        def memory(self, ptr):
+            return b"{}"
+
+        def artifacts_put(self, b):
+            return {"ptr": ""}
+
+    w = RemoteWorker.__new__(RemoteWorker)
+    # wire only the bus piece (no HTTP server in this unit test)
+    from cordum_amd.bus.tcp_bridge import BridgeClient
+    import threading
+
+    w.worker_id = "ext-1"
+    w.pool = "default"
+    w.topics = ["job.mixed"]
+    w.bus = BridgeClient(port=bridge.port)
+    w.bus.subscribe("job.mixed", queue_group="pool.default")
+    time.sleep(0.1)
+
+    def drainer():
+        while True:
+            msg = w.bus.next_message()
+            if msg is None:
+                return
+            _, pkt = msg
+            if pkt.job_request is not None:
+                ext_hits.append(pkt.job_request.job_id)
+
+    threading.Thread(target=drainer, daemon=True).start()
+
+    for i in range(12):
+        node.bus.publish("job.mixed", BusPacket(
+            protocol_version=1,
+            job_request=JobRequest(job_id=f"m{i}", topic="job.mixed")))
+    node.drain()
+    deadline = time.time() + 5
+    while time.time() < deadline and len(hits) + len(ext_hits) < 12:
+        time.sleep(0.05)
+    assert len(hits) + len(ext_hits) == 12  # exactly-once per publish
+    assert hits and ext_hits                # both members shared the load
+    w.bus.close()
