@@ -482,3 +482,28 @@ def test_job_detail_field_parity(client, node):
     assert d["workflow_id"] == "wf9" and d["run_id"] == "r9" and d["step_id"] == "s9"
     if d["state"] == "SUCCEEDED":
         assert d["result"] is not None
+
+
+def test_ws_stream_delivers_job_events(client, node):
+    """/api/v1/stream delivers protojson-shaped packets for job traffic
+    (gateway.go:2002-2049; the dashboard's live event feed)."""
+    with client.websocket_connect(
+            "/api/v1/stream",
+            headers={"X-API-Key": "test-key", "X-Principal-Id": "ws-test"}) as ws:
+        r = client.post("/api/v1/jobs", json={"topic": "job.echo", "prompt": "ws"})
+        assert r.status_code == 200
+        jid = r.json()["job_id"]
+        seen_subjects = []
+        payload = None
+        for _ in range(20):
+            msg = ws.receive_json()
+            seen_subjects.append(msg["subject"])
+            pkt = msg["packet"]
+            if pkt.get("jobRequest", {}).get("jobId") == jid:
+                payload = pkt
+            if pkt.get("jobResult", {}).get("jobId") == jid:
+                break
+        assert payload is not None, seen_subjects
+        # protojson camelCase shape
+        assert payload["jobRequest"]["topic"] == "job.echo"
+        assert any(s.startswith("sys.job.") for s in seen_subjects)
