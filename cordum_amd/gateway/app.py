@@ -21,7 +21,8 @@ import json
 import time
 from typing import Any, Dict, List, Optional
 
-from fastapi import APIRouter, Depends, FastAPI, HTTPException, Query, Request, Response
+from fastapi import (APIRouter, Depends, FastAPI, HTTPException, Query,
+                     Request, Response, WebSocket, WebSocketDisconnect)
 from fastapi.responses import JSONResponse
 
 from ..protocol import JobState
@@ -834,8 +835,11 @@ def create_app(
     app.include_router(make_packs_router(node, principal, admin), prefix="/api/v1")
 
     # ------------------------------------------------------------- WS stream
-    from fastapi import WebSocket, WebSocketDisconnect
-
+    # WebSocket/WebSocketDisconnect imported at module level: with
+    # `from __future__ import annotations` a function-local import leaves
+    # the stringified `WebSocket` annotation unresolvable and FastAPI
+    # demoted the socket parameter to a required query field, closing every
+    # connection with 1008 (found by the WS stream e2e test)
     @app.websocket("/api/v1/stream")
     async def stream(ws: WebSocket):
         pr = auth.authenticate({k: v for k, v in ws.headers.items()})
