@@ -2332,7 +2332,12 @@ void load_feedback_padded(torch::Tensor recv_widx, torch::Tensor recv_cnt,
 // Host-side fused synthetic-batch encoder (the e2e ingest stand-in).
 // The torch-op version (SyntheticEncoder.fresh) is RNG+gather bound at
 // ~1 ms/16k batch; this fused pass (counter-based splitmix64 RNG + LUT row
-// gather, at::parallel_for over jobs) does the same work at memory speed.
+// gather) does the same work at memory speed. It runs SERIAL and releases
+// the GIL: the e2e loop overlaps 2-3 encoder threads, each a fully
+// independent call — a shared at::parallel_for pool serialized those calls
+// (and holding the GIL stalled the submit loop for the whole encode).
+// Index draws use Lemire multiply-shift reduction ((u32)h * V >> 32) —
+// a plain `% V` was ~half the per-job cost.
 // Deterministic: draws depend only on (seed, step, job, stream), so every
 // backend and every re-run produces identical batches.
 // ---------------------------------------------------------------------------
@@ -2365,27 +2370,48 @@ void synthetic_fresh(torch::Tensor any_bits,   // [B,7,W] int64 (host)
     const int64_t* ql = req_lut.data_ptr<int64_t>();
     const uint64_t base = splitmix64((uint64_t)seed * 0x5851F42D4C957F2Dull
                                      ^ (uint64_t)step);
-    at::parallel_for(0, B, 512, [&](int64_t lo, int64_t hi) {
-        for (int64_t i = lo; i < hi; ++i) {
+    const uint64_t v = (uint64_t)V;
+    constexpr uint32_t RISK_CUT = 1288490189u;  // 0.30 * 2^32
+    constexpr uint32_t REQ_CUT = 429496730u;    // 0.10 * 2^32
+    if (W == 1) {
+        // fast path for the e2e bench shape (single bitset word per dim)
+        for (int64_t i = 0; i < B; ++i) {
             const uint64_t h0 = splitmix64(base ^ (uint64_t)i);
             const uint64_t h1 = splitmix64(h0);
             const uint64_t h2 = splitmix64(h1);
-            const int64_t t_idx = (int64_t)(h0 % (uint64_t)V);
-            const int64_t o_idx = (int64_t)((h0 >> 32) % (uint64_t)V);
-            const int64_t r_idx = (int64_t)(h1 % (uint64_t)V);
-            const int64_t q_idx = (int64_t)((h1 >> 32) % (uint64_t)V);
-            const bool has_risk = (h2 % 1000u) < 300;      // 30%
-            const bool has_req = ((h2 >> 32) % 1000u) < 100;  // 10%
-            int64_t* arow = any + (size_t)i * 7 * W;
-            int64_t* lrow = all + (size_t)i * 2 * W;
-            for (int64_t w = 0; w < W; ++w) {
-                arow[dim_tenant * W + w] = tl[t_idx * W + w];
-                arow[dim_topic * W + w] = ol[o_idx * W + w];
-                arow[dim_risk * W + w] = has_risk ? rl[r_idx * W + w] : 0;
-                lrow[all_requires * W + w] = has_req ? ql[q_idx * W + w] : 0;
-            }
+            const uint64_t t_idx = ((h0 & 0xffffffffull) * v) >> 32;
+            const uint64_t o_idx = ((h0 >> 32) * v) >> 32;
+            const uint64_t r_idx = ((h1 & 0xffffffffull) * v) >> 32;
+            const uint64_t q_idx = ((h1 >> 32) * v) >> 32;
+            const bool has_risk = (uint32_t)h2 < RISK_CUT;
+            const bool has_req = (uint32_t)(h2 >> 32) < REQ_CUT;
+            int64_t* arow = any + (size_t)i * 7;
+            arow[dim_tenant] = tl[t_idx];
+            arow[dim_topic] = ol[o_idx];
+            arow[dim_risk] = has_risk ? rl[r_idx] : 0;
+            all[(size_t)i * 2 + all_requires] = has_req ? ql[q_idx] : 0;
         }
-    });
+        return;
+    }
+    for (int64_t i = 0; i < B; ++i) {
+        const uint64_t h0 = splitmix64(base ^ (uint64_t)i);
+        const uint64_t h1 = splitmix64(h0);
+        const uint64_t h2 = splitmix64(h1);
+        const int64_t t_idx = (int64_t)(((h0 & 0xffffffffull) * v) >> 32);
+        const int64_t o_idx = (int64_t)(((h0 >> 32) * v) >> 32);
+        const int64_t r_idx = (int64_t)(((h1 & 0xffffffffull) * v) >> 32);
+        const int64_t q_idx = (int64_t)(((h1 >> 32) * v) >> 32);
+        const bool has_risk = (uint32_t)h2 < RISK_CUT;
+        const bool has_req = (uint32_t)(h2 >> 32) < REQ_CUT;
+        int64_t* arow = any + (size_t)i * 7 * W;
+        int64_t* lrow = all + (size_t)i * 2 * W;
+        for (int64_t w = 0; w < W; ++w) {
+            arow[dim_tenant * W + w] = tl[t_idx * W + w];
+            arow[dim_topic * W + w] = ol[o_idx * W + w];
+            arow[dim_risk * W + w] = has_risk ? rl[r_idx * W + w] : 0;
+            lrow[all_requires * W + w] = has_req ? ql[q_idx * W + w] : 0;
+        }
+    }
 }
 
 
@@ -2432,7 +2458,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("echo_execute_indexed_dyn", &echo_execute_indexed_dyn, "echo worker with device count");
     m.def("load_feedback", &load_feedback, "per-worker active-count histogram");
     m.def("run_readiness", &run_readiness, "K3 run/step readiness sweep");
-    m.def("synthetic_fresh", &synthetic_fresh, "fused host synthetic-batch encoder");
+    // GIL released: encoder threads run concurrently with the submit loop
+    m.def("synthetic_fresh", &synthetic_fresh,
+          py::call_guard<py::gil_scoped_release>(),
+          "fused host synthetic-batch encoder");
     m.def("pack_jobs_mfma_dev", &pack_jobs_mfma_dev, "device job packing for the MFMA K1");
     m.def("pack_by_dest", &pack_by_dest, "padded per-destination dispatch pack");
     m.def("pack_requeue", &pack_requeue, "redeliver last tick's requeue ring into the send segments");

@@ -201,13 +201,18 @@ def test_native_synthetic_encoder_oracle():
     V = enc.tenant_lut.shape[0]
     enc.fresh_fast(host, 7, ext)
     base = sm((enc.seed * 0x5851F42D4C957F2D) & M ^ 7)
+
+    def lemire(h32, v=V):  # index draw: multiply-shift reduction, not % V
+        return ((h32 & 0xFFFFFFFF) * v) >> 32
+
+    RISK_CUT = 1288490189  # 0.30 * 2^32
     for i in range(128):
         h0 = sm(base ^ i)
         h1 = sm(h0)
         h2 = sm(h1)
-        assert torch.equal(host.any_bits[i, DIM_TENANT], enc.tenant_lut[h0 % V])
-        assert torch.equal(host.any_bits[i, DIM_TOPIC], enc.topic_lut[(h0 >> 32) % V])
-        want = enc.risk_lut[h1 % V] if (h2 % 1000) < 300 \
+        assert torch.equal(host.any_bits[i, DIM_TENANT], enc.tenant_lut[lemire(h0)])
+        assert torch.equal(host.any_bits[i, DIM_TOPIC], enc.topic_lut[lemire(h0 >> 32)])
+        want = enc.risk_lut[lemire(h1)] if (h2 & 0xFFFFFFFF) < RISK_CUT \
             else torch.zeros_like(enc.risk_lut[0])
         assert torch.equal(host.any_bits[i, DIM_RISK], want)
     frac = float((host.any_bits[:, DIM_RISK].abs().sum(dim=1) != 0).float().mean())
