@@ -463,6 +463,38 @@ class _RefOps:
         return echo_execute_ref(ctx_arena, res_arena, int(stride))
 
 
+def alloc_batch_staging(B: int, W: int, device=None, pin: bool = False):
+    """Allocate a JobBatch whose five fields are views of ONE contiguous
+    buffer, so the e2e loop stages a whole batch with a single copy_ (one
+    DMA submission instead of six). Layout is identical on the host (pinned)
+    and device sides; int64 fields first keep every view 8-byte aligned.
+    Returns (JobBatch-of-views, flat uint8 buffer)."""
+    n64 = B * 7 * W + B * 2 * W + B * 4 * W
+    nbytes = ((n64 * 8 + 2 * B + 7) // 8) * 8
+    if pin:
+        buf = torch.zeros(nbytes, dtype=torch.uint8).pin_memory()
+    else:
+        buf = torch.zeros(nbytes, dtype=torch.uint8, device=device)
+    o = 0
+
+    def take64(*shape):
+        nonlocal o
+        n = 1
+        for s in shape:
+            n *= s
+        t = buf[o:o + n * 8].view(torch.int64).view(*shape)
+        o += n * 8
+        return t
+
+    any_b = take64(B, 7, W)
+    all_b = take64(B, 2, W)
+    mcp_b = take64(B, 4, W)
+    sec = buf[o:o + B]
+    o += B
+    used = buf[o:o + B]
+    return JobBatch(any_b, all_b, sec, mcp_b, used), buf
+
+
 class DevicePipeline:
     def __init__(
         self,
@@ -506,11 +538,22 @@ class DevicePipeline:
         self.cpol = self.compiled.to(device)
 
         # --- pre-staged synthetic job batches (ring, distinct random content)
+        # each slot's fields view one contiguous buffer so the e2e loop can
+        # re-stage the whole batch with a single fused H2D copy
         self.batches: List[JobBatch] = []
+        self.batch_bufs: List[torch.Tensor] = []
         self.payloads: List[torch.Tensor] = []
         for i in range(n_batches):
             jb = encode_synthetic_jobs(self.compiled, self.B, seed=seed * 1000 + i + rank * 77)
-            self.batches.append(jb.to(device))
+            dev_jb, dev_buf = alloc_batch_staging(self.B, self.compiled.words,
+                                                  device=device)
+            dev_jb.any_bits.copy_(jb.any_bits)
+            dev_jb.all_bits.copy_(jb.all_bits)
+            dev_jb.secrets.copy_(jb.secrets)
+            dev_jb.mcp_bits.copy_(jb.mcp_bits)
+            dev_jb.mcp_used.copy_(jb.mcp_used)
+            self.batches.append(dev_jb)
+            self.batch_bufs.append(dev_buf)
             g = torch.Generator().manual_seed(seed + i + rank * 131)
             payload = torch.randint(-(1 << 31), (1 << 31) - 1,
                                     (self.B * payload_words,), dtype=torch.int32, generator=g)
@@ -871,17 +914,13 @@ class DevicePipeline:
         nslots = 2   # device staging ring slots (graphs)
         nhost = 3    # host encode buffers: encode s+2 while H2D of s in flight
         if not hasattr(self, "_e2e_hosts2"):
-            # per-host-slot staging + per-device-slot pinned result buffers
-            def host(shape, dtype):
-                return torch.zeros(shape, dtype=dtype).pin_memory()
-
+            # per-host-slot staging + per-device-slot pinned result buffers;
+            # host staging mirrors the device slots' fused layout so one
+            # copy_ per step stages the whole batch
             Wc = self.compiled.words
-            self._e2e_hosts2 = [
-                JobBatch(host((B, 7, Wc), torch.int64), host((B, 2, Wc), torch.int64),
-                         host((B,), torch.uint8), host((B, 4, Wc), torch.int64),
-                         host((B,), torch.uint8))
-                for _ in range(nhost)
-            ]
+            pairs = [alloc_batch_staging(B, Wc, pin=True) for _ in range(nhost)]
+            self._e2e_hosts2 = [p[0] for p in pairs]
+            self._e2e_host_bufs = [p[1] for p in pairs]
             # per-host-slot encoders: two encode threads run concurrently, so
             # each buffer owns its generator (single shared generator = race)
             self._e2e_encs = [SyntheticEncoder(self.compiled, seed=101 + 13 * i + self.rank)
@@ -944,7 +983,6 @@ class DevicePipeline:
         for s in range(steps):
             futs.pop(s).result()          # encode s ready (ran ∥ device s-1)
             slot = s % nslots
-            hb = self._e2e_hosts2[s % nhost]
             # stage on the copy stream BEFORE harvesting s-1: the H2D for
             # step s then overlaps the DMA engines with graph s-1 still
             # computing on the main stream. Safe: this slot's staging
@@ -952,13 +990,9 @@ class DevicePipeline:
             # harvest(s-2) proved one iteration ago — blocking on s-1
             # first (the old order) serialized [H2D][graph][D2H] on the
             # device and cost ~40% of the step
-            jb = self.batches[slot]
             with torch.cuda.stream(self._e2e_copy_stream):
-                jb.any_bits.copy_(hb.any_bits, non_blocking=True)
-                jb.all_bits.copy_(hb.all_bits, non_blocking=True)
-                jb.secrets.copy_(hb.secrets, non_blocking=True)
-                jb.mcp_bits.copy_(hb.mcp_bits, non_blocking=True)
-                jb.mcp_used.copy_(hb.mcp_used, non_blocking=True)
+                self.batch_bufs[slot].copy_(
+                    self._e2e_host_bufs[s % nhost], non_blocking=True)
                 self.payloads[slot].copy_(
                     self._e2e_payloads[s % len(self._e2e_payloads)],
                     non_blocking=True)
