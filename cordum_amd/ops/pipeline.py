@@ -887,6 +887,74 @@ class DevicePipeline:
         st.wall_s = time.perf_counter() - t0
         return st
 
+    def _e2e_run_dist(self, steps: int):
+        """Multi-rank e2e ingest window: same costs as the single-rank
+        pipeline (fresh host encode + H2D + full tick with its collectives +
+        result egress, all timed) but with eager ticks — the padded
+        all-to-all runs inside tick() on the main stream — and a depth-2
+        encode overlap: a worker thread encodes batch s+1 (GIL-free native
+        encoder) while this thread drives tick s. Runs under gloo on CPU so
+        the world>1 path is covered by the CPU suite."""
+        from concurrent.futures import ThreadPoolExecutor
+
+        self.ensure_e2e()
+        nhost = 2
+        cuda = self.device.type == "cuda"
+        if not hasattr(self, "_d_hosts"):
+            Wc = self.compiled.words
+            pairs = [alloc_batch_staging(self.B, Wc, pin=cuda)
+                     for _ in range(nhost)]
+            self._d_hosts = [p[0] for p in pairs]
+            self._d_host_bufs = [p[1] for p in pairs]
+            self._d_encs = [SyntheticEncoder(self.compiled,
+                                             seed=211 + 7 * i + self.rank)
+                            for i in range(nhost)]
+            self._d_pool = ThreadPoolExecutor(max_workers=1)
+
+        completed = denied = 0
+        self._d_unrouted = 0  # conservation accounting for the dist tests
+        lats = [0.0] * steps
+        t_enc = [0.0] * steps
+        native = hasattr(self.ext, "synthetic_fresh")
+
+        def encode(s: int):
+            t_enc[s] = time.perf_counter()
+            hb = self._d_hosts[s % nhost]
+            enc = self._d_encs[s % nhost]
+            if native:
+                enc.fresh_fast(hb, s, self.ext)
+            else:
+                enc.fresh(hb)
+            self._e2e_payloads[s % len(self._e2e_payloads)].view(
+                self.B, self.payload_words)[:, 0] = s
+
+        fut = self._d_pool.submit(encode, 0)
+        for s in range(steps):
+            fut.result()
+            slot = s % len(self.batches)
+            # fused H2D into this tick's staging slot; stream-ordered ahead
+            # of tick()'s kernels and collectives on the main stream
+            self.batch_bufs[slot].copy_(self._d_host_bufs[s % nhost],
+                                        non_blocking=cuda)
+            self.payloads[slot].copy_(
+                self._e2e_payloads[s % len(self._e2e_payloads)],
+                non_blocking=cuda)
+            if s + 1 < steps:
+                fut = self._d_pool.submit(encode, s + 1)
+            self._tick = slot
+            st = self.tick()
+            completed += st.completed
+            denied += st.denied
+            self._d_unrouted += st.unrouted
+            # result egress: checksums + decisions leave HBM every step
+            self._e2e_sums = (self.pad_sums_back if self.world > 1
+                              else self.res_sums).cpu()
+            self._e2e_decisions = self.out_decision.cpu()
+            if cuda:
+                torch.cuda.synchronize(self.device)
+            lats[s] = time.perf_counter() - t_enc[s]
+        return completed, denied, lats
+
     def e2e_run(self, steps: int):
         """Pipelined end-to-end ingest window: encode batch t+1 on the host
         WHILE the device executes tick t (depth-2 software pipeline over the
@@ -899,7 +967,11 @@ class DevicePipeline:
         Returns (completed, denied, batch_latencies_s): latency = encode
         start -> results for that batch readable on host."""
         self.ensure_e2e()
-        if self.device.type != "cuda" or self.world > 1 or not self._fused_capable:
+        if self.world > 1:
+            # multi-rank: eager ticks (collectives aren't graph-captured)
+            # but still pipelined — encode s+1 overlaps tick s
+            return self._e2e_run_dist(steps)
+        if self.device.type != "cuda" or not self._fused_capable:
             completed = denied = 0
             lats = []
             for _ in range(steps):
