@@ -863,7 +863,10 @@ class DevicePipeline:
         t0 = time.perf_counter()
         step = self._e2e_step
         self._e2e_step += 1
-        self._e2e_enc.fresh(self._e2e_host)
+        if hasattr(self.ext, "synthetic_fresh"):
+            self._e2e_enc.fresh_fast(self._e2e_host, step, self.ext)
+        else:
+            self._e2e_enc.fresh(self._e2e_host)
         payload = self._e2e_payloads[step % len(self._e2e_payloads)]
         payload.view(self.B, self.payload_words)[:, 0] = step  # distinct content per step
         nb = self.device.type == "cuda"
