@@ -846,7 +846,12 @@ def create_app(
         if pr is None:
             await ws.close(code=4401)
             return
-        await ws.accept()
+        # when the key rode in on the subprotocol (browser WebSocket API has
+        # no headers), echo the protocol name back or the client's handshake
+        # validation rejects the upgrade (gateway.go:2154-2185)
+        offered = ws.headers.get("sec-websocket-protocol", "")
+        sub = "cordum-api-key" if offered.split(",")[0].strip() == "cordum-api-key" else None
+        await ws.accept(subprotocol=sub)
         queue: List[Any] = []
         ws_clients.append(queue)
         import asyncio

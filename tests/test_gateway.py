@@ -507,3 +507,33 @@ def test_ws_stream_delivers_job_events(client, node):
         # protojson camelCase shape
         assert payload["jobRequest"]["topic"] == "job.echo"
         assert any(s.startswith("sys.job.") for s in seen_subjects)
+
+
+def test_ws_stream_subprotocol_auth(client):
+    """Browser clients can't set headers on a WebSocket, so the API key rides
+    the subprotocol (`Sec-WebSocket-Protocol: cordum-api-key, <b64url(key)>`,
+    gateway.go:2154-2185); the server must accept AND echo the negotiated
+    subprotocol or the client-side handshake validation fails."""
+    import base64
+
+    b64 = base64.urlsafe_b64encode(b"test-key").decode().rstrip("=")
+    with client.websocket_connect(
+            "/api/v1/stream",
+            subprotocols=["cordum-api-key", b64]) as ws:
+        r = client.post("/api/v1/jobs", json={"topic": "job.echo", "prompt": "sp"},
+                        headers={"X-API-Key": "test-key"})
+        assert r.status_code == 200
+        msg = ws.receive_json()
+        assert msg["subject"].startswith("sys.job.")
+
+
+def test_ws_stream_rejects_bad_key(client):
+    """An unauthenticated WS connect must be closed, not accepted."""
+    import pytest as _pytest
+    from starlette.websockets import WebSocketDisconnect
+
+    with _pytest.raises(WebSocketDisconnect):
+        with client.websocket_connect(
+                "/api/v1/stream",
+                headers={"X-API-Key": "wrong-key"}) as ws:
+            ws.receive_json()
