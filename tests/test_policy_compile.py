@@ -217,3 +217,35 @@ def test_native_synthetic_encoder_oracle():
         assert torch.equal(host.any_bits[i, DIM_RISK], want)
     frac = float((host.any_bits[:, DIM_RISK].abs().sum(dim=1) != 0).float().mean())
     assert 0.2 < frac < 0.4
+
+
+def test_alloc_batch_staging_layout():
+    """The fused staging allocator must alias all five JobBatch fields into
+    ONE buffer (so a single copy_ stages a batch) with 8-byte-aligned int64
+    views and zero initial content; host and device layouts are the same
+    code path, so asserting the CPU side pins the wire layout."""
+    import torch
+
+    from cordum_amd.ops.pipeline import alloc_batch_staging
+
+    B, W = 64, 2
+    jb, buf = alloc_batch_staging(B, W)
+    assert buf.dtype == torch.uint8
+    base = buf.data_ptr()
+    end = base + buf.numel()
+    for t, shape, dtype in [
+        (jb.any_bits, (B, 7, W), torch.int64),
+        (jb.all_bits, (B, 2, W), torch.int64),
+        (jb.mcp_bits, (B, 4, W), torch.int64),
+        (jb.secrets, (B,), torch.uint8),
+        (jb.mcp_used, (B,), torch.uint8),
+    ]:
+        assert tuple(t.shape) == shape and t.dtype == dtype
+        assert t.is_contiguous()
+        assert base <= t.data_ptr() < end          # aliases the flat buffer
+        if dtype == torch.int64:
+            assert t.data_ptr() % 8 == 0
+        assert int(t.abs().sum() if dtype == torch.int64 else t.sum()) == 0
+    # writing a field is visible through the flat buffer (single-copy proof)
+    jb.any_bits.fill_(-1)
+    assert int((buf != 0).sum()) == B * 7 * W * 8
