@@ -987,7 +987,13 @@ class DevicePipeline:
         self._ensure_graphs()
         B, W = self.B, self.payload_words
         nslots = 2   # device staging ring slots (graphs)
-        nhost = 3    # host encode buffers: encode s+2 while H2D of s in flight
+        # host encode buffers. The reuse distance must clear the sync proof:
+        # encode s+2 is submitted BEFORE harvest(s-1), so the buffer it
+        # rewrites must have had its last H2D read by step s-2 (proven done
+        # via harvest(s-2), whose graph waited on that copy) — with 3
+        # buffers (s+2)%3 == (s-1)%3 and the encoder could race the still
+        # in-flight H2D of s-1; 4 buffers put the last reader at s-2
+        nhost = 4
         if not hasattr(self, "_e2e_hosts2"):
             # per-host-slot staging + per-device-slot pinned result buffers;
             # host staging mirrors the device slots' fused layout so one
@@ -1051,8 +1057,9 @@ class DevicePipeline:
             if s < steps:
                 futs[s] = self._e2e_pool.submit(encode, s)
 
-        # encode buffer s%3 is free to rewrite at step s+3: its H2D (step s)
-        # completed before event s, which harvest(s) synced at step s+1
+        # encode buffer s%4 is free to rewrite at step s+4: its H2D (step s)
+        # was waited on by graph s, whose completion harvest(s) proved
+        # before submit(s+4) runs (see nhost above)
         submit(0)
         submit(1)
         for s in range(steps):
