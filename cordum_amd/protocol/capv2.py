@@ -128,6 +128,10 @@ class Message:
         for name, spec in sorted(self.FIELDS.items(), key=lambda kv: kv[1].num):
             val = getattr(self, name)
             out += _encode_field(spec, val)
+        # unknown fields captured at decode re-emit verbatim (after known
+        # fields, like Go protobuf): a node relaying packets from a newer
+        # CAP client must not strip fields it doesn't model
+        out += getattr(self, "_unknown", b"")
         return bytes(out)
 
     @classmethod
@@ -136,6 +140,7 @@ class Message:
         by_num = {spec.num: (name, spec) for name, spec in cls.FIELDS.items()}
         i, n = 0, len(data)
         while i < n:
+            fstart = i
             key, i = _dec_varint(data, i)
             num, wt = key >> 3, key & 7
             if wt == _WT_VARINT:
@@ -154,7 +159,9 @@ class Message:
                 raise ValueError(f"bad wire type {wt}")
             ent = by_num.get(num)
             if ent is None:
-                continue  # unknown field: skip (forward compat)
+                # unknown field: preserve raw bytes for re-encode
+                msg._unknown = getattr(msg, "_unknown", b"") + bytes(data[fstart:i])
+                continue
             name, spec = ent
             if _WIRE_TYPE.get(spec.kind, _WT_LEN) != wt:
                 raise ValueError(

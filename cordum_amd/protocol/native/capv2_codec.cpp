@@ -190,6 +190,16 @@ static void encode_msg(const std::string& key, py::handle obj, std::string& out)
     py::object val = py::reinterpret_borrow<py::object>(obj).attr(f.name.c_str());
     encode_field(f, val, out);
   }
+  // unknown fields captured by the Python decoder re-emit verbatim after
+  // known fields (Go protobuf layout) so relays don't strip foreign fields
+  PyObject* unk = PyObject_GetAttrString(obj.ptr(), "_unknown");
+  if (unk == nullptr) {
+    PyErr_Clear();
+  } else {
+    if (PyBytes_Check(unk) && PyBytes_GET_SIZE(unk) > 0)
+      out.append(PyBytes_AS_STRING(unk), (size_t)PyBytes_GET_SIZE(unk));
+    Py_DECREF(unk);
+  }
 }
 
 // -- decode ------------------------------------------------------------------
@@ -297,7 +307,11 @@ static py::object decode_msg(const std::string& key, const unsigned char* buf,
       throw std::runtime_error("bad wire type");
     }
     auto fit = it->second.by_num.find(num);
-    if (fit == it->second.by_num.end()) continue;  // unknown: skip
+    if (fit == it->second.by_num.end())
+      // unknown field (newer CAP peer): defer the whole packet to the
+      // Python codec, which preserves unknown bytes for re-encode; the
+      // dispatcher treats this throw as "fall back"
+      throw std::runtime_error("unknown field " + std::to_string(num));
     const FieldSpec& fs = it->second.fields[fit->second];
     const int want_wt =
         (fs.kind == K_INT || fs.kind == K_SINT64 || fs.kind == K_BOOL ||

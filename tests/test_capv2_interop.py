@@ -223,3 +223,36 @@ def test_decode_protobuf_bytes(classes, type_name):
         back = classes[type_name]()
         back.ParseFromString(ours.encode())
         assert back.SerializeToString(deterministic=True) == wire
+
+
+def test_unknown_fields_survive_reencode():
+    """Forward compat: a packet from a NEWER CAP peer carrying fields this
+    build doesn't model must round-trip byte-identically (Go protobuf
+    preserves unknown fields; a relay that strips them would corrupt
+    foreign traffic). Covers the native codec's defer-to-Python path and
+    nested submessages."""
+    from cordum_amd.protocol.capv2 import BusPacket, JobRequest
+
+    def varint(n):
+        out = b""
+        while True:
+            b = n & 0x7F
+            n >>= 7
+            out += bytes([b | (0x80 if n else 0)])
+            if not n:
+                return out
+
+    base = BusPacket(protocol_version=1,
+                     job_request=JobRequest(job_id="x", topic="job.t")).encode()
+    foreign = base + varint((199 << 3) | 0) + varint(42) \
+        + varint((200 << 3) | 2) + varint(3) + b"abc"
+    rt = BusPacket.decode(foreign)
+    assert rt.job_request.job_id == "x"       # known fields still parse
+    assert rt.encode() == foreign             # unknown bytes preserved
+
+    # nested: unknown field inside the JobRequest submessage
+    jr = JobRequest(job_id="y").encode() + varint((150 << 3) | 0) + varint(7)
+    outer = varint((10 << 3) | 2) + varint(len(jr)) + jr
+    rt2 = BusPacket.decode(outer)
+    assert rt2.job_request.job_id == "y"
+    assert rt2.encode() == outer
