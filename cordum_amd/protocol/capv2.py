@@ -678,6 +678,7 @@ class PolicyCheckResponse(Message):
                     out += _tag(spec.num, _WT_LEN) + _enc_varint(len(b)) + b
             else:
                 out += _encode_field(spec, val)
+        out += getattr(self, "_unknown", b"")
         return bytes(out)
 
     @classmethod
@@ -686,6 +687,7 @@ class PolicyCheckResponse(Message):
         i, n = 0, len(data)
         by_num = {spec.num: (name, spec) for name, spec in cls.FIELDS.items()}
         while i < n:
+            fstart = i
             key, i = _dec_varint(data, i)
             num, wt = key >> 3, key & 7
             if wt == _WT_VARINT:
@@ -697,10 +699,14 @@ class PolicyCheckResponse(Message):
             elif wt == _WT_I64:
                 raw = data[i : i + 8]
                 i += 8
+            elif wt == _WT_I32:
+                raw = data[i : i + 4]
+                i += 4
             else:
                 raise ValueError(f"bad wire type {wt}")
             ent = by_num.get(num)
             if ent is None:
+                msg._unknown = getattr(msg, "_unknown", b"") + bytes(data[fstart:i])
                 continue
             name, spec = ent
             if _WIRE_TYPE.get(spec.kind, _WT_LEN) != wt:
