@@ -65,6 +65,7 @@ def create_app(
     rate_limit_rps: float = 0.0,
     rate_limit_burst: int = 100,
     dashboard_dir: str = "",
+    extensions: Optional[List[Any]] = None,
 ) -> FastAPI:
     app = FastAPI(title="cordum-mi355x gateway", version="0.1.0")
     # CORS for the dashboard (gateway.go:2051-2143)
@@ -98,9 +99,46 @@ def create_app(
     node.bus.subscribe("sys.audit.>", _ws_tap)
     node.bus.subscribe(subj.SUBJECT_WORKFLOW_EVENT, _ws_tap)
 
+    # enterprise extension seams (gateway/extensions.go:9-38): probe each
+    # extension object for the protocols it implements
+    from .extensions import (AuditEvent, AuditExporter, LicenseInfoProvider,
+                             PublicPathProvider, RouteRegistrar,
+                             path_is_public)
+
+    exts = list(extensions or [])
+    public_paths: List[str] = []
+    license_provider: Optional[Any] = None
+    for e in exts:
+        if isinstance(e, PublicPathProvider):
+            public_paths.extend(e.public_paths())
+        if isinstance(e, LicenseInfoProvider):
+            license_provider = e
+    app.state.public_paths = public_paths
+    app.state.license_provider = license_provider
+    audit_exporters = [e for e in exts if isinstance(e, AuditExporter)]
+    if audit_exporters:
+        def _audit_tap(subject: str, pkt: BusPacket):
+            alert = pkt.alert
+            ev = AuditEvent(
+                subject=subject,
+                actor=getattr(alert, "source", "") if alert else "",
+                action=getattr(alert, "severity", "") if alert else "",
+                resource=pkt.trace_id,
+                detail=_packet_json(pkt),
+            )
+            for ex in audit_exporters:
+                try:
+                    ex.export_audit(ev)
+                except Exception:
+                    pass  # a broken exporter must not break the bus
+
+        node.bus.subscribe("sys.audit.>", _audit_tap)
+
     def principal(request: Request) -> Principal:
         if not bucket.allow():
             raise HTTPException(429, "rate limited")
+        if public_paths and path_is_public(request.url.path, public_paths):
+            return Principal(id="public", tenant="", role="public")
         p = auth.authenticate(dict(request.headers))
         if p is None:
             raise HTTPException(401, "invalid api key")
@@ -388,7 +426,8 @@ def create_app(
             "store": "hbm+host",
             "workers": node.registry.count(),
             "policy_snapshot": node.safety_kernel.snapshot,
-            "license": {"edition": "oss"},
+            "license": (license_provider.license_info()
+                        if license_provider else {"edition": "oss"}),
         }
 
     # ---------------------------------------------------------------- memory
@@ -883,6 +922,11 @@ def create_app(
     from .dashboard import add_dashboard
 
     add_dashboard(app)
+    # extension-registered routes go last, like the reference's registrar
+    # hook (they may shadow nothing; the compat table is already mounted)
+    for e in exts:
+        if isinstance(e, RouteRegistrar):
+            e.register_routes(app)
     return app
 
 

@@ -537,3 +537,49 @@ def test_ws_stream_rejects_bad_key(client):
                 "/api/v1/stream",
                 headers={"X-API-Key": "wrong-key"}) as ws:
             ws.receive_json()
+
+
+def test_enterprise_extension_seams(node):
+    """The extension hook surface (gateway/extensions.go:9-38): route
+    registrar, public paths (served without auth), audit exporter, and
+    license-info provider must all take effect through create_app."""
+    from fastapi.testclient import TestClient
+
+    from cordum_amd.gateway.app import create_app
+    from cordum_amd.gateway.auth import BasicAuthProvider
+    from cordum_amd.protocol.capv2 import BusPacket, SystemAlert
+
+    audited = []
+
+    class Ext:
+        def register_routes(self, app):
+            @app.get("/ext/ping")
+            def ping():
+                return {"ext": True}
+
+        def public_paths(self):
+            return ["/api/v1/status"]
+
+        def export_audit(self, event):
+            audited.append(event)
+
+        def license_info(self):
+            return {"edition": "enterprise", "seats": 5}
+
+    app = create_app(node, auth=BasicAuthProvider(api_keys=["k"]),
+                     extensions=[Ext()])
+    c = TestClient(app)
+    # registrar route exists
+    assert c.get("/ext/ping").json() == {"ext": True}
+    # public path: no API key required; license info provider honored
+    r = c.get("/api/v1/status")
+    assert r.status_code == 200
+    assert r.json()["license"] == {"edition": "enterprise", "seats": 5}
+    # non-public path still requires auth
+    assert c.get("/api/v1/jobs").status_code == 401
+    # audit exporter receives sys.audit.> traffic
+    node.bus.publish("sys.audit.test", BusPacket(
+        trace_id="tr-1", protocol_version=1,
+        alert=SystemAlert(source="unit", severity="info")))
+    assert audited and audited[0].subject == "sys.audit.test"
+    assert audited[0].actor == "unit" and audited[0].resource == "tr-1"
