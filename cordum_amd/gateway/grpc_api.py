@@ -16,6 +16,8 @@ from concurrent import futures
 from dataclasses import dataclass, field as dc_field
 from typing import Dict, List, Optional
 
+import os
+
 import grpc
 
 from ..protocol.capv2 import (
@@ -189,11 +191,14 @@ def _auth_ok(context, api_keys) -> bool:
 
 
 def make_grpc_server(node, api_keys: Optional[List[str]] = None,
-                     max_workers: int = 8) -> grpc.Server:
+                     max_workers: int = 8, auth_tenant: str = "") -> grpc.Server:
     from ..runtime.context_engine import ContextEngine
     from ..utils.ids import new_trace_id, new_id
 
     api_keys = set(api_keys or [])
+    # auth-context tenant override (gateway.go:4149-4159): a request that
+    # omits org_id lands in the authenticated tenant, not a global default
+    auth_tenant = auth_tenant or os.environ.get("TENANT_ID", "")
     ctx_engine = ContextEngine(node.memory)
 
     def submit_job(req_bytes, context):
@@ -205,7 +210,7 @@ def make_grpc_server(node, api_keys: Optional[List[str]] = None,
         topic = req.topic or "job.default"
         if not topic.startswith("job."):
             context.abort(grpc.StatusCode.INVALID_ARGUMENT, "topic must start with job.")
-        org = req.org_id or "default"
+        org = req.org_id or auth_tenant or "default"
         job_id = new_id()
         if req.idempotency_key:
             inserted, existing = node.job_store.try_set_idempotency_key(org, req.idempotency_key, job_id)
@@ -312,8 +317,9 @@ def make_grpc_server(node, api_keys: Optional[List[str]] = None,
     return server
 
 
-def serve_grpc(node, address: str = "127.0.0.1:9090", api_keys: Optional[List[str]] = None) -> grpc.Server:
-    server = make_grpc_server(node, api_keys)
+def serve_grpc(node, address: str = "127.0.0.1:9090", api_keys: Optional[List[str]] = None,
+               auth_tenant: str = "") -> grpc.Server:
+    server = make_grpc_server(node, api_keys, auth_tenant=auth_tenant)
     server.add_insecure_port(address)
     server.start()
     return server

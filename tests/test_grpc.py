@@ -108,3 +108,28 @@ def test_context_engine_grpc(grpc_node):
             _, i = _dec_varint(raw, i)
     contents = [m.content for m in msgs]
     assert contents == ["hi", "there", "again"]
+
+
+def test_submit_tenant_from_auth_context(grpc_node):
+    """A SubmitJob that omits org_id lands in the authenticated tenant
+    (gateway.go:4149-4159), not the global default."""
+    node, stub = grpc_node
+    from cordum_amd.gateway.grpc_api import SubmitJobRequest, SubmitJobResponse, make_grpc_server
+    import grpc as _grpc
+
+    server = make_grpc_server(node, api_keys=["gkey"], auth_tenant="acme")
+    port = server.add_insecure_port("127.0.0.1:0")
+    server.start()
+    try:
+        ch = _grpc.insecure_channel(f"127.0.0.1:{port}")
+        call = ch.unary_unary(
+            "/cordum.v1.CordumApi/SubmitJob",
+            request_serializer=lambda b: b, response_deserializer=lambda b: b)
+        resp = SubmitJobResponse.decode(call(
+            SubmitJobRequest(prompt="hi", topic="job.echo").encode(),
+            metadata=(("x-api-key", "gkey"),)))
+        node.drain()
+        meta = node.job_store.get_job_meta(resp.job_id)
+        assert meta.get("tenant") == "acme"
+    finally:
+        server.stop(0)
