@@ -6,7 +6,8 @@ Oracle: core/infra/memory/redis_store.go:26-160 — `ctx:<job_id>` /
 
 The wire pointer format stays `redis://<key>` for client compatibility even
 though the backing is a host arena (and, on the data plane, an HBM payload
-arena — ops/arena.py — with the same keys).
+arena — the rotating payload arenas in ops/pipeline.py and
+ops/wf_pipeline.py — with the same keys).
 """
 from __future__ import annotations
 

@@ -5,7 +5,7 @@ condition/delay/notify + for_each have engine logic), Workflow/Step/
 WorkflowRun/StepRun/TimelineEvent shapes, retry config.
 
 Runs are plain dataclasses on the host; the batched readiness sweep
-(ops/run_table.py, K3) packs active runs' dependency bitmasks and step states
+(ops/wf_pipeline.py, K3-WF) packs active runs' dependency bitmasks and step states
 into device tensors — capped at 64 steps per workflow for the bitmask path
 (larger workflows stay host-evaluated).
 """
