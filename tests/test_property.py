@@ -616,3 +616,39 @@ def test_job_hash_strips_exactly_the_approval_binding_exclusions(job_id, topic, 
     changed3 = JobRequest.decode(base.encode())
     changed3.topic = topic + "!"
     assert job_hash(changed3) != h0
+
+
+@given(
+    job_id=st.text(min_size=0, max_size=12),
+    extra=st.lists(
+        st.tuples(st.integers(min_value=100, max_value=500),   # unknown field num
+                  st.one_of(st.integers(min_value=0, max_value=2**40),
+                            st.binary(max_size=24))),
+        max_size=4),
+)
+@settings(max_examples=60, deadline=None)
+def test_unknown_field_roundtrip_property(job_id, extra):
+    """Any combination of unknown varint/bytes fields appended to a valid
+    BusPacket must survive decode→encode byte-for-byte (forward compat for
+    relays), through whichever codec engine (native or Python) is active."""
+    from cordum_amd.protocol.capv2 import BusPacket, JobRequest
+
+    def varint(n):
+        out = b""
+        while True:
+            b = n & 0x7F
+            n >>= 7
+            out += bytes([b | (0x80 if n else 0)])
+            if not n:
+                return out
+
+    wire = BusPacket(protocol_version=1,
+                     job_request=JobRequest(job_id=job_id)).encode()
+    for num, val in extra:
+        if isinstance(val, int):
+            wire += varint((num << 3) | 0) + varint(val)
+        else:
+            wire += varint((num << 3) | 2) + varint(len(val)) + val
+    rt = BusPacket.decode(wire)
+    assert rt.job_request is not None and rt.job_request.job_id == job_id
+    assert rt.encode() == wire
