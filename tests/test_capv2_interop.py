@@ -256,3 +256,77 @@ def test_unknown_fields_survive_reencode():
     rt2 = BusPacket.decode(outer)
     assert rt2.job_request.job_id == "y"
     assert rt2.encode() == outer
+
+
+@pytest.fixture(scope="module")
+def api_classes():
+    """google.protobuf classes built from the vendored api.proto (the gRPC
+    service contract gateway/grpc_api.py hand-implements)."""
+    api_path = PROTO_PATH.parent / "api.proto"
+    enums, messages = parse_proto(api_path.read_text())
+    fdp = descriptor_pb2.FileDescriptorProto()
+    fdp.name = "api_interop.proto"
+    fdp.package = "cap.v2.interop.api"
+    fdp.syntax = "proto3"
+    for mname, fields in messages.items():
+        msg = fdp.message_type.add()
+        msg.name = mname
+        for fname, num, typ, rep in fields:
+            fld = msg.field.add()
+            fld.name = fname
+            fld.number = num
+            if typ == "map<string,string>":
+                entry = msg.nested_type.add()
+                entry.name = "".join(p.capitalize() for p in fname.split("_")) + "Entry"
+                entry.options.map_entry = True
+                for kn, kno in (("key", 1), ("value", 2)):
+                    kf = entry.field.add()
+                    kf.name = kn
+                    kf.number = kno
+                    kf.type = TYPE.TYPE_STRING
+                    kf.label = TYPE.LABEL_OPTIONAL
+                fld.type = TYPE.TYPE_MESSAGE
+                fld.type_name = f".cap.v2.interop.api.{mname}.{entry.name}"
+                fld.label = TYPE.LABEL_REPEATED
+            elif typ in SCALARS:
+                fld.type = SCALARS[typ]
+                fld.label = TYPE.LABEL_REPEATED if rep else TYPE.LABEL_OPTIONAL
+            else:
+                fld.type = TYPE.TYPE_MESSAGE
+                fld.type_name = f".cap.v2.interop.api.{typ}"
+                fld.label = TYPE.LABEL_REPEATED if rep else TYPE.LABEL_OPTIONAL
+    pool = descriptor_pool.DescriptorPool()
+    fd = pool.Add(fdp)
+    return {name: message_factory.GetMessageClass(fd.message_types_by_name[name])
+            for name in messages}
+
+
+def test_api_proto_matches_grpc_codec(api_classes):
+    """api.proto is the documented contract for the gRPC surface: encode a
+    fully-populated SubmitJobRequest with google.protobuf and decode with
+    the hand-rolled codec (and back), byte-for-byte."""
+    from cordum_amd.gateway import grpc_api as g
+
+    P = api_classes["SubmitJobRequest"]
+    pb_msg = P(prompt="p", topic="job.t", priority="batch", org_id="acme",
+               idempotency_key="ik", actor_type="human",
+               risk_tags=["a", "b"], requires=["r1"],
+               labels={"k1": "v1", "k2": "v2"}, memory_id="m1")
+    wire = pb_msg.SerializeToString(deterministic=True)
+    ours = g.SubmitJobRequest.decode(wire)
+    assert ours.prompt == "p" and ours.labels == {"k1": "v1", "k2": "v2"}
+    assert ours.risk_tags == ["a", "b"]
+    assert ours.encode() == wire
+
+    # response direction + ContextEngine shapes
+    R = api_classes["BuildWindowResponse"]
+    mine = g.BuildWindowResponse(
+        messages=[g.ModelMessagePb(role="user", content="hi"),
+                  g.ModelMessagePb(role="assistant", content="yo")],
+        input_tokens=7, output_tokens=3)
+    theirs = R()
+    theirs.ParseFromString(mine.encode())
+    assert [(m.role, m.content) for m in theirs.messages] == \
+        [("user", "hi"), ("assistant", "yo")]
+    assert theirs.input_tokens == 7 and theirs.output_tokens == 3
+    assert theirs.SerializeToString(deterministic=True) == mine.encode()
